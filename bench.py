@@ -1,0 +1,155 @@
+"""Flagship benchmark: ResNet-18 / CIFAR-10-shaped synthetic data, bf16.
+
+Measures the BASELINE.json headline metric — images/sec (whole node) for the
+synchronous PS engine — at N GPUs (N=1: the single-machine engine; N>1:
+1 PS on rank 0 + N-1 workers, batch 1024 per worker, weak scaling).
+
+Single GPU:      python bench.py --gpus 1 --steps 30 --warmup 5
+Multi GPU (driver launches):
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 --master-port P bench.py --gpus N --steps K --warmup W
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse():
+    p = argparse.ArgumentParser()
+    p.add_argument('--gpus', type=int, default=1)
+    p.add_argument('--steps', type=int, default=30)
+    p.add_argument('--warmup', type=int, default=5)
+    p.add_argument('--batch-size', type=int, default=1024,
+                   help='per-worker batch size')
+    p.add_argument('--network', type=str, default='ResNet18')
+    p.add_argument('--dataset', type=str, default='Cifar10')
+    p.add_argument('--compress-grad', type=str, default='compress')
+    p.add_argument('--bucket-mb', type=float, default=25.0)
+    p.add_argument('--no-overlap', action='store_true')
+    return p.parse_args()
+
+
+def make_batches(n_batches, bs, device, dtype, seed=1234):
+    g = torch.Generator(device='cpu').manual_seed(seed)
+    xs, ys = [], []
+    for _ in range(n_batches):
+        xs.append(torch.randn(bs, 3, 32, 32, generator=g).to(device, dtype))
+        ys.append(torch.randint(0, 10, (bs,), generator=g).to(device))
+    return xs, ys
+
+
+def main():
+    args = parse()
+    from ps_pytorch_amd.config import JobConfig
+    from ps_pytorch_amd.parallel.transport import init_distributed
+
+    world = int(os.environ.get('WORLD_SIZE', '1'))
+    rank = int(os.environ.get('RANK', '0'))
+    n_gpus = max(world, 1)
+    use_cuda = torch.cuda.is_available()
+    cfg = JobConfig(network=args.network, dataset=args.dataset,
+                    batch_size=args.batch_size, lr=0.1, momentum=0.9,
+                    max_steps=args.steps + args.warmup,
+                    compress_grad=args.compress_grad,
+                    bucket_mb=args.bucket_mb, overlap=not args.no_overlap,
+                    enable_gpu=use_cuda, eval_freq=10 ** 9,
+                    train_dir='/tmp/ps_bench_models')
+
+    if world > 1:
+        import torch.distributed as dist
+        env = init_distributed()
+        device = env['device']
+        from ps_pytorch_amd.parallel.ps import ParameterServer
+        from ps_pytorch_amd.parallel.worker import DistributedWorker
+        if rank == 0:
+            role = ParameterServer(cfg, rank, world, device)
+            role.build_model(10)
+            step = role.step
+        else:
+            role = DistributedWorker(cfg, rank, world, device)
+            role.build_model(10)
+            xs, ys = make_batches(8, args.batch_size, device,
+                                  role.compute_dtype, seed=1234 + rank)
+            it = [0]
+
+            def step():
+                i = it[0] % len(xs)
+                it[0] += 1
+                role.train_step(xs[i], ys[i])
+
+        for _ in range(args.warmup):
+            step()
+        if use_cuda:
+            torch.cuda.synchronize()
+        dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+        t0 = time.time()
+        for _ in range(args.steps):
+            step()
+        if use_cuda:
+            torch.cuda.synchronize()
+        dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+        elapsed = time.time() - t0
+        # max over ranks
+        et = torch.tensor([elapsed], dtype=torch.float64,
+                          device=device if env['backend'] == 'nccl' else 'cpu')
+        dist.all_reduce(et, op=dist.ReduceOp.MAX)
+        elapsed = float(et)
+        n_workers = world - 1
+        imgs = n_workers * args.batch_size * args.steps
+        parallelism = f"ps-dp (1 PS + {n_workers} workers)"
+        global_batch = n_workers * args.batch_size
+    else:
+        from ps_pytorch_amd.trainer import NNTrainer
+        device = torch.device('cuda', 0) if use_cuda else torch.device('cpu')
+        tr = NNTrainer(cfg, device=device)
+        tr.build_model(10)
+        xs, ys = make_batches(8, args.batch_size, device, tr.compute_dtype)
+        for i in range(args.warmup):
+            tr.train_step(xs[i % len(xs)], ys[i % len(ys)])
+        if use_cuda:
+            torch.cuda.synchronize()
+        t0 = time.time()
+        for i in range(args.steps):
+            tr.train_step(xs[i % len(xs)], ys[i % len(ys)])
+        if use_cuda:
+            torch.cuda.synchronize()
+        elapsed = time.time() - t0
+        imgs = args.batch_size * args.steps
+        parallelism = "single-gpu"
+        global_batch = args.batch_size
+
+    if rank == 0:
+        value = imgs / elapsed
+        dtype = 'bf16' if use_cuda else 'fp32'
+        print(json.dumps({
+            "metric": "images/sec (whole node), ResNet-18/CIFAR-10 synchronous PS training",
+            "value": value,
+            "unit": "images/sec",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": dtype,
+            "data": "synthetic (random-init weights, GPU-resident random batches)",
+            "config": {"model": args.network, "global_batch": global_batch,
+                       "input": "3x32x32", "per_worker_batch": args.batch_size,
+                       "parallelism": parallelism,
+                       "compress_grad": args.compress_grad,
+                       "overlap": not args.no_overlap},
+        }))
+
+
+if __name__ == '__main__':
+    main()

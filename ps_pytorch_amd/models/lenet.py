@@ -1,0 +1,35 @@
+"""LeNet for MNIST-shaped inputs.
+
+Same architecture/parameter-count surface as the reference's LeNet
+(ref: src/model_ops/lenet.py:16-37 — 8 param tensors, 431,080 params for
+1x28x28 inputs): conv(1->20,k5) -> pool -> conv(20->50,k5) -> pool ->
+fc(800->500) -> fc(500->nc).
+
+There is no hand-unrolled "split" variant here: the reference's LeNetSplit
+(per-layer Isend interleaved with backward, src/model_ops/lenet.py:39-258)
+is realized framework-wide by bucketed gradient hooks pushing RCCL ops on a
+side HIP stream (ps_pytorch_amd/parallel/worker.py) — every model gets the
+comm/compute overlap without a hand-written backward.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+class LeNet(nn.Module):
+    def __init__(self, num_classes: int = 10, in_channels: int = 1):
+        super().__init__()
+        self.conv1 = nn.Conv2d(in_channels, 20, kernel_size=5)
+        self.conv2 = nn.Conv2d(20, 50, kernel_size=5)
+        # 28x28 -> conv5 -> 24 -> pool -> 12 -> conv5 -> 8 -> pool -> 4
+        self.fc1 = nn.Linear(50 * 4 * 4, 500)
+        self.fc2 = nn.Linear(500, num_classes)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = F.max_pool2d(F.relu(self.conv1(x)), 2)
+        x = F.max_pool2d(F.relu(self.conv2(x)), 2)
+        x = x.flatten(1)
+        x = F.relu(self.fc1(x))
+        return self.fc2(x)

@@ -1,0 +1,124 @@
+"""ResNet family, CIFAR variant (3x3 stem, no maxpool), as in the reference
+model zoo (ref: src/model_ops/resnet.py:14-113 — ResNet-18 = 62 param
+tensors / 11,173,962 params on CIFAR-10).
+
+For ImageNet-shaped inputs (224x224) the stem switches to 7x7/stride-2 +
+maxpool so ResNet-50 on synthetic 3x224x224 (BASELINE config 5) has the
+standard compute shape.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+class BasicBlock(nn.Module):
+    expansion = 1
+
+    def __init__(self, in_planes: int, planes: int, stride: int = 1):
+        super().__init__()
+        self.conv1 = nn.Conv2d(in_planes, planes, 3, stride=stride, padding=1, bias=False)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.conv2 = nn.Conv2d(planes, planes, 3, stride=1, padding=1, bias=False)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.shortcut = nn.Sequential()
+        if stride != 1 or in_planes != planes * self.expansion:
+            self.shortcut = nn.Sequential(
+                nn.Conv2d(in_planes, planes * self.expansion, 1, stride=stride, bias=False),
+                nn.BatchNorm2d(planes * self.expansion),
+            )
+
+    def forward(self, x):
+        out = F.relu(self.bn1(self.conv1(x)))
+        out = self.bn2(self.conv2(out))
+        out = out + self.shortcut(x)
+        return F.relu(out)
+
+
+class Bottleneck(nn.Module):
+    expansion = 4
+
+    def __init__(self, in_planes: int, planes: int, stride: int = 1):
+        super().__init__()
+        self.conv1 = nn.Conv2d(in_planes, planes, 1, bias=False)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.conv2 = nn.Conv2d(planes, planes, 3, stride=stride, padding=1, bias=False)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.conv3 = nn.Conv2d(planes, planes * self.expansion, 1, bias=False)
+        self.bn3 = nn.BatchNorm2d(planes * self.expansion)
+        self.shortcut = nn.Sequential()
+        if stride != 1 or in_planes != planes * self.expansion:
+            self.shortcut = nn.Sequential(
+                nn.Conv2d(in_planes, planes * self.expansion, 1, stride=stride, bias=False),
+                nn.BatchNorm2d(planes * self.expansion),
+            )
+
+    def forward(self, x):
+        out = F.relu(self.bn1(self.conv1(x)))
+        out = F.relu(self.bn2(self.conv2(out)))
+        out = self.bn3(self.conv3(out))
+        out = out + self.shortcut(x)
+        return F.relu(out)
+
+
+class ResNet(nn.Module):
+    def __init__(self, block, num_blocks, num_classes: int = 10,
+                 in_channels: int = 3, imagenet_stem: bool = False):
+        super().__init__()
+        self.in_planes = 64
+        self.imagenet_stem = imagenet_stem
+        if imagenet_stem:
+            self.conv1 = nn.Conv2d(in_channels, 64, 7, stride=2, padding=3, bias=False)
+        else:
+            self.conv1 = nn.Conv2d(in_channels, 64, 3, stride=1, padding=1, bias=False)
+        self.bn1 = nn.BatchNorm2d(64)
+        self.layer1 = self._make_layer(block, 64, num_blocks[0], stride=1)
+        self.layer2 = self._make_layer(block, 128, num_blocks[1], stride=2)
+        self.layer3 = self._make_layer(block, 256, num_blocks[2], stride=2)
+        self.layer4 = self._make_layer(block, 512, num_blocks[3], stride=2)
+        self.linear = nn.Linear(512 * block.expansion, num_classes)
+
+    def _make_layer(self, block, planes, n, stride):
+        strides = [stride] + [1] * (n - 1)
+        layers = []
+        for s in strides:
+            layers.append(block(self.in_planes, planes, s))
+            self.in_planes = planes * block.expansion
+        return nn.Sequential(*layers)
+
+    def forward(self, x):
+        out = F.relu(self.bn1(self.conv1(x)))
+        if self.imagenet_stem:
+            out = F.max_pool2d(out, 3, stride=2, padding=1)
+        out = self.layer1(out)
+        out = self.layer2(out)
+        out = self.layer3(out)
+        out = self.layer4(out)
+        out = F.adaptive_avg_pool2d(out, 1).flatten(1)
+        return self.linear(out)
+
+
+def _imagenet(num_classes: int) -> bool:
+    # 1000-way default => ImageNet-shaped stem (BASELINE config 5)
+    return num_classes >= 200
+
+
+def ResNet18(num_classes=10, in_channels=3):
+    return ResNet(BasicBlock, [2, 2, 2, 2], num_classes, in_channels, _imagenet(num_classes))
+
+
+def ResNet34(num_classes=10, in_channels=3):
+    return ResNet(BasicBlock, [3, 4, 6, 3], num_classes, in_channels, _imagenet(num_classes))
+
+
+def ResNet50(num_classes=10, in_channels=3):
+    return ResNet(Bottleneck, [3, 4, 6, 3], num_classes, in_channels, _imagenet(num_classes))
+
+
+def ResNet101(num_classes=10, in_channels=3):
+    return ResNet(Bottleneck, [3, 4, 23, 3], num_classes, in_channels, _imagenet(num_classes))
+
+
+def ResNet152(num_classes=10, in_channels=3):
+    return ResNet(Bottleneck, [3, 8, 36, 3], num_classes, in_channels, _imagenet(num_classes))

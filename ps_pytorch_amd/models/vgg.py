@@ -1,0 +1,73 @@
+"""VGG family for CIFAR-shaped inputs (ref: src/model_ops/vgg.py:15-68)."""
+from __future__ import annotations
+
+import torch.nn as nn
+
+_CFG = {
+    'VGG11': [64, 'M', 128, 'M', 256, 256, 'M', 512, 512, 'M', 512, 512, 'M'],
+    'VGG13': [64, 64, 'M', 128, 128, 'M', 256, 256, 'M', 512, 512, 'M', 512, 512, 'M'],
+    'VGG16': [64, 64, 'M', 128, 128, 'M', 256, 256, 256, 'M', 512, 512, 512, 'M',
+              512, 512, 512, 'M'],
+    'VGG19': [64, 64, 'M', 128, 128, 'M', 256, 256, 256, 256, 'M', 512, 512, 512, 512,
+              'M', 512, 512, 512, 512, 'M'],
+}
+
+
+def _make_layers(cfg, in_channels: int, batch_norm: bool) -> nn.Sequential:
+    layers = []
+    c = in_channels
+    for v in cfg:
+        if v == 'M':
+            layers.append(nn.MaxPool2d(2, 2))
+        else:
+            layers.append(nn.Conv2d(c, v, 3, padding=1, bias=not batch_norm))
+            if batch_norm:
+                layers.append(nn.BatchNorm2d(v))
+            layers.append(nn.ReLU(inplace=True))
+            c = v
+    layers.append(nn.AdaptiveAvgPool2d(1))
+    return nn.Sequential(*layers)
+
+
+class VGG(nn.Module):
+    def __init__(self, name: str, num_classes: int = 10, in_channels: int = 3,
+                 batch_norm: bool = False):
+        super().__init__()
+        self.features = _make_layers(_CFG[name], in_channels, batch_norm)
+        self.classifier = nn.Linear(512, num_classes)
+
+    def forward(self, x):
+        x = self.features(x).flatten(1)
+        return self.classifier(x)
+
+
+def VGG11(num_classes=10, in_channels=3):
+    return VGG('VGG11', num_classes, in_channels, False)
+
+
+def VGG13(num_classes=10, in_channels=3):
+    return VGG('VGG13', num_classes, in_channels, False)
+
+
+def VGG16(num_classes=10, in_channels=3):
+    return VGG('VGG16', num_classes, in_channels, False)
+
+
+def VGG19(num_classes=10, in_channels=3):
+    return VGG('VGG19', num_classes, in_channels, False)
+
+
+def VGG11_BN(num_classes=10, in_channels=3):
+    return VGG('VGG11', num_classes, in_channels, True)
+
+
+def VGG13_BN(num_classes=10, in_channels=3):
+    return VGG('VGG13', num_classes, in_channels, True)
+
+
+def VGG16_BN(num_classes=10, in_channels=3):
+    return VGG('VGG16', num_classes, in_channels, True)
+
+
+def VGG19_BN(num_classes=10, in_channels=3):
+    return VGG('VGG19', num_classes, in_channels, True)

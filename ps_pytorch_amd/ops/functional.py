@@ -1,0 +1,87 @@
+"""Python-facing ops: HIP kernels on GPU, torch reference path on CPU.
+
+Every GPU entry point calls require_lib() — a GPU host without the in-tree
+libps_hip.so fails loudly (no silent eager fallback). The CPU paths double
+as the numerics references for the kernel tests (tests/test_ops_*.py).
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from . import require_lib, dtype_tag, current_stream_ptr
+
+
+def _check_flat(t: torch.Tensor, name: str, n: int) -> None:
+    if not t.is_contiguous():
+        raise ValueError(f"{name} must be contiguous")
+    if t.numel() != n:
+        raise ValueError(f"{name} numel {t.numel()} != {n}")
+
+
+def fused_sgd_step(
+    w: torch.Tensor,            # f32 master weights (flat)
+    grad_sum: torch.Tensor,     # f32 or bf16 summed grads (flat)
+    momentum_buf: torch.Tensor, # f32 (flat)
+    lr: float,
+    momentum: float = 0.0,
+    weight_decay: float = 0.0,
+    grad_scale: float = 1.0,    # usually 1/num_aggregate
+    nesterov: bool = False,
+    wire_out: Optional[torch.Tensor] = None,  # optional f32/bf16 broadcast payload
+) -> None:
+    """w/m update + optional wire re-pack, semantics of ref src/optim/sgd.py:59-92
+    (momentum buffer m = mu*m + g; w -= lr*m; nesterov variant as torch)."""
+    n = w.numel()
+    _check_flat(w, "w", n)
+    _check_flat(grad_sum, "grad_sum", n)
+    _check_flat(momentum_buf, "momentum_buf", n)
+    if wire_out is not None:
+        _check_flat(wire_out, "wire_out", n)
+
+    if w.is_cuda:
+        if n % 4:
+            raise ValueError("flat buffers must be padded to a multiple of 4")
+        lib = require_lib()
+        lib.ps_fused_sgd(
+            w.data_ptr(), grad_sum.data_ptr(), momentum_buf.data_ptr(),
+            0 if wire_out is None else wire_out.data_ptr(), n,
+            float(lr), float(momentum), float(weight_decay), float(grad_scale),
+            int(nesterov), dtype_tag(grad_sum.dtype),
+            dtype_tag(wire_out.dtype) if wire_out is not None else 0,
+            current_stream_ptr())
+        return
+
+    # CPU reference path (also the golden model for the kernel test)
+    g = grad_sum.to(torch.float32) * grad_scale
+    if weight_decay:
+        g = g.add(w, alpha=weight_decay)
+    momentum_buf.mul_(momentum).add_(g)
+    upd = g.add(momentum_buf, alpha=momentum) if nesterov else momentum_buf
+    w.add_(upd, alpha=-lr)
+    if wire_out is not None:
+        wire_out.copy_(w.to(wire_out.dtype))
+
+
+def pack_wire(dst: torch.Tensor, src: torch.Tensor) -> None:
+    """Wire pack: f32 -> bf16 (the GPU 'compression' path, ref compression.py
+    g_compress role) or plain copy when dtypes match."""
+    n = src.numel()
+    _check_flat(dst, "dst", n)
+    if dst.dtype == src.dtype:
+        dst.copy_(src)
+        return
+    if src.is_cuda:
+        lib = require_lib()
+        if src.dtype == torch.float32 and dst.dtype == torch.bfloat16:
+            lib.ps_pack_bf16(dst.data_ptr(), src.data_ptr(), n, current_stream_ptr())
+        elif src.dtype == torch.bfloat16 and dst.dtype == torch.float32:
+            lib.ps_unpack_bf16(dst.data_ptr(), src.data_ptr(), n, current_stream_ptr())
+        else:
+            raise TypeError(f"unsupported pack {src.dtype} -> {dst.dtype}")
+        return
+    dst.copy_(src.to(dst.dtype))
+
+
+unpack_wire = pack_wire  # symmetric: direction decided by dtypes

@@ -1,0 +1,51 @@
+// Vectorized wire pack/unpack (f32 <-> bf16) for gradient/weight transport.
+//
+// Role-equivalent of the reference's blosc-snappy pack/unpack
+// (ref: src/compression.py:18-46): shrink bytes-on-wire before RCCL moves
+// them over xGMI. The MI355X-native scheme is GPU-resident dtype truncation
+// (f32 -> bf16, 2x) — the wire buffer never leaves HBM and the pack runs at
+// HBM stream rate, where blosc ran on the host CPU behind a D2H copy.
+//
+// Arbitrary n: 4-wide vector main loop + scalar tail on thread 0 (buffers
+// here are bucket slices at parameter boundaries, not always 16B-sized).
+#include "common.h"
+
+__global__ __launch_bounds__(256) void pack_bf16_kernel(
+    unsigned short* __restrict__ dst, const float* __restrict__ src, long n)
+{
+    EW_IDX
+    long nvec = n >> 2;
+    for (long i = gid; i < nvec; i += stride) {
+        float4_t v = ((const float4_t*)src)[i];
+        ushort4_t o = {f32_to_bf16(v.x), f32_to_bf16(v.y), f32_to_bf16(v.z), f32_to_bf16(v.w)};
+        ((unsigned long long*)dst)[i] = *(unsigned long long*)&o;
+    }
+    if (gid == 0)
+        for (long i = nvec << 2; i < n; ++i) dst[i] = f32_to_bf16(src[i]);
+}
+
+__global__ __launch_bounds__(256) void unpack_bf16_kernel(
+    float* __restrict__ dst, const unsigned short* __restrict__ src, long n)
+{
+    EW_IDX
+    long nvec = n >> 2;
+    for (long i = gid; i < nvec; i += stride) {
+        unsigned long long u = ((const unsigned long long*)src)[i];
+        ushort4_t s = *(ushort4_t*)&u;
+        ((float4_t*)dst)[i] = {bf16_to_f32(s.x), bf16_to_f32(s.y), bf16_to_f32(s.z), bf16_to_f32(s.w)};
+    }
+    if (gid == 0)
+        for (long i = nvec << 2; i < n; ++i) dst[i] = bf16_to_f32(src[i]);
+}
+
+extern "C" void ps_pack_bf16(void* dst, const void* src, long n, void* stream) {
+    int blocks; ew_grid(n / 4, 256, &blocks);
+    hipLaunchKernelGGL(pack_bf16_kernel, dim3(blocks), dim3(256), 0, (hipStream_t)stream,
+                       (unsigned short*)dst, (const float*)src, n);
+}
+
+extern "C" void ps_unpack_bf16(void* dst, const void* src, long n, void* stream) {
+    int blocks; ew_grid(n / 4, 256, &blocks);
+    hipLaunchKernelGGL(unpack_bf16_kernel, dim3(blocks), dim3(256), 0, (hipStream_t)stream,
+                       (float*)dst, (const unsigned short*)src, n);
+}

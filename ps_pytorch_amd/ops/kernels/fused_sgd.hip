@@ -1,0 +1,108 @@
+// Fused PS-side SGD/momentum update over the flat parameter buffer.
+//
+// MI355X-native replacement for the reference's PS optimizer step
+// (ref: src/optim/sgd.py:59-92 — per-layer numpy->torch copies + 3 torch ops
+// per tensor) and the weight re-pack for broadcast (ref:
+// sync_replicas_master_nn.py:218-225 w_compress per layer). One kernel pass:
+//
+//   g  = grad_sum[i] * scale  (+ wd * w)        scale = 1/num_aggregate
+//   m  = mu * m + g
+//   w -= lr * (nesterov ? g + mu*m : m)
+//   wire[i] = cast(w)          (optional: the next broadcast's payload)
+//
+// Memory-bound: reads w, m, g; writes w, m, wire — fusing avoids 3 extra
+// round trips over the ~45 MB (ResNet-18) flat buffer at ~6.3 TB/s HBM.
+// Vectorized 4 elements/lane (float4 / ushort4 paths), grid-stride.
+#include "common.h"
+
+template <typename GV, typename WIREV, bool NESTEROV, bool HAS_WIRE, bool HAS_WD>
+__global__ __launch_bounds__(256) void fused_sgd_kernel(
+    float* __restrict__ w, const void* __restrict__ g_, float* __restrict__ m,
+    void* __restrict__ wire_, long nvec, float lr, float mu, float wd, float scale)
+{
+    const GV* __restrict__ g = (const GV*)g_;
+    WIREV* __restrict__ wire = (WIREV*)wire_;
+    long stride = (long)gridDim.x * blockDim.x;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec; i += stride) {
+        float4_t wv = ((float4_t*)w)[i];
+        float4_t mv = ((float4_t*)m)[i];
+        GV gvr = g[i];
+        float4_t gv;
+        if constexpr (sizeof(GV) == 8) {   // bf16 grads
+            ushort4_t u = *(ushort4_t*)&gvr;
+            gv = {bf16_to_f32(u.x), bf16_to_f32(u.y), bf16_to_f32(u.z), bf16_to_f32(u.w)};
+        } else {
+            gv = *(float4_t*)&gvr;
+        }
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+            float gk = gv[k] * scale;
+            if constexpr (HAS_WD) gk = fmaf(wd, wv[k], gk);
+            float mk = fmaf(mu, mv[k], gk);
+            mv[k] = mk;
+            float upd = NESTEROV ? fmaf(mu, mk, gk) : mk;
+            wv[k] = fmaf(-lr, upd, wv[k]);
+        }
+        ((float4_t*)w)[i] = wv;
+        ((float4_t*)m)[i] = mv;
+        if constexpr (HAS_WIRE) {
+            if constexpr (sizeof(WIREV) == 8) {
+                ushort4_t o = {f32_to_bf16(wv.x), f32_to_bf16(wv.y),
+                               f32_to_bf16(wv.z), f32_to_bf16(wv.w)};
+                wire[i] = *(WIREV*)&o;
+            } else {
+                wire[i] = *(WIREV*)&wv;
+            }
+        }
+    }
+}
+
+struct U64x { unsigned long long v; };   // 4 x bf16
+struct F128x { float4_t v; };            // 4 x f32
+
+template <bool N, bool W, bool D>
+static void launch_sel(float* w, const void* g, float* m, void* wire, long nvec,
+                       float lr, float mu, float wd, float scale,
+                       int g_dtype, int wire_dtype, hipStream_t s, int blocks) {
+    dim3 grid(blocks), block(256);
+    if (g_dtype == PS_BF16) {
+        if (!W || wire_dtype == PS_BF16)
+            hipLaunchKernelGGL((fused_sgd_kernel<U64x, U64x, N, W, D>), grid, block, 0, s,
+                               w, g, m, wire, nvec, lr, mu, wd, scale);
+        else
+            hipLaunchKernelGGL((fused_sgd_kernel<U64x, F128x, N, W, D>), grid, block, 0, s,
+                               w, g, m, wire, nvec, lr, mu, wd, scale);
+    } else {
+        if (!W || wire_dtype == PS_BF16)
+            hipLaunchKernelGGL((fused_sgd_kernel<F128x, U64x, N, W, D>), grid, block, 0, s,
+                               w, g, m, wire, nvec, lr, mu, wd, scale);
+        else
+            hipLaunchKernelGGL((fused_sgd_kernel<F128x, F128x, N, W, D>), grid, block, 0, s,
+                               w, g, m, wire, nvec, lr, mu, wd, scale);
+    }
+}
+
+extern "C" void ps_fused_sgd(
+    void* w, const void* g, void* m, void* wire, long n,
+    float lr, float mu, float wd, float scale, int nesterov,
+    int g_dtype, int wire_dtype, void* stream)
+{
+    // n must be a multiple of 4 (the flat buffer is padded to 4 elems by the
+    // Python side); asserted there.
+    long nvec = n / 4;
+    int blocks; ew_grid(nvec, 256, &blocks);
+    hipStream_t s = (hipStream_t)stream;
+    bool has_wire = wire != nullptr;
+    bool has_wd = wd != 0.0f;
+    if (nesterov) {
+        if (has_wire) { if (has_wd) launch_sel<true, true, true>(  (float*)w, g, (float*)m, wire, nvec, lr, mu, wd, scale, g_dtype, wire_dtype, s, blocks);
+                        else        launch_sel<true, true, false>( (float*)w, g, (float*)m, wire, nvec, lr, mu, wd, scale, g_dtype, wire_dtype, s, blocks); }
+        else          { if (has_wd) launch_sel<true, false, true>( (float*)w, g, (float*)m, wire, nvec, lr, mu, wd, scale, g_dtype, wire_dtype, s, blocks);
+                        else        launch_sel<true, false, false>((float*)w, g, (float*)m, wire, nvec, lr, mu, wd, scale, g_dtype, wire_dtype, s, blocks); }
+    } else {
+        if (has_wire) { if (has_wd) launch_sel<false, true, true>(  (float*)w, g, (float*)m, wire, nvec, lr, mu, wd, scale, g_dtype, wire_dtype, s, blocks);
+                        else        launch_sel<false, true, false>( (float*)w, g, (float*)m, wire, nvec, lr, mu, wd, scale, g_dtype, wire_dtype, s, blocks); }
+        else          { if (has_wd) launch_sel<false, false, true>( (float*)w, g, (float*)m, wire, nvec, lr, mu, wd, scale, g_dtype, wire_dtype, s, blocks);
+                        else        launch_sel<false, false, false>((float*)w, g, (float*)m, wire, nvec, lr, mu, wd, scale, g_dtype, wire_dtype, s, blocks); }
+    }
+}

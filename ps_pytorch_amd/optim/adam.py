@@ -1,0 +1,73 @@
+"""PS-side Adam on the flat master buffer.
+
+Reference parity: src/optim/adam.py:38-95 (full Adam incl. amsgrad, consuming
+wire gradients). Flat-tensor torch ops today (a handful of fused elementwise
+passes over one contiguous buffer); candidate for a single fused HIP kernel
+like FlatSGD's.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+
+class FlatAdam:
+    def __init__(self, flat_w: torch.Tensor, lr: float = 1e-3,
+                 betas=(0.9, 0.999), eps: float = 1e-8,
+                 weight_decay: float = 0.0, amsgrad: bool = False):
+        if flat_w.dtype != torch.float32:
+            raise TypeError("master weights must be f32")
+        self.w = flat_w
+        self.lr = float(lr)
+        self.beta1, self.beta2 = float(betas[0]), float(betas[1])
+        self.eps = float(eps)
+        self.weight_decay = float(weight_decay)
+        self.amsgrad = bool(amsgrad)
+        self.t = 0
+        self.exp_avg = torch.zeros_like(flat_w)
+        self.exp_avg_sq = torch.zeros_like(flat_w)
+        self.max_exp_avg_sq = torch.zeros_like(flat_w) if amsgrad else None
+
+    @torch.no_grad()
+    def step(self, grad_sum: torch.Tensor, grad_scale: float = 1.0,
+             wire_out: Optional[torch.Tensor] = None) -> None:
+        self.t += 1
+        g = grad_sum.to(torch.float32)
+        if grad_scale != 1.0:
+            g = g * grad_scale
+        if self.weight_decay:
+            g = g.add(self.w, alpha=self.weight_decay)
+        self.exp_avg.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
+        self.exp_avg_sq.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
+        bc1 = 1 - self.beta1 ** self.t
+        bc2 = 1 - self.beta2 ** self.t
+        if self.amsgrad:
+            torch.maximum(self.max_exp_avg_sq, self.exp_avg_sq,
+                          out=self.max_exp_avg_sq)
+            denom = (self.max_exp_avg_sq / bc2).sqrt_().add_(self.eps)
+        else:
+            denom = (self.exp_avg_sq / bc2).sqrt_().add_(self.eps)
+        self.w.addcdiv_(self.exp_avg, denom, value=-self.lr / bc1)
+        if wire_out is not None:
+            wire_out.copy_(self.w.to(wire_out.dtype))
+
+    def state_dict(self) -> dict:
+        return {'t': self.t, 'exp_avg': self.exp_avg,
+                'exp_avg_sq': self.exp_avg_sq,
+                'max_exp_avg_sq': self.max_exp_avg_sq,
+                'lr': self.lr, 'betas': (self.beta1, self.beta2),
+                'eps': self.eps, 'weight_decay': self.weight_decay,
+                'amsgrad': self.amsgrad}
+
+    def load_state_dict(self, sd: dict) -> None:
+        self.t = sd['t']
+        self.exp_avg.copy_(sd['exp_avg'])
+        self.exp_avg_sq.copy_(sd['exp_avg_sq'])
+        if self.amsgrad and sd['max_exp_avg_sq'] is not None:
+            self.max_exp_avg_sq.copy_(sd['max_exp_avg_sq'])
+        self.lr = sd['lr']
+        self.beta1, self.beta2 = sd['betas']
+        self.eps = sd['eps']
+        self.weight_decay = sd['weight_decay']

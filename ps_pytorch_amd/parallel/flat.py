@@ -1,0 +1,135 @@
+"""Flat contiguous parameter/gradient storage with bucket framing.
+
+This is the MI355X-native replacement for the reference's per-(layer, worker)
+tagged-message protocol (ref: sync_replicas_master_nn.py:199-237 posts
+L x (W-1) Irecvs with tag 88+layer; distributed_worker.py:254-272 sends one
+message per layer). Instead, every rank lays out all parameters in ONE flat
+buffer in REVERSE parameter order (the approximate order backward produces
+gradients), partitions it into contiguous buckets, and moves whole buckets
+with RCCL collectives — the bucket id <-> offset table is agreed at init
+(deterministic bucket framing; SURVEY.md §7 "Hard parts").
+
+Reverse-order layout means bucket 0 holds the LAST layers' params, i.e. the
+first gradients ready during backward — so overlapped per-bucket reduction
+streams in backward order with zero gather/scatter copies (grads are views
+into the flat buffer; collectives operate on contiguous slices).
+"""
+from __future__ import annotations
+
+import dataclasses
+from typing import List, Optional, Sequence
+
+import torch
+import torch.nn as nn
+
+
+@dataclasses.dataclass
+class Bucket:
+    index: int
+    start: int          # element offset into the flat buffer
+    end: int            # exclusive
+    param_ids: List[int]  # indices into FlatSpace.params covered by this bucket
+
+    @property
+    def numel(self) -> int:
+        return self.end - self.start
+
+
+class FlatSpace:
+    """Owns flat weight/grad storage for a model; params become views.
+
+    - `flat_w`: one contiguous tensor holding every parameter (reverse order).
+    - `flat_g`: matching gradient buffer; `attach_grads()` points every
+      `p.grad` at its slice so autograd accumulates in place — the "pack"
+      step of the reference (GPU->CPU copy + blosc, distributed_worker.py:259)
+      becomes a no-op.
+    - Buckets partition [0, total) contiguously, each <= bucket_bytes except
+      when a single parameter exceeds it (then the bucket grows to hold it:
+      parameters never straddle buckets).
+    """
+
+    def __init__(self, model: nn.Module, bucket_bytes: int = 25 * 1024 * 1024,
+                 dtype: Optional[torch.dtype] = None,
+                 device: Optional[torch.device] = None):
+        named = list(model.named_parameters())
+        self.names: List[str] = [n for n, _ in named][::-1]     # reverse order
+        self.params: List[nn.Parameter] = [p for _, p in named][::-1]
+        if not self.params:
+            raise ValueError("model has no parameters")
+        p0 = self.params[0]
+        self.dtype = dtype or p0.dtype
+        self.device = device or p0.device
+        self.offsets: List[int] = []
+        total = 0
+        for p in self.params:
+            self.offsets.append(total)
+            total += p.numel()
+        self.total = total
+        # pad storage to a multiple of 4 elements: full-buffer HIP kernels
+        # (fused_sgd) run purely 16B-vectorized; pad grads stay zero.
+        self.padded = (total + 3) & ~3
+
+        self.flat_w = torch.zeros(self.padded, dtype=self.dtype, device=self.device)
+        # Re-home every parameter as a view of flat_w (keeps autograd/optimizer
+        # identity: p is still the same nn.Parameter object).
+        for p, off in zip(self.params, self.offsets):
+            self.flat_w[off:off + p.numel()].copy_(
+                p.data.detach().reshape(-1).to(self.dtype))
+            p.data = self.flat_w[off:off + p.numel()].view(p.shape)
+        self.flat_g = torch.zeros(self.padded, dtype=self.dtype, device=self.device)
+
+        self.buckets = self._partition(bucket_bytes)
+
+    def _partition(self, bucket_bytes: int) -> List[Bucket]:
+        elem = self.flat_w.element_size()
+        max_elems = max(1, bucket_bytes // elem)
+        buckets: List[Bucket] = []
+        cur_ids: List[int] = []
+        cur_start = 0
+        for i, (p, off) in enumerate(zip(self.params, self.offsets)):
+            end = off + p.numel()
+            cur_ids.append(i)
+            if end - cur_start >= max_elems:
+                buckets.append(Bucket(len(buckets), cur_start, end, cur_ids))
+                cur_ids = []
+                cur_start = end
+        if cur_ids:
+            buckets.append(Bucket(len(buckets), cur_start, self.total, cur_ids))
+        return buckets
+
+    # ---- gradient plumbing ----
+
+    def attach_grads(self) -> None:
+        """Point every p.grad at its flat_g slice (autograd accumulates +=)."""
+        for p, off in zip(self.params, self.offsets):
+            p.grad = self.flat_g[off:off + p.numel()].view(p.shape)
+
+    def zero_grads(self) -> None:
+        self.flat_g.zero_()
+
+    def grad_slice(self, b: Bucket) -> torch.Tensor:
+        return self.flat_g[b.start:b.end]
+
+    def weight_slice(self, b: Bucket) -> torch.Tensor:
+        return self.flat_w[b.start:b.end]
+
+    # ---- (de)serialization helpers ----
+
+    def load_flat(self, src: torch.Tensor) -> None:
+        """Copy a flat f32/bf16 vector (same layout) into the live params."""
+        self.flat_w.copy_(src.to(self.dtype))
+
+    def state_dict_from_flat(self, flat: Optional[torch.Tensor] = None) -> dict:
+        """Reconstruct a {name: tensor} mapping (parameters only) from a flat
+        vector in THIS space's layout — used for model_step_<k> checkpoints."""
+        flat = self.flat_w if flat is None else flat
+        out = {}
+        for name, p, off in zip(self.names, self.params, self.offsets):
+            out[name] = flat[off:off + p.numel()].view(p.shape).detach().clone()
+        return out
+
+    def layout_signature(self) -> Sequence:
+        """Deterministic layout descriptor; ranks compare it at init to agree
+        on bucket framing (replaces the reference's per-layer tag contract)."""
+        return [(n, tuple(p.shape), off) for n, p, off in
+                zip(self.names, self.params, self.offsets)]

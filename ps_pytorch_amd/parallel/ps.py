@@ -1,0 +1,113 @@
+"""Parameter-server role (rank 0): global model owner + fused update.
+
+Reference parity: src/sync_replicas_master_nn.py (SyncReplicasMaster_NN) —
+per step: announce + broadcast weights, fan-in gradients, average, update,
+checkpoint. Differences by design (MI355X-first):
+
+  * The PS is a GPU rank (the reference pinned it to CPU,
+    sync_replicas_master_nn.py:131): master weights/momentum live in HBM and
+    the whole update is ONE fused HIP kernel (ops/kernels/fused_sgd.hip)
+    that also re-packs the next broadcast payload.
+  * The Waitany-drain + per-layer += aggregation loop (:157-186, :239-241)
+    is replaced by RCCL reduce(sum)-to-root per bucket: the xGMI fabric does
+    the summation work in flight; the PS contributes zeros.
+  * Partial aggregation (--num-aggregate, :179-207): in 'collective' mode
+    every worker contributes and the average divides by W-1 (documented
+    deviation, exact when there are no stragglers to drop); 'gather' mode
+    (per-worker P2P buffers) restores arrival-order selection.
+"""
+from __future__ import annotations
+
+import time
+from typing import Optional
+
+import torch
+
+from ..config import JobConfig, input_shape_of, num_classes_of
+from ..models import build_model
+from ..optim import FlatAdam, FlatSGD
+from ..parallel.flat import FlatSpace
+from ..parallel.transport import PSTransport
+from ..utils.checkpoint import save_model_step
+from ..utils.logging import get_logger, MASTER_LINE
+
+logger = get_logger('ps_pytorch_amd.ps')
+
+
+class ParameterServer:
+    def __init__(self, cfg: JobConfig, rank: int, world: int,
+                 device: torch.device, optimizer: str = 'sgd'):
+        assert rank == 0, "the PS is rank 0"
+        self.cfg = cfg
+        self.rank = rank
+        self.world = world
+        self.device = device
+        self.optimizer_name = optimizer
+        self.compute_dtype = (torch.bfloat16
+                              if (device.type == 'cuda' and cfg.compute_dtype == 'bf16')
+                              else torch.float32)
+        self.wire_dtype = (torch.bfloat16
+                           if (cfg.compress or cfg.wire_dtype == 'bf16') and device.type == 'cuda'
+                           else torch.float32)
+        self.cur_step = 0
+        self.network = None
+        self.flat: Optional[FlatSpace] = None
+        self.transport: Optional[PSTransport] = None
+
+    def build_model(self, num_classes: Optional[int] = None) -> None:
+        cfg = self.cfg
+        nc = num_classes if num_classes is not None else num_classes_of(cfg.dataset)
+        in_ch = input_shape_of(cfg.dataset)[0]
+        torch.manual_seed(cfg.seed)   # identical init across ranks
+        net = build_model(cfg.network, num_classes=nc, in_channels=in_ch)
+        net = net.to(device=self.device, dtype=self.compute_dtype)
+        self.network = net
+        self.flat = FlatSpace(net, bucket_bytes=int(cfg.bucket_mb * 2 ** 20))
+        self.transport = PSTransport(self.flat, self.wire_dtype, self.device,
+                                     self.rank, self.world)
+        # f32 master copy + optimizer state in HBM
+        self.master_w = self.flat.flat_w.detach().to(torch.float32).clone()
+        if self.optimizer_name == 'adam':
+            self.optimizer = FlatAdam(self.master_w, lr=cfg.lr)
+        else:
+            self.optimizer = FlatSGD(self.master_w, lr=cfg.lr,
+                                     momentum=cfg.momentum)
+        # first broadcast payload
+        self.transport.pack_weights_from(self.master_w)
+
+    @property
+    def grad_scale(self) -> float:
+        if self.cfg.aggregation == 'collective':
+            return 1.0 / max(self.world - 1, 1)
+        return 1.0 / max(min(self.cfg.num_aggregate, self.world - 1), 1)
+
+    def step(self) -> None:
+        """One synchronous step (mirrors DistributedWorker.train_step order)."""
+        t = self.transport
+        t.broadcast_weights()
+        t.recv_buckets(self.flat.buckets)
+        t.wait_all()
+        # fused: average-scale + momentum + update + re-pack next payload
+        self.optimizer.step(t.wire_g, grad_scale=self.grad_scale,
+                            wire_out=t.wire_w)
+        self.cur_step += 1
+
+    def start(self) -> None:
+        """Run the PS loop for max_steps (ref: sync_replicas_master_nn.py:133)."""
+        cfg = self.cfg
+        while self.cur_step < cfg.max_steps:
+            t0 = time.time()
+            self.step()
+            if self.cur_step % cfg.log_interval == 0:
+                logger.info(MASTER_LINE.format(self.cur_step, time.time() - t0))
+            # the reference's master checkpoints non-BN nets
+            # (sync_replicas_master_nn.py:194-196); BN nets are saved by
+            # worker rank 1 (see worker.py).
+            if (self.cur_step % cfg.eval_freq == 0 and
+                    not any(k in cfg.network.lower() for k in ('resnet', 'vgg'))):
+                self._save_checkpoint()
+
+    def _save_checkpoint(self) -> None:
+        sd = self.flat.state_dict_from_flat(
+            self.master_w[:self.flat.total].to(self.flat.dtype))
+        save_model_step(sd, self.cfg.train_dir, self.cur_step)

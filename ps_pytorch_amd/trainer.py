@@ -1,0 +1,110 @@
+"""Single-machine trainer — the scalability yardstick and correctness oracle.
+
+Reference parity: src/nn_ops.py:29-106 (NN_Trainer.build_model /
+train_and_validate / validate) + src/single_machine.py. Uses the SAME engine
+pieces as the distributed path (FlatSpace views + FlatSGD fused update) so
+the 1-GPU bench measures the framework, not a different code path:
+master weights f32, compute dtype bf16 on GPU (f32 on CPU), one fused
+update kernel per step.
+"""
+from __future__ import annotations
+
+import time
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .config import JobConfig, num_classes_of, input_shape_of
+from .models import build_model
+from .optim import FlatSGD, FlatAdam
+from .parallel.flat import FlatSpace
+from .utils.metrics import accuracy
+from .utils.logging import get_logger
+
+logger = get_logger('ps_pytorch_amd.trainer')
+
+
+class NNTrainer:
+    def __init__(self, cfg: JobConfig, device: Optional[torch.device] = None):
+        self.cfg = cfg
+        if device is None:
+            device = torch.device('cuda' if (cfg.enable_gpu and torch.cuda.is_available())
+                                  else 'cpu')
+        self.device = device
+        self.compute_dtype = (torch.bfloat16 if (device.type == 'cuda' and
+                                                 cfg.compute_dtype == 'bf16')
+                              else torch.float32)
+        self.network: Optional[nn.Module] = None
+        self.flat: Optional[FlatSpace] = None
+        self.optimizer = None
+        self.cur_step = 0
+
+    def build_model(self, num_classes: Optional[int] = None) -> None:
+        cfg = self.cfg
+        nc = num_classes if num_classes is not None else num_classes_of(cfg.dataset)
+        in_ch = input_shape_of(cfg.dataset)[0]
+        torch.manual_seed(cfg.seed)
+        net = build_model(cfg.network, num_classes=nc, in_channels=in_ch)
+        net = net.to(device=self.device, dtype=self.compute_dtype)
+        self.network = net
+        self.flat = FlatSpace(net, bucket_bytes=int(cfg.bucket_mb * 2 ** 20))
+        # f32 master copy + fused update; flat_g is the "wire" (local = trivially summed)
+        self.master_w = self.flat.flat_w.detach().to(torch.float32).clone()
+        self.optimizer = FlatSGD(self.master_w, lr=cfg.lr, momentum=cfg.momentum)
+        self.flat.attach_grads()
+
+    def _loss(self, data, target):
+        out = self.network(data)
+        return F.cross_entropy(out.float(), target), out
+
+    def train_step(self, data, target) -> float:
+        """forward/backward/update; returns loss."""
+        self.flat.zero_grads()
+        loss, _ = self._loss(data, target)
+        loss.backward()
+        # fused: master update + re-pack into the live (possibly bf16) params
+        self.optimizer.step(self.flat.flat_g, grad_scale=1.0,
+                            wire_out=self.flat.flat_w)
+        self.cur_step += 1
+        return float(loss.detach())
+
+    def train_and_validate(self, train_loader, test_loader,
+                           max_steps: Optional[int] = None) -> None:
+        cfg = self.cfg
+        max_steps = max_steps or cfg.max_steps
+        self.network.train()
+        for epoch in range(cfg.epochs):
+            for batch_idx, (data, target) in enumerate(train_loader):
+                t0 = time.time()
+                data = data.to(self.device, self.compute_dtype)
+                target = target.to(self.device)
+                loss = self.train_step(data, target)
+                if self.cur_step % cfg.log_interval == 0:
+                    logger.info('Step: %d, Epoch: %d, Loss: %.4f, Time: %.4f',
+                                self.cur_step, epoch, loss, time.time() - t0)
+                if self.cur_step >= max_steps:
+                    return
+            self.validate(test_loader)
+
+    @torch.no_grad()
+    def validate(self, test_loader) -> float:
+        self.network.eval()
+        tot, p1_sum, p5_sum, loss_sum = 0, 0.0, 0.0, 0.0
+        for data, target in test_loader:
+            data = data.to(self.device, self.compute_dtype)
+            target = target.to(self.device)
+            out = self.network(data)
+            loss_sum += float(F.cross_entropy(out.float(), target,
+                                              reduction='sum'))
+            k = min(5, out.shape[1])
+            p1, pk = accuracy(out.float(), target, topk=(1, k))
+            bs = target.size(0)
+            p1_sum += float(p1) * bs
+            p5_sum += float(pk) * bs
+            tot += bs
+        self.network.train()
+        logger.info('Validation: loss %.4f, prec@1 %.2f, prec@5 %.2f',
+                    loss_sum / max(tot, 1), p1_sum / max(tot, 1), p5_sum / max(tot, 1))
+        return p1_sum / max(tot, 1)

@@ -1,0 +1,22 @@
+import os
+import sys
+
+import pytest
+import torch
+
+# repo root on sys.path so `import ps_pytorch_amd` works without install
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires an AMD GPU (MI355X); run with -m gpu")
+
+
+def pytest_collection_modifyitems(config, items):
+    if torch.cuda.is_available():
+        return
+    skip = pytest.mark.skip(reason="no GPU on this host")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip)

@@ -1,0 +1,45 @@
+"""Checkpoint layout + evaluator polling (ref: distributed_evaluator.py,
+model_step_<k> NFS layout — a declared compat surface)."""
+import os
+
+import torch
+
+from ps_pytorch_amd.config import JobConfig
+from ps_pytorch_amd.data import prepare_data
+from ps_pytorch_amd.evaluator import DistributedEvaluator
+from ps_pytorch_amd.models import build_model
+from ps_pytorch_amd.utils.checkpoint import (load_model_step, model_step_path,
+                                             save_model_step)
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    torch.manual_seed(0)
+    m = build_model('LeNet', in_channels=1)
+    path = save_model_step(m, str(tmp_path), 50)
+    assert path == model_step_path(str(tmp_path), 50)
+    assert os.path.basename(path) == 'model_step_50'
+    m2 = build_model('LeNet', in_channels=1)
+    load_model_step(m2, str(tmp_path), 50)
+    for a, b in zip(m.parameters(), m2.parameters()):
+        assert torch.equal(a, b)
+
+
+def test_evaluator_consumes_checkpoints(tmp_path):
+    cfg = JobConfig(network='LeNet', dataset='MNIST', eval_freq=10,
+                    max_steps=10, train_dir=str(tmp_path),
+                    test_batch_size=64)
+    torch.manual_seed(0)
+    m = build_model('LeNet', in_channels=1)
+    save_model_step(m, str(tmp_path), 10)
+    ev = DistributedEvaluator(cfg, poll_interval=0.01, max_polls=3)
+    _, test_loader = prepare_data(cfg, device=torch.device('cpu'),
+                                  test_size=128)
+    ev.evaluate(test_loader)   # consumes step 10 then stops at max_steps
+
+
+def test_bf16_model_checkpoint_saved_as_f32(tmp_path):
+    m = build_model('LeNet', in_channels=1).to(torch.bfloat16)
+    save_model_step(m, str(tmp_path), 1)
+    sd = torch.load(model_step_path(str(tmp_path), 1), weights_only=True)
+    assert all(v.dtype == torch.float32 for v in sd.values()
+               if torch.is_floating_point(v))

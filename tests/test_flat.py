@@ -1,0 +1,76 @@
+"""FlatSpace: view aliasing, bucket framing, grad attachment."""
+import torch
+import torch.nn.functional as F
+
+from ps_pytorch_amd.models import build_model
+from ps_pytorch_amd.parallel.flat import FlatSpace
+
+
+def test_views_alias_storage():
+    m = build_model('LeNet', in_channels=1)
+    fs = FlatSpace(m)
+    # mutating flat_w must mutate the live parameters
+    fs.flat_w.fill_(0.5)
+    for p in m.parameters():
+        assert torch.all(p.data == 0.5)
+    assert fs.total == 431080
+    assert fs.padded % 4 == 0
+
+
+def test_reverse_order_layout():
+    m = build_model('LeNet', in_channels=1)
+    fs = FlatSpace(m)
+    names = fs.names
+    # reverse parameter order: fc2 bias first, conv1 weight last
+    assert 'fc2' in names[0]
+    assert 'conv1' in names[-1]
+
+
+def test_bucket_partition_covers_everything():
+    m = build_model('ResNet18')
+    fs = FlatSpace(m, bucket_bytes=4 * 1024 * 1024)
+    assert fs.buckets[0].start == 0
+    assert fs.buckets[-1].end == fs.total
+    for a, b in zip(fs.buckets, fs.buckets[1:]):
+        assert a.end == b.start
+    covered = set()
+    for b in fs.buckets:
+        covered.update(b.param_ids)
+    assert covered == set(range(len(fs.params)))
+    assert len(fs.buckets) > 1
+
+
+def test_grad_attachment_accumulates_into_flat():
+    torch.manual_seed(0)
+    m = build_model('LeNet', in_channels=1)
+    fs = FlatSpace(m)
+    fs.attach_grads()
+    x = torch.randn(4, 1, 28, 28)
+    y = torch.randint(0, 10, (4,))
+    loss = F.cross_entropy(m(x), y)
+    loss.backward()
+    assert fs.flat_g.abs().sum() > 0
+    # p.grad views alias flat_g
+    for p, off in zip(fs.params, fs.offsets):
+        assert torch.equal(p.grad.reshape(-1), fs.flat_g[off:off + p.numel()])
+    fs.zero_grads()
+    assert fs.flat_g.abs().sum() == 0
+    for p in fs.params:
+        assert p.grad.abs().sum() == 0
+
+
+def test_state_dict_from_flat_roundtrip():
+    torch.manual_seed(0)
+    m = build_model('LeNet', in_channels=1)
+    ref = {k: v.clone() for k, v in m.state_dict().items()}
+    fs = FlatSpace(m)
+    sd = fs.state_dict_from_flat()
+    for k, v in sd.items():
+        assert torch.equal(v, ref[k]), k
+
+
+def test_layout_signature_deterministic():
+    torch.manual_seed(0)
+    a = FlatSpace(build_model('ResNet18'))
+    b = FlatSpace(build_model('ResNet18'))
+    assert a.layout_signature() == b.layout_signature()

@@ -1,0 +1,85 @@
+"""GPU (MI355X) tests: HIP kernel numerics vs the fp32 torch reference,
+and end-to-end engine smoke. All marked gpu; the driver runs them on a real
+gfx950 box."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope='module')
+def lib():
+    from ps_pytorch_amd.ops import require_lib
+    return require_lib()
+
+
+def test_pack_unpack_bf16_kernel(lib):
+    from ps_pytorch_amd.ops.functional import pack_wire
+    for n in (1024, 1000003, 7):    # vector path + scalar tails
+        src = torch.randn(n, device='cuda', dtype=torch.float32)
+        dst = torch.empty(n, device='cuda', dtype=torch.bfloat16)
+        pack_wire(dst, src)
+        torch.cuda.synchronize()
+        ref = src.to(torch.bfloat16)
+        assert torch.equal(dst, ref), n
+        # unpack back
+        up = torch.empty(n, device='cuda', dtype=torch.float32)
+        pack_wire(up, dst)
+        torch.cuda.synchronize()
+        assert torch.equal(up, dst.to(torch.float32))
+
+
+@pytest.mark.parametrize('g_dtype', [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize('nesterov', [False, True])
+def test_fused_sgd_kernel_vs_cpu_reference(lib, g_dtype, nesterov):
+    from ps_pytorch_amd.ops.functional import fused_sgd_step
+    torch.manual_seed(0)
+    n = 1 << 20
+    w = torch.randn(n)
+    g = torch.randn(n).to(g_dtype)
+    m = torch.randn(n).abs()
+    wire = torch.empty(n, dtype=torch.bfloat16)
+
+    w_d, g_d, m_d = w.cuda(), g.cuda(), m.cuda()
+    wire_d = wire.cuda()
+    for _ in range(3):
+        fused_sgd_step(w_d, g_d, m_d, lr=0.1, momentum=0.9,
+                       weight_decay=1e-4, grad_scale=1.0 / 7,
+                       nesterov=nesterov, wire_out=wire_d)
+    torch.cuda.synchronize()
+
+    for _ in range(3):
+        fused_sgd_step(w, g, m, lr=0.1, momentum=0.9, weight_decay=1e-4,
+                       grad_scale=1.0 / 7, nesterov=nesterov, wire_out=None)
+    assert torch.allclose(w_d.cpu(), w, atol=1e-5, rtol=1e-5)
+    assert torch.allclose(m_d.cpu(), m, atol=1e-5, rtol=1e-5)
+    assert torch.equal(wire_d.cpu(), w_d.cpu().to(torch.bfloat16))
+
+
+def test_single_gpu_train_step_smoke():
+    from ps_pytorch_amd.config import JobConfig
+    from ps_pytorch_amd.trainer import NNTrainer
+    cfg = JobConfig(network='ResNet18', dataset='Cifar10', batch_size=64,
+                    lr=0.1, momentum=0.9, enable_gpu=True)
+    tr = NNTrainer(cfg, device=torch.device('cuda', 0))
+    tr.build_model()
+    assert tr.compute_dtype == torch.bfloat16
+    x = torch.randn(64, 3, 32, 32, device='cuda', dtype=tr.compute_dtype)
+    y = torch.randint(0, 10, (64,), device='cuda')
+    losses = [tr.train_step(x, y) for _ in range(20)]
+    torch.cuda.synchronize()
+    assert losses[-1] < losses[0], losses  # memorizes one batch quickly
+
+
+def test_lenet_gpu_loss_decreases():
+    from ps_pytorch_amd.config import JobConfig
+    from ps_pytorch_amd.trainer import NNTrainer
+    cfg = JobConfig(network='LeNet', dataset='MNIST', batch_size=128,
+                    lr=0.05, momentum=0.9, enable_gpu=True)
+    tr = NNTrainer(cfg, device=torch.device('cuda', 0))
+    tr.build_model()
+    x = torch.randn(128, 1, 28, 28, device='cuda', dtype=tr.compute_dtype)
+    y = torch.randint(0, 10, (128,), device='cuda')
+    losses = [tr.train_step(x, y) for _ in range(20)]
+    torch.cuda.synchronize()
+    assert losses[-1] < losses[0] * 0.7, losses
