@@ -41,7 +41,10 @@ def make_batches(n_batches, bs, device, dtype, seed=1234):
     g = torch.Generator(device='cpu').manual_seed(seed)
     xs, ys = [], []
     for _ in range(n_batches):
-        xs.append(torch.randn(bs, 3, 32, 32, generator=g).to(device, dtype))
+        x = torch.randn(bs, 3, 32, 32, generator=g).to(device, dtype)
+        if device.type == 'cuda':
+            x = x.contiguous(memory_format=torch.channels_last)
+        xs.append(x)
         ys.append(torch.randint(0, 10, (bs,), generator=g).to(device))
     return xs, ys
 

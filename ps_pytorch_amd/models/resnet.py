@@ -2,15 +2,22 @@
 model zoo (ref: src/model_ops/resnet.py:14-113 — ResNet-18 = 62 param
 tensors / 11,173,962 params on CIFAR-10).
 
-For ImageNet-shaped inputs (224x224) the stem switches to 7x7/stride-2 +
-maxpool so ResNet-50 on synthetic 3x224x224 (BASELINE config 5) has the
-standard compute shape.
+MI355X-native structure: every BN is a PsBatchNorm2d with its ReLU (and the
+block's residual add) fused into the BN epilogue — on GPU these run as
+hand-written NHWC CDNA4 kernels (ops/kernels/batchnorm.hip); on CPU they
+fall back to torch ops with identical math.
+
+For ImageNet-shaped inputs (224x224, num_classes >= 200) the stem switches
+to 7x7/stride-2 + maxpool so ResNet-50 on synthetic 3x224x224 (BASELINE
+config 5) has the standard compute shape.
 """
 from __future__ import annotations
 
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
+
+from ..ops.modules import PsBatchNorm2d
 
 
 class BasicBlock(nn.Module):
@@ -19,21 +26,19 @@ class BasicBlock(nn.Module):
     def __init__(self, in_planes: int, planes: int, stride: int = 1):
         super().__init__()
         self.conv1 = nn.Conv2d(in_planes, planes, 3, stride=stride, padding=1, bias=False)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = PsBatchNorm2d(planes, relu=True)
         self.conv2 = nn.Conv2d(planes, planes, 3, stride=1, padding=1, bias=False)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = PsBatchNorm2d(planes, relu=True)   # fused: relu(bn + shortcut)
         self.shortcut = nn.Sequential()
         if stride != 1 or in_planes != planes * self.expansion:
             self.shortcut = nn.Sequential(
                 nn.Conv2d(in_planes, planes * self.expansion, 1, stride=stride, bias=False),
-                nn.BatchNorm2d(planes * self.expansion),
+                PsBatchNorm2d(planes * self.expansion),
             )
 
     def forward(self, x):
-        out = F.relu(self.bn1(self.conv1(x)))
-        out = self.bn2(self.conv2(out))
-        out = out + self.shortcut(x)
-        return F.relu(out)
+        out = self.bn1(self.conv1(x))
+        return self.bn2(self.conv2(out), residual=self.shortcut(x))
 
 
 class Bottleneck(nn.Module):
@@ -42,24 +47,22 @@ class Bottleneck(nn.Module):
     def __init__(self, in_planes: int, planes: int, stride: int = 1):
         super().__init__()
         self.conv1 = nn.Conv2d(in_planes, planes, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = PsBatchNorm2d(planes, relu=True)
         self.conv2 = nn.Conv2d(planes, planes, 3, stride=stride, padding=1, bias=False)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = PsBatchNorm2d(planes, relu=True)
         self.conv3 = nn.Conv2d(planes, planes * self.expansion, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(planes * self.expansion)
+        self.bn3 = PsBatchNorm2d(planes * self.expansion, relu=True)
         self.shortcut = nn.Sequential()
         if stride != 1 or in_planes != planes * self.expansion:
             self.shortcut = nn.Sequential(
                 nn.Conv2d(in_planes, planes * self.expansion, 1, stride=stride, bias=False),
-                nn.BatchNorm2d(planes * self.expansion),
+                PsBatchNorm2d(planes * self.expansion),
             )
 
     def forward(self, x):
-        out = F.relu(self.bn1(self.conv1(x)))
-        out = F.relu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
-        out = out + self.shortcut(x)
-        return F.relu(out)
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
+        return self.bn3(self.conv3(out), residual=self.shortcut(x))
 
 
 class ResNet(nn.Module):
@@ -72,7 +75,7 @@ class ResNet(nn.Module):
             self.conv1 = nn.Conv2d(in_channels, 64, 7, stride=2, padding=3, bias=False)
         else:
             self.conv1 = nn.Conv2d(in_channels, 64, 3, stride=1, padding=1, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
+        self.bn1 = PsBatchNorm2d(64, relu=True)
         self.layer1 = self._make_layer(block, 64, num_blocks[0], stride=1)
         self.layer2 = self._make_layer(block, 128, num_blocks[1], stride=2)
         self.layer3 = self._make_layer(block, 256, num_blocks[2], stride=2)
@@ -88,7 +91,7 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x):
-        out = F.relu(self.bn1(self.conv1(x)))
+        out = self.bn1(self.conv1(x))
         if self.imagenet_stem:
             out = F.max_pool2d(out, 3, stride=2, padding=1)
         out = self.layer1(out)

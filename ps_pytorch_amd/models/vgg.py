@@ -1,7 +1,10 @@
-"""VGG family for CIFAR-shaped inputs (ref: src/model_ops/vgg.py:15-68)."""
+"""VGG family for CIFAR-shaped inputs (ref: src/model_ops/vgg.py:15-68).
+BN variants use the fused PsBatchNorm2d (BN+ReLU in one NHWC kernel pass)."""
 from __future__ import annotations
 
 import torch.nn as nn
+
+from ..ops.modules import PsBatchNorm2d
 
 _CFG = {
     'VGG11': [64, 'M', 128, 'M', 256, 256, 'M', 512, 512, 'M', 512, 512, 'M'],
@@ -22,8 +25,9 @@ def _make_layers(cfg, in_channels: int, batch_norm: bool) -> nn.Sequential:
         else:
             layers.append(nn.Conv2d(c, v, 3, padding=1, bias=not batch_norm))
             if batch_norm:
-                layers.append(nn.BatchNorm2d(v))
-            layers.append(nn.ReLU(inplace=True))
+                layers.append(PsBatchNorm2d(v, relu=True))   # fused BN+ReLU
+            else:
+                layers.append(nn.ReLU(inplace=True))
             c = v
     layers.append(nn.AdaptiveAvgPool2d(1))
     return nn.Sequential(*layers)

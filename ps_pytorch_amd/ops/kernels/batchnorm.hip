@@ -1,0 +1,381 @@
+// Fused BatchNorm(+residual)(+ReLU) forward/backward for NHWC activations.
+//
+// Replaces torch's NCHW bf16 batch-norm kernels, which dominated the eager
+// baseline profile (58% of step time: profiles/r01_step01_*.md), and folds
+// the ResNet residual add + ReLU into the BN epilogue (removing the separate
+// elementwise add/clamp passes). Reference-role parity: the cuDNN/THNN
+// BN+ReLU ops the reference drives via nn.BatchNorm2d.
+//
+// Layout: x is [M][C] with C innermost (NHWC / channels_last), C % 8 == 0.
+// Stats/reductions accumulate in f32; io is bf16 or f32 (template).
+//
+// forward (training):
+//   k1 stats_partial : per-channel sum(x), sum(x^2) -> ws atomics
+//   k2 fwd_finalize  : mean/invstd, running-stat update, scale/shift
+//   k3 normalize     : y = relu(x*scale + shift + residual?)
+// forward (eval): k2' eval_finalize (from running stats) + k3.
+// backward:
+//   k4 bwd_reduce    : per-channel sum(dy_eff), sum(dy_eff * xhat)
+//   k5 bwd_finalize  : dgamma/dbeta + per-channel dx coefficients a,b,c
+//   k6 bwd_dx        : dx = a*dy_eff + b*(x-mean) + c ; dres = dy_eff
+// where dy_eff = relu ? (y>0 ? dy : 0) : dy.
+//
+// ws layout (f32): fwd [0:C] sum, [C:2C] sumsq, [2C:3C] scale, [3C:4C] shift
+//                  bwd [0:C] sum_dy, [C:2C] sum_dy_xhat,
+//                      [2C:3C] a, [3C:4C] b, [4C:5C] c
+#include "common.h"
+
+typedef float f32x8 __attribute__((ext_vector_type(8)));
+typedef unsigned short u16x8 __attribute__((ext_vector_type(8)));
+
+template <typename T> struct VecIO;
+template <> struct VecIO<unsigned short> {          // bf16
+    static __device__ __forceinline__ f32x8 load(const unsigned short* p) {
+        u16x8 v = *(const u16x8*)p;
+        f32x8 o;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) o[i] = bf16_to_f32(v[i]);
+        return o;
+    }
+    static __device__ __forceinline__ void store(unsigned short* p, f32x8 v) {
+        u16x8 o;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) o[i] = f32_to_bf16(v[i]);
+        *(u16x8*)p = o;
+    }
+};
+template <> struct VecIO<float> {
+    static __device__ __forceinline__ f32x8 load(const float* p) {
+        return *(const f32x8*)p;
+    }
+    static __device__ __forceinline__ void store(float* p, f32x8 v) {
+        *(f32x8*)p = v;
+    }
+};
+
+__global__ __launch_bounds__(256) void zero_ws_kernel(float* ws, int n) {
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n) ws[i] = 0.f;
+}
+
+// ---------------- k1: partial stats ----------------
+// thread t handles channel-octet cg = t % G (G = C/8) of row r = t / G;
+// block strides rows. LDS tree-reduce across the R rows in the block, then
+// one atomicAdd octet per (block, cg).
+template <typename T>
+__global__ __launch_bounds__(256) void bn_stats_kernel(
+    const T* __restrict__ x, float* __restrict__ ws, long M, int C)
+{
+    const int G = C >> 3;
+    const int R = 256 / G;                  // rows per block-iteration
+    const int cg = threadIdx.x % G;
+    const int row_in_blk = threadIdx.x / G;
+    __shared__ float s_sum[256][8];
+    __shared__ float s_sq[256][8];
+    f32x8 sum = {0, 0, 0, 0, 0, 0, 0, 0};
+    f32x8 sq = {0, 0, 0, 0, 0, 0, 0, 0};
+    long row0 = (long)blockIdx.x * R + row_in_blk;
+    long stride = (long)gridDim.x * R;
+    for (long r = row0; r < M; r += stride) {
+        f32x8 v = VecIO<T>::load(x + r * C + cg * 8);
+        sum += v;
+        sq += v * v;
+    }
+#pragma unroll
+    for (int i = 0; i < 8; ++i) { s_sum[threadIdx.x][i] = sum[i]; s_sq[threadIdx.x][i] = sq[i]; }
+    __syncthreads();
+    for (int s = R >> 1; s > 0; s >>= 1) {
+        if (row_in_blk < s) {
+            int a = threadIdx.x, b = (row_in_blk + s) * G + cg;
+#pragma unroll
+            for (int i = 0; i < 8; ++i) { s_sum[a][i] += s_sum[b][i]; s_sq[a][i] += s_sq[b][i]; }
+        }
+        __syncthreads();
+    }
+    if (row_in_blk == 0) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+            atomicAdd(&ws[cg * 8 + i], s_sum[threadIdx.x][i]);
+            atomicAdd(&ws[C + cg * 8 + i], s_sq[threadIdx.x][i]);
+        }
+    }
+}
+
+// ---------------- k2: forward finalize ----------------
+template <typename PT>   // param dtype (gamma/beta)
+__global__ __launch_bounds__(256) void bn_fwd_finalize_kernel(
+    float* __restrict__ ws, const PT* __restrict__ gamma,
+    const PT* __restrict__ beta, float* __restrict__ running_mean,
+    float* __restrict__ running_var, float* __restrict__ save_mean,
+    float* __restrict__ save_invstd, long M, int C, float momentum, float eps)
+{
+    int c = blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    float mean = ws[c] / (float)M;
+    float var = fmaxf(ws[C + c] / (float)M - mean * mean, 0.f);
+    float invstd = rsqrtf(var + eps);
+    save_mean[c] = mean;
+    save_invstd[c] = invstd;
+    if (running_mean) {
+        float unbiased = var * (float)M / (float)(M > 1 ? M - 1 : 1);
+        running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+        running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    }
+    float g = (float)gamma[c], b = (float)beta[c];
+    float scale = g * invstd;
+    ws[2 * C + c] = scale;
+    ws[3 * C + c] = b - mean * scale;
+}
+
+template <typename PT>
+__global__ __launch_bounds__(256) void bn_eval_finalize_kernel(
+    float* __restrict__ ws, const PT* __restrict__ gamma,
+    const PT* __restrict__ beta, const float* __restrict__ running_mean,
+    const float* __restrict__ running_var, int C, float eps)
+{
+    int c = blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    float invstd = rsqrtf(running_var[c] + eps);
+    float scale = (float)gamma[c] * invstd;
+    ws[2 * C + c] = scale;
+    ws[3 * C + c] = (float)beta[c] - running_mean[c] * scale;
+}
+
+// ---------------- k3: normalize (+residual)(+relu) ----------------
+template <typename T, bool RELU, bool RES>
+__global__ __launch_bounds__(256) void bn_normalize_kernel(
+    const T* __restrict__ x, T* __restrict__ y, const T* __restrict__ res,
+    const float* __restrict__ ws, long M, int C)
+{
+    const int G = C >> 3;
+    long nvec = M * G;
+    EW_IDX
+    for (long i = gid; i < nvec; i += stride) {
+        int cg = (int)(i % G);
+        f32x8 scale = *(const f32x8*)&ws[2 * C + cg * 8];
+        f32x8 shift = *(const f32x8*)&ws[3 * C + cg * 8];
+        f32x8 v = VecIO<T>::load(x + i * 8);
+        v = v * scale + shift;
+        if constexpr (RES) v += VecIO<T>::load(res + i * 8);
+        if constexpr (RELU)
+#pragma unroll
+            for (int k = 0; k < 8; ++k) v[k] = fmaxf(v[k], 0.f);
+        VecIO<T>::store(y + i * 8, v);
+    }
+}
+
+// ---------------- k4: backward reduce ----------------
+template <typename T, bool RELU>
+__global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
+    const T* __restrict__ x, const T* __restrict__ y, const T* __restrict__ dy,
+    const float* __restrict__ save_mean, const float* __restrict__ save_invstd,
+    float* __restrict__ ws, long M, int C)
+{
+    const int G = C >> 3;
+    const int R = 256 / G;
+    const int cg = threadIdx.x % G;
+    const int row_in_blk = threadIdx.x / G;
+    __shared__ float s_dy[256][8];
+    __shared__ float s_dyx[256][8];
+    f32x8 mean = *(const f32x8*)&save_mean[cg * 8];
+    f32x8 invstd = *(const f32x8*)&save_invstd[cg * 8];
+    f32x8 sum_dy = {0, 0, 0, 0, 0, 0, 0, 0};
+    f32x8 sum_dyx = {0, 0, 0, 0, 0, 0, 0, 0};
+    long row0 = (long)blockIdx.x * R + row_in_blk;
+    long stride = (long)gridDim.x * R;
+    for (long r = row0; r < M; r += stride) {
+        long o = r * C + cg * 8;
+        f32x8 d = VecIO<T>::load(dy + o);
+        if constexpr (RELU) {
+            f32x8 yo = VecIO<T>::load(y + o);
+#pragma unroll
+            for (int k = 0; k < 8; ++k) d[k] = yo[k] > 0.f ? d[k] : 0.f;
+        }
+        f32x8 xv = VecIO<T>::load(x + o);
+        sum_dy += d;
+        sum_dyx += d * (xv - mean) * invstd;
+    }
+#pragma unroll
+    for (int i = 0; i < 8; ++i) { s_dy[threadIdx.x][i] = sum_dy[i]; s_dyx[threadIdx.x][i] = sum_dyx[i]; }
+    __syncthreads();
+    for (int s = R >> 1; s > 0; s >>= 1) {
+        if (row_in_blk < s) {
+            int a = threadIdx.x, b = (row_in_blk + s) * G + cg;
+#pragma unroll
+            for (int i = 0; i < 8; ++i) { s_dy[a][i] += s_dy[b][i]; s_dyx[a][i] += s_dyx[b][i]; }
+        }
+        __syncthreads();
+    }
+    if (row_in_blk == 0) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+            atomicAdd(&ws[cg * 8 + i], s_dy[threadIdx.x][i]);
+            atomicAdd(&ws[C + cg * 8 + i], s_dyx[threadIdx.x][i]);
+        }
+    }
+}
+
+// ---------------- k5: backward finalize ----------------
+template <typename PT>
+__global__ __launch_bounds__(256) void bn_bwd_finalize_kernel(
+    float* __restrict__ ws, const PT* __restrict__ gamma,
+    const float* __restrict__ save_mean, const float* __restrict__ save_invstd,
+    PT* __restrict__ dgamma, PT* __restrict__ dbeta, long M, int C)
+{
+    int c = blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    float sum_dy = ws[c], sum_dyx = ws[C + c];
+    float invstd = save_invstd[c];
+    dgamma[c] = (PT)sum_dyx;
+    dbeta[c] = (PT)sum_dy;
+    float a = (float)gamma[c] * invstd;
+    ws[2 * C + c] = a;
+    ws[3 * C + c] = -a * invstd * sum_dyx / (float)M;   // b: coeff of (x-mean)
+    ws[4 * C + c] = -a * sum_dy / (float)M;             // c: constant
+}
+
+// ---------------- k6: backward dx (+dres) ----------------
+template <typename T, bool RELU, bool DRES>
+__global__ __launch_bounds__(256) void bn_bwd_dx_kernel(
+    const T* __restrict__ x, const T* __restrict__ y, const T* __restrict__ dy,
+    T* __restrict__ dx, T* __restrict__ dres,
+    const float* __restrict__ save_mean, const float* __restrict__ ws,
+    long M, int C)
+{
+    const int G = C >> 3;
+    long nvec = M * G;
+    EW_IDX
+    for (long i = gid; i < nvec; i += stride) {
+        int cg = (int)(i % G);
+        f32x8 mean = *(const f32x8*)&save_mean[cg * 8];
+        f32x8 a = *(const f32x8*)&ws[2 * C + cg * 8];
+        f32x8 b = *(const f32x8*)&ws[3 * C + cg * 8];
+        f32x8 c = *(const f32x8*)&ws[4 * C + cg * 8];
+        long o = i * 8;
+        f32x8 d = VecIO<T>::load(dy + o);
+        if constexpr (RELU) {
+            f32x8 yo = VecIO<T>::load(y + o);
+#pragma unroll
+            for (int k = 0; k < 8; ++k) d[k] = yo[k] > 0.f ? d[k] : 0.f;
+        }
+        if constexpr (DRES) VecIO<T>::store(dres + o, d);
+        f32x8 xv = VecIO<T>::load(x + o);
+        VecIO<T>::store(dx + o, a * d + b * (xv - mean) + c);
+    }
+}
+
+// ---------------- C API ----------------
+static inline int stats_blocks(long M, int C) {
+    int R = 256 / (C >> 3);
+    long want = (M + R - 1) / R;
+    return (int)(want < 768 ? (want > 0 ? want : 1) : 768);
+}
+
+template <typename T, typename PT>
+static void bn_fwd_t(const void* x, void* y, const void* gamma, const void* beta,
+                     void* rmean, void* rvar, void* smean, void* sinvstd,
+                     void* ws, const void* res, long M, long C,
+                     float momentum, float eps, int training, int relu,
+                     hipStream_t s)
+{
+    int Ci = (int)C;
+    float* wsf = (float*)ws;
+    dim3 b256(256);
+    if (training) {
+        hipLaunchKernelGGL(zero_ws_kernel, dim3((2 * Ci + 255) / 256), b256, 0, s, wsf, 2 * Ci);
+        hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(stats_blocks(M, Ci)), b256, 0, s,
+                           (const T*)x, wsf, M, Ci);
+        hipLaunchKernelGGL((bn_fwd_finalize_kernel<PT>), dim3((Ci + 255) / 256), b256, 0, s,
+                           wsf, (const PT*)gamma, (const PT*)beta, (float*)rmean,
+                           (float*)rvar, (float*)smean, (float*)sinvstd, M, Ci,
+                           momentum, eps);
+    } else {
+        hipLaunchKernelGGL((bn_eval_finalize_kernel<PT>), dim3((Ci + 255) / 256), b256, 0, s,
+                           wsf, (const PT*)gamma, (const PT*)beta,
+                           (const float*)rmean, (const float*)rvar, Ci, eps);
+    }
+    long nvec = M * (C >> 3);
+    int blocks; ew_grid(nvec, 256, &blocks);
+    if (relu) {
+        if (res) hipLaunchKernelGGL((bn_normalize_kernel<T, true, true>), dim3(blocks), b256, 0, s,
+                                    (const T*)x, (T*)y, (const T*)res, wsf, M, Ci);
+        else     hipLaunchKernelGGL((bn_normalize_kernel<T, true, false>), dim3(blocks), b256, 0, s,
+                                    (const T*)x, (T*)y, nullptr, wsf, M, Ci);
+    } else {
+        if (res) hipLaunchKernelGGL((bn_normalize_kernel<T, false, true>), dim3(blocks), b256, 0, s,
+                                    (const T*)x, (T*)y, (const T*)res, wsf, M, Ci);
+        else     hipLaunchKernelGGL((bn_normalize_kernel<T, false, false>), dim3(blocks), b256, 0, s,
+                                    (const T*)x, (T*)y, nullptr, wsf, M, Ci);
+    }
+}
+
+extern "C" void ps_bn_fwd(const void* x, void* y, const void* gamma,
+                          const void* beta, void* rmean, void* rvar,
+                          void* smean, void* sinvstd, void* ws, const void* res,
+                          long M, long C, float momentum, float eps,
+                          int training, int relu, int dtype, void* stream)
+{
+    hipStream_t s = (hipStream_t)stream;
+    if (dtype == PS_BF16)
+        bn_fwd_t<unsigned short, unsigned short>(x, y, gamma, beta, rmean, rvar,
+                                                 smean, sinvstd, ws, res, M, C,
+                                                 momentum, eps, training, relu, s);
+    else
+        bn_fwd_t<float, float>(x, y, gamma, beta, rmean, rvar, smean, sinvstd,
+                               ws, res, M, C, momentum, eps, training, relu, s);
+}
+
+template <typename T, typename PT>
+static void bn_bwd_t(const void* x, const void* y, const void* dy,
+                     const void* gamma, const void* smean, const void* sinvstd,
+                     void* dx, void* dgamma, void* dbeta, void* dres, void* ws,
+                     long M, long C, int relu, hipStream_t s)
+{
+    int Ci = (int)C;
+    float* wsf = (float*)ws;
+    dim3 b256(256);
+    hipLaunchKernelGGL(zero_ws_kernel, dim3((2 * Ci + 255) / 256), b256, 0, s, wsf, 2 * Ci);
+    if (relu)
+        hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, true>), dim3(stats_blocks(M, Ci)), b256, 0, s,
+                           (const T*)x, (const T*)y, (const T*)dy,
+                           (const float*)smean, (const float*)sinvstd, wsf, M, Ci);
+    else
+        hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), dim3(stats_blocks(M, Ci)), b256, 0, s,
+                           (const T*)x, (const T*)y, (const T*)dy,
+                           (const float*)smean, (const float*)sinvstd, wsf, M, Ci);
+    hipLaunchKernelGGL((bn_bwd_finalize_kernel<PT>), dim3((Ci + 255) / 256), b256, 0, s,
+                       wsf, (const PT*)gamma, (const float*)smean,
+                       (const float*)sinvstd, (PT*)dgamma, (PT*)dbeta, M, Ci);
+    long nvec = M * (C >> 3);
+    int blocks; ew_grid(nvec, 256, &blocks);
+    if (relu) {
+        if (dres) hipLaunchKernelGGL((bn_bwd_dx_kernel<T, true, true>), dim3(blocks), b256, 0, s,
+                                     (const T*)x, (const T*)y, (const T*)dy, (T*)dx, (T*)dres,
+                                     (const float*)smean, wsf, M, Ci);
+        else      hipLaunchKernelGGL((bn_bwd_dx_kernel<T, true, false>), dim3(blocks), b256, 0, s,
+                                     (const T*)x, (const T*)y, (const T*)dy, (T*)dx, nullptr,
+                                     (const float*)smean, wsf, M, Ci);
+    } else {
+        if (dres) hipLaunchKernelGGL((bn_bwd_dx_kernel<T, false, true>), dim3(blocks), b256, 0, s,
+                                     (const T*)x, (const T*)y, (const T*)dy, (T*)dx, (T*)dres,
+                                     (const float*)smean, wsf, M, Ci);
+        else      hipLaunchKernelGGL((bn_bwd_dx_kernel<T, false, false>), dim3(blocks), b256, 0, s,
+                                     (const T*)x, (const T*)y, (const T*)dy, (T*)dx, nullptr,
+                                     (const float*)smean, wsf, M, Ci);
+    }
+}
+
+extern "C" void ps_bn_bwd(const void* x, const void* y, const void* dy,
+                          const void* gamma, const void* smean,
+                          const void* sinvstd, void* dx, void* dgamma,
+                          void* dbeta, void* dres, void* ws, long M, long C,
+                          int relu, int dtype, void* stream)
+{
+    hipStream_t s = (hipStream_t)stream;
+    if (dtype == PS_BF16)
+        bn_bwd_t<unsigned short, unsigned short>(x, y, dy, gamma, smean, sinvstd,
+                                                 dx, dgamma, dbeta, dres, ws, M, C, relu, s);
+    else
+        bn_bwd_t<float, float>(x, y, dy, gamma, smean, sinvstd, dx, dgamma,
+                               dbeta, dres, ws, M, C, relu, s);
+}

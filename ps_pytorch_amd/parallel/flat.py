@@ -23,6 +23,19 @@ import torch
 import torch.nn as nn
 
 
+def prep_model(net: nn.Module, device: torch.device,
+               dtype: torch.dtype) -> nn.Module:
+    """Move a model to its compute placement, MI355X-style: on GPU, conv nets
+    go channels_last end-to-end (NHWC is the native layout for both the
+    MIOpen MFMA conv kernels and our fused BN kernels — the eager NCHW path
+    spent 10% of step time in batched_transpose, profiles/r01_step01)."""
+    net = net.to(device=device, dtype=dtype)
+    if device.type == 'cuda' and any(isinstance(m, nn.Conv2d)
+                                     for m in net.modules()):
+        net = net.to(memory_format=torch.channels_last)
+    return net
+
+
 @dataclasses.dataclass
 class Bucket:
     index: int
@@ -69,13 +82,23 @@ class FlatSpace:
         # (fused_sgd) run purely 16B-vectorized; pad grads stay zero.
         self.padded = (total + 3) & ~3
 
+        # Per-param storage layout: 4D channels_last params are stored in
+        # their (O,H,W,I) storage order and re-exposed as strided views, so
+        # the flat buffer IS the wire/kernel layout with zero copies. A param
+        # is treated channels_last iff its incoming data is (and is not plain
+        # contiguous, e.g. 1x1 kernels which are both).
+        self._cl: List[bool] = [
+            (p.data.dim() == 4
+             and p.data.is_contiguous(memory_format=torch.channels_last)
+             and not p.data.is_contiguous())
+            for p in self.params]
+
         self.flat_w = torch.zeros(self.padded, dtype=self.dtype, device=self.device)
         # Re-home every parameter as a view of flat_w (keeps autograd/optimizer
         # identity: p is still the same nn.Parameter object).
-        for p, off in zip(self.params, self.offsets):
-            self.flat_w[off:off + p.numel()].copy_(
-                p.data.detach().reshape(-1).to(self.dtype))
-            p.data = self.flat_w[off:off + p.numel()].view(p.shape)
+        for pid, (p, off) in enumerate(zip(self.params, self.offsets)):
+            self._view(self.flat_w, p, off, pid).copy_(p.data.detach().to(self.dtype))
+            p.data = self._view(self.flat_w, p, off, pid)
         self.flat_g = torch.zeros(self.padded, dtype=self.dtype, device=self.device)
 
         self.buckets = self._partition(bucket_bytes)
@@ -97,12 +120,21 @@ class FlatSpace:
             buckets.append(Bucket(len(buckets), cur_start, self.total, cur_ids))
         return buckets
 
+    def _view(self, flat: torch.Tensor, p: nn.Parameter, off: int,
+              pid: int) -> torch.Tensor:
+        """Slice of `flat` shaped/strided like p (honoring channels_last)."""
+        sl = flat[off:off + p.numel()]
+        if self._cl[pid]:
+            O, I, H, W = p.shape
+            return sl.view(O, H, W, I).permute(0, 3, 1, 2)
+        return sl.view(p.shape)
+
     # ---- gradient plumbing ----
 
     def attach_grads(self) -> None:
         """Point every p.grad at its flat_g slice (autograd accumulates +=)."""
-        for p, off in zip(self.params, self.offsets):
-            p.grad = self.flat_g[off:off + p.numel()].view(p.shape)
+        for pid, (p, off) in enumerate(zip(self.params, self.offsets)):
+            p.grad = self._view(self.flat_g, p, off, pid)
 
     def zero_grads(self) -> None:
         self.flat_g.zero_()
@@ -124,12 +156,13 @@ class FlatSpace:
         vector in THIS space's layout — used for model_step_<k> checkpoints."""
         flat = self.flat_w if flat is None else flat
         out = {}
-        for name, p, off in zip(self.names, self.params, self.offsets):
-            out[name] = flat[off:off + p.numel()].view(p.shape).detach().clone()
+        for pid, (name, p, off) in enumerate(zip(self.names, self.params,
+                                                 self.offsets)):
+            out[name] = self._view(flat, p, off, pid).detach().clone()
         return out
 
     def layout_signature(self) -> Sequence:
         """Deterministic layout descriptor; ranks compare it at init to agree
         on bucket framing (replaces the reference's per-layer tag contract)."""
-        return [(n, tuple(p.shape), off) for n, p, off in
-                zip(self.names, self.params, self.offsets)]
+        return [(n, tuple(p.shape), off, cl) for n, p, off, cl in
+                zip(self.names, self.params, self.offsets, self._cl)]

@@ -26,7 +26,7 @@ import torch
 from ..config import JobConfig, input_shape_of, num_classes_of
 from ..models import build_model
 from ..optim import FlatAdam, FlatSGD
-from ..parallel.flat import FlatSpace
+from ..parallel.flat import FlatSpace, prep_model
 from ..parallel.transport import PSTransport
 from ..utils.checkpoint import save_model_step
 from ..utils.logging import get_logger, MASTER_LINE
@@ -60,7 +60,7 @@ class ParameterServer:
         in_ch = input_shape_of(cfg.dataset)[0]
         torch.manual_seed(cfg.seed)   # identical init across ranks
         net = build_model(cfg.network, num_classes=nc, in_channels=in_ch)
-        net = net.to(device=self.device, dtype=self.compute_dtype)
+        net = prep_model(net, self.device, self.compute_dtype)
         self.network = net
         self.flat = FlatSpace(net, bucket_bytes=int(cfg.bucket_mb * 2 ** 20))
         self.transport = PSTransport(self.flat, self.wire_dtype, self.device,

@@ -20,7 +20,7 @@ import torch.nn.functional as F
 
 from ..config import JobConfig, input_shape_of, num_classes_of
 from ..models import build_model
-from ..parallel.flat import FlatSpace
+from ..parallel.flat import FlatSpace, prep_model
 from ..parallel.transport import PSTransport
 from ..utils.checkpoint import save_model_step
 from ..utils.logging import get_logger, worker_log_line
@@ -54,7 +54,7 @@ class DistributedWorker:
         in_ch = input_shape_of(cfg.dataset)[0]
         torch.manual_seed(cfg.seed)   # same init on every rank
         net = build_model(cfg.network, num_classes=nc, in_channels=in_ch)
-        net = net.to(device=self.device, dtype=self.compute_dtype)
+        net = prep_model(net, self.device, self.compute_dtype)
         self.network = net
         self.flat = FlatSpace(net, bucket_bytes=int(cfg.bucket_mb * 2 ** 20))
         self.flat.attach_grads()
@@ -122,6 +122,8 @@ class DistributedWorker:
 
     def train_step(self, data: torch.Tensor, target: torch.Tensor):
         """One synchronous PS step; returns detached loss tensor."""
+        if data.dim() == 4 and data.is_cuda:
+            data = data.contiguous(memory_format=torch.channels_last)
         self.fetch_weights()
         self.flat.zero_grads()
         self._reset_bucket_state()

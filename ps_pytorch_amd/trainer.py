@@ -19,7 +19,7 @@ import torch.nn.functional as F
 from .config import JobConfig, num_classes_of, input_shape_of
 from .models import build_model
 from .optim import FlatSGD, FlatAdam
-from .parallel.flat import FlatSpace
+from .parallel.flat import FlatSpace, prep_model
 from .utils.metrics import accuracy
 from .utils.logging import get_logger
 
@@ -47,7 +47,7 @@ class NNTrainer:
         in_ch = input_shape_of(cfg.dataset)[0]
         torch.manual_seed(cfg.seed)
         net = build_model(cfg.network, num_classes=nc, in_channels=in_ch)
-        net = net.to(device=self.device, dtype=self.compute_dtype)
+        net = prep_model(net, self.device, self.compute_dtype)
         self.network = net
         self.flat = FlatSpace(net, bucket_bytes=int(cfg.bucket_mb * 2 ** 20))
         # f32 master copy + fused update; flat_g is the "wire" (local = trivially summed)
@@ -61,6 +61,8 @@ class NNTrainer:
 
     def train_step(self, data, target) -> float:
         """forward/backward/update; returns loss."""
+        if data.dim() == 4 and data.is_cuda:
+            data = data.contiguous(memory_format=torch.channels_last)
         self.flat.zero_grads()
         loss, _ = self._loss(data, target)
         loss.backward()
