@@ -28,6 +28,15 @@
 typedef float f32x8 __attribute__((ext_vector_type(8)));
 typedef unsigned short u16x8 __attribute__((ext_vector_type(8)));
 
+// scalar param-dtype conversion (bf16 is carried as unsigned short — a
+// plain (float) cast would convert the BIT PATTERN, not the value)
+template <typename PT> __device__ __forceinline__ float pt_to_f32(PT v);
+template <> __device__ __forceinline__ float pt_to_f32<float>(float v) { return v; }
+template <> __device__ __forceinline__ float pt_to_f32<unsigned short>(unsigned short v) { return bf16_to_f32(v); }
+template <typename PT> __device__ __forceinline__ PT pt_from_f32(float v);
+template <> __device__ __forceinline__ float pt_from_f32<float>(float v) { return v; }
+template <> __device__ __forceinline__ unsigned short pt_from_f32<unsigned short>(float v) { return f32_to_bf16(v); }
+
 template <typename T> struct VecIO;
 template <> struct VecIO<unsigned short> {          // bf16
     static __device__ __forceinline__ f32x8 load(const unsigned short* p) {
@@ -121,7 +130,7 @@ __global__ __launch_bounds__(256) void bn_fwd_finalize_kernel(
         running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
         running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
     }
-    float g = (float)gamma[c], b = (float)beta[c];
+    float g = pt_to_f32<PT>(gamma[c]), b = pt_to_f32<PT>(beta[c]);
     float scale = g * invstd;
     ws[2 * C + c] = scale;
     ws[3 * C + c] = b - mean * scale;
@@ -136,9 +145,9 @@ __global__ __launch_bounds__(256) void bn_eval_finalize_kernel(
     int c = blockIdx.x * blockDim.x + threadIdx.x;
     if (c >= C) return;
     float invstd = rsqrtf(running_var[c] + eps);
-    float scale = (float)gamma[c] * invstd;
+    float scale = pt_to_f32<PT>(gamma[c]) * invstd;
     ws[2 * C + c] = scale;
-    ws[3 * C + c] = (float)beta[c] - running_mean[c] * scale;
+    ws[3 * C + c] = pt_to_f32<PT>(beta[c]) - running_mean[c] * scale;
 }
 
 // ---------------- k3: normalize (+residual)(+relu) ----------------
@@ -226,9 +235,9 @@ __global__ __launch_bounds__(256) void bn_bwd_finalize_kernel(
     if (c >= C) return;
     float sum_dy = ws[c], sum_dyx = ws[C + c];
     float invstd = save_invstd[c];
-    dgamma[c] = (PT)sum_dyx;
-    dbeta[c] = (PT)sum_dy;
-    float a = (float)gamma[c] * invstd;
+    dgamma[c] = pt_from_f32<PT>(sum_dyx);
+    dbeta[c] = pt_from_f32<PT>(sum_dy);
+    float a = pt_to_f32<PT>(gamma[c]) * invstd;
     ws[2 * C + c] = a;
     ws[3 * C + c] = -a * invstd * sum_dyx / (float)M;   // b: coeff of (x-mean)
     ws[4 * C + c] = -a * sum_dy / (float)M;             // c: constant
