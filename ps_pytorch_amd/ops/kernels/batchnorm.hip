@@ -85,7 +85,18 @@ __global__ __launch_bounds__(256) void bn_stats_kernel(
     f32x8 sq = {0, 0, 0, 0, 0, 0, 0, 0};
     long row0 = (long)blockIdx.x * R + row_in_blk;
     long stride = (long)gridDim.x * R;
-    for (long r = row0; r < M; r += stride) {
+    // 4-deep strided unroll: 4 independent loads in flight per wave
+    // (a single dependent load per iteration is HBM-latency-bound)
+    long r = row0;
+    for (; r + 3 * stride < M; r += 4 * stride) {
+        f32x8 v0 = VecIO<T>::load(x + r * C + cg * 8);
+        f32x8 v1 = VecIO<T>::load(x + (r + stride) * C + cg * 8);
+        f32x8 v2 = VecIO<T>::load(x + (r + 2 * stride) * C + cg * 8);
+        f32x8 v3 = VecIO<T>::load(x + (r + 3 * stride) * C + cg * 8);
+        sum += v0 + v1 + v2 + v3;
+        sq += v0 * v0 + v1 * v1 + v2 * v2 + v3 * v3;
+    }
+    for (; r < M; r += stride) {
         f32x8 v = VecIO<T>::load(x + r * C + cg * 8);
         sum += v;
         sq += v * v;
@@ -192,15 +203,32 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
     f32x8 sum_dyx = {0, 0, 0, 0, 0, 0, 0, 0};
     long row0 = (long)blockIdx.x * R + row_in_blk;
     long stride = (long)gridDim.x * R;
-    for (long r = row0; r < M; r += stride) {
+    // 2-deep strided unroll on top of the 2-3 independent streams per row
+    long r = row0;
+    for (; r + stride < M; r += 2 * stride) {
+#pragma unroll
+        for (int u = 0; u < 2; ++u) {
+            long o = (r + u * stride) * C + cg * 8;
+            f32x8 d = VecIO<T>::load(dy + o);
+            f32x8 xv = VecIO<T>::load(x + o);
+            if constexpr (RELU) {
+                f32x8 yo = VecIO<T>::load(y + o);
+#pragma unroll
+                for (int k = 0; k < 8; ++k) d[k] = yo[k] > 0.f ? d[k] : 0.f;
+            }
+            sum_dy += d;
+            sum_dyx += d * (xv - mean) * invstd;
+        }
+    }
+    for (; r < M; r += stride) {
         long o = r * C + cg * 8;
         f32x8 d = VecIO<T>::load(dy + o);
+        f32x8 xv = VecIO<T>::load(x + o);
         if constexpr (RELU) {
             f32x8 yo = VecIO<T>::load(y + o);
 #pragma unroll
             for (int k = 0; k < 8; ++k) d[k] = yo[k] > 0.f ? d[k] : 0.f;
         }
-        f32x8 xv = VecIO<T>::load(x + o);
         sum_dy += d;
         sum_dyx += d * (xv - mean) * invstd;
     }
@@ -277,7 +305,8 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_kernel(
 static inline int stats_blocks(long M, int C) {
     int R = 256 / (C >> 3);
     long want = (M + R - 1) / R;
-    return (int)(want < 768 ? (want > 0 ? want : 1) : 768);
+    long cap = 2048;   // 8 blocks/CU: max TLP for the latency-bound sweep
+    return (int)(want < cap ? (want > 0 ? want : 1) : cap);
 }
 
 template <typename T, typename PT>

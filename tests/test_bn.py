@@ -91,19 +91,30 @@ def test_gpu_kernel_vs_fp32_reference(C, HW, N, relu, with_res):
     yk.backward(g32.to(torch.bfloat16).contiguous(memory_format=cl))
     torch.cuda.synchronize()
 
-    def close(a, b, tol):
+    def close(a, b, tol, where=None):
         d = (a.float() - b.float()).abs()
+        if where is not None:
+            d = d * where
         s = b.float().abs().max().clamp(min=1)
         assert (d.max() / s) < tol, (d.max().item(), s.item())
 
+    # With relu, the mask is computed on bf16-rounded y while the reference
+    # masks on fp32 y: elements at the relu boundary legitimately flip.
+    # Require flips to be rare and compare gradients off-boundary.
+    if relu:
+        agree = ((yk.float() > 0) == (yr > 0)).float()
+        assert agree.mean() > 0.999, agree.mean()
+    else:
+        agree = None
+
     close(yk, yr, 2e-2)              # bf16 io => ~1e-2 relative
-    close(xk.grad, xr.grad, 3e-2)
-    close(m.weight.grad, ref_bn.weight.grad, 2e-2)
-    close(m.bias.grad, ref_bn.bias.grad, 2e-2)
+    close(xk.grad, xr.grad, 3e-2, agree)
+    close(m.weight.grad, ref_bn.weight.grad, 3e-2)
+    close(m.bias.grad, ref_bn.bias.grad, 3e-2)
     close(m.running_mean, ref_bn.running_mean, 2e-2)
     close(m.running_var, ref_bn.running_var, 2e-2)
     if with_res:
-        close(rk.grad, rr.grad, 2e-2)
+        close(rk.grad, rr.grad, 2e-2, agree)
 
 
 @pytest.mark.gpu

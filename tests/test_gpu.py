@@ -60,15 +60,17 @@ def test_single_gpu_train_step_smoke():
     from ps_pytorch_amd.config import JobConfig
     from ps_pytorch_amd.trainer import NNTrainer
     cfg = JobConfig(network='ResNet18', dataset='Cifar10', batch_size=64,
-                    lr=0.1, momentum=0.9, enable_gpu=True)
+                    lr=0.02, momentum=0.9, enable_gpu=True)
     tr = NNTrainer(cfg, device=torch.device('cuda', 0))
     tr.build_model()
     assert tr.compute_dtype == torch.bfloat16
+    torch.manual_seed(11)
     x = torch.randn(64, 3, 32, 32, device='cuda', dtype=tr.compute_dtype)
     y = torch.randint(0, 10, (64,), device='cuda')
-    losses = [tr.train_step(x, y) for _ in range(20)]
+    losses = [tr.train_step(x, y) for _ in range(30)]
     torch.cuda.synchronize()
-    assert losses[-1] < losses[0], losses  # memorizes one batch quickly
+    # memorizes one batch; min over the tail is robust to oscillation
+    assert min(losses[-10:]) < losses[0] * 0.5, losses
 
 
 def test_lenet_gpu_loss_decreases():
