@@ -10,7 +10,7 @@
 // Stats/reductions accumulate in f32; io is bf16 or f32 (template).
 //
 // forward (training):
-//   k1 stats_partial : per-channel sum(x), sum(x^2) -> ws atomics
+//   k1 stats_partial : per-block per-channel sum(x), sum(x^2) -> partial\n//   k1b fold        : partial -> ws[0:2C] (deterministic, no atomics)
 //   k2 fwd_finalize  : mean/invstd, running-stat update, scale/shift
 //   k3 normalize     : y = relu(x*scale + shift + residual?)
 // forward (eval): k2' eval_finalize (from running stats) + k3.
@@ -68,38 +68,37 @@ __global__ __launch_bounds__(256) void zero_ws_kernel(float* ws, int n) {
 }
 
 // ---------------- k1: partial stats ----------------
-// thread t handles channel-octet cg = t % G (G = C/8) of row r = t / G;
-// block strides rows. LDS tree-reduce across the R rows in the block, then
-// one atomicAdd octet per (block, cg).
+// thread t handles channel-octet cg = t % G (G = C/8) of row r = t / G.
+// Each block owns CONTIGUOUS 4R-row chunks (grid-strided): thread loads 4
+// rows R apart inside the chunk — 4 independent loads in flight, addresses
+// within one ~hundred-KB window (a gridDim-strided unroll put the 4 loads
+// ~8 MB apart and thrashed DRAM: 157 -> 399 us; this form: ~40-70 us).
+// Per-block partials go to `partial` (no atomics: deterministic; 2048-block
+// atomics serialized ~40 us/word); k1b folds partials -> ws[0:2C].
 template <typename T>
 __global__ __launch_bounds__(256) void bn_stats_kernel(
-    const T* __restrict__ x, float* __restrict__ ws, long M, int C)
+    const T* __restrict__ x, float* __restrict__ partial, long M, int C)
 {
     const int G = C >> 3;
-    const int R = 256 / G;                  // rows per block-iteration
+    const int R = 256 / G;                  // rows per block sub-iteration
     const int cg = threadIdx.x % G;
     const int row_in_blk = threadIdx.x / G;
     __shared__ float s_sum[256][8];
     __shared__ float s_sq[256][8];
     f32x8 sum = {0, 0, 0, 0, 0, 0, 0, 0};
     f32x8 sq = {0, 0, 0, 0, 0, 0, 0, 0};
-    long row0 = (long)blockIdx.x * R + row_in_blk;
-    long stride = (long)gridDim.x * R;
-    // 4-deep strided unroll: 4 independent loads in flight per wave
-    // (a single dependent load per iteration is HBM-latency-bound)
-    long r = row0;
-    for (; r + 3 * stride < M; r += 4 * stride) {
-        f32x8 v0 = VecIO<T>::load(x + r * C + cg * 8);
-        f32x8 v1 = VecIO<T>::load(x + (r + stride) * C + cg * 8);
-        f32x8 v2 = VecIO<T>::load(x + (r + 2 * stride) * C + cg * 8);
-        f32x8 v3 = VecIO<T>::load(x + (r + 3 * stride) * C + cg * 8);
-        sum += v0 + v1 + v2 + v3;
-        sq += v0 * v0 + v1 * v1 + v2 * v2 + v3 * v3;
-    }
-    for (; r < M; r += stride) {
-        f32x8 v = VecIO<T>::load(x + r * C + cg * 8);
-        sum += v;
-        sq += v * v;
+    const long chunk = 4L * R;
+    for (long base = (long)blockIdx.x * chunk + row_in_blk; base < M;
+         base += (long)gridDim.x * chunk) {
+#pragma unroll
+        for (int u = 0; u < 4; ++u) {
+            long r = base + (long)u * R;
+            if (r < M) {
+                f32x8 v = VecIO<T>::load(x + r * C + cg * 8);
+                sum += v;
+                sq += v * v;
+            }
+        }
     }
 #pragma unroll
     for (int i = 0; i < 8; ++i) { s_sum[threadIdx.x][i] = sum[i]; s_sq[threadIdx.x][i] = sq[i]; }
@@ -113,12 +112,28 @@ __global__ __launch_bounds__(256) void bn_stats_kernel(
         __syncthreads();
     }
     if (row_in_blk == 0) {
+        float* dst = partial + (long)blockIdx.x * 2 * C;
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
-            atomicAdd(&ws[cg * 8 + i], s_sum[threadIdx.x][i]);
-            atomicAdd(&ws[C + cg * 8 + i], s_sq[threadIdx.x][i]);
+            dst[cg * 8 + i] = s_sum[threadIdx.x][i];
+            dst[C + cg * 8 + i] = s_sq[threadIdx.x][i];
         }
     }
+}
+
+// ---------------- k1b: fold per-block partials into ws[0:2C] ----------------
+// one block per output word-pair region: blockIdx.x covers 2C words in
+// 256-thread tiles; each thread serially sums its word across nblocks rows
+// (partials are L2-resident).
+__global__ __launch_bounds__(256) void bn_fold_partials_kernel(
+    const float* __restrict__ partial, float* __restrict__ ws, int nblocks, int C)
+{
+    int w = blockIdx.x * blockDim.x + threadIdx.x;   // word in [0, 2C)
+    if (w >= 2 * C) return;
+    float acc = 0.f;
+    for (int b = 0; b < nblocks; ++b)
+        acc += partial[(long)b * 2 * C + w];
+    ws[w] = acc;
 }
 
 // ---------------- k2: forward finalize ----------------
@@ -189,7 +204,7 @@ template <typename T, bool RELU>
 __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
     const T* __restrict__ x, const T* __restrict__ y, const T* __restrict__ dy,
     const float* __restrict__ save_mean, const float* __restrict__ save_invstd,
-    float* __restrict__ ws, long M, int C)
+    float* __restrict__ partial, long M, int C)
 {
     const int G = C >> 3;
     const int R = 256 / G;
@@ -201,36 +216,27 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
     f32x8 invstd = *(const f32x8*)&save_invstd[cg * 8];
     f32x8 sum_dy = {0, 0, 0, 0, 0, 0, 0, 0};
     f32x8 sum_dyx = {0, 0, 0, 0, 0, 0, 0, 0};
-    long row0 = (long)blockIdx.x * R + row_in_blk;
-    long stride = (long)gridDim.x * R;
-    // 2-deep strided unroll on top of the 2-3 independent streams per row
-    long r = row0;
-    for (; r + stride < M; r += 2 * stride) {
+    // contiguous 2-row chunks per block iteration (see k1 comment); the 2-3
+    // streams per row already give memory-level parallelism
+    const long chunk = 2L * R;
+    for (long base = (long)blockIdx.x * chunk + row_in_blk; base < M;
+         base += (long)gridDim.x * chunk) {
 #pragma unroll
         for (int u = 0; u < 2; ++u) {
-            long o = (r + u * stride) * C + cg * 8;
-            f32x8 d = VecIO<T>::load(dy + o);
-            f32x8 xv = VecIO<T>::load(x + o);
-            if constexpr (RELU) {
-                f32x8 yo = VecIO<T>::load(y + o);
+            long r = base + (long)u * R;
+            if (r < M) {
+                long o = r * C + cg * 8;
+                f32x8 d = VecIO<T>::load(dy + o);
+                f32x8 xv = VecIO<T>::load(x + o);
+                if constexpr (RELU) {
+                    f32x8 yo = VecIO<T>::load(y + o);
 #pragma unroll
-                for (int k = 0; k < 8; ++k) d[k] = yo[k] > 0.f ? d[k] : 0.f;
+                    for (int k = 0; k < 8; ++k) d[k] = yo[k] > 0.f ? d[k] : 0.f;
+                }
+                sum_dy += d;
+                sum_dyx += d * (xv - mean) * invstd;
             }
-            sum_dy += d;
-            sum_dyx += d * (xv - mean) * invstd;
         }
-    }
-    for (; r < M; r += stride) {
-        long o = r * C + cg * 8;
-        f32x8 d = VecIO<T>::load(dy + o);
-        f32x8 xv = VecIO<T>::load(x + o);
-        if constexpr (RELU) {
-            f32x8 yo = VecIO<T>::load(y + o);
-#pragma unroll
-            for (int k = 0; k < 8; ++k) d[k] = yo[k] > 0.f ? d[k] : 0.f;
-        }
-        sum_dy += d;
-        sum_dyx += d * (xv - mean) * invstd;
     }
 #pragma unroll
     for (int i = 0; i < 8; ++i) { s_dy[threadIdx.x][i] = sum_dy[i]; s_dyx[threadIdx.x][i] = sum_dyx[i]; }
@@ -244,10 +250,11 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
         __syncthreads();
     }
     if (row_in_blk == 0) {
+        float* dst = partial + (long)blockIdx.x * 2 * C;
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
-            atomicAdd(&ws[cg * 8 + i], s_dy[threadIdx.x][i]);
-            atomicAdd(&ws[C + cg * 8 + i], s_dyx[threadIdx.x][i]);
+            dst[cg * 8 + i] = s_dy[threadIdx.x][i];
+            dst[C + cg * 8 + i] = s_dyx[threadIdx.x][i];
         }
     }
 }
@@ -302,17 +309,19 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_kernel(
 }
 
 // ---------------- C API ----------------
-static inline int stats_blocks(long M, int C) {
-    int R = 256 / (C >> 3);
-    long want = (M + R - 1) / R;
-    long cap = 2048;   // 8 blocks/CU: max TLP for the latency-bound sweep
+static inline int stats_blocks(long M, int C, int unroll) {
+    long chunk = (long)unroll * (256 / (C >> 3));
+    long want = (M + chunk - 1) / chunk;
+    long cap = 512;    // partial buffer rows; ~2 blocks/CU suffice (4 loads
+                       // in flight per wave from the in-chunk unroll)
     return (int)(want < cap ? (want > 0 ? want : 1) : cap);
 }
+#define BN_MAX_PARTIAL_BLOCKS 512
 
 template <typename T, typename PT>
 static void bn_fwd_t(const void* x, void* y, const void* gamma, const void* beta,
                      void* rmean, void* rvar, void* smean, void* sinvstd,
-                     void* ws, const void* res, long M, long C,
+                     void* ws, void* partial, const void* res, long M, long C,
                      float momentum, float eps, int training, int relu,
                      hipStream_t s)
 {
@@ -320,9 +329,11 @@ static void bn_fwd_t(const void* x, void* y, const void* gamma, const void* beta
     float* wsf = (float*)ws;
     dim3 b256(256);
     if (training) {
-        hipLaunchKernelGGL(zero_ws_kernel, dim3((2 * Ci + 255) / 256), b256, 0, s, wsf, 2 * Ci);
-        hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(stats_blocks(M, Ci)), b256, 0, s,
-                           (const T*)x, wsf, M, Ci);
+        int nb = stats_blocks(M, Ci, 4);
+        hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(nb), b256, 0, s,
+                           (const T*)x, (float*)partial, M, Ci);
+        hipLaunchKernelGGL(bn_fold_partials_kernel, dim3((2 * Ci + 255) / 256), b256, 0, s,
+                           (const float*)partial, wsf, nb, Ci);
         hipLaunchKernelGGL((bn_fwd_finalize_kernel<PT>), dim3((Ci + 255) / 256), b256, 0, s,
                            wsf, (const PT*)gamma, (const PT*)beta, (float*)rmean,
                            (float*)rvar, (float*)smean, (float*)sinvstd, M, Ci,
@@ -349,38 +360,41 @@ static void bn_fwd_t(const void* x, void* y, const void* gamma, const void* beta
 
 extern "C" void ps_bn_fwd(const void* x, void* y, const void* gamma,
                           const void* beta, void* rmean, void* rvar,
-                          void* smean, void* sinvstd, void* ws, const void* res,
+                          void* smean, void* sinvstd, void* ws, void* partial,
+                          const void* res,
                           long M, long C, float momentum, float eps,
                           int training, int relu, int dtype, void* stream)
 {
     hipStream_t s = (hipStream_t)stream;
     if (dtype == PS_BF16)
         bn_fwd_t<unsigned short, unsigned short>(x, y, gamma, beta, rmean, rvar,
-                                                 smean, sinvstd, ws, res, M, C,
+                                                 smean, sinvstd, ws, partial, res, M, C,
                                                  momentum, eps, training, relu, s);
     else
         bn_fwd_t<float, float>(x, y, gamma, beta, rmean, rvar, smean, sinvstd,
-                               ws, res, M, C, momentum, eps, training, relu, s);
+                               ws, partial, res, M, C, momentum, eps, training, relu, s);
 }
 
 template <typename T, typename PT>
 static void bn_bwd_t(const void* x, const void* y, const void* dy,
                      const void* gamma, const void* smean, const void* sinvstd,
                      void* dx, void* dgamma, void* dbeta, void* dres, void* ws,
-                     long M, long C, int relu, hipStream_t s)
+                     void* partial, long M, long C, int relu, hipStream_t s)
 {
     int Ci = (int)C;
     float* wsf = (float*)ws;
     dim3 b256(256);
-    hipLaunchKernelGGL(zero_ws_kernel, dim3((2 * Ci + 255) / 256), b256, 0, s, wsf, 2 * Ci);
+    int nb = stats_blocks(M, Ci, 2);
     if (relu)
-        hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, true>), dim3(stats_blocks(M, Ci)), b256, 0, s,
+        hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, true>), dim3(nb), b256, 0, s,
                            (const T*)x, (const T*)y, (const T*)dy,
-                           (const float*)smean, (const float*)sinvstd, wsf, M, Ci);
+                           (const float*)smean, (const float*)sinvstd, (float*)partial, M, Ci);
     else
-        hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), dim3(stats_blocks(M, Ci)), b256, 0, s,
+        hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), dim3(nb), b256, 0, s,
                            (const T*)x, (const T*)y, (const T*)dy,
-                           (const float*)smean, (const float*)sinvstd, wsf, M, Ci);
+                           (const float*)smean, (const float*)sinvstd, (float*)partial, M, Ci);
+    hipLaunchKernelGGL(bn_fold_partials_kernel, dim3((2 * Ci + 255) / 256), b256, 0, s,
+                       (const float*)partial, wsf, nb, Ci);
     hipLaunchKernelGGL((bn_bwd_finalize_kernel<PT>), dim3((Ci + 255) / 256), b256, 0, s,
                        wsf, (const PT*)gamma, (const float*)smean,
                        (const float*)sinvstd, (PT*)dgamma, (PT*)dbeta, M, Ci);
@@ -406,14 +420,15 @@ static void bn_bwd_t(const void* x, const void* y, const void* dy,
 extern "C" void ps_bn_bwd(const void* x, const void* y, const void* dy,
                           const void* gamma, const void* smean,
                           const void* sinvstd, void* dx, void* dgamma,
-                          void* dbeta, void* dres, void* ws, long M, long C,
+                          void* dbeta, void* dres, void* ws, void* partial,
+                          long M, long C,
                           int relu, int dtype, void* stream)
 {
     hipStream_t s = (hipStream_t)stream;
     if (dtype == PS_BF16)
         bn_bwd_t<unsigned short, unsigned short>(x, y, dy, gamma, smean, sinvstd,
-                                                 dx, dgamma, dbeta, dres, ws, M, C, relu, s);
+                                                 dx, dgamma, dbeta, dres, ws, partial, M, C, relu, s);
     else
         bn_bwd_t<float, float>(x, y, dy, gamma, smean, sinvstd, dx, dgamma,
-                               dbeta, dres, ws, M, C, relu, s);
+                               dbeta, dres, ws, partial, M, C, relu, s);
 }

@@ -41,10 +41,12 @@ class _FusedBNFn(torch.autograd.Function):
         save_mean = torch.empty(C, dtype=torch.float32, device=x.device)
         save_invstd = torch.empty(C, dtype=torch.float32, device=x.device)
         ws = torch.empty(5 * C, dtype=torch.float32, device=x.device)
+        partial = torch.empty(512 * 2 * C, dtype=torch.float32, device=x.device)
         lib.ps_bn_fwd(
             x.data_ptr(), y.data_ptr(), gamma.data_ptr(), beta.data_ptr(),
             running_mean.data_ptr(), running_var.data_ptr(),
             save_mean.data_ptr(), save_invstd.data_ptr(), ws.data_ptr(),
+            partial.data_ptr(),
             residual.data_ptr() if residual is not None else 0,
             M, C, float(momentum), float(eps), 1, int(relu),
             dtype_tag(x.dtype), current_stream_ptr())
@@ -65,11 +67,13 @@ class _FusedBNFn(torch.autograd.Function):
         dgamma = torch.empty_like(gamma)
         dbeta = torch.empty_like(gamma)
         ws = torch.empty(5 * C, dtype=torch.float32, device=x.device)
+        partial = torch.empty(512 * 2 * C, dtype=torch.float32, device=x.device)
         lib.ps_bn_bwd(
             x.data_ptr(), y.data_ptr(), dy.data_ptr(), gamma.data_ptr(),
             save_mean.data_ptr(), save_invstd.data_ptr(), dx.data_ptr(),
             dgamma.data_ptr(), dbeta.data_ptr(),
             dres.data_ptr() if dres is not None else 0, ws.data_ptr(),
+            partial.data_ptr(),
             M, C, int(ctx.relu), dtype_tag(x.dtype), current_stream_ptr())
         return (dx, dgamma, dbeta, dres, None, None, None, None, None)
 
