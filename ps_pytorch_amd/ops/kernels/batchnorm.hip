@@ -122,18 +122,25 @@ __global__ __launch_bounds__(256) void bn_stats_kernel(
 }
 
 // ---------------- k1b: fold per-block partials into ws[0:2C] ----------------
-// one block per output word-pair region: blockIdx.x covers 2C words in
-// 256-thread tiles; each thread serially sums its word across nblocks rows
-// (partials are L2-resident).
+// one 256-thread block per output WORD: threads stride the nblocks rows of
+// that word's column, LDS tree-reduce to one value. (A thread-per-word
+// version ran 2 waves on the whole chip summing 512 values serially —
+// 118 us; this one is a few us.)
 __global__ __launch_bounds__(256) void bn_fold_partials_kernel(
     const float* __restrict__ partial, float* __restrict__ ws, int nblocks, int C)
 {
-    int w = blockIdx.x * blockDim.x + threadIdx.x;   // word in [0, 2C)
-    if (w >= 2 * C) return;
+    __shared__ float s[256];
+    const int w = blockIdx.x;                 // word in [0, 2C)
     float acc = 0.f;
-    for (int b = 0; b < nblocks; ++b)
+    for (int b = threadIdx.x; b < nblocks; b += 256)
         acc += partial[(long)b * 2 * C + w];
-    ws[w] = acc;
+    s[threadIdx.x] = acc;
+    __syncthreads();
+    for (int k = 128; k > 0; k >>= 1) {
+        if ((int)threadIdx.x < k) s[threadIdx.x] += s[threadIdx.x + k];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) ws[w] = s[0];
 }
 
 // ---------------- k2: forward finalize ----------------
@@ -332,7 +339,7 @@ static void bn_fwd_t(const void* x, void* y, const void* gamma, const void* beta
         int nb = stats_blocks(M, Ci, 4);
         hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(nb), b256, 0, s,
                            (const T*)x, (float*)partial, M, Ci);
-        hipLaunchKernelGGL(bn_fold_partials_kernel, dim3((2 * Ci + 255) / 256), b256, 0, s,
+        hipLaunchKernelGGL(bn_fold_partials_kernel, dim3(2 * Ci), b256, 0, s,
                            (const float*)partial, wsf, nb, Ci);
         hipLaunchKernelGGL((bn_fwd_finalize_kernel<PT>), dim3((Ci + 255) / 256), b256, 0, s,
                            wsf, (const PT*)gamma, (const PT*)beta, (float*)rmean,
@@ -393,7 +400,7 @@ static void bn_bwd_t(const void* x, const void* y, const void* dy,
         hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), dim3(nb), b256, 0, s,
                            (const T*)x, (const T*)y, (const T*)dy,
                            (const float*)smean, (const float*)sinvstd, (float*)partial, M, Ci);
-    hipLaunchKernelGGL(bn_fold_partials_kernel, dim3((2 * Ci + 255) / 256), b256, 0, s,
+    hipLaunchKernelGGL(bn_fold_partials_kernel, dim3(2 * Ci), b256, 0, s,
                        (const float*)partial, wsf, nb, Ci);
     hipLaunchKernelGGL((bn_bwd_finalize_kernel<PT>), dim3((Ci + 255) / 256), b256, 0, s,
                        wsf, (const PT*)gamma, (const float*)smean,
