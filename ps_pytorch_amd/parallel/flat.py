@@ -33,6 +33,9 @@ def prep_model(net: nn.Module, device: torch.device,
     if device.type == 'cuda' and any(isinstance(m, nn.Conv2d)
                                      for m in net.modules()):
         net = net.to(memory_format=torch.channels_last)
+        # fixed shapes every step: let MIOpen search once for the best
+        # MFMA conv kernel per shape
+        torch.backends.cudnn.benchmark = True
     return net
 
 
