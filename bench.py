@@ -119,13 +119,15 @@ def main():
         tr = NNTrainer(cfg, device=device)
         tr.build_model(10)
         xs, ys = make_batches(8, args.batch_size, device, tr.compute_dtype)
+        graphed = use_cuda and tr.enable_graph(xs[0], ys[0])
+        step_fn = tr.graph_step if graphed else tr.train_step
         for i in range(args.warmup):
-            tr.train_step(xs[i % len(xs)], ys[i % len(ys)])
+            step_fn(xs[i % len(xs)], ys[i % len(ys)])
         if use_cuda:
             torch.cuda.synchronize()
         t0 = time.time()
         for i in range(args.steps):
-            tr.train_step(xs[i % len(xs)], ys[i % len(ys)])
+            step_fn(xs[i % len(xs)], ys[i % len(ys)])
         if use_cuda:
             torch.cuda.synchronize()
         elapsed = time.time() - t0
