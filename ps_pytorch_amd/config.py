@@ -41,7 +41,9 @@ def add_fit_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
     parser.add_argument('--network', type=str, default='LeNet', metavar='N',
                         help='LeNet | ResNet18/34/50/101/152 | VGG11/13/16/19[_BN]')
     parser.add_argument('--mode', type=str, default='normal', metavar='N',
-                        help='normal | kill : straggler handling mode')
+                        help='normal | kill | timeout : straggler handling '
+                             '(kill = PS signal on quota, ref backward_signal_kill; '
+                             'timeout = local --kill-threshold abort)')
     parser.add_argument('--kill-threshold', type=float, default=7.0, metavar='KT',
                         help='timeout threshold (s) that triggers straggler kill')
     parser.add_argument('--dataset', type=str, default='MNIST', metavar='N',

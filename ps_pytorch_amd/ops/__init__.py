@@ -46,6 +46,12 @@ def _try_load() -> None:
                                      ctypes.c_long, ctypes.c_void_p]
         lib.ps_unpack_bf16.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                        ctypes.c_long, ctypes.c_void_p]
+        lib.ps_pack_q8.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                   ctypes.c_long, ctypes.c_int, ctypes.c_void_p]
+        lib.ps_unpack_q8.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                     ctypes.c_long, ctypes.c_int, ctypes.c_void_p]
+        lib.ps_acc.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                               ctypes.c_long, ctypes.c_int, ctypes.c_void_p]
         lib.ps_bn_fwd.argtypes = [ctypes.c_void_p] * 11 + [
             ctypes.c_long, ctypes.c_long, ctypes.c_float, ctypes.c_float,
             ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p]

@@ -85,3 +85,84 @@ def pack_wire(dst: torch.Tensor, src: torch.Tensor) -> None:
 
 
 unpack_wire = pack_wire  # symmetric: direction decided by dtypes
+
+
+# ---- block-scaled int8 codec (the blosc-role GPU compressor) ----
+# Scheme/layout documented in ops/kernels/quant.hip: 256-elem blocks,
+# per-block f32 scale = max|x|/127, payload = [int8 x align4(n)][f32 x nblk].
+# The CPU path below is the numerics REFERENCE for the HIP kernels and the
+# gloo test path; both round identically (rint == torch.round, half-to-even).
+
+Q8_BLOCK = 256
+
+
+def q8_layout(n: int):
+    """(scales_byte_offset, total_payload_bytes) for n f32 values."""
+    qb = (n + 3) & ~3
+    nblk = (n + Q8_BLOCK - 1) // Q8_BLOCK
+    return qb, qb + 4 * nblk
+
+
+def pack_q8(payload: torch.Tensor, src: torch.Tensor) -> None:
+    """Compress flat f32 `src` into the uint8 `payload` buffer (4x vs f32)."""
+    n = src.numel()
+    qb, tot = q8_layout(n)
+    if payload.dtype != torch.uint8 or payload.numel() != tot:
+        raise ValueError(f"payload must be uint8[{tot}], got "
+                         f"{payload.dtype}[{payload.numel()}]")
+    if src.dtype not in (torch.float32, torch.bfloat16):
+        raise TypeError("q8 source must be f32 or bf16")
+    _check_flat(src, "src", n)
+    if src.is_cuda:
+        require_lib().ps_pack_q8(payload.data_ptr(), src.data_ptr(), n,
+                                 dtype_tag(src.dtype), current_stream_ptr())
+        return
+    nblk = (n + Q8_BLOCK - 1) // Q8_BLOCK
+    x = torch.zeros(nblk * Q8_BLOCK, dtype=torch.float32)
+    x[:n] = src.to(torch.float32)
+    xb = x.view(nblk, Q8_BLOCK)
+    m = xb.abs().amax(dim=1)
+    inv = torch.where(m > 0, 127.0 / m, torch.zeros_like(m))
+    q = torch.clamp(torch.round(xb * inv[:, None]), -127, 127).to(torch.int8)
+    payload[:n] = q.view(-1)[:n].view(torch.uint8)
+    payload[qb:qb + 4 * nblk].view(torch.float32).copy_(m / 127.0)
+
+
+def unpack_q8(dst: torch.Tensor, payload: torch.Tensor,
+              accumulate: bool = False) -> None:
+    """Decompress `payload` into f32 `dst` (dst += values if accumulate)."""
+    n = dst.numel()
+    qb, tot = q8_layout(n)
+    if payload.numel() != tot:
+        raise ValueError(f"payload size {payload.numel()} != {tot}")
+    if dst.dtype != torch.float32:
+        raise TypeError("q8 destination must be f32")
+    _check_flat(dst, "dst", n)
+    if dst.is_cuda:
+        require_lib().ps_unpack_q8(dst.data_ptr(), payload.data_ptr(), n,
+                                   int(accumulate), current_stream_ptr())
+        return
+    nblk = (n + Q8_BLOCK - 1) // Q8_BLOCK
+    q = payload[:n].view(torch.int8).to(torch.float32)
+    scales = payload[qb:qb + 4 * nblk].view(torch.float32)
+    vals = q * scales.repeat_interleave(Q8_BLOCK)[:n]
+    if accumulate:
+        dst.add_(vals)
+    else:
+        dst.copy_(vals)
+
+
+def acc_into(acc: torch.Tensor, src: torch.Tensor) -> None:
+    """acc += src (f32 acc; f32/bf16 src) — PS fan-in accumulate for
+    uncompressed gather mode (the reference's Waitany += loop,
+    sync_replicas_master_nn.py:157-186, on-device)."""
+    n = acc.numel()
+    _check_flat(acc, "acc", n)
+    _check_flat(src, "src", n)
+    if acc.dtype != torch.float32:
+        raise TypeError("accumulator must be f32")
+    if acc.is_cuda:
+        require_lib().ps_acc(acc.data_ptr(), src.data_ptr(), n,
+                             dtype_tag(src.dtype), current_stream_ptr())
+        return
+    acc.add_(src.to(torch.float32))

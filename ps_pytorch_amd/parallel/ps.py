@@ -27,7 +27,7 @@ from ..config import JobConfig, input_shape_of, num_classes_of
 from ..models import build_model
 from ..optim import FlatAdam, FlatSGD
 from ..parallel.flat import FlatSpace, prep_model
-from ..parallel.transport import PSTransport
+from ..parallel.transport import ControlPlane, PSTransport
 from ..utils.checkpoint import save_model_step
 from ..utils.logging import get_logger, MASTER_LINE
 
@@ -63,8 +63,15 @@ class ParameterServer:
         net = prep_model(net, self.device, self.compute_dtype)
         self.network = net
         self.flat = FlatSpace(net, bucket_bytes=int(cfg.bucket_mb * 2 ** 20))
+        # --compress-grad encodes per mode: gather -> block-scaled int8
+        # payloads (4x, quant.hip); collective -> bf16 wire (2x) since a
+        # summed int8 payload can't ride an in-flight ncclReduce.
         self.transport = PSTransport(self.flat, self.wire_dtype, self.device,
-                                     self.rank, self.world)
+                                     self.rank, self.world,
+                                     mode=cfg.aggregation,
+                                     compress=cfg.compress)
+        self.ctrl = (ControlPlane(self.rank, self.world)
+                     if cfg.mode == 'kill' else None)
         # f32 master copy + optimizer state in HBM
         self.master_w = self.flat.flat_w.detach().to(torch.float32).clone()
         if self.optimizer_name == 'adam':
@@ -85,10 +92,17 @@ class ParameterServer:
         """One synchronous step (mirrors DistributedWorker.train_step order)."""
         t = self.transport
         t.broadcast_weights()
-        t.recv_buckets(self.flat.buckets)
-        t.wait_all()
+        if t.mode == 'gather':
+            t.post_gather_recvs()
+            on_quota = self.ctrl.signal if self.ctrl is not None else None
+            t.drain_arrivals(self.cfg.num_aggregate, on_quota=on_quota)
+            grad = t.acc_g
+        else:
+            t.recv_buckets(self.flat.buckets)
+            t.wait_all()
+            grad = t.wire_g
         # fused: average-scale + momentum + update + re-pack next payload
-        self.optimizer.step(t.wire_g, grad_scale=self.grad_scale,
+        self.optimizer.step(grad, grad_scale=self.grad_scale,
                             wire_out=t.wire_w)
         self.cur_step += 1
 

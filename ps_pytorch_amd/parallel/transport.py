@@ -7,20 +7,32 @@ bucketed collectives on ONE flat buffer over `torch.distributed`
 
   step announce (tag 10)         -> implicit in the weight broadcast
   weight bcast  (tag 11+l, x L)  -> ONE ncclBroadcast of the flat wire buffer
-  grad push     (tag 88+l, x LW) -> per-bucket ncclReduce(sum) to rank 0,
-                                    issued in bucket order (matching by
-                                    program order replaces tag matching)
-  kill (tag 77) / control        -> small gloo side-channel (RCCL collectives
-                                    cannot be probed or cancelled)
+  grad push     (tag 88+l, x LW) -> two aggregation modes:
+     'collective': per-bucket ncclReduce(sum) to rank 0, issued in bucket
+        order (program-order matching replaces tags); xGMI sums in flight.
+     'gather': per-bucket ncclSend to rank 0 into per-worker staging
+        buffers; the PS accumulates IN ARRIVAL ORDER with HIP kernels
+        (ps_acc / ps_unpack_q8 acc) — this restores the reference's
+        Waitany-drain semantics (sync_replicas_master_nn.py:157-186) and
+        enables --num-aggregate first-k selection and the compressed
+        payload (summed payloads can't ride an in-flight reduction).
+  compression                    -> gather mode sends block-scaled int8
+        payloads (ops/kernels/quant.hip, 4x smaller than f32) — the GPU
+        re-expression of blosc g_compress (ref compression.py:18-46).
+  kill (tag 77) / control        -> gloo side-channel (ControlPlane below):
+        RCCL collectives cannot be probed or cancelled, so the straggler
+        kill signal travels host-side while payload matching stays intact
+        (killed workers send zero payloads for their remaining buckets).
 
 Weight broadcast from rank 0 can use all 7 outgoing xGMI links (every peer
-is one hop); gradient fan-in as reduce-to-root is bound by the PS's 7
+is one hop); gradient fan-in (either mode) is bound by the PS's 7
 incoming links (~1 TB/s aggregate) — see SURVEY.md §5.
 """
 from __future__ import annotations
 
 import datetime
 import os
+import time
 from typing import List, Optional
 
 import torch
@@ -60,21 +72,74 @@ class PSTransport:
       wire_w : flat weights as broadcast payload (wire dtype)
       wire_g : flat gradient accumulator in wire dtype (the reduce target on
                the PS; the reduce source on workers)
+    Gather mode adds:
+      payload   (worker) / stages[w] (PS) : per-bucket P2P gradient payloads
+               (q8-compressed bytes, or wire-dtype values)
+      acc_g (PS) : f32 arrival-order accumulator consumed by the optimizer
     """
 
     def __init__(self, flat: FlatSpace, wire_dtype: torch.dtype,
                  device: torch.device, rank: int, world: int,
-                 group: Optional[dist.ProcessGroup] = None):
+                 group: Optional[dist.ProcessGroup] = None,
+                 mode: str = 'collective', compress: bool = False):
+        if mode not in ('collective', 'gather'):
+            raise ValueError(f"unknown aggregation mode {mode!r}")
         self.flat = flat
         self.rank = rank
         self.world = world
         self.group = group
         self.device = device
         self.wire_dtype = wire_dtype
+        self.mode = mode
+        self.compress = bool(compress) and mode == 'gather'
         n = flat.padded
         self.wire_w = torch.zeros(n, dtype=wire_dtype, device=device)
         self.wire_g = torch.zeros(n, dtype=wire_dtype, device=device)
         self._works: List[dist.Work] = []
+
+        if mode == 'gather':
+            if self.compress:
+                # per-bucket q8 payload framing inside ONE byte buffer
+                self._pl_off: List[tuple] = []
+                cur = 0
+                for b in flat.buckets:
+                    _, tot = ops_f.q8_layout(b.numel)
+                    self._pl_off.append((cur, tot))
+                    cur += tot
+                self.payload_bytes = cur
+                if rank == PS_RANK:
+                    self.stages = [torch.zeros(cur, dtype=torch.uint8, device=device)
+                                   for _ in range(world - 1)]
+                else:
+                    self.payload = torch.zeros(cur, dtype=torch.uint8, device=device)
+                    self._zero_payload = torch.zeros(
+                        max(t for _, t in self._pl_off), dtype=torch.uint8,
+                        device=device)
+            else:
+                if rank == PS_RANK:
+                    self.stages = [torch.zeros(n, dtype=wire_dtype, device=device)
+                                   for _ in range(world - 1)]
+                else:
+                    self._zero_payload = torch.zeros(
+                        max(b.numel for b in flat.buckets), dtype=wire_dtype,
+                        device=device)
+            if rank == PS_RANK:
+                self.acc_g = torch.zeros(n, dtype=torch.float32, device=device)
+            self._pending: dict = {}     # (bucket_idx, worker) -> Work
+            if dist.is_initialized() and dist.get_backend(group) == 'nccl':
+                self._warmup_p2p()
+
+    def _warmup_p2p(self) -> None:
+        """Open every PS<->worker NCCL P2P channel in one deterministic
+        global order before the first real step (lazy per-pair communicator
+        creation in mismatched orders can deadlock)."""
+        t = torch.zeros(1, device=self.device)
+        for w in range(1, self.world):
+            if self.rank == PS_RANK:
+                dist.recv(t, src=w, group=self.group)
+            elif self.rank == w:
+                dist.send(t, dst=PS_RANK, group=self.group)
+        dist.barrier(group=self.group)
 
     @property
     def num_workers(self) -> int:
@@ -99,20 +164,59 @@ class PSTransport:
     def grad_wire_slice(self, b: Bucket) -> torch.Tensor:
         return self.wire_g[b.start:b.end]
 
-    def push_bucket(self, b: Bucket) -> dist.Work:
-        """Worker: pack this bucket's grads to wire dtype and reduce to PS.
-        Must be called in bucket-index order on every rank (program-order
-        matching)."""
-        src = self.flat.grad_slice(b)
-        wire = self.grad_wire_slice(b)
-        if wire.dtype != src.dtype:
-            ops_f.pack_wire(wire, src)
+    def push_bucket(self, b: Bucket, killed: bool = False) -> dist.Work:
+        """Worker: move this bucket's grads toward the PS. Must be called in
+        bucket-index order on every rank (program-order matching).
+
+        collective: pack to wire dtype, ncclReduce(sum) to rank 0.
+        gather:     pack (q8 or wire dtype), ncclSend to rank 0.
+        killed=True sends a zero payload without touching the grads (the
+        straggler-abort path: matching stays intact, compute is skipped)."""
+        if self.mode == 'collective':
+            src = self.flat.grad_slice(b)
+            wire = self.grad_wire_slice(b)
+            if killed:
+                wire.zero_()
+            elif wire.dtype != src.dtype:
+                ops_f.pack_wire(wire, src)
+            else:
+                wire.copy_(src)
+            w = dist.reduce(wire, dst=PS_RANK, op=dist.ReduceOp.SUM,
+                            group=self.group, async_op=True)
+        elif self.compress:
+            if killed:
+                payload = self._zero_payload[:self._pl_off[b.index][1]]
+            else:
+                payload = self._worker_payload_slice(b)
+                ops_f.pack_q8(payload, self.flat.grad_slice(b))
+            w = dist.isend(payload, dst=PS_RANK, group=self.group, tag=b.index)
         else:
-            wire.copy_(src)
-        w = dist.reduce(wire, dst=PS_RANK, op=dist.ReduceOp.SUM,
-                        group=self.group, async_op=True)
+            if killed:
+                wire = self._zero_payload[:b.numel]
+            else:
+                src = self.flat.grad_slice(b)
+                wire = self.grad_wire_slice(b)
+                if wire.dtype != src.dtype:
+                    ops_f.pack_wire(wire, src)
+                else:
+                    wire.copy_(src)
+            w = dist.isend(wire, dst=PS_RANK, group=self.group, tag=b.index)
         self._works.append(w)
         return w
+
+    def _worker_payload_slice(self, b: Bucket) -> torch.Tensor:
+        off, nbytes = self._pl_off[b.index]
+        return self.payload[off:off + nbytes]
+
+    def _stage_slice(self, worker: int, b: Bucket) -> torch.Tensor:
+        """PS: worker w's staging region for bucket b (w is 1-based rank)."""
+        st = self.stages[worker - 1]
+        if self.compress:
+            off, nbytes = self._pl_off[b.index]
+            return st[off:off + nbytes]
+        return st[b.start:b.end]
+
+    # -- collective mode, PS side --
 
     def recv_buckets(self, buckets: List[Bucket]) -> None:
         """PS: contribute zeros and post all reduces (async, in order)."""
@@ -123,6 +227,91 @@ class PSTransport:
                             async_op=True)
             self._works.append(w)
 
+    # -- gather mode, PS side --
+
+    def post_gather_recvs(self) -> None:
+        """PS: post one irecv per (bucket, worker). Posting order is
+        bucket-major to mirror workers' send order per peer."""
+        assert self.rank == PS_RANK and self.mode == 'gather'
+        for b in self.flat.buckets:
+            for w in range(1, self.world):
+                self._pending[(b.index, w)] = dist.irecv(
+                    self._stage_slice(w, b), src=w, group=self.group,
+                    tag=b.index)
+
+    def drain_arrivals(self, num_aggregate: int, on_quota=None) -> List[int]:
+        """PS: poll posted recvs; accumulate the FIRST `num_aggregate`
+        arrivals per bucket into acc_g (f32) in arrival order — the
+        reference's Waitany-drain (sync_replicas_master_nn.py:157-186) with
+        its --num-aggregate first-k selection (:179-207). Later arrivals are
+        drained (matching) but discarded. `on_quota()` fires once, the
+        moment every bucket has its quota while some recvs are still in
+        flight (the straggler-kill trigger). Returns per-bucket counts."""
+        assert self.rank == PS_RANK and self.mode == 'gather'
+        buckets = self.flat.buckets
+        k = max(1, min(num_aggregate, self.world - 1))
+        self.acc_g.zero_()
+        counts = [0] * len(buckets)
+        left = {w: len(buckets) for w in range(1, self.world)}
+        quota_fired = False
+        for bi, w in self._arrivals():
+            left[w] -= 1
+            if counts[bi] < k:
+                b = buckets[bi]
+                if self.compress:
+                    ops_f.unpack_q8(self.acc_g[b.start:b.end],
+                                    self._stage_slice(w, b), accumulate=True)
+                else:
+                    ops_f.acc_into(self.acc_g[b.start:b.end],
+                                   self._stage_slice(w, b))
+                counts[bi] += 1
+            if (on_quota is not None and not quota_fired
+                    and all(c >= k for c in counts)):
+                quota_fired = True
+                on_quota(sorted(w for w, n in left.items() if n > 0))
+        if on_quota is not None and not quota_fired:
+            on_quota([])
+        return counts
+
+    def _arrivals(self):
+        """Yield (bucket_idx, worker) in completion order, draining
+        self._pending. NCCL work objects expose real event-backed
+        is_completed() — poll them. Gloo only materializes completion
+        inside wait(), so one waiter thread per worker blocks through that
+        worker's buckets in FIFO send order and feeds a completion queue
+        (cross-worker arrival order is what the queue observes)."""
+        if dist.get_backend(self.group) == 'nccl':
+            while self._pending:
+                done = [key for key, wk in self._pending.items()
+                        if wk.is_completed()]
+                if not done:
+                    time.sleep(0)
+                    continue
+                for key in done:
+                    self._pending.pop(key)
+                    yield key
+            return
+        import queue as _queue
+        import threading
+        q: _queue.Queue = _queue.Queue()
+        by_worker: dict = {}
+        for (bi, w), wk in sorted(self._pending.items()):
+            by_worker.setdefault(w, []).append((bi, wk))
+        def waiter(w, items):
+            for bi, wk in items:
+                wk.wait()
+                q.put((bi, w))
+        threads = [threading.Thread(target=waiter, args=(w, items), daemon=True)
+                   for w, items in by_worker.items()]
+        for t in threads:
+            t.start()
+        total = len(self._pending)
+        self._pending.clear()
+        for _ in range(total):
+            yield q.get()
+        for t in threads:
+            t.join()
+
     def wait_all(self) -> None:
         for w in self._works:
             w.wait()
@@ -131,3 +320,83 @@ class PSTransport:
     def barrier(self) -> None:
         if dist.is_initialized():
             dist.barrier(group=self.group)
+
+
+class StepKilled(Exception):
+    """Raised inside a worker's backward to abort the rest of the step
+    (the reference's mid-backward straggler abort,
+    resnet_split.py:503-615 backward_signal_kill)."""
+
+
+class ControlPlane:
+    """Host-side gloo channel for the straggler-kill protocol.
+
+    The reference polls MPI tag 77 between layers of its hand-unrolled
+    backward (resnet_split.py:513-523 Iprobe). RCCL kernels can't be probed
+    or cancelled, so the signal travels on a gloo side-group: the PS sends
+    EXACTLY ONE byte to every worker per step — 1 as soon as every bucket
+    hit its --num-aggregate quota while that worker still has sends in
+    flight ("abort your remaining backward"), else 0 at step end. Workers
+    post the matching irecv at step start and poll it from autograd hooks;
+    the one-send/one-recv-per-step pairing keeps matching stall-free and
+    stale-signal-proof (no step-stamped tags needed: FIFO matching + the
+    end-of-step wait ensure a message can only belong to the current step,
+    the property resnet_split.py:25-42 generate_tag emulated)."""
+
+    KILL_TAG = 77
+
+    def __init__(self, rank: int, world: int):
+        self.rank = rank
+        self.world = world
+        self.group = dist.new_group(backend='gloo')
+        if rank == PS_RANK:
+            self._bufs = [torch.zeros(1, dtype=torch.uint8)
+                          for _ in range(world)]
+        else:
+            self._buf = torch.zeros(1, dtype=torch.uint8)
+            self._work: Optional[dist.Work] = None
+
+    # -- PS side --
+
+    def signal(self, kill_ranks) -> None:
+        """Send this step's verdict to every worker (1 = abort rest)."""
+        kill = set(kill_ranks)
+        works = []
+        for w in range(1, self.world):
+            self._bufs[w].fill_(1 if w in kill else 0)
+            works.append(dist.isend(self._bufs[w], dst=w, group=self.group,
+                                    tag=self.KILL_TAG))
+        for wk in works:
+            wk.wait()
+
+    # -- worker side --
+    # gloo only materializes completion inside wait() (is_completed() stays
+    # false), so a waiter thread blocks on the verdict and flips an Event
+    # the hooks can poll without blocking.
+
+    def post(self) -> None:
+        import threading
+        self._buf.zero_()
+        work = dist.irecv(self._buf, src=PS_RANK, group=self.group,
+                          tag=self.KILL_TAG)
+        self._evt = threading.Event()
+
+        def waiter(wk, evt):
+            wk.wait()
+            evt.set()
+
+        self._thread = threading.Thread(target=waiter, args=(work, self._evt),
+                                        daemon=True)
+        self._thread.start()
+
+    def killed(self) -> bool:
+        """Non-blocking poll (called from backward hooks)."""
+        return (getattr(self, '_evt', None) is not None
+                and self._evt.is_set() and int(self._buf) == 1)
+
+    def finish(self) -> None:
+        """End of step: consume this step's verdict (always arrives)."""
+        if getattr(self, '_thread', None) is not None:
+            self._thread.join()
+            self._thread = None
+            self._evt = None

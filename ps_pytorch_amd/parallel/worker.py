@@ -21,7 +21,7 @@ import torch.nn.functional as F
 from ..config import JobConfig, input_shape_of, num_classes_of
 from ..models import build_model
 from ..parallel.flat import FlatSpace, prep_model
-from ..parallel.transport import PSTransport
+from ..parallel.transport import ControlPlane, PSTransport, StepKilled
 from ..utils.checkpoint import save_model_step
 from ..utils.logging import get_logger, worker_log_line
 
@@ -59,7 +59,15 @@ class DistributedWorker:
         self.flat = FlatSpace(net, bucket_bytes=int(cfg.bucket_mb * 2 ** 20))
         self.flat.attach_grads()
         self.transport = PSTransport(self.flat, self.wire_dtype, self.device,
-                                     self.rank, self.world)
+                                     self.rank, self.world,
+                                     mode=cfg.aggregation,
+                                     compress=cfg.compress)
+        # straggler handling (ref resnet_split.py:503-728): 'kill' polls the
+        # PS's gloo signal from backward hooks; 'timeout' aborts locally past
+        # --kill-threshold seconds. Both raise StepKilled mid-backward and
+        # send zero payloads for the un-pushed buckets (matching intact).
+        self.ctrl = ControlPlane(self.rank, self.world) if cfg.mode == 'kill' else None
+        self._step_start = 0.0
         if cfg.overlap:
             self._install_hooks()
 
@@ -82,12 +90,23 @@ class DistributedWorker:
 
     def _make_hook(self, pid: int):
         def hook(_param):
+            self._check_abort()
             bi = self._param_bucket[pid]
             self._pending[bi] -= 1
             if self._pending[bi] == 0:
                 self._ready[bi] = True
                 self._flush_ready()
         return hook
+
+    def _check_abort(self) -> None:
+        """Raise StepKilled if the PS said stop (kill mode) or the step ran
+        past --kill-threshold (timeout mode)."""
+        if self.ctrl is not None and self.ctrl.killed():
+            raise StepKilled(f"rank {self.rank}: PS kill signal")
+        if (self.cfg.mode == 'timeout'
+                and time.time() - self._step_start > self.cfg.kill_threshold):
+            raise StepKilled(f"rank {self.rank}: step exceeded "
+                             f"{self.cfg.kill_threshold}s")
 
     def _flush_ready(self) -> None:
         while (self._next_launch < len(self._ready)
@@ -96,10 +115,10 @@ class DistributedWorker:
             self._next_launch += 1
 
     def _reset_bucket_state(self) -> None:
+        self._next_launch = 0
         if self.cfg.overlap:
             self._pending = list(self._bucket_nparams)
             self._ready = [False] * len(self.flat.buckets)
-            self._next_launch = 0
 
     # ---- per-step protocol (must mirror ParameterServer.step order) ----
 
@@ -107,32 +126,57 @@ class DistributedWorker:
         self.transport.broadcast_weights()
         self.transport.unpack_weights_into(self.flat.flat_w)
 
-    def push_gradients(self) -> None:
+    def push_gradients(self, killed: bool = False) -> None:
         if self.cfg.overlap:
-            self._flush_ready()
+            if not killed:
+                self._flush_ready()
             if self._next_launch < len(self.flat.buckets):
-                # params that never got grads (unused in graph): push anyway
+                # remaining buckets: zero payloads after an abort, else
+                # params that never got grads (unused in graph)
                 for b in self.flat.buckets[self._next_launch:]:
-                    self.transport.push_bucket(b)
+                    self.transport.push_bucket(b, killed=killed)
                 self._next_launch = len(self.flat.buckets)
         else:
-            for b in self.flat.buckets:
-                self.transport.push_bucket(b)
+            for b in self.flat.buckets[self._next_launch:]:
+                self.transport.push_bucket(b, killed=killed)
+            self._next_launch = len(self.flat.buckets)
         self.transport.wait_all()
 
+    def _forward_backward(self, data, target):
+        """forward + backward with the straggler-abort protocol around it.
+        Returns (loss_or_None, killed)."""
+        if self.ctrl is not None:
+            self.ctrl.post()
+        self._step_start = time.time()
+        self.f_dur = self.b_dur = 0.0
+        try:
+            t0 = time.time()
+            out = self.network(data)
+            loss = F.cross_entropy(out.float(), target)
+            self.f_dur = time.time() - t0
+            t0 = time.time()
+            loss.backward()
+            self.b_dur = time.time() - t0
+            if not self.cfg.overlap:
+                self._check_abort()
+            return loss.detach(), False
+        except StepKilled as e:
+            logger.info('%s — aborting rest of backward', e)
+            return None, True
+
     def train_step(self, data: torch.Tensor, target: torch.Tensor):
-        """One synchronous PS step; returns detached loss tensor."""
+        """One synchronous PS step; returns detached loss (None if killed)."""
         if data.dim() == 4 and data.is_cuda:
             data = data.contiguous(memory_format=torch.channels_last)
         self.fetch_weights()
         self.flat.zero_grads()
         self._reset_bucket_state()
-        out = self.network(data)
-        loss = F.cross_entropy(out.float(), target)
-        loss.backward()
-        self.push_gradients()
+        loss, killed = self._forward_backward(data, target)
+        self.push_gradients(killed=killed)
+        if self.ctrl is not None:
+            self.ctrl.finish()
         self.cur_step += 1
-        return loss.detach()
+        return loss
 
     # ---- training loop ----
 
@@ -153,23 +197,20 @@ class DistributedWorker:
                 fetch_dur = time.time() - t0
                 self.flat.zero_grads()
                 self._reset_bucket_state()
+                loss, killed = self._forward_backward(data, target)
                 t0 = time.time()
-                out = self.network(data)
-                loss = F.cross_entropy(out.float(), target)
-                f_dur = time.time() - t0
-                t0 = time.time()
-                loss.backward()
-                b_dur = time.time() - t0
-                t0 = time.time()
-                self.push_gradients()
+                self.push_gradients(killed=killed)
+                if self.ctrl is not None:
+                    self.ctrl.finish()
                 comm_dur = time.time() - t0
                 self.cur_step += 1
                 if self.cur_step % cfg.log_interval == 0:
                     logger.info(worker_log_line(
                         self.rank, self.cur_step, epoch,
-                        batch_idx * cfg.batch_size, n_total, float(loss),
-                        time.time() - iter_start, fetch_dur, f_dur, b_dur,
-                        comm_dur))
+                        batch_idx * cfg.batch_size, n_total,
+                        float('nan') if loss is None else float(loss),
+                        time.time() - iter_start, fetch_dur, self.f_dur,
+                        self.b_dur, comm_dur))
                 # checkpoint division of labor mirrors the reference
                 # (distributed_worker.py:175-177): rank 1 saves nets with BN
                 # buffers (ResNet/VGG) so running stats come from a worker.
