@@ -34,6 +34,10 @@ def parse():
     p.add_argument('--compress-grad', type=str, default='compress')
     p.add_argument('--bucket-mb', type=float, default=25.0)
     p.add_argument('--no-overlap', action='store_true')
+    p.add_argument('--aggregation', type=str, default='collective',
+                   help='collective | gather (PS engine fan-in mode)')
+    p.add_argument('--engine', type=str, default='ps',
+                   help='ps | allreduce (N>1 only)')
     return p.parse_args()
 
 
@@ -63,6 +67,7 @@ def main():
                     max_steps=args.steps + args.warmup,
                     compress_grad=args.compress_grad,
                     bucket_mb=args.bucket_mb, overlap=not args.no_overlap,
+                    aggregation=args.aggregation, engine=args.engine,
                     enable_gpu=use_cuda, eval_freq=10 ** 9,
                     train_dir='/tmp/ps_bench_models')
 
@@ -72,7 +77,19 @@ def main():
         device = env['device']
         from ps_pytorch_amd.parallel.ps import ParameterServer
         from ps_pytorch_amd.parallel.worker import DistributedWorker
-        if rank == 0:
+        if args.engine == 'allreduce':
+            from ps_pytorch_amd.parallel.allreduce import AllReduceTrainer
+            role = AllReduceTrainer(cfg, rank, world, device)
+            role.build_model(10)
+            xs, ys = make_batches(8, args.batch_size, device,
+                                  role.compute_dtype, seed=1234 + rank)
+            it = [0]
+
+            def step():
+                i = it[0] % len(xs)
+                it[0] += 1
+                role.train_step(xs[i], ys[i])
+        elif rank == 0:
             role = ParameterServer(cfg, rank, world, device)
             role.build_model(10)
             step = role.step
@@ -109,9 +126,13 @@ def main():
                           device=device if env['backend'] == 'nccl' else 'cpu')
         dist.all_reduce(et, op=dist.ReduceOp.MAX)
         elapsed = float(et)
-        n_workers = world - 1
+        if args.engine == 'allreduce':
+            n_workers = world
+            parallelism = f"allreduce-dp ({world} ranks)"
+        else:
+            n_workers = world - 1
+            parallelism = f"ps-dp (1 PS + {n_workers} workers)"
         imgs = n_workers * args.batch_size * args.steps
-        parallelism = f"ps-dp (1 PS + {n_workers} workers)"
         global_batch = n_workers * args.batch_size
     else:
         from ps_pytorch_amd.trainer import NNTrainer

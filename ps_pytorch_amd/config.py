@@ -73,6 +73,11 @@ def add_fit_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
     parser.add_argument('--aggregation', type=str, default='collective',
                         help='collective (reduce-to-root) | gather (per-worker P2P, '
                              'enables arrival-order --num-aggregate selection)')
+    parser.add_argument('--engine', type=str, default='ps',
+                        help='ps (1 PS + N-1 workers, the reference design) | '
+                             'allreduce (collective DP on all N ranks — the '
+                             'reference\'s vendored data_parallel_dist.py, built '
+                             'properly)')
     return parser
 
 
@@ -108,6 +113,7 @@ class JobConfig:
     bucket_mb: float = 25.0
     overlap: bool = True
     aggregation: str = 'collective'
+    engine: str = 'ps'
 
     @property
     def compress(self) -> bool:

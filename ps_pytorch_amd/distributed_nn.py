@@ -30,6 +30,22 @@ def main(argv=None) -> None:
         raise SystemExit("distributed_nn needs world_size >= 2 "
                          "(1 PS + >=1 worker); use single_machine.py for 1 rank")
     nc = num_classes_of(cfg.dataset)
+    if cfg.engine == 'allreduce':
+        from .parallel.allreduce import AllReduceTrainer
+        tr = AllReduceTrainer(cfg, rank, world, device)
+        tr.build_model(nc)
+        train_loader, _ = prepare_data(cfg, rank=rank, num_shards=world,
+                                       device=device, dtype=tr.compute_dtype)
+        step = 0
+        while step < cfg.max_steps:
+            for data, target in train_loader:
+                if step >= cfg.max_steps:
+                    break
+                loss = tr.train_step(data, target)
+                step += 1
+                if rank == 0 and step % cfg.log_interval == 0:
+                    logger.info('AllReduce step %d loss %.4f', step, float(loss))
+        return
     if rank == 0:
         ps = ParameterServer(cfg, rank, world, device)
         ps.build_model(nc)
