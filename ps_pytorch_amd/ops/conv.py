@@ -1,0 +1,104 @@
+"""PsConv2d — NHWC implicit-GEMM MFMA convolution (ops/kernels/conv.hip).
+
+Replaces the MIOpen path the reference reaches through nn.Conv2d
+(ref: src/model_ops/resnet.py:19-97, lenet.py:19-33). GPU training path is
+the hand-written CDNA4 kernel set (fwd / dgrad / split-K wgrad, fp32
+accumulate, deterministic reduce); CPU and unsupported shapes fall back to
+torch's conv, which doubles as the numerics reference in tests.
+
+PsConv2d subclasses nn.Conv2d, so parameter shapes, init and the
+state_dict surface are identical (evaluator/checkpoint compatible).
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from . import require_lib, current_stream_ptr
+
+_CL = torch.channels_last
+
+
+def _supported(x: torch.Tensor, w: torch.Tensor, stride, padding,
+               dilation, groups) -> bool:
+    if not (x.is_cuda and x.dim() == 4 and x.dtype == torch.bfloat16):
+        return False
+    if groups != 1 or dilation[0] != 1 or dilation[1] != 1:
+        return False
+    if stride[0] != stride[1] or padding[0] != padding[1]:
+        return False
+    R, S = w.shape[2], w.shape[3]
+    return R <= 7 and S <= 7
+
+
+def _wgrad_split(M: int, K: int, C: int, R: int, S: int) -> int:
+    """Pick the split-K factor: enough blocks to fill 256 CUs (~8 blocks/CU)
+    without exploding the f32 partial buffer."""
+    tiles = ((K + 63) // 64) * R * S * ((C + 63) // 64)
+    want = max(1, 2048 // max(tiles, 1))
+    max_split = max(1, M // 64)
+    return max(1, min(want, max_split, 64))
+
+
+class _ConvFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b, stride, pad):
+        lib = require_lib()
+        x = x.contiguous(memory_format=_CL)
+        wc = w.contiguous(memory_format=_CL)
+        Nb, C, H, W = x.shape
+        K, _, R, S = w.shape
+        P = (H + 2 * pad - R) // stride + 1
+        Q = (W + 2 * pad - S) // stride + 1
+        out = torch.empty((Nb, K, P, Q), dtype=x.dtype, device=x.device) \
+            .contiguous(memory_format=_CL)
+        lib.ps_conv_fwd(x.data_ptr(), wc.data_ptr(),
+                        b.data_ptr() if b is not None else 0, out.data_ptr(),
+                        Nb, H, W, C, K, P, Q, R, S, stride, pad,
+                        current_stream_ptr())
+        ctx.save_for_backward(x, wc)
+        ctx.conf = (stride, pad, b is not None)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        lib = require_lib()
+        x, w = ctx.saved_tensors
+        stride, pad, has_bias = ctx.conf
+        dout = dout.contiguous(memory_format=_CL)
+        Nb, C, H, W = x.shape
+        K, _, R, S = w.shape
+        P, Q = dout.shape[2], dout.shape[3]
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = torch.empty_like(x).contiguous(memory_format=_CL)
+            lib.ps_conv_dgrad(dout.data_ptr(), w.data_ptr(), dx.data_ptr(),
+                              Nb, H, W, C, K, P, Q, R, S, stride, pad,
+                              current_stream_ptr())
+        if ctx.needs_input_grad[1]:
+            M = Nb * P * Q
+            split = _wgrad_split(M, K, C, R, S)
+            partial = torch.empty(split * K * R * S * C,
+                                  dtype=torch.float32, device=x.device)
+            dw = torch.empty_like(w).contiguous(memory_format=_CL)
+            lib.ps_conv_wgrad(dout.data_ptr(), x.data_ptr(),
+                              partial.data_ptr(), dw.data_ptr(),
+                              Nb, H, W, C, K, P, Q, R, S, stride, pad,
+                              split, current_stream_ptr())
+        if has_bias and ctx.needs_input_grad[2]:
+            db = torch.empty(K, dtype=dout.dtype, device=dout.device)
+            lib.ps_conv_bias_grad(db.data_ptr(), dout.data_ptr(),
+                                  Nb * P * Q, K, current_stream_ptr())
+        return dx, dw, db, None, None
+
+
+class PsConv2d(nn.Conv2d):
+    """nn.Conv2d whose GPU bf16 path runs the in-tree CDNA4 kernels."""
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if _supported(x, self.weight, self.stride, self.padding,
+                      self.dilation, self.groups):
+            return _ConvFn.apply(x, self.weight, self.bias,
+                                 self.stride[0], self.padding[0])
+        return super().forward(x)

@@ -1,0 +1,82 @@
+"""PsConv2d MFMA kernels vs plain-torch fp32 reference (SURVEY.md §4 rule:
+numerics tests for a HIP kernel compare it against a torch fp32 reference
+of the same op). Covers the ResNet-18/CIFAR shape family, the 1x1-stride-2
+downsample, the C=3 stem, and LeNet's 5x5 no-pad convs with bias."""
+import pytest
+import torch
+import torch.nn.functional as F
+
+from ps_pytorch_amd.ops.conv import _ConvFn
+
+pytestmark = pytest.mark.gpu
+
+_CL = torch.channels_last
+
+SHAPES = [
+    # Nb, C, H, W, K, R, stride, pad, bias
+    (8, 64, 32, 32, 64, 3, 1, 1, False),     # layer1 3x3
+    (8, 64, 32, 32, 128, 3, 2, 1, False),    # layer2 downsample 3x3 s2
+    (8, 128, 16, 16, 128, 3, 1, 1, False),
+    (8, 64, 32, 32, 128, 1, 2, 0, False),    # 1x1 s2 shortcut
+    (4, 256, 8, 8, 512, 3, 2, 1, False),
+    (4, 3, 32, 32, 64, 3, 1, 1, False),      # stem C=3 (scalar path)
+    (4, 1, 28, 28, 20, 5, 1, 0, True),       # LeNet conv1
+    (4, 20, 12, 12, 50, 5, 1, 0, True),      # LeNet conv2 (C,K % 8 != 0)
+    (3, 16, 9, 7, 24, 3, 1, 1, False),       # ragged M (tile tails)
+]
+
+
+def _rel_err(a: torch.Tensor, b: torch.Tensor) -> float:
+    d = (a.float() - b.float()).abs().max()
+    return float(d / b.float().abs().max().clamp_min(1e-6))
+
+
+@pytest.mark.parametrize("shape", SHAPES)
+def test_conv_fwd_bwd_matches_torch(shape):
+    Nb, C, H, W, K, R, stride, pad, bias = shape
+    g = torch.Generator().manual_seed(hash(shape) % (2 ** 31))
+    x = torch.randn(Nb, C, H, W, generator=g).to('cuda', torch.bfloat16) \
+        .contiguous(memory_format=_CL).requires_grad_(True)
+    w = (torch.randn(K, C, R, R, generator=g) / (R * R * C) ** 0.5) \
+        .to('cuda', torch.bfloat16).contiguous(memory_format=_CL) \
+        .requires_grad_(True)
+    b = (torch.randn(K, generator=g).to('cuda', torch.bfloat16)
+         .requires_grad_(True) if bias else None)
+
+    out = _ConvFn.apply(x, w, b, stride, pad)
+    dout = torch.randn(out.shape, generator=g).to('cuda', torch.bfloat16) \
+        .contiguous(memory_format=_CL)
+    out.backward(dout)
+
+    # fp32 reference on the SAME bf16-rounded values
+    xr = x.detach().float().requires_grad_(True)
+    wr = w.detach().float().requires_grad_(True)
+    br = b.detach().float().requires_grad_(True) if bias else None
+    outr = F.conv2d(xr, wr, br, stride=stride, padding=pad)
+    outr.backward(dout.float())
+
+    assert out.shape == outr.shape
+    assert _rel_err(out, outr) < 0.03, f"fwd {_rel_err(out, outr)}"
+    assert _rel_err(x.grad, xr.grad) < 0.03, f"dgrad {_rel_err(x.grad, xr.grad)}"
+    assert _rel_err(w.grad, wr.grad) < 0.03, f"wgrad {_rel_err(w.grad, wr.grad)}"
+    if bias:
+        assert _rel_err(b.grad, br.grad) < 0.03, "bias grad"
+
+
+def test_conv_wgrad_deterministic():
+    """Two identical backward passes give bit-identical dw (fixed-order
+    split-K reduce, no atomics) — PS replicas must agree bitwise."""
+    g = torch.Generator().manual_seed(7)
+    x = torch.randn(8, 64, 32, 32, generator=g).to('cuda', torch.bfloat16) \
+        .contiguous(memory_format=_CL)
+    w = torch.randn(64, 64, 3, 3, generator=g).mul(0.05) \
+        .to('cuda', torch.bfloat16).contiguous(memory_format=_CL)
+    dout = torch.randn(8, 64, 32, 32, generator=g).to('cuda', torch.bfloat16) \
+        .contiguous(memory_format=_CL)
+    grads = []
+    for _ in range(2):
+        wv = w.clone().requires_grad_(True)
+        out = _ConvFn.apply(x, wv, None, 1, 1)
+        out.backward(dout)
+        grads.append(wv.grad.clone())
+    assert torch.equal(grads[0], grads[1])
