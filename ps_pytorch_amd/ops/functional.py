@@ -122,10 +122,14 @@ def pack_q8(payload: torch.Tensor, src: torch.Tensor) -> None:
     x[:n] = src.to(torch.float32)
     xb = x.view(nblk, Q8_BLOCK)
     m = xb.abs().amax(dim=1)
-    inv = torch.where(m > 0, 127.0 / m, torch.zeros_like(m))
+    # NB: divide f32-tensor by f32-tensor — torch computes python-scalar /
+    # tensor in f64 and casts, which is 1 ulp off strict f32 division and
+    # breaks bitwise parity with the GPU kernel's IEEE f32 divide.
+    c127 = torch.full_like(m, 127.0)
+    inv = torch.where(m > 0, c127 / m, torch.zeros_like(m))
     q = torch.clamp(torch.round(xb * inv[:, None]), -127, 127).to(torch.int8)
     payload[:n] = q.view(-1)[:n].view(torch.uint8)
-    payload[qb:qb + 4 * nblk].view(torch.float32).copy_(m / 127.0)
+    payload[qb:qb + 4 * nblk].view(torch.float32).copy_(m / c127)
 
 
 def unpack_q8(dst: torch.Tensor, payload: torch.Tensor,
