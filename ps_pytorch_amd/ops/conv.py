@@ -28,6 +28,8 @@ def _supported(x: torch.Tensor, w: torch.Tensor, stride, padding,
         return False
     if stride[0] != stride[1] or padding[0] != padding[1]:
         return False
+    if stride[0] not in (1, 2):
+        return False
     R, S = w.shape[2], w.shape[3]
     return R <= 7 and S <= 7
 
@@ -38,7 +40,7 @@ def _wgrad_split(M: int, K: int, C: int, R: int, S: int) -> int:
     tiles = ((K + 63) // 64) * R * S * ((C + 63) // 64)
     want = max(1, 2048 // max(tiles, 1))
     max_split = max(1, M // 64)
-    return max(1, min(want, max_split, 64))
+    return max(1, min(want, max_split, 256))
 
 
 class _ConvFn(torch.autograd.Function):
@@ -72,8 +74,12 @@ class _ConvFn(torch.autograd.Function):
         P, Q = dout.shape[2], dout.shape[3]
         dx = dw = db = None
         if ctx.needs_input_grad[0]:
+            # dgrad wants wT[R,S,C,K] in memory so its B stage is the same
+            # contiguous-in-contraction load as fwd (no in-kernel transpose);
+            # one small permute-copy per backward (<5 MB even for ResNet-50).
+            wt = w.permute(2, 3, 1, 0).contiguous()
             dx = torch.empty_like(x).contiguous(memory_format=_CL)
-            lib.ps_conv_dgrad(dout.data_ptr(), w.data_ptr(), dx.data_ptr(),
+            lib.ps_conv_dgrad(dout.data_ptr(), wt.data_ptr(), dx.data_ptr(),
                               Nb, H, W, C, K, P, Q, R, S, stride, pad,
                               current_stream_ptr())
         if ctx.needs_input_grad[1]:

@@ -7,21 +7,36 @@
 //
 // Formulation (NHWC, all tensors bf16, fp32 accumulate):
 //   fwd  : out[n,p,q,k] = sum_{r,s,c} in[n, p*st-pad+r, q*st-pad+s, c] * w[k,r,s,c]
-//          GEMM  M=N*P*Q (pixels) x N=K, inner = R*S*C.
-//          Weight [K][R][S][C] is ALREADY the B^T layout the mfma B-fragment
-//          wants (lane j=lane&15 reads its k-slice contiguously), so both
-//          operands stage as [row][k] tiles with NO transpose.
-//   dgrad: dx[n,h,w,c] = sum_{k,r,s} dout[n,(h+pad-r)/st,(w+pad-s)/st,k] * w[k,r,s,c]
-//          GEMM  M=N*H*W x N=C, inner = K per (r,s); B tile needs a
-//          transpose stage (w rows are k-major).
+//          GEMM  M=N*P*Q (pixels) x N=K, contraction R*S*C.
+//   dgrad: dx[n,h,w,c] = sum_{k,r,s} dout[n,(h+pad-r)/st,(w+pad-s)/st,k] * wT[r,s,c,k]
+//          GEMM  M=N*H*W x N=C, contraction R*S*K.  The host passes the
+//          weight PRE-TRANSPOSED to [R,S,C,K] (one tiny permute per backward)
+//          so the B stage is the same contiguous-in-contraction load as fwd —
+//          no in-kernel transpose.
 //   wgrad: dw[k,r,s,c] = sum_{n,p,q} dout[n,p,q,k] * in[n,p*st-pad+r,...,c]
-//          GEMM  M=K x N=R*S*C, inner = N*P*Q (huge) -> split-K over pixel
-//          chunks into f32 partials + deterministic slab reduce (fixed
+//          GEMM  M=K x N=R*S*C, contraction = N*P*Q (huge) -> split-K over
+//          pixel chunks into f32 partials + deterministic slab reduce (fixed
 //          order, no atomics — replicas must stay bit-identical).
 //
-// Tiles: fwd/dgrad 128x64 (BK=64), wgrad 64x64; 4 waves/block;
-// mfma_f32_16x16x32_bf16; LDS rows padded 64->72 elems (144 B) to break
-// the power-of-2 column-read conflict (guide §6 G4).
+// Performance structure (guide §5/§6: T14 register-staged pipeline):
+//   * fwd/dgrad: 4-wave blocks, wave tile 64x64 (4x4 mfma_f32_16x16x32_bf16),
+//     block tile 256x64 (N<=64 output channels) or 128x128; BK=64; ONE
+//     register staging set, 2 LDS buffers, one barrier per k-step: write
+//     tile t+1 to LDS after the barrier, immediately issue loads of t+2,
+//     then MFMA tile t while the loads fly.
+//   * stride is a template parameter — no integer div/mod in inner loops
+//     (stride-2 dgrad parity checks become bit ops).
+//   * wgrad: contraction runs over pixels; both operands need an
+//     [out-idx][pixel] LDS image, i.e. a transpose on store. The store
+//     scatter is XOR-swizzled on 8-element granules (g ^= (row&7)^(row>>3))
+//     so the 8 lanes sharing one pixel column hit 8 different banks; reads
+//     unswizzle the same way (2-way conflict worst case on ds_read_b128).
+//     Pixel->(n,p,q) decode is shifts when P*Q and Q are powers of two
+//     (every CIFAR shape), runtime div otherwise.
+//   * wgrad grid is XCD-swizzled: all blocks covering the SAME pixel chunk
+//     (same split id, all (r,s)/k/c tiles) are placed on the SAME XCD
+//     (block b runs on XCD b%8, microarch guide) so the 9x re-read of
+//     dout/in per (r,s) comes from that XCD's L2, not HBM.
 #include "common.h"
 
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
@@ -41,268 +56,247 @@ __device__ __forceinline__ V16 zero16() {
     V16 v; v.u4 = make_uint4(0, 0, 0, 0); return v;
 }
 
-// ---------------------------------------------------------------- forward
-
-struct ConvLds {
-    unsigned short A[128][LDSP];   // [pixel][kg]  (kg = c within (r,s))
-    unsigned short B[64][LDSP];    // [out-ch][kg]
-};
-
-__global__ __launch_bounds__(256) void conv_fwd_kernel(
-    const unsigned short* __restrict__ in,   // [Nb,H,W,C]
-    const unsigned short* __restrict__ wgt,  // [K,R,S,C]
-    const unsigned short* __restrict__ bias, // [K] or null
-    unsigned short* __restrict__ out,        // [Nb,P,Q,K]
-    int Nb, int H, int W, int C, int K, int P, int Q,
-    int R, int S, int stride, int pad)
-{
-    __shared__ __attribute__((aligned(16))) ConvLds lds;
-    const int M = Nb * P * Q;
-    const int tiles_n = (K + 63) >> 6;
-    const int tile_m = blockIdx.x / tiles_n;
-    const int tile_n = blockIdx.x % tiles_n;
-    const int m0 = tile_m * 128, n0 = tile_n * 64;
-    const int t = threadIdx.x;
-    const int RSC = R * S * C;
-
-    // per-thread A staging descriptors: rows t/8 + {0,32,64,96}
-    long abase[4]; int aph[4], apw[4];
-#pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
-        int i = (t >> 3) + 32 * rr;
-        int m = m0 + i;
-        if (m < M) {
-            int n = m / (P * Q), rem = m % (P * Q);
-            int p = rem / Q, q = rem % Q;
-            aph[rr] = p * stride - pad;
-            apw[rr] = q * stride - pad;
-            abase[rr] = ((long)(n * H + aph[rr]) * W + apw[rr]) * C;
-        } else { aph[rr] = INT_MIN / 2; apw[rr] = 0; abase[rr] = 0; }
+// 16B load with a tail guard for channel counts that are not multiples of 8
+// (LeNet C=1/20/50); ResNet/VGG take the single vector load.
+__device__ __forceinline__ V16 load16(const unsigned short* src, int cc, int Cn) {
+    V16 v;
+    if (cc + 8 <= Cn && ((Cn | cc) & 7) == 0) {
+        v.u4 = *(const uint4*)src;
+    } else {
+        v = zero16();
+        for (int u = 0; u < 8; ++u)
+            if (cc + u < Cn) v.us[u] = src[u];
     }
-    const int cc8 = (t & 7) * 8;   // this thread's 8-elem column chunk
-
-    const int lane = t & 63, wid = t >> 6;
-    const int wm = wid >> 1, wn = wid & 1;        // 2x2 wave grid
-    const int fr = lane & 15, fq = lane >> 4;
-    f32x4_t acc[4][2];
-#pragma unroll
-    for (int i = 0; i < 4; ++i)
-#pragma unroll
-        for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
-
-    for (int r = 0; r < R; ++r)
-    for (int s = 0; s < S; ++s)
-    for (int c0 = 0; c0 < C; c0 += 64) {
-        // ---- stage A (input gather): 4 rows x 16B per thread ----
-#pragma unroll
-        for (int rr = 0; rr < 4; ++rr) {
-            int i = (t >> 3) + 32 * rr;
-            int hh = aph[rr] + r, ww = apw[rr] + s;
-            bool v = hh >= 0 && hh < H && ww >= 0 && ww < W;
-            int cc = c0 + cc8;
-            V16 val = zero16();
-            if (v) {
-                const unsigned short* src = in + abase[rr] + (long)(r * W + s) * C + cc;
-                if ((C & 7) == 0 && cc + 8 <= C) val.u4 = *(const uint4*)src;
-                else for (int u = 0; u < 8; ++u)
-                    if (cc + u < C) val.us[u] = src[u];
-            }
-            *(uint4*)&lds.A[i][cc8] = val.u4;
-        }
-        // ---- stage B (weights, no transpose): 2 rows x 16B ----
-#pragma unroll
-        for (int rr = 0; rr < 2; ++rr) {
-            int j = (t >> 3) + 32 * rr;
-            int k = n0 + j;
-            int cc = c0 + cc8;
-            V16 val = zero16();
-            if (k < K) {
-                const unsigned short* src = wgt + (long)k * RSC + (r * S + s) * C + cc;
-                if ((C & 7) == 0 && cc + 8 <= C) val.u4 = *(const uint4*)src;
-                else for (int u = 0; u < 8; ++u)
-                    if (cc + u < C) val.us[u] = src[u];
-            }
-            *(uint4*)&lds.B[j][cc8] = val.u4;
-        }
-        __syncthreads();
-#pragma unroll
-        for (int kk = 0; kk < 2; ++kk) {
-            bf16x8_t a[4], b[2];
-#pragma unroll
-            for (int mi = 0; mi < 4; ++mi)
-                a[mi] = *(const bf16x8_t*)&lds.A[wm * 64 + mi * 16 + fr][kk * 32 + fq * 8];
-#pragma unroll
-            for (int nj = 0; nj < 2; ++nj)
-                b[nj] = *(const bf16x8_t*)&lds.B[wn * 32 + nj * 16 + fr][kk * 32 + fq * 8];
-#pragma unroll
-            for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-                for (int nj = 0; nj < 2; ++nj)
-                    acc[mi][nj] = MFMA_BF16(a[mi], b[nj], acc[mi][nj]);
-        }
-        __syncthreads();
-    }
-
-    // ---- epilogue: C/D map col=lane&15 (k), row=(lane>>4)*4+e (m) ----
-#pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
-#pragma unroll
-    for (int nj = 0; nj < 2; ++nj) {
-        int ok = n0 + wn * 32 + nj * 16 + fr;
-        float bv = (bias && ok < K) ? bf16_to_f32(bias[ok]) : 0.f;
-#pragma unroll
-        for (int e = 0; e < 4; ++e) {
-            int om = m0 + wm * 64 + mi * 16 + fq * 4 + e;
-            if (om < M && ok < K)
-                out[(long)om * K + ok] = f32_to_bf16(acc[mi][nj][e] + bv);
-        }
-    }
+    return v;
 }
 
-// ---------------------------------------------------------------- dgrad
+// ------------------------------------------------------------ fwd / dgrad
+//
+// One templated GEMM core. TM x TN block tile, 256 threads:
+//   TN==64 : 4 waves stacked along M (TM=256), wave tile 64x64
+//   TN==128: 2x2 wave grid (TM=128), wave tile 64x64
+// DGRAD=false: rows are output pixels, cols K, contraction (r,s,c).
+// DGRAD=true : rows are input pixels, cols C, contraction (r,s,k),
+//              weights given as wT[R,S,C,K].
 
-__global__ __launch_bounds__(256) void conv_dgrad_kernel(
-    const unsigned short* __restrict__ dout, // [Nb,P,Q,K]
-    const unsigned short* __restrict__ wgt,  // [K,R,S,C]
-    unsigned short* __restrict__ dx,         // [Nb,H,W,C]
+template <int TM, int TN, int STRIDE, bool DGRAD>
+struct FwdLds {
+    unsigned short A[2][TM][LDSP];
+    unsigned short B[2][TN][LDSP];
+};
+
+template <int TM, int TN, int STRIDE, bool DGRAD>
+__global__ __launch_bounds__(256) void conv_gemm_kernel(
+    const unsigned short* __restrict__ src,  // fwd: in [Nb,H,W,C]; dgrad: dout [Nb,P,Q,K]
+    const unsigned short* __restrict__ wgt,  // fwd: w [K,R,S,C]; dgrad: wT [R,S,C,K]
+    const unsigned short* __restrict__ bias, // [K] or null (fwd only)
+    unsigned short* __restrict__ dst,        // fwd: out [Nb,P,Q,K]; dgrad: dx [Nb,H,W,C]
     int Nb, int H, int W, int C, int K, int P, int Q,
-    int R, int S, int stride, int pad)
+    int R, int S, int pad)
 {
-    __shared__ __attribute__((aligned(16))) ConvLds lds;   // A:[pixel][k] B:[c][k]
-    const int M = Nb * H * W;
-    const int tiles_n = (C + 63) >> 6;
-    const int tile_m = blockIdx.x / tiles_n;
+    __shared__ __attribute__((aligned(16))) FwdLds<TM, TN, STRIDE, DGRAD> lds;
+    constexpr int AR = TM / 32;        // A rows staged per thread
+    constexpr int BR = TN / 32;        // B rows staged per thread
+    const int Cin = DGRAD ? K : C;     // contraction channel count
+    const int Nout = DGRAD ? C : K;    // output column count
+    const long M = DGRAD ? (long)Nb * H * W : (long)Nb * P * Q;
+    const int tiles_n = (Nout + TN - 1) / TN;
+    const long tile_m = blockIdx.x / tiles_n;
     const int tile_n = blockIdx.x % tiles_n;
-    const int m0 = tile_m * 128, n0 = tile_n * 64;
+    const long m0 = tile_m * TM;
+    const int n0 = tile_n * TN;
     const int t = threadIdx.x;
-    const int RSC = R * S * C;
+    const int trow = t >> 3;           // 0..31
+    const int cc8 = (t & 7) * 8;       // this thread's 8-elem contraction chunk
 
-    // per-thread pixel descriptors (input coords)
-    int an[4], ah[4], aw[4];
+    // ---- per-row pixel descriptors (computed once) ----
+    long abase[AR]; int ax[AR], ay[AR];   // fwd: h0,w0; dgrad: h,w (+n folded in abase)
 #pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
-        int m = m0 + (t >> 3) + 32 * rr;
+    for (int rr = 0; rr < AR; ++rr) {
+        long m = m0 + trow + 32 * rr;
         if (m < M) {
-            an[rr] = m / (H * W);
-            int rem = m % (H * W);
-            ah[rr] = rem / W; aw[rr] = rem % W;
-        } else an[rr] = -1;
+            if constexpr (!DGRAD) {
+                int n = (int)(m / ((long)P * Q)); int rem = (int)(m % ((long)P * Q));
+                int p = rem / Q, q = rem % Q;
+                ax[rr] = p * STRIDE - pad;       // input h origin
+                ay[rr] = q * STRIDE - pad;       // input w origin
+                abase[rr] = ((long)(n * H + ax[rr]) * W + ay[rr]) * C;
+            } else {
+                int n = (int)(m / ((long)H * W)); int rem = (int)(m % ((long)H * W));
+                ax[rr] = rem / W; ay[rr] = rem % W;
+                abase[rr] = (long)n * P * Q * K;
+            }
+        } else { ax[rr] = INT_MIN / 2; ay[rr] = 0; abase[rr] = 0; }
     }
-    const int cc8 = (t & 7) * 8;
 
     const int lane = t & 63, wid = t >> 6;
-    const int wm = wid >> 1, wn = wid & 1;
+    const int wm = (TN == 64) ? wid : (wid >> 1);
+    const int wn = (TN == 64) ? 0 : (wid & 1);
     const int fr = lane & 15, fq = lane >> 4;
-    f32x4_t acc[4][2];
+
+    f32x4_t acc[4][4];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
-        for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+        for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-    for (int r = 0; r < R; ++r)
-    for (int s = 0; s < S; ++s)
-    for (int k0 = 0; k0 < K; k0 += 64) {
-        // ---- stage A: dout gather, contiguous in k ----
+    const int KC = (Cin + 63) >> 6;          // contraction chunks per (r,s)
+    const int nsteps = R * S * KC;
+    int lr = 0, ls = 0, lc = 0;              // load-pointer step state
+
+    V16 areg[AR], breg[BR];
+
+    // stage loads for contraction step (lr, ls, lc*64) into registers
+    auto load_step = [&]() {
+        const int c0 = lc << 6;
 #pragma unroll
-        for (int rr = 0; rr < 4; ++rr) {
-            int i = (t >> 3) + 32 * rr;
-            V16 val = zero16();
-            if (an[rr] >= 0) {
-                int ph = ah[rr] + pad - r, pw = aw[rr] + pad - s;
-                if (ph >= 0 && pw >= 0 && ph % stride == 0 && pw % stride == 0) {
-                    int p = ph / stride, q = pw / stride;
-                    if (p < P && q < Q) {
-                        int kk = k0 + cc8;
-                        const unsigned short* src =
-                            dout + ((long)(an[rr] * P + p) * Q + q) * K + kk;
-                        if ((K & 7) == 0 && kk + 8 <= K) val.u4 = *(const uint4*)src;
-                        else for (int u = 0; u < 8; ++u)
-                            if (kk + u < K) val.us[u] = src[u];
-                    }
+        for (int rr = 0; rr < AR; ++rr) {
+            if constexpr (!DGRAD) {
+                int hh = ax[rr] + lr, ww = ay[rr] + ls;
+                bool v = hh >= 0 && hh < H && ww >= 0 && ww < W;
+                areg[rr] = v ? load16(src + abase[rr] + ((long)lr * W + ls) * C
+                                      + c0 + cc8, c0 + cc8, C)
+                             : zero16();
+            } else {
+                int ph = ax[rr] + pad - lr, pw = ay[rr] + pad - ls;
+                bool v;
+                int p, q;
+                if constexpr (STRIDE == 1) {
+                    p = ph; q = pw;
+                    v = ax[rr] >= 0 && p >= 0 && p < P && q >= 0 && q < Q;
+                } else {
+                    v = ax[rr] >= 0 && ph >= 0 && pw >= 0
+                        && (ph & 1) == 0 && (pw & 1) == 0;
+                    p = ph >> 1; q = pw >> 1;
+                    v = v && p < P && q < Q;
                 }
+                areg[rr] = v ? load16(src + abase[rr] + ((long)p * Q + q) * K
+                                      + c0 + cc8, c0 + cc8, K)
+                             : zero16();
             }
-            *(uint4*)&lds.A[i][cc8] = val.u4;
         }
-        // ---- stage B: w[k][r][s][c] -> lds.B[c][k] (transpose) ----
 #pragma unroll
-        for (int rr = 0; rr < 2; ++rr) {
-            int kidx = (t >> 3) + 32 * rr;      // k within tile
-            int k = k0 + kidx;
-            int c = n0 + cc8;                   // 8 consecutive c
-            V16 val = zero16();
-            if (k < K) {
-                const unsigned short* src = wgt + (long)k * RSC + (r * S + s) * C + c;
-                for (int u = 0; u < 8; ++u)
-                    if (c + u < C) val.us[u] = src[u];
-            }
-#pragma unroll
-            for (int u = 0; u < 8; ++u)
-                lds.B[cc8 + u][kidx] = val.us[u];
+        for (int rr = 0; rr < BR; ++rr) {
+            int j = trow + 32 * rr;            // row within tile
+            int col = n0 + j;                  // output channel / input channel
+            if (col < Nout) {
+                const unsigned short* wsrc;
+                if constexpr (!DGRAD)
+                    wsrc = wgt + (((long)col * R + lr) * S + ls) * C + c0 + cc8;
+                else
+                    wsrc = wgt + (((long)(lr * S + ls) * C) + col) * (long)K + c0 + cc8;
+                breg[rr] = load16(wsrc, c0 + cc8, Cin);
+            } else breg[rr] = zero16();
         }
-        __syncthreads();
+        if (++lc == KC) { lc = 0; if (++ls == S) { ls = 0; ++lr; } }
+    };
+
+    auto write_lds = [&](int buf) {
+#pragma unroll
+        for (int rr = 0; rr < AR; ++rr)
+            *(uint4*)&lds.A[buf][trow + 32 * rr][cc8] = areg[rr].u4;
+#pragma unroll
+        for (int rr = 0; rr < BR; ++rr)
+            *(uint4*)&lds.B[buf][trow + 32 * rr][cc8] = breg[rr].u4;
+    };
+
+    auto mfma_step = [&](int buf) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
-            bf16x8_t a[4], b[2];
+            bf16x8_t a[4], b[4];
 #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
-                a[mi] = *(const bf16x8_t*)&lds.A[wm * 64 + mi * 16 + fr][kk * 32 + fq * 8];
+                a[mi] = *(const bf16x8_t*)
+                    &lds.A[buf][wm * 64 + mi * 16 + fr][kk * 32 + fq * 8];
 #pragma unroll
-            for (int nj = 0; nj < 2; ++nj)
-                b[nj] = *(const bf16x8_t*)&lds.B[wn * 32 + nj * 16 + fr][kk * 32 + fq * 8];
+            for (int nj = 0; nj < 4; ++nj)
+                b[nj] = *(const bf16x8_t*)
+                    &lds.B[buf][wn * 64 + nj * 16 + fr][kk * 32 + fq * 8];
 #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-                for (int nj = 0; nj < 2; ++nj)
+                for (int nj = 0; nj < 4; ++nj)
                     acc[mi][nj] = MFMA_BF16(a[mi], b[nj], acc[mi][nj]);
         }
+    };
+
+    // ---- pipeline: 2 LDS buffers, one reg set, one barrier per step ----
+    load_step();
+    write_lds(0);
+    if (nsteps > 1) load_step();
+    __syncthreads();
+    for (int it = 0; it < nsteps; ++it) {
+        if (it + 1 < nsteps) {
+            write_lds((it + 1) & 1);
+            if (it + 2 < nsteps) load_step();
+        }
+        mfma_step(it & 1);
         __syncthreads();
     }
 
+    // ---- epilogue ----
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
+    for (int nj = 0; nj < 4; ++nj) {
+        int col = n0 + wn * 64 + nj * 16 + fr;
+        float bv = (!DGRAD && bias && col < Nout) ? bf16_to_f32(bias[col]) : 0.f;
 #pragma unroll
-    for (int nj = 0; nj < 2; ++nj)
+        for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-    for (int e = 0; e < 4; ++e) {
-        int om = m0 + wm * 64 + mi * 16 + fq * 4 + e;
-        int oc = n0 + wn * 32 + nj * 16 + fr;
-        if (om < M && oc < C)
-            dx[(long)om * C + oc] = f32_to_bf16(acc[mi][nj][e]);
+        for (int e = 0; e < 4; ++e) {
+            long om = m0 + wm * 64 + mi * 16 + fq * 4 + e;
+            if (om < M && col < Nout)
+                dst[om * Nout + col] = f32_to_bf16(acc[mi][nj][e] + bv);
+        }
     }
 }
 
 // ---------------------------------------------------------------- wgrad
 
+// XOR-swizzled transposed LDS image: element (row, mcol) of a 64x64 tile
+// lives at [row][ swz(row, mcol>>3)*8 + (mcol&7) ], pitch 64 (no pad — the
+// swizzle does the bank spreading).
+__device__ __forceinline__ int wg_swz(int row, int g) {
+    return g ^ (row & 7) ^ ((row >> 3) & 7);
+}
+
 struct WgradLds {
-    unsigned short A[64][LDSP];   // [k][pixel]
-    unsigned short B[64][LDSP];   // [c][pixel]
+    unsigned short A[2][64][64];   // [k][pixel]
+    unsigned short B[2][64][64];   // [c][pixel]
 };
 
-// One block: 64 k x 64 rsc output tile for ONE (r,s), summing a pixel range
-// [split_id*chunk, ...) of length `chunk`; f32 partial out[split][k][rsc].
+// One block: 64 k x 64 c output tile for ONE (r,s), summing the pixel range
+// [sid*chunk, ...) of length `chunk`; f32 partial out[sid][K][R*S*C].
+// POW2: P*Q and Q are powers of two (shift decode); else runtime div.
+template <int STRIDE, bool POW2>
 __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     const unsigned short* __restrict__ dout,  // [Nb,P,Q,K]
     const unsigned short* __restrict__ in,    // [Nb,H,W,C]
     float* __restrict__ partial,              // [SPLIT][K][R*S*C]
     int Nb, int H, int W, int C, int K, int P, int Q,
-    int R, int S, int stride, int pad, int split, int chunk)
+    int R, int S, int pad, int split, int chunk,
+    int l2pq, int l2q, int per_xcd)
 {
     __shared__ __attribute__((aligned(16))) WgradLds lds;
-    const int M = Nb * P * Q;
+    const long M = (long)Nb * P * Q;
     const int RSC = R * S * C;
     const int tiles_k = (K + 63) >> 6;
     const int tiles_c = (C + 63) >> 6;
-    // grid: [tiles_k * R*S * tiles_c * split]
-    int b = blockIdx.x;
+    // XCD swizzle: logical index l groups all tiles of one pixel chunk on
+    // one XCD (physical block b runs on XCD b%8).
+    long l = (long)(blockIdx.x & 7) * per_xcd + (blockIdx.x >> 3);
+    const long nlog = (long)tiles_k * tiles_c * R * S * split;
+    if (l >= nlog) return;
+    int b = (int)l;
     const int k0 = (b % tiles_k) * 64; b /= tiles_k;
+    const int c0 = (b % tiles_c) * 64; b /= tiles_c;
     const int rs = b % (R * S); b /= (R * S);
     const int r = rs / S, s = rs % S;
-    const int c0 = (b % tiles_c) * 64; b /= tiles_c;
     const int sid = b;
-    const int mbeg = sid * chunk;
-    const int mend = min(mbeg + chunk, M);
+    const long mbeg = (long)sid * chunk;
+    const long mend = (mbeg + chunk < M) ? mbeg + chunk : M;
 
     const int t = threadIdx.x;
+    const int trow = t >> 3;
     const int cc8 = (t & 7) * 8;
     const int lane = t & 63, wid = t >> 6;
     const int wm = wid >> 1, wn = wid & 1;
@@ -313,64 +307,89 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
 #pragma unroll
         for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-    for (int mb = mbeg; mb < mend; mb += 64) {
-        // each thread stages 2 pixels (m = mb + t/8 + {0,32}) for A and B
+    const int nsteps = (int)((mend - mbeg + 63) >> 6);
+    long lm = mbeg;                     // load pointer
+    V16 areg[2], breg[2];
+
+    auto load_step = [&]() {
 #pragma unroll
         for (int rr = 0; rr < 2; ++rr) {
-            int midx = (t >> 3) + 32 * rr;
-            int m = mb + midx;
+            long m = lm + trow + 32 * rr;
             V16 av = zero16(), bv = zero16();
             if (m < mend) {
-                int n = m / (P * Q), rem = m % (P * Q);
-                int p = rem / Q, q = rem % Q;
-                // A: dout[m][k0+cc8 ..] contiguous in k
-                {
-                    int kk = k0 + cc8;
-                    const unsigned short* src = dout + (long)m * K + kk;
-                    if ((K & 7) == 0 && kk + 8 <= K) av.u4 = *(const uint4*)src;
-                    else for (int u = 0; u < 8; ++u)
-                        if (kk + u < K) av.us[u] = src[u];
+                int n, p, q;
+                if constexpr (POW2) {
+                    n = (int)(m >> l2pq);
+                    int rem = (int)m & ((1 << l2pq) - 1);
+                    p = rem >> l2q; q = rem & ((1 << l2q) - 1);
+                } else {
+                    n = (int)(m / ((long)P * Q));
+                    int rem = (int)(m % ((long)P * Q));
+                    p = rem / Q; q = rem % Q;
                 }
-                // B: in[n, p*st-pad+r, q*st-pad+s, c0+cc8..] contiguous in c
-                {
-                    int hh = p * stride - pad + r, ww = q * stride - pad + s;
-                    if (hh >= 0 && hh < H && ww >= 0 && ww < W) {
-                        int cc = c0 + cc8;
-                        const unsigned short* src =
-                            in + ((long)(n * H + hh) * W + ww) * C + cc;
-                        if ((C & 7) == 0 && cc + 8 <= C) bv.u4 = *(const uint4*)src;
-                        else for (int u = 0; u < 8; ++u)
-                            if (cc + u < C) bv.us[u] = src[u];
-                    }
-                }
+                av = load16(dout + m * K + k0 + cc8, k0 + cc8, K);
+                int hh = p * STRIDE - pad + r, ww = q * STRIDE - pad + s;
+                if (hh >= 0 && hh < H && ww >= 0 && ww < W)
+                    bv = load16(in + ((long)(n * H + hh) * W + ww) * C + c0 + cc8,
+                                c0 + cc8, C);
             }
-            // transpose scatter into [k][m] / [c][m]
+            areg[rr] = av; breg[rr] = bv;
+        }
+        lm += 64;
+    };
+
+    auto write_lds = [&](int buf) {
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr) {
+            int midx = trow + 32 * rr;
+            int g0 = midx >> 3, o = midx & 7;
 #pragma unroll
             for (int u = 0; u < 8; ++u) {
-                lds.A[cc8 + u][midx] = av.us[u];
-                lds.B[cc8 + u][midx] = bv.us[u];
+                int row = cc8 + u;
+                int g = wg_swz(row, g0);
+                lds.A[buf][row][g * 8 + o] = areg[rr].us[u];
+                lds.B[buf][row][g * 8 + o] = breg[rr].us[u];
             }
         }
-        __syncthreads();
+    };
+
+    auto mfma_step = [&](int buf) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
+            int gg = kk * 4 + fq;
             bf16x8_t a[2], bfr[2];
 #pragma unroll
-            for (int mi = 0; mi < 2; ++mi)
-                a[mi] = *(const bf16x8_t*)&lds.A[wm * 32 + mi * 16 + fr][kk * 32 + fq * 8];
+            for (int mi = 0; mi < 2; ++mi) {
+                int row = wm * 32 + mi * 16 + fr;
+                a[mi] = *(const bf16x8_t*)&lds.A[buf][row][wg_swz(row, gg) * 8];
+            }
 #pragma unroll
-            for (int nj = 0; nj < 2; ++nj)
-                bfr[nj] = *(const bf16x8_t*)&lds.B[wn * 32 + nj * 16 + fr][kk * 32 + fq * 8];
+            for (int nj = 0; nj < 2; ++nj) {
+                int row = wn * 32 + nj * 16 + fr;
+                bfr[nj] = *(const bf16x8_t*)&lds.B[buf][row][wg_swz(row, gg) * 8];
+            }
 #pragma unroll
             for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
                 for (int nj = 0; nj < 2; ++nj)
                     acc[mi][nj] = MFMA_BF16(a[mi], bfr[nj], acc[mi][nj]);
         }
+    };
+
+    load_step();
+    write_lds(0);
+    if (nsteps > 1) load_step();
+    __syncthreads();
+    for (int it = 0; it < nsteps; ++it) {
+        if (it + 1 < nsteps) {
+            write_lds((it + 1) & 1);
+            if (it + 2 < nsteps) load_step();
+        }
+        mfma_step(it & 1);
         __syncthreads();
     }
 
-    float* dst = partial + (long)sid * K * RSC;
+    float* dstp = partial + (long)sid * K * RSC;
 #pragma unroll
     for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
@@ -380,7 +399,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
         int k = k0 + wm * 32 + mi * 16 + fq * 4 + e;
         int c = c0 + wn * 32 + nj * 16 + fr;
         if (k < K && c < C)
-            dst[(long)k * RSC + (r * S + s) * C + c] = acc[mi][nj][e];
+            dstp[(long)k * RSC + (r * S + s) * C + c] = acc[mi][nj][e];
     }
 }
 
@@ -419,30 +438,49 @@ __global__ __launch_bounds__(256) void colsum_kernel(
 
 // ---------------------------------------------------------------- C API
 
+#define DISPATCH_GEMM(TM, TN, DG)                                             \
+    do {                                                                      \
+        long M_ = DG ? (long)Nb * H * W : (long)Nb * P * Q;                   \
+        int Nout_ = DG ? C : K;                                               \
+        long grid = ((M_ + TM - 1) / TM) * ((Nout_ + TN - 1) / TN);           \
+        if (stride == 1)                                                      \
+            hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, 1, DG>),             \
+                dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,        \
+                (const unsigned short*)src, (const unsigned short*)wgt,       \
+                (const unsigned short*)bias, (unsigned short*)dst,            \
+                Nb, H, W, C, K, P, Q, R, S, pad);                             \
+        else                                                                  \
+            hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, 2, DG>),             \
+                dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,        \
+                (const unsigned short*)src, (const unsigned short*)wgt,       \
+                (const unsigned short*)bias, (unsigned short*)dst,            \
+                Nb, H, W, C, K, P, Q, R, S, pad);                             \
+    } while (0)
+
 extern "C" void ps_conv_fwd(
-    const void* in, const void* wgt, const void* bias, void* out,
+    const void* src, const void* wgt, const void* bias, void* dst,
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int stride, int pad, void* strm)
 {
-    long M = (long)Nb * P * Q;
-    int grid = (int)((M + 127) / 128) * ((K + 63) / 64);
-    hipLaunchKernelGGL(conv_fwd_kernel, dim3(grid), dim3(256), 0, (hipStream_t)strm,
-                       (const unsigned short*)in, (const unsigned short*)wgt,
-                       (const unsigned short*)bias, (unsigned short*)out,
-                       Nb, H, W, C, K, P, Q, R, S, stride, pad);
+    if (K >= 128) DISPATCH_GEMM(128, 128, false);
+    else          DISPATCH_GEMM(256, 64, false);
 }
 
+// wgt here is the TRANSPOSED weight wT[R,S,C,K] (host permutes once per
+// backward; ~us for the largest ResNet tensor).
 extern "C" void ps_conv_dgrad(
-    const void* dout, const void* wgt, void* dx,
+    const void* src, const void* wgt, void* dst,
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int stride, int pad, void* strm)
 {
-    long M = (long)Nb * H * W;
-    int grid = (int)((M + 127) / 128) * ((C + 63) / 64);
-    hipLaunchKernelGGL(conv_dgrad_kernel, dim3(grid), dim3(256), 0, (hipStream_t)strm,
-                       (const unsigned short*)dout, (const unsigned short*)wgt,
-                       (unsigned short*)dx,
-                       Nb, H, W, C, K, P, Q, R, S, stride, pad);
+    const void* bias = nullptr;
+    if (C >= 128) DISPATCH_GEMM(128, 128, true);
+    else          DISPATCH_GEMM(256, 64, true);
+}
+
+static inline int ilog2_exact(long v) {
+    int l = 0; while ((1L << l) < v) ++l;
+    return ((1L << l) == v) ? l : -1;
 }
 
 extern "C" void ps_conv_wgrad(
@@ -454,11 +492,20 @@ extern "C" void ps_conv_wgrad(
     long chunk64 = (M + (long)split * 64 - 1) / ((long)split * 64);
     int chunk = (int)(chunk64 * 64);
     int tiles_k = (K + 63) / 64, tiles_c = (C + 63) / 64;
-    int grid = tiles_k * R * S * tiles_c * split;
-    hipLaunchKernelGGL(conv_wgrad_kernel, dim3(grid), dim3(256), 0, (hipStream_t)strm,
-                       (const unsigned short*)dout, (const unsigned short*)in,
-                       (float*)partial_f32,
-                       Nb, H, W, C, K, P, Q, R, S, stride, pad, split, chunk);
+    long nlog = (long)tiles_k * tiles_c * R * S * split;
+    int per_xcd = (int)((nlog + 7) / 8);
+    long grid = (long)per_xcd * 8;
+    int l2pq = ilog2_exact((long)P * Q), l2q = ilog2_exact(Q);
+    bool pow2 = l2pq >= 0 && l2q >= 0;
+#define WG_LAUNCH(ST, PW)                                                     \
+    hipLaunchKernelGGL((conv_wgrad_kernel<ST, PW>), dim3((unsigned)grid),     \
+        dim3(256), 0, (hipStream_t)strm,                                      \
+        (const unsigned short*)dout, (const unsigned short*)in,               \
+        (float*)partial_f32, Nb, H, W, C, K, P, Q, R, S, pad, split, chunk,   \
+        l2pq, l2q, per_xcd)
+    if (stride == 1) { if (pow2) WG_LAUNCH(1, true); else WG_LAUNCH(1, false); }
+    else             { if (pow2) WG_LAUNCH(2, true); else WG_LAUNCH(2, false); }
+#undef WG_LAUNCH
     long n = (long)K * R * S * C;
     int blocks; ew_grid(n, 256, &blocks);
     hipLaunchKernelGGL(reduce_slabs_kernel, dim3(blocks), dim3(256), 0,
