@@ -85,7 +85,11 @@ struct FwdLds {
     unsigned short B[2][TN][LDSP];
 };
 
-template <int TM, int TN, int STRIDE, bool DGRAD>
+// SMALL=true: the whole R*S*C contraction fits one 64-chunk (ResNet stem
+// 3*3*3=27, LeNet conv1 5*5*1=25) — flatten (r,s,c) into the contraction
+// axis via a per-lane gather table and run ONE k-step instead of R*S, so
+// the MFMA utilization is RSC/64 of a full tile instead of C/64 per step.
+template <int TM, int TN, int STRIDE, bool DGRAD, bool SMALL = false>
 __global__ __launch_bounds__(256) void conv_gemm_kernel(
     const unsigned short* __restrict__ src,  // fwd: in [Nb,H,W,C]; dgrad: dout [Nb,P,Q,K]
     const unsigned short* __restrict__ wgt,  // fwd: w [K,R,S,C]; dgrad: wT [R,S,C,K]
@@ -129,20 +133,39 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         } else { ax[rr] = INT_MIN / 2; ay[rr] = 0; abase[rr] = 0; }
     }
 
+    // wave grid: every wave owns a 64-row sub-tile; columns split when the
+    // block is narrower than 4 waves' worth of rows.
+    constexpr int WGM = TM / 64;             // waves along M
+    constexpr int WGN = 4 / WGM;             // waves along N
+    constexpr int NJ = TN / WGN / 16;        // 16-col mfma tiles per wave
     const int lane = t & 63, wid = t >> 6;
-    const int wm = (TN == 64) ? wid : (wid >> 1);
-    const int wn = (TN == 64) ? 0 : (wid & 1);
+    const int wm = (WGN == 1) ? wid : (wid >> 1);
+    const int wn = (WGN == 1) ? 0 : (wid & 1);
     const int fr = lane & 15, fq = lane >> 4;
 
-    f32x4_t acc[4][4];
+    f32x4_t acc[4][NJ];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
 #pragma unroll
-        for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+        for (int j = 0; j < NJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
     const int KC = (Cin + 63) >> 6;          // contraction chunks per (r,s)
-    const int nsteps = R * S * KC;
+    const int nsteps = SMALL ? 1 : R * S * KC;
     int lr = 0, ls = 0, lc = 0;              // load-pointer step state
+
+    // SMALL: per-lane flattened (r,s,c) gather table for the 8 contraction
+    // elements this thread stages (element e = cc8+u -> (e/(S*C), (e/C)%S,
+    // e%C)); a handful of integer divides once per kernel.
+    int tre[SMALL ? 8 : 1], tse[SMALL ? 8 : 1], tce[SMALL ? 8 : 1];
+    if constexpr (SMALL) {
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+            int e = cc8 + u;
+            tre[u] = e / (S * C);
+            tse[u] = (e / C) % S;
+            tce[u] = e % C;
+        }
+    }
 
     V16 areg[AR], breg[BR];
 
@@ -151,7 +174,20 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         const int c0 = lc << 6;
 #pragma unroll
         for (int rr = 0; rr < AR; ++rr) {
-            if constexpr (!DGRAD) {
+            if constexpr (SMALL) {
+                const int RSC = R * S * C;
+                V16 v = zero16();
+                bool rowv = ax[rr] > INT_MIN / 4;
+#pragma unroll
+                for (int u = 0; u < 8; ++u) {
+                    int hh = ax[rr] + tre[u], ww = ay[rr] + tse[u];
+                    if (rowv && cc8 + u < RSC
+                        && hh >= 0 && hh < H && ww >= 0 && ww < W)
+                        v.us[u] = src[abase[rr]
+                                      + ((long)tre[u] * W + tse[u]) * C + tce[u]];
+                }
+                areg[rr] = v;
+            } else if constexpr (!DGRAD) {
                 int hh = ax[rr] + lr, ww = ay[rr] + ls;
                 bool v = hh >= 0 && hh < H && ww >= 0 && ww < W;
                 areg[rr] = v ? load16(src + abase[rr] + ((long)lr * W + ls) * C
@@ -180,12 +216,17 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
             int j = trow + 32 * rr;            // row within tile
             int col = n0 + j;                  // output channel / input channel
             if (col < Nout) {
-                const unsigned short* wsrc;
-                if constexpr (!DGRAD)
-                    wsrc = wgt + (((long)col * R + lr) * S + ls) * C + c0 + cc8;
-                else
-                    wsrc = wgt + (((long)(lr * S + ls) * C) + col) * (long)K + c0 + cc8;
-                breg[rr] = load16(wsrc, c0 + cc8, Cin);
+                if constexpr (SMALL) {
+                    breg[rr] = load16(wgt + (long)col * R * S * C + cc8,
+                                      cc8, R * S * C);
+                } else {
+                    const unsigned short* wsrc;
+                    if constexpr (!DGRAD)
+                        wsrc = wgt + (((long)col * R + lr) * S + ls) * C + c0 + cc8;
+                    else
+                        wsrc = wgt + (((long)(lr * S + ls) * C) + col) * (long)K + c0 + cc8;
+                    breg[rr] = load16(wsrc, c0 + cc8, Cin);
+                }
             } else breg[rr] = zero16();
         }
         if (++lc == KC) { lc = 0; if (++ls == S) { ls = 0; ++lr; } }
@@ -203,19 +244,19 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
     auto mfma_step = [&](int buf) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
-            bf16x8_t a[4], b[4];
+            bf16x8_t a[4], b[NJ];
 #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
                 a[mi] = *(const bf16x8_t*)
                     &lds.A[buf][wm * 64 + mi * 16 + fr][kk * 32 + fq * 8];
 #pragma unroll
-            for (int nj = 0; nj < 4; ++nj)
+            for (int nj = 0; nj < NJ; ++nj)
                 b[nj] = *(const bf16x8_t*)
-                    &lds.B[buf][wn * 64 + nj * 16 + fr][kk * 32 + fq * 8];
+                    &lds.B[buf][wn * (NJ * 16) + nj * 16 + fr][kk * 32 + fq * 8];
 #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-                for (int nj = 0; nj < 4; ++nj)
+                for (int nj = 0; nj < NJ; ++nj)
                     acc[mi][nj] = MFMA_BF16(a[mi], b[nj], acc[mi][nj]);
         }
     };
@@ -236,8 +277,8 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
 
     // ---- epilogue ----
 #pragma unroll
-    for (int nj = 0; nj < 4; ++nj) {
-        int col = n0 + wn * 64 + nj * 16 + fr;
+    for (int nj = 0; nj < NJ; ++nj) {
+        int col = n0 + wn * (NJ * 16) + nj * 16 + fr;
         float bv = (!DGRAD && bias && col < Nout) ? bf16_to_f32(bias[col]) : 0.f;
 #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
@@ -246,6 +287,149 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
             long om = m0 + wm * 64 + mi * 16 + fq * 4 + e;
             if (om < M && col < Nout)
                 dst[om * Nout + col] = f32_to_bf16(acc[mi][nj][e] + bv);
+        }
+    }
+}
+
+// ------------------------------------------------- dgrad, stride 2
+//
+// Stride-2 dgrad by parity class: dx pixels (h,w) with h%2==hp, w%2==wp form
+// a regular subgrid, and only taps with r == (hp+pad) mod 2 and
+// s == (wp+pad) mod 2 contribute — so instead of predicating 3/4 of the MFMA
+// work to zero (what a direct stride-2 GEMM does), launch 4 kernels, one per
+// class, each contracting only its valid (r,s) subset at full tile density.
+template <int TM, int TN>
+__global__ __launch_bounds__(256) void conv_dgrad2_kernel(
+    const unsigned short* __restrict__ dout, // [Nb,P,Q,K]
+    const unsigned short* __restrict__ wgt,  // wT [R,S,C,K]
+    unsigned short* __restrict__ dx,         // [Nb,H,W,C]
+    int Nb, int H, int W, int C, int K, int P, int Q,
+    int R, int S, int pad, int hp, int wp, int Hc, int Wc)
+{
+    __shared__ __attribute__((aligned(16))) FwdLds<TM, TN, 2, true> lds;
+    constexpr int AR = TM / 32;
+    constexpr int BR = TN / 32;
+    const long M = (long)Nb * Hc * Wc;
+    const int tiles_n = (C + TN - 1) / TN;
+    const long m0 = (blockIdx.x / tiles_n) * (long)TM;
+    const int n0 = (blockIdx.x % tiles_n) * TN;
+    const int t = threadIdx.x;
+    const int trow = t >> 3;
+    const int cc8 = (t & 7) * 8;
+
+    long abase[AR]; int ax[AR], ay[AR];      // h, w of the dx pixel
+#pragma unroll
+    for (int rr = 0; rr < AR; ++rr) {
+        long m = m0 + trow + 32 * rr;
+        if (m < M) {
+            int n = (int)(m / ((long)Hc * Wc)); int rem = (int)(m % ((long)Hc * Wc));
+            ax[rr] = hp + 2 * (rem / Wc); ay[rr] = wp + 2 * (rem % Wc);
+            abase[rr] = (long)n * P * Q * K;
+        } else { ax[rr] = INT_MIN / 2; ay[rr] = 0; abase[rr] = 0; }
+    }
+
+    constexpr int WGM = TM / 64;
+    constexpr int WGN = 4 / WGM;
+    constexpr int NJ = TN / WGN / 16;
+    const int lane = t & 63, wid = t >> 6;
+    const int wm = (WGN == 1) ? wid : (wid >> 1);
+    const int wn = (WGN == 1) ? 0 : (wid & 1);
+    const int fr = lane & 15, fq = lane >> 4;
+    f32x4_t acc[4][NJ];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < NJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    const int r0 = (hp + pad) & 1, s0 = (wp + pad) & 1;
+    const int nr = (R > r0) ? ((R - r0 + 1) >> 1) : 0;
+    const int ns = (S > s0) ? ((S - s0 + 1) >> 1) : 0;
+    const int KC = (K + 63) >> 6;
+    const int nsteps = nr * ns * KC;
+    int lri = 0, lsi = 0, lkc = 0;
+    V16 areg[AR], breg[BR];
+
+    auto load_step = [&]() {
+        const int r = r0 + 2 * lri, s = s0 + 2 * lsi;
+        const int k0 = lkc << 6;
+#pragma unroll
+        for (int rr = 0; rr < AR; ++rr) {
+            int ph = ax[rr] + pad - r, pw = ay[rr] + pad - s;   // even by construction
+            int p = ph >> 1, q = pw >> 1;
+            bool v = ax[rr] >= 0 && ph >= 0 && pw >= 0 && p < P && q < Q;
+            areg[rr] = v ? load16(dout + abase[rr] + ((long)p * Q + q) * K
+                                  + k0 + cc8, k0 + cc8, K)
+                         : zero16();
+        }
+#pragma unroll
+        for (int rr = 0; rr < BR; ++rr) {
+            int col = n0 + trow + 32 * rr;
+            breg[rr] = (col < C)
+                ? load16(wgt + (((long)(r * S + s) * C) + col) * (long)K + k0 + cc8,
+                         k0 + cc8, K)
+                : zero16();
+        }
+        if (++lkc == KC) { lkc = 0; if (++lsi == ns) { lsi = 0; ++lri; } }
+    };
+
+    auto write_lds = [&](int buf) {
+#pragma unroll
+        for (int rr = 0; rr < AR; ++rr)
+            *(uint4*)&lds.A[buf][trow + 32 * rr][cc8] = areg[rr].u4;
+#pragma unroll
+        for (int rr = 0; rr < BR; ++rr)
+            *(uint4*)&lds.B[buf][trow + 32 * rr][cc8] = breg[rr].u4;
+    };
+
+    auto mfma_step = [&](int buf) {
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+            bf16x8_t a[4], b[NJ];
+#pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+                a[mi] = *(const bf16x8_t*)
+                    &lds.A[buf][wm * 64 + mi * 16 + fr][kk * 32 + fq * 8];
+#pragma unroll
+            for (int nj = 0; nj < NJ; ++nj)
+                b[nj] = *(const bf16x8_t*)
+                    &lds.B[buf][wn * (NJ * 16) + nj * 16 + fr][kk * 32 + fq * 8];
+#pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+                for (int nj = 0; nj < NJ; ++nj)
+                    acc[mi][nj] = MFMA_BF16(a[mi], b[nj], acc[mi][nj]);
+        }
+    };
+
+    if (nsteps > 0) {
+        load_step();
+        write_lds(0);
+        if (nsteps > 1) load_step();
+        __syncthreads();
+        for (int it = 0; it < nsteps; ++it) {
+            if (it + 1 < nsteps) {
+                write_lds((it + 1) & 1);
+                if (it + 2 < nsteps) load_step();
+            }
+            mfma_step(it & 1);
+            __syncthreads();
+        }
+    }
+
+#pragma unroll
+    for (int nj = 0; nj < NJ; ++nj) {
+        int col = n0 + wn * (NJ * 16) + nj * 16 + fr;
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int e = 0; e < 4; ++e) {
+            long om = m0 + wm * 64 + mi * 16 + fq * 4 + e;
+            if (om < M && col < C) {
+                int n = (int)(om / ((long)Hc * Wc));
+                int rem = (int)(om % ((long)Hc * Wc));
+                int h = hp + 2 * (rem / Wc), w = wp + 2 * (rem % Wc);
+                dx[((long)(n * H + h) * W + w) * C + col] = f32_to_bf16(acc[mi][nj][e]);
+            }
         }
     }
 }
@@ -438,23 +622,16 @@ __global__ __launch_bounds__(256) void colsum_kernel(
 
 // ---------------------------------------------------------------- C API
 
-#define DISPATCH_GEMM(TM, TN, DG)                                             \
+#define LAUNCH_GEMM(TM, TN, ST, DG, SM)                                       \
     do {                                                                      \
         long M_ = DG ? (long)Nb * H * W : (long)Nb * P * Q;                   \
         int Nout_ = DG ? C : K;                                               \
         long grid = ((M_ + TM - 1) / TM) * ((Nout_ + TN - 1) / TN);           \
-        if (stride == 1)                                                      \
-            hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, 1, DG>),             \
-                dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,        \
-                (const unsigned short*)src, (const unsigned short*)wgt,       \
-                (const unsigned short*)bias, (unsigned short*)dst,            \
-                Nb, H, W, C, K, P, Q, R, S, pad);                             \
-        else                                                                  \
-            hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, 2, DG>),             \
-                dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,        \
-                (const unsigned short*)src, (const unsigned short*)wgt,       \
-                (const unsigned short*)bias, (unsigned short*)dst,            \
-                Nb, H, W, C, K, P, Q, R, S, pad);                             \
+        hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, ST, DG, SM>),            \
+            dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,            \
+            (const unsigned short*)src, (const unsigned short*)wgt,           \
+            (const unsigned short*)bias, (unsigned short*)dst,                \
+            Nb, H, W, C, K, P, Q, R, S, pad);                                 \
     } while (0)
 
 extern "C" void ps_conv_fwd(
@@ -462,8 +639,16 @@ extern "C" void ps_conv_fwd(
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int stride, int pad, void* strm)
 {
-    if (K >= 128) DISPATCH_GEMM(128, 128, false);
-    else          DISPATCH_GEMM(256, 64, false);
+    if (R * S * C <= 64) {          // flattened-contraction stem path
+        if (stride == 1) LAUNCH_GEMM(128, 64, 1, false, true);
+        else             LAUNCH_GEMM(128, 64, 2, false, true);
+    } else if (K >= 128) {
+        if (stride == 1) LAUNCH_GEMM(128, 128, 1, false, false);
+        else             LAUNCH_GEMM(128, 128, 2, false, false);
+    } else {
+        if (stride == 1) LAUNCH_GEMM(128, 64, 1, false, false);
+        else             LAUNCH_GEMM(128, 64, 2, false, false);
+    }
 }
 
 // wgt here is the TRANSPOSED weight wT[R,S,C,K] (host permutes once per
@@ -474,8 +659,27 @@ extern "C" void ps_conv_dgrad(
     int R, int S, int stride, int pad, void* strm)
 {
     const void* bias = nullptr;
-    if (C >= 128) DISPATCH_GEMM(128, 128, true);
-    else          DISPATCH_GEMM(256, 64, true);
+    if (stride == 1) {
+        if (C >= 128) LAUNCH_GEMM(128, 128, 1, true, false);
+        else          LAUNCH_GEMM(128, 64, 1, true, false);
+    } else {
+        // 4 parity-class launches (see conv_dgrad2_kernel)
+        const unsigned short* dout = (const unsigned short*)src;
+        for (int hp = 0; hp < 2 && hp < H; ++hp)
+        for (int wp = 0; wp < 2 && wp < W; ++wp) {
+            int Hc = (H - hp + 1) >> 1, Wc = (W - wp + 1) >> 1;
+            long M_ = (long)Nb * Hc * Wc;
+#define DG2(TM, TN)                                                           \
+            hipLaunchKernelGGL((conv_dgrad2_kernel<TM, TN>),                  \
+                dim3((unsigned)(((M_ + TM - 1) / TM) * ((C + TN - 1) / TN))), \
+                dim3(256), 0, (hipStream_t)strm, dout,                        \
+                (const unsigned short*)wgt, (unsigned short*)dst,             \
+                Nb, H, W, C, K, P, Q, R, S, pad, hp, wp, Hc, Wc)
+            if (C >= 128) DG2(128, 128);
+            else          DG2(128, 64);
+#undef DG2
+        }
+    }
 }
 
 static inline int ilog2_exact(long v) {
