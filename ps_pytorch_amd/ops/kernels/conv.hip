@@ -79,11 +79,20 @@ __device__ __forceinline__ V16 load16(const unsigned short* src, int cc, int Cn)
 // DGRAD=true : rows are input pixels, cols C, contraction (r,s,k),
 //              weights given as wT[R,S,C,K].
 
-template <int TM, int TN, int STRIDE, bool DGRAD>
+// TN==64 tiles drop the 72-pad and XOR-swizzle 8-element granules instead
+// (same bank spreading, 11% less LDS) — that fits 2 blocks/CU at 256x64 and
+// 3 at 128x64, which the padded layout cannot.
+template <int TM, int TN>
 struct FwdLds {
-    unsigned short A[2][TM][LDSP];
-    unsigned short B[2][TN][LDSP];
+    static constexpr int PITCH = (TN == 64) ? 64 : LDSP;
+    unsigned short A[2][TM][PITCH];
+    unsigned short B[2][TN][PITCH];
 };
+
+template <bool SWZ>
+__device__ __forceinline__ int gswz(int row, int g) {
+    return SWZ ? (g ^ (row & 7) ^ ((row >> 3) & 7)) : g;
+}
 
 // SMALL=true: the whole R*S*C contraction fits one 64-chunk (ResNet stem
 // 3*3*3=27, LeNet conv1 5*5*1=25) — flatten (r,s,c) into the contraction
@@ -98,7 +107,8 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int pad)
 {
-    __shared__ __attribute__((aligned(16))) FwdLds<TM, TN, STRIDE, DGRAD> lds;
+    __shared__ __attribute__((aligned(16))) FwdLds<TM, TN> lds;
+    constexpr bool SWZ = (TN == 64);
     constexpr int AR = TM / 32;        // A rows staged per thread
     constexpr int BR = TN / 32;        // B rows staged per thread
     const int Cin = DGRAD ? K : C;     // contraction channel count
@@ -234,11 +244,15 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
 
     auto write_lds = [&](int buf) {
 #pragma unroll
-        for (int rr = 0; rr < AR; ++rr)
-            *(uint4*)&lds.A[buf][trow + 32 * rr][cc8] = areg[rr].u4;
+        for (int rr = 0; rr < AR; ++rr) {
+            int row = trow + 32 * rr;
+            *(uint4*)&lds.A[buf][row][gswz<SWZ>(row, t & 7) * 8] = areg[rr].u4;
+        }
 #pragma unroll
-        for (int rr = 0; rr < BR; ++rr)
-            *(uint4*)&lds.B[buf][trow + 32 * rr][cc8] = breg[rr].u4;
+        for (int rr = 0; rr < BR; ++rr) {
+            int row = trow + 32 * rr;
+            *(uint4*)&lds.B[buf][row][gswz<SWZ>(row, t & 7) * 8] = breg[rr].u4;
+        }
     };
 
     auto mfma_step = [&](int buf) {
@@ -246,13 +260,17 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         for (int kk = 0; kk < 2; ++kk) {
             bf16x8_t a[4], b[NJ];
 #pragma unroll
-            for (int mi = 0; mi < 4; ++mi)
+            for (int mi = 0; mi < 4; ++mi) {
+                int row = wm * 64 + mi * 16 + fr;
                 a[mi] = *(const bf16x8_t*)
-                    &lds.A[buf][wm * 64 + mi * 16 + fr][kk * 32 + fq * 8];
+                    &lds.A[buf][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
+            }
 #pragma unroll
-            for (int nj = 0; nj < NJ; ++nj)
+            for (int nj = 0; nj < NJ; ++nj) {
+                int row = wn * (NJ * 16) + nj * 16 + fr;
                 b[nj] = *(const bf16x8_t*)
-                    &lds.B[buf][wn * (NJ * 16) + nj * 16 + fr][kk * 32 + fq * 8];
+                    &lds.B[buf][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
+            }
 #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
@@ -306,7 +324,8 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int pad, int hp, int wp, int Hc, int Wc)
 {
-    __shared__ __attribute__((aligned(16))) FwdLds<TM, TN, 2, true> lds;
+    __shared__ __attribute__((aligned(16))) FwdLds<TM, TN> lds;
+    constexpr bool SWZ = (TN == 64);
     constexpr int AR = TM / 32;
     constexpr int BR = TN / 32;
     const long M = (long)Nb * Hc * Wc;
@@ -374,11 +393,15 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
 
     auto write_lds = [&](int buf) {
 #pragma unroll
-        for (int rr = 0; rr < AR; ++rr)
-            *(uint4*)&lds.A[buf][trow + 32 * rr][cc8] = areg[rr].u4;
+        for (int rr = 0; rr < AR; ++rr) {
+            int row = trow + 32 * rr;
+            *(uint4*)&lds.A[buf][row][gswz<SWZ>(row, t & 7) * 8] = areg[rr].u4;
+        }
 #pragma unroll
-        for (int rr = 0; rr < BR; ++rr)
-            *(uint4*)&lds.B[buf][trow + 32 * rr][cc8] = breg[rr].u4;
+        for (int rr = 0; rr < BR; ++rr) {
+            int row = trow + 32 * rr;
+            *(uint4*)&lds.B[buf][row][gswz<SWZ>(row, t & 7) * 8] = breg[rr].u4;
+        }
     };
 
     auto mfma_step = [&](int buf) {
@@ -386,13 +409,17 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
         for (int kk = 0; kk < 2; ++kk) {
             bf16x8_t a[4], b[NJ];
 #pragma unroll
-            for (int mi = 0; mi < 4; ++mi)
+            for (int mi = 0; mi < 4; ++mi) {
+                int row = wm * 64 + mi * 16 + fr;
                 a[mi] = *(const bf16x8_t*)
-                    &lds.A[buf][wm * 64 + mi * 16 + fr][kk * 32 + fq * 8];
+                    &lds.A[buf][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
+            }
 #pragma unroll
-            for (int nj = 0; nj < NJ; ++nj)
+            for (int nj = 0; nj < NJ; ++nj) {
+                int row = wn * (NJ * 16) + nj * 16 + fr;
                 b[nj] = *(const bf16x8_t*)
-                    &lds.B[buf][wn * (NJ * 16) + nj * 16 + fr][kk * 32 + fq * 8];
+                    &lds.B[buf][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
+            }
 #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
@@ -443,15 +470,18 @@ __device__ __forceinline__ int wg_swz(int row, int g) {
     return g ^ (row & 7) ^ ((row >> 3) & 7);
 }
 
+template <int TK>
 struct WgradLds {
-    unsigned short A[2][64][64];   // [k][pixel]
+    unsigned short A[2][TK][64];   // [k][pixel]
     unsigned short B[2][64][64];   // [c][pixel]
 };
 
-// One block: 64 k x 64 c output tile for ONE (r,s), summing the pixel range
+// One block: TK k x 64 c output tile for ONE (r,s), summing the pixel range
 // [sid*chunk, ...) of length `chunk`; f32 partial out[sid][K][R*S*C].
+// TK=128 for K>=128 layers (halves the scatter cost per MFMA and the
+// dout re-reads); TK=64 otherwise.
 // POW2: P*Q and Q are powers of two (shift decode); else runtime div.
-template <int STRIDE, bool POW2>
+template <int STRIDE, bool POW2, int TK>
 __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     const unsigned short* __restrict__ dout,  // [Nb,P,Q,K]
     const unsigned short* __restrict__ in,    // [Nb,H,W,C]
@@ -460,10 +490,12 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     int R, int S, int pad, int split, int chunk,
     int l2pq, int l2q, int per_xcd)
 {
-    __shared__ __attribute__((aligned(16))) WgradLds lds;
+    __shared__ __attribute__((aligned(16))) WgradLds<TK> lds;
+    constexpr int KH = TK / 64;        // 64-wide k sub-chunks per tile
+    constexpr int MI = TK / 32;        // 16-row mfma tiles per wave (k dim)
     const long M = (long)Nb * P * Q;
     const int RSC = R * S * C;
-    const int tiles_k = (K + 63) >> 6;
+    const int tiles_k = (K + TK - 1) / TK;
     const int tiles_c = (C + 63) >> 6;
     // XCD swizzle: logical index l groups all tiles of one pixel chunk on
     // one XCD (physical block b runs on XCD b%8).
@@ -471,7 +503,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     const long nlog = (long)tiles_k * tiles_c * R * S * split;
     if (l >= nlog) return;
     int b = (int)l;
-    const int k0 = (b % tiles_k) * 64; b /= tiles_k;
+    const int k0 = (b % tiles_k) * TK; b /= tiles_k;
     const int c0 = (b % tiles_c) * 64; b /= tiles_c;
     const int rs = b % (R * S); b /= (R * S);
     const int r = rs / S, s = rs % S;
@@ -485,21 +517,23 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     const int lane = t & 63, wid = t >> 6;
     const int wm = wid >> 1, wn = wid & 1;
     const int fr = lane & 15, fq = lane >> 4;
-    f32x4_t acc[2][2];
+    f32x4_t acc[MI][2];
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
+    for (int i = 0; i < MI; ++i)
 #pragma unroll
         for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
     const int nsteps = (int)((mend - mbeg + 63) >> 6);
     long lm = mbeg;                     // load pointer
-    V16 areg[2], breg[2];
+    V16 areg[2][KH], breg[2];
 
     auto load_step = [&]() {
 #pragma unroll
         for (int rr = 0; rr < 2; ++rr) {
             long m = lm + trow + 32 * rr;
-            V16 av = zero16(), bv = zero16();
+            V16 bv = zero16();
+#pragma unroll
+            for (int h = 0; h < KH; ++h) areg[rr][h] = zero16();
             if (m < mend) {
                 int n, p, q;
                 if constexpr (POW2) {
@@ -511,13 +545,16 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
                     int rem = (int)(m % ((long)P * Q));
                     p = rem / Q; q = rem % Q;
                 }
-                av = load16(dout + m * K + k0 + cc8, k0 + cc8, K);
+#pragma unroll
+                for (int h = 0; h < KH; ++h)
+                    areg[rr][h] = load16(dout + m * K + k0 + h * 64 + cc8,
+                                         k0 + h * 64 + cc8, K);
                 int hh = p * STRIDE - pad + r, ww = q * STRIDE - pad + s;
                 if (hh >= 0 && hh < H && ww >= 0 && ww < W)
                     bv = load16(in + ((long)(n * H + hh) * W + ww) * C + c0 + cc8,
                                 c0 + cc8, C);
             }
-            areg[rr] = av; breg[rr] = bv;
+            breg[rr] = bv;
         }
         lm += 64;
     };
@@ -531,7 +568,9 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
             for (int u = 0; u < 8; ++u) {
                 int row = cc8 + u;
                 int g = wg_swz(row, g0);
-                lds.A[buf][row][g * 8 + o] = areg[rr].us[u];
+#pragma unroll
+                for (int h = 0; h < KH; ++h)
+                    lds.A[buf][h * 64 + row][g * 8 + o] = areg[rr][h].us[u];
                 lds.B[buf][row][g * 8 + o] = breg[rr].us[u];
             }
         }
@@ -541,10 +580,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
             int gg = kk * 4 + fq;
-            bf16x8_t a[2], bfr[2];
+            bf16x8_t a[MI], bfr[2];
 #pragma unroll
-            for (int mi = 0; mi < 2; ++mi) {
-                int row = wm * 32 + mi * 16 + fr;
+            for (int mi = 0; mi < MI; ++mi) {
+                int row = wm * (MI * 16) + mi * 16 + fr;
                 a[mi] = *(const bf16x8_t*)&lds.A[buf][row][wg_swz(row, gg) * 8];
             }
 #pragma unroll
@@ -553,7 +592,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
                 bfr[nj] = *(const bf16x8_t*)&lds.B[buf][row][wg_swz(row, gg) * 8];
             }
 #pragma unroll
-            for (int mi = 0; mi < 2; ++mi)
+            for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
                 for (int nj = 0; nj < 2; ++nj)
                     acc[mi][nj] = MFMA_BF16(a[mi], bfr[nj], acc[mi][nj]);
@@ -575,12 +614,12 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
 
     float* dstp = partial + (long)sid * K * RSC;
 #pragma unroll
-    for (int mi = 0; mi < 2; ++mi)
+    for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
     for (int nj = 0; nj < 2; ++nj)
 #pragma unroll
     for (int e = 0; e < 4; ++e) {
-        int k = k0 + wm * 32 + mi * 16 + fq * 4 + e;
+        int k = k0 + wm * (MI * 16) + mi * 16 + fq * 4 + e;
         int c = c0 + wn * 32 + nj * 16 + fr;
         if (k < K && c < C)
             dstp[(long)k * RSC + (r * S + s) * C + c] = acc[mi][nj][e];
@@ -639,15 +678,15 @@ extern "C" void ps_conv_fwd(
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int stride, int pad, void* strm)
 {
-    if (R * S * C <= 64) {          // flattened-contraction stem path
+    if (R * S > 1 && R * S * C <= 64) {   // flattened-contraction stem path
         if (stride == 1) LAUNCH_GEMM(128, 64, 1, false, true);
         else             LAUNCH_GEMM(128, 64, 2, false, true);
     } else if (K >= 128) {
         if (stride == 1) LAUNCH_GEMM(128, 128, 1, false, false);
         else             LAUNCH_GEMM(128, 128, 2, false, false);
     } else {
-        if (stride == 1) LAUNCH_GEMM(128, 64, 1, false, false);
-        else             LAUNCH_GEMM(128, 64, 2, false, false);
+        if (stride == 1) LAUNCH_GEMM(256, 64, 1, false, false);
+        else             LAUNCH_GEMM(256, 64, 2, false, false);
     }
 }
 
@@ -663,10 +702,23 @@ extern "C" void ps_conv_dgrad(
         if (C >= 128) LAUNCH_GEMM(128, 128, 1, true, false);
         else          LAUNCH_GEMM(128, 64, 1, true, false);
     } else {
-        // 4 parity-class launches (see conv_dgrad2_kernel)
+        // 4 parity-class launches (see conv_dgrad2_kernel); classes with no
+        // contributing (r,s) — e.g. 3 of 4 for a 1x1 stride-2 conv — are
+        // covered by one bulk memset instead of scattered zero stores.
         const unsigned short* dout = (const unsigned short*)src;
+        bool any_empty = false;
         for (int hp = 0; hp < 2 && hp < H; ++hp)
         for (int wp = 0; wp < 2 && wp < W; ++wp) {
+            int r0 = (hp + pad) & 1, s0 = (wp + pad) & 1;
+            if (R <= r0 || S <= s0) any_empty = true;
+        }
+        if (any_empty)
+            (void)hipMemsetAsync(dst, 0, (long)Nb * H * W * C * 2,
+                                 (hipStream_t)strm);
+        for (int hp = 0; hp < 2 && hp < H; ++hp)
+        for (int wp = 0; wp < 2 && wp < W; ++wp) {
+            int r0 = (hp + pad) & 1, s0 = (wp + pad) & 1;
+            if (R <= r0 || S <= s0) continue;
             int Hc = (H - hp + 1) >> 1, Wc = (W - wp + 1) >> 1;
             long M_ = (long)Nb * Hc * Wc;
 #define DG2(TM, TN)                                                           \
@@ -695,20 +747,24 @@ extern "C" void ps_conv_wgrad(
     long M = (long)Nb * P * Q;
     long chunk64 = (M + (long)split * 64 - 1) / ((long)split * 64);
     int chunk = (int)(chunk64 * 64);
-    int tiles_k = (K + 63) / 64, tiles_c = (C + 63) / 64;
+    int TK = (K >= 128) ? 128 : 64;
+    int tiles_k = (K + TK - 1) / TK, tiles_c = (C + 63) / 64;
     long nlog = (long)tiles_k * tiles_c * R * S * split;
     int per_xcd = (int)((nlog + 7) / 8);
     long grid = (long)per_xcd * 8;
     int l2pq = ilog2_exact((long)P * Q), l2q = ilog2_exact(Q);
     bool pow2 = l2pq >= 0 && l2q >= 0;
-#define WG_LAUNCH(ST, PW)                                                     \
-    hipLaunchKernelGGL((conv_wgrad_kernel<ST, PW>), dim3((unsigned)grid),     \
-        dim3(256), 0, (hipStream_t)strm,                                      \
+#define WG_LAUNCH(ST, PW, TKV)                                                \
+    hipLaunchKernelGGL((conv_wgrad_kernel<ST, PW, TKV>),                      \
+        dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,                \
         (const unsigned short*)dout, (const unsigned short*)in,               \
         (float*)partial_f32, Nb, H, W, C, K, P, Q, R, S, pad, split, chunk,   \
         l2pq, l2q, per_xcd)
-    if (stride == 1) { if (pow2) WG_LAUNCH(1, true); else WG_LAUNCH(1, false); }
-    else             { if (pow2) WG_LAUNCH(2, true); else WG_LAUNCH(2, false); }
+#define WG_TK(ST, PW) do { if (TK == 128) WG_LAUNCH(ST, PW, 128);             \
+                           else WG_LAUNCH(ST, PW, 64); } while (0)
+    if (stride == 1) { if (pow2) WG_TK(1, true); else WG_TK(1, false); }
+    else             { if (pow2) WG_TK(2, true); else WG_TK(2, false); }
+#undef WG_TK
 #undef WG_LAUNCH
     long n = (long)K * R * S * C;
     int blocks; ew_grid(n, 256, &blocks);
