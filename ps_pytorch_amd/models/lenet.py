@@ -15,14 +15,16 @@ from __future__ import annotations
 
 import torch
 import torch.nn as nn
+
+from ..ops.conv import PsConv2d
 import torch.nn.functional as F
 
 
 class LeNet(nn.Module):
     def __init__(self, num_classes: int = 10, in_channels: int = 1):
         super().__init__()
-        self.conv1 = nn.Conv2d(in_channels, 20, kernel_size=5)
-        self.conv2 = nn.Conv2d(20, 50, kernel_size=5)
+        self.conv1 = PsConv2d(in_channels, 20, kernel_size=5)
+        self.conv2 = PsConv2d(20, 50, kernel_size=5)
         # 28x28 -> conv5 -> 24 -> pool -> 12 -> conv5 -> 8 -> pool -> 4
         self.fc1 = nn.Linear(50 * 4 * 4, 500)
         self.fc2 = nn.Linear(500, num_classes)

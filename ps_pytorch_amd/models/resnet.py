@@ -17,6 +17,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.conv import PsConv2d
 from ..ops.modules import PsBatchNorm2d
 
 
@@ -25,14 +26,14 @@ class BasicBlock(nn.Module):
 
     def __init__(self, in_planes: int, planes: int, stride: int = 1):
         super().__init__()
-        self.conv1 = nn.Conv2d(in_planes, planes, 3, stride=stride, padding=1, bias=False)
+        self.conv1 = PsConv2d(in_planes, planes, 3, stride=stride, padding=1, bias=False)
         self.bn1 = PsBatchNorm2d(planes, relu=True)
-        self.conv2 = nn.Conv2d(planes, planes, 3, stride=1, padding=1, bias=False)
+        self.conv2 = PsConv2d(planes, planes, 3, stride=1, padding=1, bias=False)
         self.bn2 = PsBatchNorm2d(planes, relu=True)   # fused: relu(bn + shortcut)
         self.shortcut = nn.Sequential()
         if stride != 1 or in_planes != planes * self.expansion:
             self.shortcut = nn.Sequential(
-                nn.Conv2d(in_planes, planes * self.expansion, 1, stride=stride, bias=False),
+                PsConv2d(in_planes, planes * self.expansion, 1, stride=stride, bias=False),
                 PsBatchNorm2d(planes * self.expansion),
             )
 
@@ -46,16 +47,16 @@ class Bottleneck(nn.Module):
 
     def __init__(self, in_planes: int, planes: int, stride: int = 1):
         super().__init__()
-        self.conv1 = nn.Conv2d(in_planes, planes, 1, bias=False)
+        self.conv1 = PsConv2d(in_planes, planes, 1, bias=False)
         self.bn1 = PsBatchNorm2d(planes, relu=True)
-        self.conv2 = nn.Conv2d(planes, planes, 3, stride=stride, padding=1, bias=False)
+        self.conv2 = PsConv2d(planes, planes, 3, stride=stride, padding=1, bias=False)
         self.bn2 = PsBatchNorm2d(planes, relu=True)
-        self.conv3 = nn.Conv2d(planes, planes * self.expansion, 1, bias=False)
+        self.conv3 = PsConv2d(planes, planes * self.expansion, 1, bias=False)
         self.bn3 = PsBatchNorm2d(planes * self.expansion, relu=True)
         self.shortcut = nn.Sequential()
         if stride != 1 or in_planes != planes * self.expansion:
             self.shortcut = nn.Sequential(
-                nn.Conv2d(in_planes, planes * self.expansion, 1, stride=stride, bias=False),
+                PsConv2d(in_planes, planes * self.expansion, 1, stride=stride, bias=False),
                 PsBatchNorm2d(planes * self.expansion),
             )
 
@@ -72,9 +73,9 @@ class ResNet(nn.Module):
         self.in_planes = 64
         self.imagenet_stem = imagenet_stem
         if imagenet_stem:
-            self.conv1 = nn.Conv2d(in_channels, 64, 7, stride=2, padding=3, bias=False)
+            self.conv1 = PsConv2d(in_channels, 64, 7, stride=2, padding=3, bias=False)
         else:
-            self.conv1 = nn.Conv2d(in_channels, 64, 3, stride=1, padding=1, bias=False)
+            self.conv1 = PsConv2d(in_channels, 64, 3, stride=1, padding=1, bias=False)
         self.bn1 = PsBatchNorm2d(64, relu=True)
         self.layer1 = self._make_layer(block, 64, num_blocks[0], stride=1)
         self.layer2 = self._make_layer(block, 128, num_blocks[1], stride=2)

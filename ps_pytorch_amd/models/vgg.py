@@ -4,6 +4,7 @@ from __future__ import annotations
 
 import torch.nn as nn
 
+from ..ops.conv import PsConv2d
 from ..ops.modules import PsBatchNorm2d
 
 _CFG = {
@@ -23,7 +24,7 @@ def _make_layers(cfg, in_channels: int, batch_norm: bool) -> nn.Sequential:
         if v == 'M':
             layers.append(nn.MaxPool2d(2, 2))
         else:
-            layers.append(nn.Conv2d(c, v, 3, padding=1, bias=not batch_norm))
+            layers.append(PsConv2d(c, v, 3, padding=1, bias=not batch_norm))
             if batch_norm:
                 layers.append(PsBatchNorm2d(v, relu=True))   # fused BN+ReLU
             else:

@@ -11,6 +11,8 @@ state_dict surface are identical (evaluator/checkpoint compatible).
 """
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
@@ -19,9 +21,14 @@ from . import require_lib, current_stream_ptr
 
 _CL = torch.channels_last
 
+# A/B kill-switch: PS_CONV=0 routes every conv through torch/MIOpen.
+_ENABLED = os.environ.get('PS_CONV', '1') != '0'
+
 
 def _supported(x: torch.Tensor, w: torch.Tensor, stride, padding,
                dilation, groups) -> bool:
+    if not _ENABLED:
+        return False
     if not (x.is_cuda and x.dim() == 4 and x.dtype == torch.bfloat16):
         return False
     if groups != 1 or dilation[0] != 1 or dilation[1] != 1:
@@ -37,7 +44,11 @@ def _supported(x: torch.Tensor, w: torch.Tensor, stride, padding,
 def _wgrad_split(M: int, K: int, C: int, R: int, S: int) -> int:
     """Pick the split-K factor: enough blocks to fill 256 CUs (~8 blocks/CU)
     without exploding the f32 partial buffer."""
-    tiles = ((K + 63) // 64) * R * S * ((C + 63) // 64)
+    if R * S > 1 and R * S * C <= 64:       # flattened small-RSC kernel
+        tiles = (K + 63) // 64
+    else:
+        tk = 128 if K >= 128 else 64
+        tiles = ((K + tk - 1) // tk) * R * S * ((C + 63) // 64)
     want = max(1, 2048 // max(tiles, 1))
     max_split = max(1, M // 64)
     return max(1, min(want, max_split, 256))

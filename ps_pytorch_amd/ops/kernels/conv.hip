@@ -626,6 +626,151 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     }
 }
 
+// Small-RSC wgrad (ResNet stem 3x3x3=27, LeNet conv1 5x5x1=25): the whole
+// flattened (r,s,c) axis fits one 64-column tile, so one block covers every
+// tap in a single pass over its pixel chunk — vs the generic kernel's
+// R*S separate 64-c tiles at C/64 utilization each.
+template <int STRIDE, bool POW2>
+__global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
+    const unsigned short* __restrict__ dout,  // [Nb,P,Q,K]
+    const unsigned short* __restrict__ in,    // [Nb,H,W,C]
+    float* __restrict__ partial,              // [SPLIT][K][R*S*C]
+    int Nb, int H, int W, int C, int K, int P, int Q,
+    int R, int S, int pad, int split, int chunk,
+    int l2pq, int l2q)
+{
+    __shared__ __attribute__((aligned(16))) WgradLds<64> lds;
+    const long M = (long)Nb * P * Q;
+    const int RSC = R * S * C;
+    const int tiles_k = (K + 63) >> 6;
+    int b = blockIdx.x;
+    const int k0 = (b % tiles_k) * 64; b /= tiles_k;
+    const int sid = b;
+    const long mbeg = (long)sid * chunk;
+    const long mend = (mbeg + chunk < M) ? mbeg + chunk : M;
+
+    const int t = threadIdx.x;
+    const int trow = t >> 3;
+    const int cc8 = (t & 7) * 8;
+    const int lane = t & 63, wid = t >> 6;
+    const int wm = wid >> 1, wn = wid & 1;
+    const int fr = lane & 15, fq = lane >> 4;
+    f32x4_t acc[2][2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    // per-lane flattened (r,s,c) gather table (as the SMALL fwd path)
+    int tre[8], tse[8], tce[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+        int e = cc8 + u;
+        tre[u] = e / (S * C);
+        tse[u] = (e / C) % S;
+        tce[u] = e % C;
+    }
+
+    const int nsteps = (int)((mend - mbeg + 63) >> 6);
+    long lm = mbeg;
+    V16 areg[2], breg[2];
+
+    auto load_step = [&]() {
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr) {
+            long m = lm + trow + 32 * rr;
+            V16 av = zero16(), bv = zero16();
+            if (m < mend) {
+                int n, p, q;
+                if constexpr (POW2) {
+                    n = (int)(m >> l2pq);
+                    int rem = (int)m & ((1 << l2pq) - 1);
+                    p = rem >> l2q; q = rem & ((1 << l2q) - 1);
+                } else {
+                    n = (int)(m / ((long)P * Q));
+                    int rem = (int)(m % ((long)P * Q));
+                    p = rem / Q; q = rem % Q;
+                }
+                av = load16(dout + m * K + k0 + cc8, k0 + cc8, K);
+                int h0 = p * STRIDE - pad, w0 = q * STRIDE - pad;
+                long nb = (long)n * H * W * C;
+#pragma unroll
+                for (int u = 0; u < 8; ++u) {
+                    int hh = h0 + tre[u], ww = w0 + tse[u];
+                    if (cc8 + u < RSC && hh >= 0 && hh < H && ww >= 0 && ww < W)
+                        bv.us[u] = in[nb + ((long)hh * W + ww) * C + tce[u]];
+                }
+            }
+            areg[rr] = av; breg[rr] = bv;
+        }
+        lm += 64;
+    };
+
+    auto write_lds = [&](int buf) {
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr) {
+            int midx = trow + 32 * rr;
+            int g0 = midx >> 3, o = midx & 7;
+#pragma unroll
+            for (int u = 0; u < 8; ++u) {
+                int row = cc8 + u;
+                int g = wg_swz(row, g0);
+                lds.A[buf][row][g * 8 + o] = areg[rr].us[u];
+                lds.B[buf][row][g * 8 + o] = breg[rr].us[u];
+            }
+        }
+    };
+
+    auto mfma_step = [&](int buf) {
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+            int gg = kk * 4 + fq;
+            bf16x8_t a[2], bfr[2];
+#pragma unroll
+            for (int mi = 0; mi < 2; ++mi) {
+                int row = wm * 32 + mi * 16 + fr;
+                a[mi] = *(const bf16x8_t*)&lds.A[buf][row][wg_swz(row, gg) * 8];
+            }
+#pragma unroll
+            for (int nj = 0; nj < 2; ++nj) {
+                int row = wn * 32 + nj * 16 + fr;
+                bfr[nj] = *(const bf16x8_t*)&lds.B[buf][row][wg_swz(row, gg) * 8];
+            }
+#pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+                for (int nj = 0; nj < 2; ++nj)
+                    acc[mi][nj] = MFMA_BF16(a[mi], bfr[nj], acc[mi][nj]);
+        }
+    };
+
+    load_step();
+    write_lds(0);
+    if (nsteps > 1) load_step();
+    __syncthreads();
+    for (int it = 0; it < nsteps; ++it) {
+        if (it + 1 < nsteps) {
+            write_lds((it + 1) & 1);
+            if (it + 2 < nsteps) load_step();
+        }
+        mfma_step(it & 1);
+        __syncthreads();
+    }
+
+    float* dstp = partial + (long)sid * K * RSC;
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int nj = 0; nj < 2; ++nj)
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+        int k = k0 + wm * 32 + mi * 16 + fq * 4 + e;
+        int col = wn * 32 + nj * 16 + fr;
+        if (k < K && col < RSC)
+            dstp[(long)k * RSC + col] = acc[mi][nj][e];
+    }
+}
+
 // deterministic slab reduce: dw[e] = sum_s partial[s][e] (fixed order)
 __global__ __launch_bounds__(256) void reduce_slabs_kernel(
     unsigned short* __restrict__ dw, const float* __restrict__ partial,
@@ -747,6 +892,27 @@ extern "C" void ps_conv_wgrad(
     long M = (long)Nb * P * Q;
     long chunk64 = (M + (long)split * 64 - 1) / ((long)split * 64);
     int chunk = (int)(chunk64 * 64);
+    if (R * S > 1 && R * S * C <= 64) {     // flattened stem/LeNet path
+        int tiles_k = (K + 63) / 64;
+        long grid_s = (long)tiles_k * split;
+        int l2pq_ = ilog2_exact((long)P * Q), l2q_ = ilog2_exact(Q);
+        bool pw = l2pq_ >= 0 && l2q_ >= 0;
+#define WGS(ST, PW)                                                           \
+        hipLaunchKernelGGL((conv_wgrad_small_kernel<ST, PW>),                 \
+            dim3((unsigned)grid_s), dim3(256), 0, (hipStream_t)strm,          \
+            (const unsigned short*)dout, (const unsigned short*)in,           \
+            (float*)partial_f32, Nb, H, W, C, K, P, Q, R, S, pad, split,      \
+            chunk, l2pq_, l2q_)
+        if (stride == 1) { if (pw) WGS(1, true); else WGS(1, false); }
+        else             { if (pw) WGS(2, true); else WGS(2, false); }
+#undef WGS
+        long n_ = (long)K * R * S * C;
+        int blocks_; ew_grid(n_, 256, &blocks_);
+        hipLaunchKernelGGL(reduce_slabs_kernel, dim3(blocks_), dim3(256), 0,
+                           (hipStream_t)strm, (unsigned short*)dw,
+                           (const float*)partial_f32, n_, split);
+        return;
+    }
     int TK = (K >= 128) ? 128 : 64;
     int tiles_k = (K + TK - 1) / TK, tiles_c = (C + 63) / 64;
     long nlog = (long)tiles_k * tiles_c * R * S * split;
