@@ -242,35 +242,60 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         if (++lc == KC) { lc = 0; if (++ls == S) { ls = 0; ++lr; } }
     };
 
-    auto write_lds = [&](int buf) {
+    // Precomputed LDS pointers (PMC-measured: recomputing the swizzled
+    // addresses per step made the kernels VALU-bound — ~130 VALU per 8 MFMA).
+    // Buffer 1 / array B live at constexpr ushort offsets, folded into the
+    // ds_read/ds_write immediate when `buf` is a literal at the call site.
+    constexpr int PITCH = FwdLds<TM, TN>::PITCH;
+    constexpr int A1 = TM * PITCH;                 // lds.A[1] offset
+    constexpr int B1 = TN * PITCH;                 // lds.B[1] - lds.B[0]
+    unsigned short* wrA[AR];
+    unsigned short* wrB[BR];
 #pragma unroll
-        for (int rr = 0; rr < AR; ++rr) {
-            int row = trow + 32 * rr;
-            *(uint4*)&lds.A[buf][row][gswz<SWZ>(row, t & 7) * 8] = areg[rr].u4;
+    for (int rr = 0; rr < AR; ++rr) {
+        int row = trow + 32 * rr;
+        wrA[rr] = &lds.A[0][row][gswz<SWZ>(row, t & 7) * 8];
+    }
+#pragma unroll
+    for (int rr = 0; rr < BR; ++rr) {
+        int row = trow + 32 * rr;
+        wrB[rr] = &lds.B[0][row][gswz<SWZ>(row, t & 7) * 8];
+    }
+    const unsigned short* rdA[2][4];
+    const unsigned short* rdB[2][NJ];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+            int row = wm * 64 + mi * 16 + fr;
+            rdA[kk][mi] = &lds.A[0][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
         }
 #pragma unroll
-        for (int rr = 0; rr < BR; ++rr) {
-            int row = trow + 32 * rr;
-            *(uint4*)&lds.B[buf][row][gswz<SWZ>(row, t & 7) * 8] = breg[rr].u4;
+        for (int nj = 0; nj < NJ; ++nj) {
+            int row = wn * (NJ * 16) + nj * 16 + fr;
+            rdB[kk][nj] = &lds.B[0][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
         }
+    }
+
+    auto write_lds = [&](int buf) {     // call with a LITERAL buf only
+#pragma unroll
+        for (int rr = 0; rr < AR; ++rr)
+            *(uint4*)(wrA[rr] + (buf ? A1 : 0)) = areg[rr].u4;
+#pragma unroll
+        for (int rr = 0; rr < BR; ++rr)
+            *(uint4*)(wrB[rr] + (buf ? B1 : 0)) = breg[rr].u4;
     };
 
-    auto mfma_step = [&](int buf) {
+    auto mfma_step = [&](int buf) {     // call with a LITERAL buf only
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
             bf16x8_t a[4], b[NJ];
 #pragma unroll
-            for (int mi = 0; mi < 4; ++mi) {
-                int row = wm * 64 + mi * 16 + fr;
-                a[mi] = *(const bf16x8_t*)
-                    &lds.A[buf][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
-            }
+            for (int mi = 0; mi < 4; ++mi)
+                a[mi] = *(const bf16x8_t*)(rdA[kk][mi] + (buf ? A1 : 0));
 #pragma unroll
-            for (int nj = 0; nj < NJ; ++nj) {
-                int row = wn * (NJ * 16) + nj * 16 + fr;
-                b[nj] = *(const bf16x8_t*)
-                    &lds.B[buf][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
-            }
+            for (int nj = 0; nj < NJ; ++nj)
+                b[nj] = *(const bf16x8_t*)(rdB[kk][nj] + (buf ? B1 : 0));
 #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
@@ -279,17 +304,25 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         }
     };
 
-    // ---- pipeline: 2 LDS buffers, one reg set, one barrier per step ----
+    // ---- pipeline: 2 LDS buffers, one reg set, one barrier per step,
+    // unrolled x2 so the buffer select is a compile-time immediate ----
     load_step();
     write_lds(0);
     if (nsteps > 1) load_step();
     __syncthreads();
-    for (int it = 0; it < nsteps; ++it) {
+    for (int it = 0; it < nsteps; it += 2) {
         if (it + 1 < nsteps) {
-            write_lds((it + 1) & 1);
+            write_lds(1);
             if (it + 2 < nsteps) load_step();
         }
-        mfma_step(it & 1);
+        mfma_step(0);
+        __syncthreads();
+        if (it + 1 >= nsteps) break;
+        if (it + 2 < nsteps) {
+            write_lds(0);
+            if (it + 3 < nsteps) load_step();
+        }
+        mfma_step(1);
         __syncthreads();
     }
 
@@ -391,35 +424,56 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
         if (++lkc == KC) { lkc = 0; if (++lsi == ns) { lsi = 0; ++lri; } }
     };
 
-    auto write_lds = [&](int buf) {
+    constexpr int PITCH = FwdLds<TM, TN>::PITCH;
+    constexpr int A1 = TM * PITCH;
+    constexpr int B1 = TN * PITCH;
+    unsigned short* wrA[AR];
+    unsigned short* wrB[BR];
 #pragma unroll
-        for (int rr = 0; rr < AR; ++rr) {
-            int row = trow + 32 * rr;
-            *(uint4*)&lds.A[buf][row][gswz<SWZ>(row, t & 7) * 8] = areg[rr].u4;
+    for (int rr = 0; rr < AR; ++rr) {
+        int row = trow + 32 * rr;
+        wrA[rr] = &lds.A[0][row][gswz<SWZ>(row, t & 7) * 8];
+    }
+#pragma unroll
+    for (int rr = 0; rr < BR; ++rr) {
+        int row = trow + 32 * rr;
+        wrB[rr] = &lds.B[0][row][gswz<SWZ>(row, t & 7) * 8];
+    }
+    const unsigned short* rdA[2][4];
+    const unsigned short* rdB[2][NJ];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+            int row = wm * 64 + mi * 16 + fr;
+            rdA[kk][mi] = &lds.A[0][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
         }
 #pragma unroll
-        for (int rr = 0; rr < BR; ++rr) {
-            int row = trow + 32 * rr;
-            *(uint4*)&lds.B[buf][row][gswz<SWZ>(row, t & 7) * 8] = breg[rr].u4;
+        for (int nj = 0; nj < NJ; ++nj) {
+            int row = wn * (NJ * 16) + nj * 16 + fr;
+            rdB[kk][nj] = &lds.B[0][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
         }
+    }
+
+    auto write_lds = [&](int buf) {     // literal buf only
+#pragma unroll
+        for (int rr = 0; rr < AR; ++rr)
+            *(uint4*)(wrA[rr] + (buf ? A1 : 0)) = areg[rr].u4;
+#pragma unroll
+        for (int rr = 0; rr < BR; ++rr)
+            *(uint4*)(wrB[rr] + (buf ? B1 : 0)) = breg[rr].u4;
     };
 
-    auto mfma_step = [&](int buf) {
+    auto mfma_step = [&](int buf) {     // literal buf only
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
             bf16x8_t a[4], b[NJ];
 #pragma unroll
-            for (int mi = 0; mi < 4; ++mi) {
-                int row = wm * 64 + mi * 16 + fr;
-                a[mi] = *(const bf16x8_t*)
-                    &lds.A[buf][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
-            }
+            for (int mi = 0; mi < 4; ++mi)
+                a[mi] = *(const bf16x8_t*)(rdA[kk][mi] + (buf ? A1 : 0));
 #pragma unroll
-            for (int nj = 0; nj < NJ; ++nj) {
-                int row = wn * (NJ * 16) + nj * 16 + fr;
-                b[nj] = *(const bf16x8_t*)
-                    &lds.B[buf][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
-            }
+            for (int nj = 0; nj < NJ; ++nj)
+                b[nj] = *(const bf16x8_t*)(rdB[kk][nj] + (buf ? B1 : 0));
 #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
@@ -433,12 +487,19 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
         write_lds(0);
         if (nsteps > 1) load_step();
         __syncthreads();
-        for (int it = 0; it < nsteps; ++it) {
+        for (int it = 0; it < nsteps; it += 2) {
             if (it + 1 < nsteps) {
-                write_lds((it + 1) & 1);
+                write_lds(1);
                 if (it + 2 < nsteps) load_step();
             }
-            mfma_step(it & 1);
+            mfma_step(0);
+            __syncthreads();
+            if (it + 1 >= nsteps) break;
+            if (it + 2 < nsteps) {
+                write_lds(0);
+                if (it + 3 < nsteps) load_step();
+            }
+            mfma_step(1);
             __syncthreads();
         }
     }
@@ -526,6 +587,11 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     const int nsteps = (int)((mend - mbeg + 63) >> 6);
     long lm = mbeg;                     // load pointer
     V16 areg[2][KH], breg[2];
+    // hoisted dout pointers (advance by 64*K per step)
+    const unsigned short* aptr[2];
+#pragma unroll
+    for (int rr = 0; rr < 2; ++rr)
+        aptr[rr] = dout + (mbeg + trow + 32 * rr) * K + k0 + cc8;
 
     auto load_step = [&]() {
 #pragma unroll
@@ -547,50 +613,73 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
                 }
 #pragma unroll
                 for (int h = 0; h < KH; ++h)
-                    areg[rr][h] = load16(dout + m * K + k0 + h * 64 + cc8,
-                                         k0 + h * 64 + cc8, K);
+                    areg[rr][h] = load16(aptr[rr] + h * 64, k0 + h * 64 + cc8, K);
                 int hh = p * STRIDE - pad + r, ww = q * STRIDE - pad + s;
                 if (hh >= 0 && hh < H && ww >= 0 && ww < W)
                     bv = load16(in + ((long)(n * H + hh) * W + ww) * C + c0 + cc8,
                                 c0 + cc8, C);
             }
             breg[rr] = bv;
+            aptr[rr] += (long)64 * K;
         }
         lm += 64;
     };
 
-    auto write_lds = [&](int buf) {
+    // precomputed swizzled scatter/read pointers (all offsets constexpr)
+    constexpr int A1 = TK * 64;          // lds.A[1] - lds.A[0]   (ushorts)
+    constexpr int BOF = 2 * TK * 64;     // lds.B[0] - lds.A[0]
+    constexpr int BB1 = 64 * 64;         // lds.B[1] - lds.B[0]
+    unsigned short* wr[2][8];
 #pragma unroll
-        for (int rr = 0; rr < 2; ++rr) {
-            int midx = trow + 32 * rr;
-            int g0 = midx >> 3, o = midx & 7;
+    for (int rr = 0; rr < 2; ++rr) {
+        int midx = trow + 32 * rr;
+        int g0 = midx >> 3, o = midx & 7;
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+            int row = cc8 + u;
+            wr[rr][u] = &lds.A[0][row][wg_swz(row, g0) * 8 + o];
+        }
+    }
+    const unsigned short* rdA[2][MI];
+    const unsigned short* rdB[2][2];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+        int gg = kk * 4 + fq;
+#pragma unroll
+        for (int mi = 0; mi < MI; ++mi) {
+            int row = wm * (MI * 16) + mi * 16 + fr;
+            rdA[kk][mi] = &lds.A[0][row][wg_swz(row, gg) * 8];
+        }
+#pragma unroll
+        for (int nj = 0; nj < 2; ++nj) {
+            int row = wn * 32 + nj * 16 + fr;
+            rdB[kk][nj] = &lds.B[0][row][wg_swz(row, gg) * 8];
+        }
+    }
+
+    auto write_lds = [&](int buf) {     // literal buf only
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr)
 #pragma unroll
             for (int u = 0; u < 8; ++u) {
-                int row = cc8 + u;
-                int g = wg_swz(row, g0);
+                unsigned short* p = wr[rr][u] + (buf ? A1 : 0);
 #pragma unroll
                 for (int h = 0; h < KH; ++h)
-                    lds.A[buf][h * 64 + row][g * 8 + o] = areg[rr][h].us[u];
-                lds.B[buf][row][g * 8 + o] = breg[rr].us[u];
+                    p[h * 64 * 64] = areg[rr][h].us[u];
+                (p + BOF + (buf ? BB1 - A1 : 0))[0] = breg[rr].us[u];
             }
-        }
     };
 
-    auto mfma_step = [&](int buf) {
+    auto mfma_step = [&](int buf) {     // literal buf only
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
-            int gg = kk * 4 + fq;
             bf16x8_t a[MI], bfr[2];
 #pragma unroll
-            for (int mi = 0; mi < MI; ++mi) {
-                int row = wm * (MI * 16) + mi * 16 + fr;
-                a[mi] = *(const bf16x8_t*)&lds.A[buf][row][wg_swz(row, gg) * 8];
-            }
+            for (int mi = 0; mi < MI; ++mi)
+                a[mi] = *(const bf16x8_t*)(rdA[kk][mi] + (buf ? A1 : 0));
 #pragma unroll
-            for (int nj = 0; nj < 2; ++nj) {
-                int row = wn * 32 + nj * 16 + fr;
-                bfr[nj] = *(const bf16x8_t*)&lds.B[buf][row][wg_swz(row, gg) * 8];
-            }
+            for (int nj = 0; nj < 2; ++nj)
+                bfr[nj] = *(const bf16x8_t*)(rdB[kk][nj] + (buf ? BB1 : 0));
 #pragma unroll
             for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
@@ -603,12 +692,19 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     write_lds(0);
     if (nsteps > 1) load_step();
     __syncthreads();
-    for (int it = 0; it < nsteps; ++it) {
+    for (int it = 0; it < nsteps; it += 2) {
         if (it + 1 < nsteps) {
-            write_lds((it + 1) & 1);
+            write_lds(1);
             if (it + 2 < nsteps) load_step();
         }
-        mfma_step(it & 1);
+        mfma_step(0);
+        __syncthreads();
+        if (it + 1 >= nsteps) break;
+        if (it + 2 < nsteps) {
+            write_lds(0);
+            if (it + 3 < nsteps) load_step();
+        }
+        mfma_step(1);
         __syncthreads();
     }
 
@@ -661,14 +757,15 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
 #pragma unroll
         for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-    // per-lane flattened (r,s,c) gather table (as the SMALL fwd path)
-    int tre[8], tse[8], tce[8];
+    // per-lane flattened (r,s,c) gather table (as the SMALL fwd path);
+    // toff = the lane's constant element offset from the pixel origin
+    int tre[8], tse[8], toff[8];
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
         int e = cc8 + u;
         tre[u] = e / (S * C);
         tse[u] = (e / C) % S;
-        tce[u] = e % C;
+        toff[u] = (tre[u] * W + tse[u]) * C + e % C;
     }
 
     const int nsteps = (int)((mend - mbeg + 63) >> 6);
@@ -693,12 +790,13 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
                 }
                 av = load16(dout + m * K + k0 + cc8, k0 + cc8, K);
                 int h0 = p * STRIDE - pad, w0 = q * STRIDE - pad;
-                long nb = (long)n * H * W * C;
+                const unsigned short* pix =
+                    in + (long)n * H * W * C + ((long)h0 * W + w0) * C;
 #pragma unroll
                 for (int u = 0; u < 8; ++u) {
                     int hh = h0 + tre[u], ww = w0 + tse[u];
                     if (cc8 + u < RSC && hh >= 0 && hh < H && ww >= 0 && ww < W)
-                        bv.us[u] = in[nb + ((long)hh * W + ww) * C + tce[u]];
+                        bv.us[u] = pix[toff[u]];
                 }
             }
             areg[rr] = av; breg[rr] = bv;
@@ -706,36 +804,57 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
         lm += 64;
     };
 
-    auto write_lds = [&](int buf) {
+    constexpr int A1 = 64 * 64;          // lds.A[1] - lds.A[0] (ushorts)
+    constexpr int BOF = 2 * 64 * 64;     // lds.B[0] - lds.A[0]
+    unsigned short* wr[2][8];
 #pragma unroll
-        for (int rr = 0; rr < 2; ++rr) {
-            int midx = trow + 32 * rr;
-            int g0 = midx >> 3, o = midx & 7;
+    for (int rr = 0; rr < 2; ++rr) {
+        int midx = trow + 32 * rr;
+        int g0 = midx >> 3, o = midx & 7;
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+            int row = cc8 + u;
+            wr[rr][u] = &lds.A[0][row][wg_swz(row, g0) * 8 + o];
+        }
+    }
+    const unsigned short* rdA[2][2];
+    const unsigned short* rdB[2][2];
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+        int gg = kk * 4 + fq;
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi) {
+            int row = wm * 32 + mi * 16 + fr;
+            rdA[kk][mi] = &lds.A[0][row][wg_swz(row, gg) * 8];
+        }
+#pragma unroll
+        for (int nj = 0; nj < 2; ++nj) {
+            int row = wn * 32 + nj * 16 + fr;
+            rdB[kk][nj] = &lds.B[0][row][wg_swz(row, gg) * 8];
+        }
+    }
+
+    auto write_lds = [&](int buf) {     // literal buf only
+#pragma unroll
+        for (int rr = 0; rr < 2; ++rr)
 #pragma unroll
             for (int u = 0; u < 8; ++u) {
-                int row = cc8 + u;
-                int g = wg_swz(row, g0);
-                lds.A[buf][row][g * 8 + o] = areg[rr].us[u];
-                lds.B[buf][row][g * 8 + o] = breg[rr].us[u];
+                unsigned short* p = wr[rr][u] + (buf ? A1 : 0);
+                p[0] = areg[rr].us[u];
+                p[BOF] = breg[rr].us[u];
             }
-        }
     };
 
-    auto mfma_step = [&](int buf) {
+    auto mfma_step = [&](int buf) {     // literal buf only
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
-            int gg = kk * 4 + fq;
             bf16x8_t a[2], bfr[2];
 #pragma unroll
-            for (int mi = 0; mi < 2; ++mi) {
-                int row = wm * 32 + mi * 16 + fr;
-                a[mi] = *(const bf16x8_t*)&lds.A[buf][row][wg_swz(row, gg) * 8];
-            }
+            for (int mi = 0; mi < 2; ++mi)
+                a[mi] = *(const bf16x8_t*)(rdA[kk][mi] + (buf ? A1 : 0));
 #pragma unroll
-            for (int nj = 0; nj < 2; ++nj) {
-                int row = wn * 32 + nj * 16 + fr;
-                bfr[nj] = *(const bf16x8_t*)&lds.B[buf][row][wg_swz(row, gg) * 8];
-            }
+            for (int nj = 0; nj < 2; ++nj)
+                bfr[nj] = *(const bf16x8_t*)(rdB[kk][nj] + (buf ? A1 : 0));
 #pragma unroll
             for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
@@ -748,12 +867,19 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
     write_lds(0);
     if (nsteps > 1) load_step();
     __syncthreads();
-    for (int it = 0; it < nsteps; ++it) {
+    for (int it = 0; it < nsteps; it += 2) {
         if (it + 1 < nsteps) {
-            write_lds((it + 1) & 1);
+            write_lds(1);
             if (it + 2 < nsteps) load_step();
         }
-        mfma_step(it & 1);
+        mfma_step(0);
+        __syncthreads();
+        if (it + 1 >= nsteps) break;
+        if (it + 2 < nsteps) {
+            write_lds(0);
+            if (it + 3 < nsteps) load_step();
+        }
+        mfma_step(1);
         __syncthreads();
     }
 
