@@ -242,60 +242,35 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         if (++lc == KC) { lc = 0; if (++ls == S) { ls = 0; ++lr; } }
     };
 
-    // Precomputed LDS pointers (PMC-measured: recomputing the swizzled
-    // addresses per step made the kernels VALU-bound — ~130 VALU per 8 MFMA).
-    // Buffer 1 / array B live at constexpr ushort offsets, folded into the
-    // ds_read/ds_write immediate when `buf` is a literal at the call site.
-    constexpr int PITCH = FwdLds<TM, TN>::PITCH;
-    constexpr int A1 = TM * PITCH;                 // lds.A[1] offset
-    constexpr int B1 = TN * PITCH;                 // lds.B[1] - lds.B[0]
-    unsigned short* wrA[AR];
-    unsigned short* wrB[BR];
+    auto write_lds = [&](int buf) {
 #pragma unroll
-    for (int rr = 0; rr < AR; ++rr) {
-        int row = trow + 32 * rr;
-        wrA[rr] = &lds.A[0][row][gswz<SWZ>(row, t & 7) * 8];
-    }
-#pragma unroll
-    for (int rr = 0; rr < BR; ++rr) {
-        int row = trow + 32 * rr;
-        wrB[rr] = &lds.B[0][row][gswz<SWZ>(row, t & 7) * 8];
-    }
-    const unsigned short* rdA[2][4];
-    const unsigned short* rdB[2][NJ];
-#pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
-#pragma unroll
-        for (int mi = 0; mi < 4; ++mi) {
-            int row = wm * 64 + mi * 16 + fr;
-            rdA[kk][mi] = &lds.A[0][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
+        for (int rr = 0; rr < AR; ++rr) {
+            int row = trow + 32 * rr;
+            *(uint4*)&lds.A[buf][row][gswz<SWZ>(row, t & 7) * 8] = areg[rr].u4;
         }
 #pragma unroll
-        for (int nj = 0; nj < NJ; ++nj) {
-            int row = wn * (NJ * 16) + nj * 16 + fr;
-            rdB[kk][nj] = &lds.B[0][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
+        for (int rr = 0; rr < BR; ++rr) {
+            int row = trow + 32 * rr;
+            *(uint4*)&lds.B[buf][row][gswz<SWZ>(row, t & 7) * 8] = breg[rr].u4;
         }
-    }
-
-    auto write_lds = [&](int buf) {     // call with a LITERAL buf only
-#pragma unroll
-        for (int rr = 0; rr < AR; ++rr)
-            *(uint4*)(wrA[rr] + (buf ? A1 : 0)) = areg[rr].u4;
-#pragma unroll
-        for (int rr = 0; rr < BR; ++rr)
-            *(uint4*)(wrB[rr] + (buf ? B1 : 0)) = breg[rr].u4;
     };
 
-    auto mfma_step = [&](int buf) {     // call with a LITERAL buf only
+    auto mfma_step = [&](int buf) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
             bf16x8_t a[4], b[NJ];
 #pragma unroll
-            for (int mi = 0; mi < 4; ++mi)
-                a[mi] = *(const bf16x8_t*)(rdA[kk][mi] + (buf ? A1 : 0));
+            for (int mi = 0; mi < 4; ++mi) {
+                int row = wm * 64 + mi * 16 + fr;
+                a[mi] = *(const bf16x8_t*)
+                    &lds.A[buf][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
+            }
 #pragma unroll
-            for (int nj = 0; nj < NJ; ++nj)
-                b[nj] = *(const bf16x8_t*)(rdB[kk][nj] + (buf ? B1 : 0));
+            for (int nj = 0; nj < NJ; ++nj) {
+                int row = wn * (NJ * 16) + nj * 16 + fr;
+                b[nj] = *(const bf16x8_t*)
+                    &lds.B[buf][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
+            }
 #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
@@ -304,25 +279,17 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         }
     };
 
-    // ---- pipeline: 2 LDS buffers, one reg set, one barrier per step,
-    // unrolled x2 so the buffer select is a compile-time immediate ----
+    // ---- pipeline: 2 LDS buffers, one reg set, one barrier per step ----
     load_step();
     write_lds(0);
     if (nsteps > 1) load_step();
     __syncthreads();
-    for (int it = 0; it < nsteps; it += 2) {
+    for (int it = 0; it < nsteps; ++it) {
         if (it + 1 < nsteps) {
-            write_lds(1);
+            write_lds((it + 1) & 1);
             if (it + 2 < nsteps) load_step();
         }
-        mfma_step(0);
-        __syncthreads();
-        if (it + 1 >= nsteps) break;
-        if (it + 2 < nsteps) {
-            write_lds(0);
-            if (it + 3 < nsteps) load_step();
-        }
-        mfma_step(1);
+        mfma_step(it & 1);
         __syncthreads();
     }
 
@@ -424,56 +391,35 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
         if (++lkc == KC) { lkc = 0; if (++lsi == ns) { lsi = 0; ++lri; } }
     };
 
-    constexpr int PITCH = FwdLds<TM, TN>::PITCH;
-    constexpr int A1 = TM * PITCH;
-    constexpr int B1 = TN * PITCH;
-    unsigned short* wrA[AR];
-    unsigned short* wrB[BR];
+    auto write_lds = [&](int buf) {
 #pragma unroll
-    for (int rr = 0; rr < AR; ++rr) {
-        int row = trow + 32 * rr;
-        wrA[rr] = &lds.A[0][row][gswz<SWZ>(row, t & 7) * 8];
-    }
-#pragma unroll
-    for (int rr = 0; rr < BR; ++rr) {
-        int row = trow + 32 * rr;
-        wrB[rr] = &lds.B[0][row][gswz<SWZ>(row, t & 7) * 8];
-    }
-    const unsigned short* rdA[2][4];
-    const unsigned short* rdB[2][NJ];
-#pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
-#pragma unroll
-        for (int mi = 0; mi < 4; ++mi) {
-            int row = wm * 64 + mi * 16 + fr;
-            rdA[kk][mi] = &lds.A[0][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
+        for (int rr = 0; rr < AR; ++rr) {
+            int row = trow + 32 * rr;
+            *(uint4*)&lds.A[buf][row][gswz<SWZ>(row, t & 7) * 8] = areg[rr].u4;
         }
 #pragma unroll
-        for (int nj = 0; nj < NJ; ++nj) {
-            int row = wn * (NJ * 16) + nj * 16 + fr;
-            rdB[kk][nj] = &lds.B[0][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
+        for (int rr = 0; rr < BR; ++rr) {
+            int row = trow + 32 * rr;
+            *(uint4*)&lds.B[buf][row][gswz<SWZ>(row, t & 7) * 8] = breg[rr].u4;
         }
-    }
-
-    auto write_lds = [&](int buf) {     // literal buf only
-#pragma unroll
-        for (int rr = 0; rr < AR; ++rr)
-            *(uint4*)(wrA[rr] + (buf ? A1 : 0)) = areg[rr].u4;
-#pragma unroll
-        for (int rr = 0; rr < BR; ++rr)
-            *(uint4*)(wrB[rr] + (buf ? B1 : 0)) = breg[rr].u4;
     };
 
-    auto mfma_step = [&](int buf) {     // literal buf only
+    auto mfma_step = [&](int buf) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
             bf16x8_t a[4], b[NJ];
 #pragma unroll
-            for (int mi = 0; mi < 4; ++mi)
-                a[mi] = *(const bf16x8_t*)(rdA[kk][mi] + (buf ? A1 : 0));
+            for (int mi = 0; mi < 4; ++mi) {
+                int row = wm * 64 + mi * 16 + fr;
+                a[mi] = *(const bf16x8_t*)
+                    &lds.A[buf][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
+            }
 #pragma unroll
-            for (int nj = 0; nj < NJ; ++nj)
-                b[nj] = *(const bf16x8_t*)(rdB[kk][nj] + (buf ? B1 : 0));
+            for (int nj = 0; nj < NJ; ++nj) {
+                int row = wn * (NJ * 16) + nj * 16 + fr;
+                b[nj] = *(const bf16x8_t*)
+                    &lds.B[buf][row][gswz<SWZ>(row, kk * 4 + fq) * 8];
+            }
 #pragma unroll
             for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
@@ -487,19 +433,12 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
         write_lds(0);
         if (nsteps > 1) load_step();
         __syncthreads();
-        for (int it = 0; it < nsteps; it += 2) {
+        for (int it = 0; it < nsteps; ++it) {
             if (it + 1 < nsteps) {
-                write_lds(1);
+                write_lds((it + 1) & 1);
                 if (it + 2 < nsteps) load_step();
             }
-            mfma_step(0);
-            __syncthreads();
-            if (it + 1 >= nsteps) break;
-            if (it + 2 < nsteps) {
-                write_lds(0);
-                if (it + 3 < nsteps) load_step();
-            }
-            mfma_step(1);
+            mfma_step(it & 1);
             __syncthreads();
         }
     }
