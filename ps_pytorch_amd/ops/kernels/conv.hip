@@ -123,24 +123,53 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
     const int trow = t >> 3;           // 0..31
     const int cc8 = (t & 7) * 8;       // this thread's 8-elem contraction chunk
 
-    // ---- per-row pixel descriptors (computed once) ----
-    long abase[AR]; int ax[AR], ay[AR];   // fwd: h0,w0; dgrad: h,w (+n folded in abase)
+    // ---- per-row state (computed once) ----
+    // !SMALL: running global pointer pA (advanced by a wave-uniform delta
+    // each step — no per-step address rebuild, PMC showed the kernel was
+    // VALU-bound on it) + a validity bitmask over the R*S taps, shifted
+    // right once per (r,s) advance so "valid now" is always bit 0.
+    // SMALL: classic origin descriptors for the one-step gather.
+    const unsigned short* pA[AR];
+    unsigned long long vm[AR];
+    long abase[SMALL ? AR : 1]; int ax[SMALL ? AR : 1], ay[SMALL ? AR : 1];
 #pragma unroll
     for (int rr = 0; rr < AR; ++rr) {
         long m = m0 + trow + 32 * rr;
+        pA[rr] = src; vm[rr] = 0;
+        if constexpr (SMALL) { ax[rr] = INT_MIN / 2; ay[rr] = 0; abase[rr] = 0; }
         if (m < M) {
             if constexpr (!DGRAD) {
                 int n = (int)(m / ((long)P * Q)); int rem = (int)(m % ((long)P * Q));
                 int p = rem / Q, q = rem % Q;
-                ax[rr] = p * STRIDE - pad;       // input h origin
-                ay[rr] = q * STRIDE - pad;       // input w origin
-                abase[rr] = ((long)(n * H + ax[rr]) * W + ay[rr]) * C;
+                int h0 = p * STRIDE - pad, w0 = q * STRIDE - pad;
+                if constexpr (SMALL) {
+                    ax[rr] = h0; ay[rr] = w0;
+                    abase[rr] = ((long)(n * H + h0) * W + w0) * C;
+                } else {
+                    pA[rr] = src + ((long)(n * H + h0) * W + w0) * C + cc8;
+                    unsigned long long msk = 0;
+                    for (int r = 0; r < R; ++r)
+                        for (int s = 0; s < S; ++s)
+                            if (h0 + r >= 0 && h0 + r < H
+                                && w0 + s >= 0 && w0 + s < W)
+                                msk |= 1ull << (r * S + s);
+                    vm[rr] = msk;
+                }
             } else {
                 int n = (int)(m / ((long)H * W)); int rem = (int)(m % ((long)H * W));
-                ax[rr] = rem / W; ay[rr] = rem % W;
-                abase[rr] = (long)n * P * Q * K;
+                int h = rem / W, w = rem % W;
+                // stride-1 only (stride-2 dgrad runs conv_dgrad2_kernel)
+                int p0 = h + pad, q0 = w + pad;
+                pA[rr] = src + (long)n * P * Q * K + ((long)p0 * Q + q0) * K + cc8;
+                unsigned long long msk = 0;
+                for (int r = 0; r < R; ++r)
+                    for (int s = 0; s < S; ++s)
+                        if (p0 - r >= 0 && p0 - r < P
+                            && q0 - s >= 0 && q0 - s < Q)
+                            msk |= 1ull << (r * S + s);
+                vm[rr] = msk;
             }
-        } else { ax[rr] = INT_MIN / 2; ay[rr] = 0; abase[rr] = 0; }
+        }
     }
 
     // wave grid: every wave owns a 64-row sub-tile; columns split when the
@@ -179,9 +208,27 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
 
     V16 areg[AR], breg[BR];
 
-    // stage loads for contraction step (lr, ls, lc*64) into registers
+    // running B pointers + fixed column-valid flags
+    const unsigned short* pB[BR];
+    bool colv[BR];
+#pragma unroll
+    for (int rr = 0; rr < BR; ++rr) {
+        int col = n0 + trow + 32 * rr;
+        colv[rr] = col < Nout;
+        if (!colv[rr]) col = 0;
+        if constexpr (SMALL)
+            pB[rr] = wgt + (long)col * R * S * C + cc8;
+        else if constexpr (!DGRAD)
+            pB[rr] = wgt + (long)col * R * S * C + cc8;   // (r,s,c0)=(0,0,0)
+        else
+            pB[rr] = wgt + (long)col * K + cc8;
+    }
+    const int KCm1 = (KC - 1) * 64;
+
+    // stage loads for contraction step (lr, ls, lc*64) into registers,
+    // then advance every pointer by the wave-uniform step delta
     auto load_step = [&]() {
-        const int c0 = lc << 6;
+        const int cbase = (lc << 6) + cc8;     // contraction offset (tail guard)
 #pragma unroll
         for (int rr = 0; rr < AR; ++rr) {
             if constexpr (SMALL) {
@@ -197,49 +244,41 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
                                       + ((long)tre[u] * W + tse[u]) * C + tce[u]];
                 }
                 areg[rr] = v;
-            } else if constexpr (!DGRAD) {
-                int hh = ax[rr] + lr, ww = ay[rr] + ls;
-                bool v = hh >= 0 && hh < H && ww >= 0 && ww < W;
-                areg[rr] = v ? load16(src + abase[rr] + ((long)lr * W + ls) * C
-                                      + c0 + cc8, c0 + cc8, C)
-                             : zero16();
             } else {
-                int ph = ax[rr] + pad - lr, pw = ay[rr] + pad - ls;
-                bool v;
-                int p, q;
-                if constexpr (STRIDE == 1) {
-                    p = ph; q = pw;
-                    v = ax[rr] >= 0 && p >= 0 && p < P && q >= 0 && q < Q;
-                } else {
-                    v = ax[rr] >= 0 && ph >= 0 && pw >= 0
-                        && (ph & 1) == 0 && (pw & 1) == 0;
-                    p = ph >> 1; q = pw >> 1;
-                    v = v && p < P && q < Q;
-                }
-                areg[rr] = v ? load16(src + abase[rr] + ((long)p * Q + q) * K
-                                      + c0 + cc8, c0 + cc8, K)
-                             : zero16();
+                areg[rr] = (vm[rr] & 1)
+                    ? load16(pA[rr], cbase, Cin) : zero16();
             }
         }
 #pragma unroll
         for (int rr = 0; rr < BR; ++rr) {
-            int j = trow + 32 * rr;            // row within tile
-            int col = n0 + j;                  // output channel / input channel
-            if (col < Nout) {
-                if constexpr (SMALL) {
-                    breg[rr] = load16(wgt + (long)col * R * S * C + cc8,
-                                      cc8, R * S * C);
-                } else {
-                    const unsigned short* wsrc;
-                    if constexpr (!DGRAD)
-                        wsrc = wgt + (((long)col * R + lr) * S + ls) * C + c0 + cc8;
-                    else
-                        wsrc = wgt + (((long)(lr * S + ls) * C) + col) * (long)K + c0 + cc8;
-                    breg[rr] = load16(wsrc, c0 + cc8, Cin);
-                }
-            } else breg[rr] = zero16();
+            if constexpr (SMALL)
+                breg[rr] = colv[rr] ? load16(pB[rr], cc8, R * S * C) : zero16();
+            else
+                breg[rr] = colv[rr] ? load16(pB[rr], cbase, Cin) : zero16();
         }
-        if (++lc == KC) { lc = 0; if (++ls == S) { ls = 0; ++lr; } }
+        if constexpr (!SMALL) {
+            long dA, dB;
+            bool rs_adv = (++lc == KC);
+            if (rs_adv) {
+                lc = 0;
+                bool rwrap = (++ls == S);
+                if (rwrap) { ls = 0; ++lr; }
+                if constexpr (!DGRAD) {
+                    dA = (long)(rwrap ? (W - S + 1) : 1) * C - KCm1;
+                    dB = (long)C - KCm1;
+                } else {
+                    dA = (long)(rwrap ? (S - 1 - Q) : -1) * K - KCm1;
+                    dB = (long)C * K - KCm1;
+                }
+            } else { dA = 64; dB = 64; }
+#pragma unroll
+            for (int rr = 0; rr < AR; ++rr) {
+                pA[rr] += dA;
+                if (rs_adv) vm[rr] >>= 1;
+            }
+#pragma unroll
+            for (int rr = 0; rr < BR; ++rr) pB[rr] += dB;
+        }
     };
 
     auto write_lds = [&](int buf) {
