@@ -375,15 +375,31 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
     const int trow = t >> 3;
     const int cc8 = (t & 7) * 8;
 
-    long abase[AR]; int ax[AR], ay[AR];      // h, w of the dx pixel
+    const int r0 = (hp + pad) & 1, s0 = (wp + pad) & 1;
+    const int nr = (R > r0) ? ((R - r0 + 1) >> 1) : 0;
+    const int ns = (S > s0) ? ((S - s0 + 1) >> 1) : 0;
+
+    // affine-walked dout pointers + (ri,si) tap-validity masks (cf. the
+    // conv_gemm_kernel staging — same VALU-bound fix)
+    const unsigned short* pA[AR];
+    unsigned int vm[AR];
 #pragma unroll
     for (int rr = 0; rr < AR; ++rr) {
         long m = m0 + trow + 32 * rr;
+        pA[rr] = dout; vm[rr] = 0;
         if (m < M) {
             int n = (int)(m / ((long)Hc * Wc)); int rem = (int)(m % ((long)Hc * Wc));
-            ax[rr] = hp + 2 * (rem / Wc); ay[rr] = wp + 2 * (rem % Wc);
-            abase[rr] = (long)n * P * Q * K;
-        } else { ax[rr] = INT_MIN / 2; ay[rr] = 0; abase[rr] = 0; }
+            int h = hp + 2 * (rem / Wc), w = wp + 2 * (rem % Wc);
+            int p1 = (h + pad - r0) >> 1, q1 = (w + pad - s0) >> 1;
+            pA[rr] = dout + (long)n * P * Q * K + ((long)p1 * Q + q1) * K + cc8;
+            unsigned int msk = 0;
+            for (int ri = 0; ri < nr; ++ri)
+                for (int si = 0; si < ns; ++si)
+                    if (p1 - ri >= 0 && p1 - ri < P
+                        && q1 - si >= 0 && q1 - si < Q)
+                        msk |= 1u << (ri * ns + si);
+            vm[rr] = msk;
+        }
     }
 
     constexpr int WGM = TM / 64;
@@ -399,35 +415,48 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
 #pragma unroll
         for (int j = 0; j < NJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-    const int r0 = (hp + pad) & 1, s0 = (wp + pad) & 1;
-    const int nr = (R > r0) ? ((R - r0 + 1) >> 1) : 0;
-    const int ns = (S > s0) ? ((S - s0 + 1) >> 1) : 0;
     const int KC = (K + 63) >> 6;
+    const int KCm1 = (KC - 1) * 64;
     const int nsteps = nr * ns * KC;
-    int lri = 0, lsi = 0, lkc = 0;
+    int lsi = 0, lkc = 0;
     V16 areg[AR], breg[BR];
 
+    // running B pointers at (r0, s0, k0=0)
+    const unsigned short* pB[BR];
+    bool colv[BR];
+#pragma unroll
+    for (int rr = 0; rr < BR; ++rr) {
+        int col = n0 + trow + 32 * rr;
+        colv[rr] = col < C;
+        pB[rr] = wgt + (((long)(r0 * S + s0) * C) + (colv[rr] ? col : 0))
+                 * (long)K + cc8;
+    }
+
     auto load_step = [&]() {
-        const int r = r0 + 2 * lri, s = s0 + 2 * lsi;
-        const int k0 = lkc << 6;
+        const int kbase = (lkc << 6) + cc8;
+#pragma unroll
+        for (int rr = 0; rr < AR; ++rr)
+            areg[rr] = (vm[rr] & 1) ? load16(pA[rr], kbase, K) : zero16();
+#pragma unroll
+        for (int rr = 0; rr < BR; ++rr)
+            breg[rr] = colv[rr] ? load16(pB[rr], kbase, K) : zero16();
+        long dA, dB;
+        bool rs_adv = (++lkc == KC);
+        if (rs_adv) {
+            lkc = 0;
+            bool rwrap = (++lsi == ns);
+            if (rwrap) lsi = 0;
+            // s += 2 -> q -= 1; r += 2 -> p -= 1, q += (ns-1)
+            dA = (long)(rwrap ? (ns - 1) * (long)K - (long)Q * K : -(long)K) - KCm1;
+            dB = (long)(rwrap ? 2 * S - 2 * (ns - 1) : 2) * C * (long)K - KCm1;
+        } else { dA = 64; dB = 64; }
 #pragma unroll
         for (int rr = 0; rr < AR; ++rr) {
-            int ph = ax[rr] + pad - r, pw = ay[rr] + pad - s;   // even by construction
-            int p = ph >> 1, q = pw >> 1;
-            bool v = ax[rr] >= 0 && ph >= 0 && pw >= 0 && p < P && q < Q;
-            areg[rr] = v ? load16(dout + abase[rr] + ((long)p * Q + q) * K
-                                  + k0 + cc8, k0 + cc8, K)
-                         : zero16();
+            pA[rr] += dA;
+            if (rs_adv) vm[rr] >>= 1;
         }
 #pragma unroll
-        for (int rr = 0; rr < BR; ++rr) {
-            int col = n0 + trow + 32 * rr;
-            breg[rr] = (col < C)
-                ? load16(wgt + (((long)(r * S + s) * C) + col) * (long)K + k0 + cc8,
-                         k0 + cc8, K)
-                : zero16();
-        }
-        if (++lkc == KC) { lkc = 0; if (++lsi == ns) { lsi = 0; ++lri; } }
+        for (int rr = 0; rr < BR; ++rr) pB[rr] += dB;
     };
 
     auto write_lds = [&](int buf) {
