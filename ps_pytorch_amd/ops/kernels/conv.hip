@@ -538,11 +538,29 @@ __device__ __forceinline__ int wg_swz(int row, int g) {
     return g ^ (row & 7) ^ ((row >> 3) & 7);
 }
 
+// Natural-layout wgrad LDS: both operands stored [pixel][channel] exactly as
+// loaded (vector ds_write_b128 staging — no transpose scatter), reorganized
+// into 16-channel subtiles of 32-B row stride so fragments come back through
+// gfx950's hardware transpose read `ds_read_b64_tr_b16` (guide T10): a
+// 16-lane group reads a [4-pixel][16-chan] block, 4 contiguous bf16 per lane
+// at its own 8-B-aligned address, and each lane receives its CHANNEL's 4
+// pixel values — i.e. the [chan][pixel] MFMA fragment, transposed in HW.
+// Subtile stride is padded +16 elements so the 8 staging writes of one
+// 8-lane group land on 8 distinct bank classes.
+#define WG_SUB 1040                    // 64*16 + 16 pad (ushort elements)
 template <int TK>
 struct WgradLds {
-    unsigned short A[2][TK][64];   // [k][pixel]
-    unsigned short B[2][64][64];   // [c][pixel]
+    unsigned short A[2][TK / 16][WG_SUB];   // [k-subtile][pixel*16+koff]
+    unsigned short B[2][4][WG_SUB];         // [c-subtile][pixel*16+coff]
 };
+
+// one hardware transpose read: 4 contiguous bf16 at 8-B-aligned LDS byte
+// offset `off`; the 16-lane group's lanes get their column's 4 values.
+__device__ __forceinline__ unsigned long long ds_tr16(unsigned off) {
+    unsigned long long v;
+    asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(v) : "v"(off));
+    return v;
+}
 
 // One block: TK k x 64 c output tile for ONE (r,s), summing the pixel range
 // [sid*chunk, ...) of length `chunk`; f32 partial out[sid][K][R*S*C].
@@ -632,66 +650,93 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
         lm += 64;
     };
 
-    // precomputed swizzled scatter/read pointers (all offsets constexpr)
-    constexpr int A1 = TK * 64;          // lds.A[1] - lds.A[0]   (ushorts)
-    constexpr int BOF = 2 * TK * 64;     // lds.B[0] - lds.A[0]
-    constexpr int BB1 = 64 * 64;         // lds.B[1] - lds.B[0]
-    unsigned short* wr[2][8];
+    // ---- natural-store / transpose-read staging (see WgradLds) ----
+    // byte offsets within the LDS struct; buffer 1 deltas are constexpr
+    constexpr unsigned A1 = (TK / 16) * WG_SUB * 2;     // lds.A[1] - lds.A[0]
+    constexpr unsigned BOF = 2 * A1;                    // lds.B[0] - lds.A[0]
+    constexpr unsigned BB1 = 4 * WG_SUB * 2;            // lds.B[1] - lds.B[0]
+    // staging write pointers: pixel midx's 16B lands in subtile cc8>>4 at
+    // column offset cc8&15 (a 16B quantum never straddles a subtile)
+    unsigned short* wrA[2][KH];
+    unsigned short* wrB[2];
 #pragma unroll
     for (int rr = 0; rr < 2; ++rr) {
         int midx = trow + 32 * rr;
-        int g0 = midx >> 3, o = midx & 7;
 #pragma unroll
-        for (int u = 0; u < 8; ++u) {
-            int row = cc8 + u;
-            wr[rr][u] = &lds.A[0][row][wg_swz(row, g0) * 8 + o];
+        for (int h = 0; h < KH; ++h) {
+            int kk8 = h * 64 + cc8;
+            wrA[rr][h] = &lds.A[0][kk8 >> 4][midx * 16 + (kk8 & 15)];
         }
+        wrB[rr] = &lds.B[0][cc8 >> 4][midx * 16 + (cc8 & 15)];
     }
-    const unsigned short* rdA[2][MI];
-    const unsigned short* rdB[2][2];
+    // transpose-read offsets: frag (mi,kk,half i) = block rows
+    // m = kk*32 + fq*8 + i*4 .. +4, cols = 16-chan subtile; lane fr reads
+    // 4 contiguous bf16 of row m + (fr>>2), quad fr&3, and receives its
+    // channel's column.
+    unsigned roA[2][MI][2];
+    unsigned roB[2][2][2];
 #pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
-        int gg = kk * 4 + fq;
+    for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-        for (int mi = 0; mi < MI; ++mi) {
-            int row = wm * (MI * 16) + mi * 16 + fr;
-            rdA[kk][mi] = &lds.A[0][row][wg_swz(row, gg) * 8];
+        for (int i = 0; i < 2; ++i) {
+            int m = kk * 32 + fq * 8 + i * 4 + (fr >> 2);
+            int qo = (fr & 3) * 4;
+#pragma unroll
+            for (int mi = 0; mi < MI; ++mi)
+                roA[kk][mi][i] = (unsigned)((char*)&lds.A[0][wm * MI + mi]
+                                            [m * 16 + qo] - (char*)&lds);
+#pragma unroll
+            for (int nj = 0; nj < 2; ++nj)
+                roB[kk][nj][i] = (unsigned)((char*)&lds.B[0][wn * 2 + nj]
+                                            [m * 16 + qo] - (char*)&lds);
         }
-#pragma unroll
-        for (int nj = 0; nj < 2; ++nj) {
-            int row = wn * 32 + nj * 16 + fr;
-            rdB[kk][nj] = &lds.B[0][row][wg_swz(row, gg) * 8];
-        }
-    }
 
     auto write_lds = [&](int buf) {     // literal buf only
 #pragma unroll
-        for (int rr = 0; rr < 2; ++rr)
+        for (int rr = 0; rr < 2; ++rr) {
 #pragma unroll
-            for (int u = 0; u < 8; ++u) {
-                unsigned short* p = wr[rr][u] + (buf ? A1 : 0);
-#pragma unroll
-                for (int h = 0; h < KH; ++h)
-                    p[h * 64 * 64] = areg[rr][h].us[u];
-                (p + BOF + (buf ? BB1 - A1 : 0))[0] = breg[rr].us[u];
-            }
+            for (int h = 0; h < KH; ++h)
+                *(uint4*)((char*)wrA[rr][h] + (buf ? A1 : 0)) = areg[rr][h].u4;
+            *(uint4*)((char*)wrB[rr] + (buf ? BB1 : 0)) = breg[rr].u4;
+        }
     };
 
+    union U64x8 { unsigned long long q[2]; bf16x8_t v; };
     auto mfma_step = [&](int buf) {     // literal buf only
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
-            bf16x8_t a[MI], bfr[2];
+            U64x8 a[MI], bfr[2];
 #pragma unroll
-            for (int mi = 0; mi < MI; ++mi)
-                a[mi] = *(const bf16x8_t*)(rdA[kk][mi] + (buf ? A1 : 0));
+            for (int mi = 0; mi < MI; ++mi) {
+                a[mi].q[0] = ds_tr16(roA[kk][mi][0] + (buf ? A1 : 0));
+                a[mi].q[1] = ds_tr16(roA[kk][mi][1] + (buf ? A1 : 0));
+            }
 #pragma unroll
-            for (int nj = 0; nj < 2; ++nj)
-                bfr[nj] = *(const bf16x8_t*)(rdB[kk][nj] + (buf ? BB1 : 0));
+            for (int nj = 0; nj < 2; ++nj) {
+                bfr[nj].q[0] = ds_tr16(roB[kk][nj][0] + (buf ? BB1 : 0));
+                bfr[nj].q[1] = ds_tr16(roB[kk][nj][1] + (buf ? BB1 : 0));
+            }
+            // one wait for the whole batch; the value operands pin the
+            // mfma uses behind it
+            if constexpr (MI == 2)
+                asm volatile("s_waitcnt lgkmcnt(0)"
+                             : "+v"(a[0].q[0]), "+v"(a[0].q[1]),
+                               "+v"(a[1].q[0]), "+v"(a[1].q[1]),
+                               "+v"(bfr[0].q[0]), "+v"(bfr[0].q[1]),
+                               "+v"(bfr[1].q[0]), "+v"(bfr[1].q[1]));
+            else
+                asm volatile("s_waitcnt lgkmcnt(0)"
+                             : "+v"(a[0].q[0]), "+v"(a[0].q[1]),
+                               "+v"(a[1].q[0]), "+v"(a[1].q[1]),
+                               "+v"(a[2].q[0]), "+v"(a[2].q[1]),
+                               "+v"(a[3].q[0]), "+v"(a[3].q[1]),
+                               "+v"(bfr[0].q[0]), "+v"(bfr[0].q[1]),
+                               "+v"(bfr[1].q[0]), "+v"(bfr[1].q[1]));
 #pragma unroll
             for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
                 for (int nj = 0; nj < 2; ++nj)
-                    acc[mi][nj] = MFMA_BF16(a[mi], bfr[nj], acc[mi][nj]);
+                    acc[mi][nj] = MFMA_BF16(a[mi].v, bfr[nj].v, acc[mi][nj]);
         }
     };
 
