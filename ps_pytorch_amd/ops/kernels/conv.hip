@@ -787,7 +787,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
     int R, int S, int pad, int split, int chunk,
     int l2pq, int l2q)
 {
-    __shared__ __attribute__((aligned(16))) WgradLds<64> lds;
+    __shared__ __attribute__((aligned(16))) struct {
+        unsigned short A[2][64][64];   // [k][pixel], XOR-swizzled granules
+        unsigned short B[2][64][64];   // [rsc][pixel]
+    } lds;
     const long M = (long)Nb * P * Q;
     const int RSC = R * S * C;
     const int tiles_k = (K + 63) >> 6;
