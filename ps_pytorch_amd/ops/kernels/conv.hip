@@ -787,10 +787,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
     int R, int S, int pad, int split, int chunk,
     int l2pq, int l2q)
 {
-    __shared__ __attribute__((aligned(16))) struct {
-        unsigned short A[2][64][64];   // [k][pixel], XOR-swizzled granules
-        unsigned short B[2][64][64];   // [rsc][pixel]
-    } lds;
+    __shared__ __attribute__((aligned(16))) WgradLds<64> lds;
     const long M = (long)Nb * P * Q;
     const int RSC = R * S * C;
     const int tiles_k = (K + 63) >> 6;
@@ -859,62 +856,68 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
         lm += 64;
     };
 
-    constexpr int A1 = 64 * 64;          // lds.A[1] - lds.A[0] (ushorts)
-    constexpr int BOF = 2 * 64 * 64;     // lds.B[0] - lds.A[0]
-    unsigned short* wr[2][8];
+    // natural-store / transpose-read staging, as conv_wgrad_kernel (TK=64)
+    constexpr unsigned A1 = 4 * WG_SUB * 2;     // lds.A[1] - lds.A[0] (bytes)
+    constexpr unsigned BB1 = A1;                // lds.B[1] - lds.B[0]
+    unsigned short* wrA[2];
+    unsigned short* wrB[2];
 #pragma unroll
     for (int rr = 0; rr < 2; ++rr) {
         int midx = trow + 32 * rr;
-        int g0 = midx >> 3, o = midx & 7;
-#pragma unroll
-        for (int u = 0; u < 8; ++u) {
-            int row = cc8 + u;
-            wr[rr][u] = &lds.A[0][row][wg_swz(row, g0) * 8 + o];
-        }
+        wrA[rr] = &lds.A[0][cc8 >> 4][midx * 16 + (cc8 & 15)];
+        wrB[rr] = &lds.B[0][cc8 >> 4][midx * 16 + (cc8 & 15)];
     }
-    const unsigned short* rdA[2][2];
-    const unsigned short* rdB[2][2];
+    unsigned roA[2][2][2];
+    unsigned roB[2][2][2];
 #pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
-        int gg = kk * 4 + fq;
+    for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-        for (int mi = 0; mi < 2; ++mi) {
-            int row = wm * 32 + mi * 16 + fr;
-            rdA[kk][mi] = &lds.A[0][row][wg_swz(row, gg) * 8];
+        for (int i = 0; i < 2; ++i) {
+            int m = kk * 32 + fq * 8 + i * 4 + (fr >> 2);
+            int qo = (fr & 3) * 4;
+#pragma unroll
+            for (int mi = 0; mi < 2; ++mi)
+                roA[kk][mi][i] = (unsigned)((char*)&lds.A[0][wm * 2 + mi]
+                                            [m * 16 + qo] - (char*)&lds);
+#pragma unroll
+            for (int nj = 0; nj < 2; ++nj)
+                roB[kk][nj][i] = (unsigned)((char*)&lds.B[0][wn * 2 + nj]
+                                            [m * 16 + qo] - (char*)&lds);
         }
-#pragma unroll
-        for (int nj = 0; nj < 2; ++nj) {
-            int row = wn * 32 + nj * 16 + fr;
-            rdB[kk][nj] = &lds.B[0][row][wg_swz(row, gg) * 8];
-        }
-    }
 
     auto write_lds = [&](int buf) {     // literal buf only
 #pragma unroll
-        for (int rr = 0; rr < 2; ++rr)
-#pragma unroll
-            for (int u = 0; u < 8; ++u) {
-                unsigned short* p = wr[rr][u] + (buf ? A1 : 0);
-                p[0] = areg[rr].us[u];
-                p[BOF] = breg[rr].us[u];
-            }
+        for (int rr = 0; rr < 2; ++rr) {
+            *(uint4*)((char*)wrA[rr] + (buf ? A1 : 0)) = areg[rr].u4;
+            *(uint4*)((char*)wrB[rr] + (buf ? BB1 : 0)) = breg[rr].u4;
+        }
     };
 
+    union U64x8 { unsigned long long q[2]; bf16x8_t v; };
     auto mfma_step = [&](int buf) {     // literal buf only
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
-            bf16x8_t a[2], bfr[2];
+            U64x8 a[2], bfr[2];
 #pragma unroll
-            for (int mi = 0; mi < 2; ++mi)
-                a[mi] = *(const bf16x8_t*)(rdA[kk][mi] + (buf ? A1 : 0));
+            for (int mi = 0; mi < 2; ++mi) {
+                a[mi].q[0] = ds_tr16(roA[kk][mi][0] + (buf ? A1 : 0));
+                a[mi].q[1] = ds_tr16(roA[kk][mi][1] + (buf ? A1 : 0));
+            }
 #pragma unroll
-            for (int nj = 0; nj < 2; ++nj)
-                bfr[nj] = *(const bf16x8_t*)(rdB[kk][nj] + (buf ? A1 : 0));
+            for (int nj = 0; nj < 2; ++nj) {
+                bfr[nj].q[0] = ds_tr16(roB[kk][nj][0] + (buf ? BB1 : 0));
+                bfr[nj].q[1] = ds_tr16(roB[kk][nj][1] + (buf ? BB1 : 0));
+            }
+            asm volatile("s_waitcnt lgkmcnt(0)"
+                         : "+v"(a[0].q[0]), "+v"(a[0].q[1]),
+                           "+v"(a[1].q[0]), "+v"(a[1].q[1]),
+                           "+v"(bfr[0].q[0]), "+v"(bfr[0].q[1]),
+                           "+v"(bfr[1].q[0]), "+v"(bfr[1].q[1]));
 #pragma unroll
             for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
                 for (int nj = 0; nj < 2; ++nj)
-                    acc[mi][nj] = MFMA_BF16(a[mi], bfr[nj], acc[mi][nj]);
+                    acc[mi][nj] = MFMA_BF16(a[mi].v, bfr[nj].v, acc[mi][nj]);
         }
     };
 
