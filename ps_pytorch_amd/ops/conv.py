@@ -49,7 +49,7 @@ def _wgrad_split(M: int, K: int, C: int, R: int, S: int) -> int:
     else:
         tk = 128 if K >= 128 else 64
         tiles = ((K + tk - 1) // tk) * R * S * ((C + 63) // 64)
-    want = max(1, 2048 // max(tiles, 1))
+    want = max(1, 1024 // max(tiles, 1))
     max_split = max(1, M // 64)
     return max(1, min(want, max_split, 256))
 
@@ -87,8 +87,10 @@ class _ConvFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             # dgrad wants wT[R,S,C,K] in memory so its B stage is the same
             # contiguous-in-contraction load as fwd (no in-kernel transpose);
-            # one small permute-copy per backward (<5 MB even for ResNet-50).
-            wt = w.permute(2, 3, 1, 0).contiguous()
+            # one tiled-transpose kernel launch per backward (~us).
+            wt = torch.empty(R * S * C * K, dtype=w.dtype, device=w.device)
+            lib.ps_wt_transpose(wt.data_ptr(), w.data_ptr(), K, R * S * C,
+                                current_stream_ptr())
             dx = torch.empty_like(x).contiguous(memory_format=_CL)
             lib.ps_conv_dgrad(dout.data_ptr(), wt.data_ptr(), dx.data_ptr(),
                               Nb, H, W, C, K, P, Q, R, S, stride, pad,

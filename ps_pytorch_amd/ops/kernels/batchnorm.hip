@@ -189,20 +189,31 @@ __global__ __launch_bounds__(256) void bn_normalize_kernel(
     const T* __restrict__ x, T* __restrict__ y, const T* __restrict__ res,
     const float* __restrict__ ws, long M, int C)
 {
+    // fixed channel-octet per thread: scale/shift loaded ONCE, rows strided
+    // (the i%G form re-read 64 B of coefficients per 16 B of payload)
     const int G = C >> 3;
-    long nvec = M * G;
-    EW_IDX
-    for (long i = gid; i < nvec; i += stride) {
-        int cg = (int)(i % G);
-        f32x8 scale = *(const f32x8*)&ws[2 * C + cg * 8];
-        f32x8 shift = *(const f32x8*)&ws[3 * C + cg * 8];
-        f32x8 v = VecIO<T>::load(x + i * 8);
-        v = v * scale + shift;
-        if constexpr (RES) v += VecIO<T>::load(res + i * 8);
-        if constexpr (RELU)
+    const int R = 256 / G;
+    const int cg = threadIdx.x % G;
+    const int rib = threadIdx.x / G;
+    const f32x8 scale = *(const f32x8*)&ws[2 * C + cg * 8];
+    const f32x8 shift = *(const f32x8*)&ws[3 * C + cg * 8];
+    const long chunk = 4L * R;
+    for (long base = (long)blockIdx.x * chunk + rib; base < M;
+         base += (long)gridDim.x * chunk) {
 #pragma unroll
-            for (int k = 0; k < 8; ++k) v[k] = fmaxf(v[k], 0.f);
-        VecIO<T>::store(y + i * 8, v);
+        for (int u = 0; u < 4; ++u) {
+            long r = base + (long)u * R;
+            if (r < M) {
+                long o = r * C + cg * 8;
+                f32x8 v = VecIO<T>::load(x + o);
+                v = v * scale + shift;
+                if constexpr (RES) v += VecIO<T>::load(res + o);
+                if constexpr (RELU)
+#pragma unroll
+                    for (int k = 0; k < 8; ++k) v[k] = fmaxf(v[k], 0.f);
+                VecIO<T>::store(y + o, v);
+            }
+        }
     }
 }
 
@@ -223,13 +234,12 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
     f32x8 invstd = *(const f32x8*)&save_invstd[cg * 8];
     f32x8 sum_dy = {0, 0, 0, 0, 0, 0, 0, 0};
     f32x8 sum_dyx = {0, 0, 0, 0, 0, 0, 0, 0};
-    // contiguous 2-row chunks per block iteration (see k1 comment); the 2-3
-    // streams per row already give memory-level parallelism
-    const long chunk = 2L * R;
+    // contiguous 4-row chunks per block iteration (see k1 comment)
+    const long chunk = 4L * R;
     for (long base = (long)blockIdx.x * chunk + row_in_blk; base < M;
          base += (long)gridDim.x * chunk) {
 #pragma unroll
-        for (int u = 0; u < 2; ++u) {
+        for (int u = 0; u < 4; ++u) {
             long r = base + (long)u * R;
             if (r < M) {
                 long o = r * C + cg * 8;
@@ -293,25 +303,34 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_kernel(
     const float* __restrict__ save_mean, const float* __restrict__ ws,
     long M, int C)
 {
+    // fixed channel-octet per thread, coefficients hoisted (cf. normalize)
     const int G = C >> 3;
-    long nvec = M * G;
-    EW_IDX
-    for (long i = gid; i < nvec; i += stride) {
-        int cg = (int)(i % G);
-        f32x8 mean = *(const f32x8*)&save_mean[cg * 8];
-        f32x8 a = *(const f32x8*)&ws[2 * C + cg * 8];
-        f32x8 b = *(const f32x8*)&ws[3 * C + cg * 8];
-        f32x8 c = *(const f32x8*)&ws[4 * C + cg * 8];
-        long o = i * 8;
-        f32x8 d = VecIO<T>::load(dy + o);
-        if constexpr (RELU) {
-            f32x8 yo = VecIO<T>::load(y + o);
+    const int R = 256 / G;
+    const int cg = threadIdx.x % G;
+    const int rib = threadIdx.x / G;
+    const f32x8 mean = *(const f32x8*)&save_mean[cg * 8];
+    const f32x8 a = *(const f32x8*)&ws[2 * C + cg * 8];
+    const f32x8 b = *(const f32x8*)&ws[3 * C + cg * 8];
+    const f32x8 c = *(const f32x8*)&ws[4 * C + cg * 8];
+    const long chunk = 4L * R;
+    for (long base = (long)blockIdx.x * chunk + rib; base < M;
+         base += (long)gridDim.x * chunk) {
 #pragma unroll
-            for (int k = 0; k < 8; ++k) d[k] = yo[k] > 0.f ? d[k] : 0.f;
+        for (int u = 0; u < 4; ++u) {
+            long r = base + (long)u * R;
+            if (r < M) {
+                long o = r * C + cg * 8;
+                f32x8 d = VecIO<T>::load(dy + o);
+                if constexpr (RELU) {
+                    f32x8 yo = VecIO<T>::load(y + o);
+#pragma unroll
+                    for (int k = 0; k < 8; ++k) d[k] = yo[k] > 0.f ? d[k] : 0.f;
+                }
+                if constexpr (DRES) VecIO<T>::store(dres + o, d);
+                f32x8 xv = VecIO<T>::load(x + o);
+                VecIO<T>::store(dx + o, a * d + b * (xv - mean) + c);
+            }
         }
-        if constexpr (DRES) VecIO<T>::store(dres + o, d);
-        f32x8 xv = VecIO<T>::load(x + o);
-        VecIO<T>::store(dx + o, a * d + b * (xv - mean) + c);
     }
 }
 
@@ -319,11 +338,11 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_kernel(
 static inline int stats_blocks(long M, int C, int unroll) {
     long chunk = (long)unroll * (256 / (C >> 3));
     long want = (M + chunk - 1) / chunk;
-    long cap = 512;    // partial buffer rows; ~2 blocks/CU suffice (4 loads
-                       // in flight per wave from the in-chunk unroll)
+    long cap = 1024;   // partial buffer rows; 4 blocks/CU (the in-chunk
+                       // unroll alone does not cover HBM latency)
     return (int)(want < cap ? (want > 0 ? want : 1) : cap);
 }
-#define BN_MAX_PARTIAL_BLOCKS 512
+#define BN_MAX_PARTIAL_BLOCKS 1024
 
 template <typename T, typename PT>
 static void bn_fwd_t(const void* x, void* y, const void* gamma, const void* beta,
@@ -350,8 +369,8 @@ static void bn_fwd_t(const void* x, void* y, const void* gamma, const void* beta
                            wsf, (const PT*)gamma, (const PT*)beta,
                            (const float*)rmean, (const float*)rvar, Ci, eps);
     }
-    long nvec = M * (C >> 3);
-    int blocks; ew_grid(nvec, 256, &blocks);
+    int blocks = stats_blocks(M, Ci, 4);
+    if (blocks > 2048) blocks = 2048;
     if (relu) {
         if (res) hipLaunchKernelGGL((bn_normalize_kernel<T, true, true>), dim3(blocks), b256, 0, s,
                                     (const T*)x, (T*)y, (const T*)res, wsf, M, Ci);
@@ -391,7 +410,7 @@ static void bn_bwd_t(const void* x, const void* y, const void* dy,
     int Ci = (int)C;
     float* wsf = (float*)ws;
     dim3 b256(256);
-    int nb = stats_blocks(M, Ci, 2);
+    int nb = stats_blocks(M, Ci, 4);
     if (relu)
         hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, true>), dim3(nb), b256, 0, s,
                            (const T*)x, (const T*)y, (const T*)dy,
@@ -405,8 +424,8 @@ static void bn_bwd_t(const void* x, const void* y, const void* dy,
     hipLaunchKernelGGL((bn_bwd_finalize_kernel<PT>), dim3((Ci + 255) / 256), b256, 0, s,
                        wsf, (const PT*)gamma, (const float*)smean,
                        (const float*)sinvstd, (PT*)dgamma, (PT*)dbeta, M, Ci);
-    long nvec = M * (C >> 3);
-    int blocks; ew_grid(nvec, 256, &blocks);
+    int blocks = stats_blocks(M, Ci, 4);
+    if (blocks > 2048) blocks = 2048;
     if (relu) {
         if (dres) hipLaunchKernelGGL((bn_bwd_dx_kernel<T, true, true>), dim3(blocks), b256, 0, s,
                                      (const T*)x, (const T*)y, (const T*)dy, (T*)dx, (T*)dres,

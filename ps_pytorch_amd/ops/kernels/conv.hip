@@ -1130,3 +1130,39 @@ extern "C" void ps_conv_bias_grad(
                        (hipStream_t)strm, (unsigned short*)db,
                        (const unsigned short*)dout, M, K);
 }
+
+// [K][RSC] -> [RSC][K] bf16 transpose for the dgrad wT operand (torch's
+// permute().contiguous() on a channels-last weight ran as a ~45us gather
+// kernel; this is a plain 32x32 LDS-tiled transpose at stream rate).
+__global__ __launch_bounds__(256) void wt_transpose_kernel(
+    const unsigned short* __restrict__ w, unsigned short* __restrict__ wt,
+    int K, int RSC)
+{
+    __shared__ unsigned short tile[32][33];
+    int tiles_e = (RSC + 31) >> 5;
+    int k0 = (blockIdx.x / tiles_e) << 5;
+    int e0 = (blockIdx.x % tiles_e) << 5;
+    int tx = threadIdx.x & 31, ty = threadIdx.x >> 5;   // 32 x 8
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+        int k = k0 + ty + 8 * i, e = e0 + tx;
+        tile[ty + 8 * i][tx] =
+            (k < K && e < RSC) ? w[(long)k * RSC + e] : (unsigned short)0;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+        int e = e0 + ty + 8 * i, k = k0 + tx;
+        if (e < RSC && k < K)
+            wt[(long)e * K + k] = tile[tx][ty + 8 * i];
+    }
+}
+
+extern "C" void ps_wt_transpose(void* wt, const void* w, int K, int RSC,
+                                void* strm)
+{
+    long grid = (long)((K + 31) / 32) * ((RSC + 31) / 32);
+    hipLaunchKernelGGL(wt_transpose_kernel, dim3((unsigned)grid), dim3(256), 0,
+                       (hipStream_t)strm, (const unsigned short*)w,
+                       (unsigned short*)wt, K, RSC);
+}

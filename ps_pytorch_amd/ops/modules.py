@@ -41,7 +41,7 @@ class _FusedBNFn(torch.autograd.Function):
         save_mean = torch.empty(C, dtype=torch.float32, device=x.device)
         save_invstd = torch.empty(C, dtype=torch.float32, device=x.device)
         ws = torch.empty(5 * C, dtype=torch.float32, device=x.device)
-        partial = torch.empty(512 * 2 * C, dtype=torch.float32, device=x.device)
+        partial = torch.empty(1024 * 2 * C, dtype=torch.float32, device=x.device)
         lib.ps_bn_fwd(
             x.data_ptr(), y.data_ptr(), gamma.data_ptr(), beta.data_ptr(),
             running_mean.data_ptr(), running_var.data_ptr(),
@@ -67,7 +67,7 @@ class _FusedBNFn(torch.autograd.Function):
         dgamma = torch.empty_like(gamma)
         dbeta = torch.empty_like(gamma)
         ws = torch.empty(5 * C, dtype=torch.float32, device=x.device)
-        partial = torch.empty(512 * 2 * C, dtype=torch.float32, device=x.device)
+        partial = torch.empty(1024 * 2 * C, dtype=torch.float32, device=x.device)
         lib.ps_bn_bwd(
             x.data_ptr(), y.data_ptr(), dy.data_ptr(), gamma.data_ptr(),
             save_mean.data_ptr(), save_invstd.data_ptr(), dx.data_ptr(),
