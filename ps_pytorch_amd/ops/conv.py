@@ -64,8 +64,10 @@ class _ConvFn(torch.autograd.Function):
         K, _, R, S = w.shape
         P = (H + 2 * pad - R) // stride + 1
         Q = (W + 2 * pad - S) // stride + 1
-        out = torch.empty((Nb, K, P, Q), dtype=x.dtype, device=x.device) \
-            .contiguous(memory_format=_CL)
+        # NB: allocate channels_last DIRECTLY — empty().contiguous(CL) runs a
+        # full transposing copy of uninitialized memory (~190us at l1 size)
+        out = torch.empty((Nb, K, P, Q), dtype=x.dtype, device=x.device,
+                          memory_format=_CL)
         lib.ps_conv_fwd(x.data_ptr(), wc.data_ptr(),
                         b.data_ptr() if b is not None else 0, out.data_ptr(),
                         Nb, H, W, C, K, P, Q, R, S, stride, pad,
