@@ -955,17 +955,32 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
     }
 }
 
-// deterministic slab reduce: dw[e] = sum_s partial[s][e] (fixed order)
+// deterministic slab reduce: dw[e] = sum_s partial[s][e] (fixed order);
+// float4 lanes — the scalar version ran ~18x off stream rate.
 __global__ __launch_bounds__(256) void reduce_slabs_kernel(
     unsigned short* __restrict__ dw, const float* __restrict__ partial,
     long n, int nslab)
 {
     EW_IDX
-    for (long i = gid; i < n; i += stride) {
-        float acc = 0.f;
-        for (int s = 0; s < nslab; ++s) acc += partial[(long)s * n + i];
-        dw[i] = f32_to_bf16(acc);
+    long nv = n >> 2;
+    for (long i = gid; i < nv; i += stride) {
+        f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
+        for (int s = 0; s < nslab; ++s) {
+            f32x4_t v = *(const f32x4_t*)(partial + (long)s * n + i * 4);
+#pragma unroll
+            for (int k = 0; k < 4; ++k) acc[k] += v[k];
+        }
+        ushort4_t o;
+#pragma unroll
+        for (int k = 0; k < 4; ++k) o[k] = f32_to_bf16(acc[k]);
+        *(ushort4_t*)(dw + i * 4) = o;
     }
+    if (gid == 0)
+        for (long i = nv << 2; i < n; ++i) {
+            float acc = 0.f;
+            for (int s = 0; s < nslab; ++s) acc += partial[(long)s * n + i];
+            dw[i] = f32_to_bf16(acc);
+        }
 }
 
 // column sum for bias grad: db[k] = sum_m dout[m][k] (one block per 64 k,
