@@ -41,15 +41,19 @@ def parse():
     return p.parse_args()
 
 
-def make_batches(n_batches, bs, device, dtype, seed=1234):
+def make_batches(n_batches, bs, device, dtype, seed=1234,
+                 dataset='Cifar10'):
+    from ps_pytorch_amd.config import input_shape_of, num_classes_of
+    shape = input_shape_of(dataset)
+    ncls = num_classes_of(dataset)
     g = torch.Generator(device='cpu').manual_seed(seed)
     xs, ys = [], []
     for _ in range(n_batches):
-        x = torch.randn(bs, 3, 32, 32, generator=g).to(device, dtype)
+        x = torch.randn(bs, *shape, generator=g).to(device, dtype)
         if device.type == 'cuda':
             x = x.contiguous(memory_format=torch.channels_last)
         xs.append(x)
-        ys.append(torch.randint(0, 10, (bs,), generator=g).to(device))
+        ys.append(torch.randint(0, ncls, (bs,), generator=g).to(device))
     return xs, ys
 
 
@@ -58,6 +62,9 @@ def main():
     from ps_pytorch_amd.config import JobConfig
     from ps_pytorch_amd.parallel.transport import init_distributed
 
+    from ps_pytorch_amd.config import num_classes_of, input_shape_of
+    ncls = num_classes_of(args.dataset)
+    ishape = input_shape_of(args.dataset)
     world = int(os.environ.get('WORLD_SIZE', '1'))
     rank = int(os.environ.get('RANK', '0'))
     n_gpus = max(world, 1)
@@ -80,9 +87,10 @@ def main():
         if args.engine == 'allreduce':
             from ps_pytorch_amd.parallel.allreduce import AllReduceTrainer
             role = AllReduceTrainer(cfg, rank, world, device)
-            role.build_model(10)
+            role.build_model(ncls)
             xs, ys = make_batches(8, args.batch_size, device,
-                                  role.compute_dtype, seed=1234 + rank)
+                                  role.compute_dtype, seed=1234 + rank,
+                                  dataset=args.dataset)
             it = [0]
 
             def step():
@@ -91,13 +99,14 @@ def main():
                 role.train_step(xs[i], ys[i])
         elif rank == 0:
             role = ParameterServer(cfg, rank, world, device)
-            role.build_model(10)
+            role.build_model(ncls)
             step = role.step
         else:
             role = DistributedWorker(cfg, rank, world, device)
-            role.build_model(10)
+            role.build_model(ncls)
             xs, ys = make_batches(8, args.batch_size, device,
-                                  role.compute_dtype, seed=1234 + rank)
+                                  role.compute_dtype, seed=1234 + rank,
+                                  dataset=args.dataset)
             it = [0]
 
             def step():
@@ -138,8 +147,9 @@ def main():
         from ps_pytorch_amd.trainer import NNTrainer
         device = torch.device('cuda', 0) if use_cuda else torch.device('cpu')
         tr = NNTrainer(cfg, device=device)
-        tr.build_model(10)
-        xs, ys = make_batches(8, args.batch_size, device, tr.compute_dtype)
+        tr.build_model(ncls)
+        xs, ys = make_batches(8, args.batch_size, device, tr.compute_dtype,
+                              dataset=args.dataset)
         graphed = use_cuda and tr.enable_graph(xs[0], ys[0])
         step_fn = tr.graph_step if graphed else tr.train_step
         for i in range(args.warmup):
@@ -173,7 +183,7 @@ def main():
             "dtype": dtype,
             "data": "synthetic (random-init weights, GPU-resident random batches)",
             "config": {"model": args.network, "global_batch": global_batch,
-                       "input": "3x32x32", "per_worker_batch": args.batch_size,
+                       "input": "x".join(str(d) for d in ishape), "per_worker_batch": args.batch_size,
                        "parallelism": parallelism,
                        "compress_grad": args.compress_grad,
                        "overlap": not args.no_overlap},
