@@ -109,8 +109,11 @@ class _ConvFn(torch.autograd.Function):
                               split, current_stream_ptr())
         if has_bias and ctx.needs_input_grad[2]:
             db = torch.empty(K, dtype=dout.dtype, device=dout.device)
+            bpart = torch.empty(512 * K, dtype=torch.float32,
+                                device=dout.device)
             lib.ps_conv_bias_grad(db.data_ptr(), dout.data_ptr(),
-                                  Nb * P * Q, K, current_stream_ptr())
+                                  bpart.data_ptr(), Nb * P * Q, K,
+                                  current_stream_ptr())
         return dx, dw, db, None, None
 
 

@@ -983,24 +983,38 @@ __global__ __launch_bounds__(256) void reduce_slabs_kernel(
         }
 }
 
-// column sum for bias grad: db[k] = sum_m dout[m][k] (one block per 64 k,
-// two-stage in-block tree over a grid-stride m loop; deterministic)
-__global__ __launch_bounds__(256) void colsum_kernel(
-    unsigned short* __restrict__ db, const unsigned short* __restrict__ dout,
+// column sum for bias grad: db[k] = sum_m dout[m][k], two-stage —
+// many blocks write per-block partials (the 1-block version serialized the
+// whole M-scan on one CU: 17 ms for LeNet), then one block folds them in
+// fixed order (deterministic).
+__global__ __launch_bounds__(256) void colsum_part_kernel(
+    float* __restrict__ partial, const unsigned short* __restrict__ dout,
     long M, int K)
 {
     __shared__ float red[256];
-    int k = blockIdx.x * 64 + (threadIdx.x & 63);
-    int part = threadIdx.x >> 6;          // 4 m-partitions
+    int k = threadIdx.x & 63;
+    int sub = threadIdx.x >> 6;           // 4 m-partitions per block
     float acc = 0.f;
     if (k < K)
-        for (long m = part; m < M; m += 4)
+        for (long m = (long)blockIdx.x * 4 + sub; m < M;
+             m += (long)gridDim.x * 4)
             acc += bf16_to_f32(dout[m * K + k]);
     red[threadIdx.x] = acc;
     __syncthreads();
-    if (part == 0 && k < K)
-        db[k] = f32_to_bf16(red[threadIdx.x] + red[threadIdx.x + 64] +
-                            red[threadIdx.x + 128] + red[threadIdx.x + 192]);
+    if (sub == 0 && k < K)
+        partial[(long)blockIdx.x * K + k] =
+            red[k] + red[64 + k] + red[128 + k] + red[192 + k];
+}
+
+__global__ __launch_bounds__(256) void colsum_fold_kernel(
+    unsigned short* __restrict__ db, const float* __restrict__ partial,
+    int nblk, int K)
+{
+    int k = blockIdx.x * 256 + threadIdx.x;
+    if (k >= K) return;
+    float acc = 0.f;
+    for (int b = 0; b < nblk; ++b) acc += partial[(long)b * K + k];
+    db[k] = f32_to_bf16(acc);
 }
 
 // ---------------------------------------------------------------- C API
@@ -1138,12 +1152,17 @@ extern "C" void ps_conv_wgrad(
                        (const float*)partial_f32, n, split);
 }
 
+// partial_f32 must hold 512*K floats.
 extern "C" void ps_conv_bias_grad(
-    void* db, const void* dout, long M, int K, void* strm)
+    void* db, const void* dout, void* partial_f32, long M, int K, void* strm)
 {
-    hipLaunchKernelGGL(colsum_kernel, dim3((K + 63) / 64), dim3(256), 0,
-                       (hipStream_t)strm, (unsigned short*)db,
+    int nblk = (int)((M + 3) / 4 < 512 ? (M + 3) / 4 : 512);
+    hipLaunchKernelGGL(colsum_part_kernel, dim3(nblk), dim3(256), 0,
+                       (hipStream_t)strm, (float*)partial_f32,
                        (const unsigned short*)dout, M, K);
+    hipLaunchKernelGGL(colsum_fold_kernel, dim3((K + 255) / 256), dim3(256), 0,
+                       (hipStream_t)strm, (unsigned short*)db,
+                       (const float*)partial_f32, nblk, K);
 }
 
 // [K][RSC] -> [RSC][K] bf16 transpose for the dgrad wT operand (torch's
