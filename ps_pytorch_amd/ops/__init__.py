@@ -59,6 +59,9 @@ def _try_load() -> None:
             ctypes.c_long, ctypes.c_int, ctypes.c_void_p]
         lib.ps_wt_transpose.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                         ctypes.c_int, ctypes.c_int, ctypes.c_void_p]
+        lib.ps_fused_adam.argtypes = [ctypes.c_void_p] * 6 + [
+            ctypes.c_long] + [ctypes.c_float] * 7 + [
+            ctypes.c_int, ctypes.c_int, ctypes.c_void_p]
         lib.ps_bn_fwd.argtypes = [ctypes.c_void_p] * 11 + [
             ctypes.c_long, ctypes.c_long, ctypes.c_float, ctypes.c_float,
             ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p]

@@ -64,6 +64,52 @@ def fused_sgd_step(
         wire_out.copy_(w.to(wire_out.dtype))
 
 
+
+def fused_adam_step(
+    w: torch.Tensor, grad_sum: torch.Tensor, exp_avg: torch.Tensor,
+    exp_avg_sq: torch.Tensor, t: int, lr: float, beta1: float, beta2: float,
+    eps: float, weight_decay: float = 0.0, grad_scale: float = 1.0,
+    max_exp_avg_sq: Optional[torch.Tensor] = None,
+    wire_out: Optional[torch.Tensor] = None,
+) -> None:
+    """Fused flat Adam (ref src/optim/adam.py:48-95 semantics, incl.
+    amsgrad); one kernel pass on GPU, torch ops on CPU."""
+    n = w.numel()
+    _check_flat(w, "w", n)
+    _check_flat(grad_sum, "grad_sum", n)
+    bc1 = 1.0 - beta1 ** t
+    bc2 = 1.0 - beta2 ** t
+    if w.is_cuda:
+        if n % 4:
+            raise ValueError("flat buffers must be padded to a multiple of 4")
+        lib = require_lib()
+        lib.ps_fused_adam(
+            w.data_ptr(), grad_sum.data_ptr(), exp_avg.data_ptr(),
+            exp_avg_sq.data_ptr(),
+            0 if max_exp_avg_sq is None else max_exp_avg_sq.data_ptr(),
+            0 if wire_out is None else wire_out.data_ptr(), n,
+            float(lr / bc1), float(beta1), float(beta2), float(1.0 / bc2),
+            float(eps), float(weight_decay), float(grad_scale),
+            dtype_tag(grad_sum.dtype),
+            dtype_tag(wire_out.dtype) if wire_out is not None else 0,
+            current_stream_ptr())
+        return
+    g = grad_sum.to(torch.float32)
+    if grad_scale != 1.0:
+        g = g * grad_scale
+    if weight_decay:
+        g = g.add(w, alpha=weight_decay)
+    exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
+    exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+    if max_exp_avg_sq is not None:
+        torch.maximum(max_exp_avg_sq, exp_avg_sq, out=max_exp_avg_sq)
+        denom = (max_exp_avg_sq / bc2).sqrt_().add_(eps)
+    else:
+        denom = (exp_avg_sq / bc2).sqrt_().add_(eps)
+    w.addcdiv_(exp_avg, denom, value=-lr / bc1)
+    if wire_out is not None:
+        wire_out.copy_(w.to(wire_out.dtype))
+
 def pack_wire(dst: torch.Tensor, src: torch.Tensor) -> None:
     """Wire pack: f32 -> bf16 (the GPU 'compression' path, ref compression.py
     g_compress role) or plain copy when dtypes match."""

@@ -106,3 +106,102 @@ extern "C" void ps_fused_sgd(
                         else        launch_sel<false, false, false>((float*)w, g, (float*)m, wire, nvec, lr, mu, wd, scale, g_dtype, wire_dtype, s, blocks); }
     }
 }
+
+// ---------------------------------------------------------------- Adam
+//
+// Fused PS-side Adam over the flat buffer (ref: src/optim/adam.py:48-95,
+// incl. amsgrad), one pass:
+//   g  = grad_sum * scale (+ wd*w)
+//   m  = b1*m + (1-b1)*g ;  v = b2*v + (1-b2)*g^2
+//   [vm = max(vm, v)]
+//   w -= (lr/bc1) * m / (sqrt(v[max]/bc2) + eps)
+//   wire = cast(w)                     (optional broadcast payload)
+// bc1/bc2 bias corrections are host-computed scalars per step.
+
+template <typename GV, typename WIREV, bool AMS, bool HAS_WIRE, bool HAS_WD>
+__global__ __launch_bounds__(256) void fused_adam_kernel(
+    float* __restrict__ w, const void* __restrict__ g_, float* __restrict__ m,
+    float* __restrict__ v, float* __restrict__ vmax, void* __restrict__ wire_,
+    long nvec, float lr_bc1, float b1, float b2, float inv_bc2, float eps,
+    float wd, float scale)
+{
+    const GV* __restrict__ g = (const GV*)g_;
+    WIREV* __restrict__ wire = (WIREV*)wire_;
+    long stride = (long)gridDim.x * blockDim.x;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nvec; i += stride) {
+        float4_t wv = ((float4_t*)w)[i];
+        float4_t mv = ((float4_t*)m)[i];
+        float4_t vv = ((float4_t*)v)[i];
+        GV gvr = g[i];
+        float4_t gv;
+        if constexpr (sizeof(GV) == 8) {
+            ushort4_t u = *(ushort4_t*)&gvr;
+            gv = {bf16_to_f32(u.x), bf16_to_f32(u.y), bf16_to_f32(u.z), bf16_to_f32(u.w)};
+        } else {
+            gv = *(float4_t*)&gvr;
+        }
+        float4_t xv;
+        if constexpr (AMS) xv = ((float4_t*)vmax)[i];
+#pragma unroll
+        for (int k = 0; k < 4; ++k) {
+            float gk = gv[k] * scale;
+            if constexpr (HAS_WD) gk = fmaf(wd, wv[k], gk);
+            float mk = b1 * mv[k] + (1.f - b1) * gk;
+            float vk = b2 * vv[k] + (1.f - b2) * gk * gk;
+            mv[k] = mk; vv[k] = vk;
+            float veff = vk;
+            if constexpr (AMS) {
+                veff = fmaxf(xv[k], vk);
+                xv[k] = veff;
+            }
+            float denom = sqrtf(veff * inv_bc2) + eps;
+            wv[k] = wv[k] - lr_bc1 * mk / denom;
+        }
+        ((float4_t*)w)[i] = wv;
+        ((float4_t*)m)[i] = mv;
+        ((float4_t*)v)[i] = vv;
+        if constexpr (AMS) ((float4_t*)vmax)[i] = xv;
+        if constexpr (HAS_WIRE) {
+            if constexpr (sizeof(WIREV) == 8) {
+                ushort4_t o = {f32_to_bf16(wv.x), f32_to_bf16(wv.y),
+                               f32_to_bf16(wv.z), f32_to_bf16(wv.w)};
+                wire[i] = *(WIREV*)&o;
+            } else {
+                wire[i] = *(WIREV*)&wv;
+            }
+        }
+    }
+}
+
+template <bool A, bool W, bool D>
+static void adam_sel(float* w, const void* g, float* m, float* v, float* vm,
+                     void* wire, long nvec, float lr_bc1, float b1, float b2,
+                     float inv_bc2, float eps, float wd, float scale,
+                     int g_dtype, int wire_dtype, hipStream_t s, int blocks) {
+    dim3 grid(blocks), block(256);
+#define AD(GT, WT) hipLaunchKernelGGL((fused_adam_kernel<GT, WT, A, W, D>),   \
+        grid, block, 0, s, w, g, m, v, vm, wire, nvec, lr_bc1, b1, b2,        \
+        inv_bc2, eps, wd, scale)
+    if (g_dtype == PS_BF16) { if (!W || wire_dtype == PS_BF16) AD(U64x, U64x); else AD(U64x, F128x); }
+    else                    { if (!W || wire_dtype == PS_BF16) AD(F128x, U64x); else AD(F128x, F128x); }
+#undef AD
+}
+
+extern "C" void ps_fused_adam(
+    void* w, const void* g, void* m, void* v, void* vmax, void* wire, long n,
+    float lr_bc1, float b1, float b2, float inv_bc2, float eps, float wd,
+    float scale, int g_dtype, int wire_dtype, void* stream)
+{
+    long nvec = n / 4;
+    int blocks; ew_grid(nvec, 256, &blocks);
+    hipStream_t s = (hipStream_t)stream;
+    bool ams = vmax != nullptr, hw = wire != nullptr, hd = wd != 0.0f;
+#define SEL(A, W, D) adam_sel<A, W, D>((float*)w, g, (float*)m, (float*)v,    \
+        (float*)vmax, wire, nvec, lr_bc1, b1, b2, inv_bc2, eps, wd, scale,    \
+        g_dtype, wire_dtype, s, blocks)
+    if (ams) { if (hw) { if (hd) SEL(true, true, true); else SEL(true, true, false); }
+               else    { if (hd) SEL(true, false, true); else SEL(true, false, false); } }
+    else     { if (hw) { if (hd) SEL(false, true, true); else SEL(false, true, false); }
+               else    { if (hd) SEL(false, false, true); else SEL(false, false, false); } }
+#undef SEL
+}

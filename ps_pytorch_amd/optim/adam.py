@@ -1,9 +1,10 @@
 """PS-side Adam on the flat master buffer.
 
 Reference parity: src/optim/adam.py:38-95 (full Adam incl. amsgrad, consuming
-wire gradients). Flat-tensor torch ops today (a handful of fused elementwise
-passes over one contiguous buffer); candidate for a single fused HIP kernel
-like FlatSGD's.
+wire gradients). GPU path: ONE fused HIP kernel over the flat buffer
+(ops/kernels/fused_sgd.hip: ps_fused_adam) — m/v update, bias-corrected
+step, optional amsgrad and wire re-pack in a single pass; CPU path: the
+equivalent torch ops (the numerics reference).
 """
 from __future__ import annotations
 
@@ -33,25 +34,12 @@ class FlatAdam:
     @torch.no_grad()
     def step(self, grad_sum: torch.Tensor, grad_scale: float = 1.0,
              wire_out: Optional[torch.Tensor] = None) -> None:
+        from ..ops.functional import fused_adam_step
         self.t += 1
-        g = grad_sum.to(torch.float32)
-        if grad_scale != 1.0:
-            g = g * grad_scale
-        if self.weight_decay:
-            g = g.add(self.w, alpha=self.weight_decay)
-        self.exp_avg.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
-        self.exp_avg_sq.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
-        bc1 = 1 - self.beta1 ** self.t
-        bc2 = 1 - self.beta2 ** self.t
-        if self.amsgrad:
-            torch.maximum(self.max_exp_avg_sq, self.exp_avg_sq,
-                          out=self.max_exp_avg_sq)
-            denom = (self.max_exp_avg_sq / bc2).sqrt_().add_(self.eps)
-        else:
-            denom = (self.exp_avg_sq / bc2).sqrt_().add_(self.eps)
-        self.w.addcdiv_(self.exp_avg, denom, value=-self.lr / bc1)
-        if wire_out is not None:
-            wire_out.copy_(self.w.to(wire_out.dtype))
+        fused_adam_step(self.w, grad_sum, self.exp_avg, self.exp_avg_sq,
+                        self.t, self.lr, self.beta1, self.beta2, self.eps,
+                        self.weight_decay, grad_scale,
+                        self.max_exp_avg_sq, wire_out)
 
     def state_dict(self) -> dict:
         return {'t': self.t, 'exp_avg': self.exp_avg,

@@ -56,6 +56,40 @@ def test_fused_sgd_kernel_vs_cpu_reference(lib, g_dtype, nesterov):
     assert torch.equal(wire_d.cpu(), w_d.cpu().to(torch.bfloat16))
 
 
+@pytest.mark.parametrize('g_dtype', [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize('amsgrad', [False, True])
+def test_fused_adam_kernel_vs_cpu_reference(lib, g_dtype, amsgrad):
+    from ps_pytorch_amd.ops.functional import fused_adam_step
+    torch.manual_seed(1)
+    n = 1 << 20
+    w = torch.randn(n)
+    g = torch.randn(n).to(g_dtype)
+    m = torch.zeros(n)
+    v = torch.zeros(n)
+    vm = torch.zeros(n) if amsgrad else None
+    wire = torch.empty(n, dtype=torch.bfloat16)
+
+    w_d, g_d, m_d, v_d = w.cuda(), g.cuda(), m.cuda(), v.cuda()
+    vm_d = vm.cuda() if amsgrad else None
+    wire_d = wire.cuda()
+    for t in range(1, 4):
+        fused_adam_step(w_d, g_d, m_d, v_d, t, lr=1e-3, beta1=0.9,
+                        beta2=0.999, eps=1e-8, weight_decay=1e-4,
+                        grad_scale=1.0 / 7, max_exp_avg_sq=vm_d,
+                        wire_out=wire_d)
+    torch.cuda.synchronize()
+    for t in range(1, 4):
+        fused_adam_step(w, g, m, v, t, lr=1e-3, beta1=0.9, beta2=0.999,
+                        eps=1e-8, weight_decay=1e-4, grad_scale=1.0 / 7,
+                        max_exp_avg_sq=vm, wire_out=None)
+    assert torch.allclose(w_d.cpu(), w, atol=1e-5, rtol=1e-5)
+    assert torch.allclose(m_d.cpu(), m, atol=1e-5, rtol=1e-5)
+    assert torch.allclose(v_d.cpu(), v, atol=1e-5, rtol=1e-5)
+    if amsgrad:
+        assert torch.allclose(vm_d.cpu(), vm, atol=1e-5, rtol=1e-5)
+    assert torch.equal(wire_d.cpu(), w_d.cpu().to(torch.bfloat16))
+
+
 def test_single_gpu_train_step_smoke():
     from ps_pytorch_amd.config import JobConfig
     from ps_pytorch_amd.trainer import NNTrainer
