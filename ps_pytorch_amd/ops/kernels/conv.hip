@@ -56,11 +56,16 @@ __device__ __forceinline__ V16 zero16() {
     V16 v; v.u4 = make_uint4(0, 0, 0, 0); return v;
 }
 
-// 16B load with a tail guard for channel counts that are not multiples of 8
-// (LeNet C=1/20/50); ResNet/VGG take the single vector load.
+// 16B load; AL=true (channel counts divisible by 8 — every ResNet/VGG
+// shape) is a single unconditional vector load, AL=false takes the guarded
+// scalar path (LeNet C=1/20/50). The per-call branch of a runtime check
+// showed up as ~1/3 of the kernels' scalar-pipe traffic.
+template <bool AL>
 __device__ __forceinline__ V16 load16(const unsigned short* src, int cc, int Cn) {
     V16 v;
-    if (cc + 8 <= Cn && ((Cn | cc) & 7) == 0) {
+    if constexpr (AL) {
+        v.u4 = *(const uint4*)src;
+    } else if (cc + 8 <= Cn && ((Cn | cc) & 7) == 0) {
         v.u4 = *(const uint4*)src;
     } else {
         v = zero16();
@@ -98,7 +103,7 @@ __device__ __forceinline__ int gswz(int row, int g) {
 // 3*3*3=27, LeNet conv1 5*5*1=25) — flatten (r,s,c) into the contraction
 // axis via a per-lane gather table and run ONE k-step instead of R*S, so
 // the MFMA utilization is RSC/64 of a full tile instead of C/64 per step.
-template <int TM, int TN, int STRIDE, bool DGRAD, bool SMALL = false>
+template <int TM, int TN, int STRIDE, bool DGRAD, bool SMALL = false, bool AL = true>
 __global__ __launch_bounds__(256) void conv_gemm_kernel(
     const unsigned short* __restrict__ src,  // fwd: in [Nb,H,W,C]; dgrad: dout [Nb,P,Q,K]
     const unsigned short* __restrict__ wgt,  // fwd: w [K,R,S,C]; dgrad: wT [R,S,C,K]
@@ -246,15 +251,15 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
                 areg[rr] = v;
             } else {
                 areg[rr] = (vm[rr] & 1)
-                    ? load16(pA[rr], cbase, Cin) : zero16();
+                    ? load16<AL>(pA[rr], cbase, Cin) : zero16();
             }
         }
 #pragma unroll
         for (int rr = 0; rr < BR; ++rr) {
             if constexpr (SMALL)
-                breg[rr] = colv[rr] ? load16(pB[rr], cc8, R * S * C) : zero16();
+                breg[rr] = colv[rr] ? load16<AL>(pB[rr], cc8, R * S * C) : zero16();
             else
-                breg[rr] = colv[rr] ? load16(pB[rr], cbase, Cin) : zero16();
+                breg[rr] = colv[rr] ? load16<AL>(pB[rr], cbase, Cin) : zero16();
         }
         if constexpr (!SMALL) {
             long dA, dB;
@@ -355,7 +360,7 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
 // s == (wp+pad) mod 2 contribute — so instead of predicating 3/4 of the MFMA
 // work to zero (what a direct stride-2 GEMM does), launch 4 kernels, one per
 // class, each contracting only its valid (r,s) subset at full tile density.
-template <int TM, int TN>
+template <int TM, int TN, bool AL = true>
 __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
     const unsigned short* __restrict__ dout, // [Nb,P,Q,K]
     const unsigned short* __restrict__ wgt,  // wT [R,S,C,K]
@@ -436,10 +441,10 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
         const int kbase = (lkc << 6) + cc8;
 #pragma unroll
         for (int rr = 0; rr < AR; ++rr)
-            areg[rr] = (vm[rr] & 1) ? load16(pA[rr], kbase, K) : zero16();
+            areg[rr] = (vm[rr] & 1) ? load16<AL>(pA[rr], kbase, K) : zero16();
 #pragma unroll
         for (int rr = 0; rr < BR; ++rr)
-            breg[rr] = colv[rr] ? load16(pB[rr], kbase, K) : zero16();
+            breg[rr] = colv[rr] ? load16<AL>(pB[rr], kbase, K) : zero16();
         long dA, dB;
         bool rs_adv = (++lkc == KC);
         if (rs_adv) {
@@ -554,12 +559,15 @@ struct WgradLds {
     unsigned short B[2][4][WG_SUB];         // [c-subtile][pixel*16+coff]
 };
 
-// one hardware transpose read: 4 contiguous bf16 at 8-B-aligned LDS byte
-// offset `off`; the 16-lane group's lanes get their column's 4 values.
-__device__ __forceinline__ unsigned long long ds_tr16(unsigned off) {
-    unsigned long long v;
-    asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(v) : "v"(off));
-    return v;
+// one hardware transpose read: 4 contiguous bf16 at an 8-B-aligned LDS
+// address; the 16-lane group's lanes get their column's 4 values. The
+// compiler-modeled builtin (ck_tile idiom) — an inline-asm version with a
+// tied-operand waitcnt barrier cost ~150 VALU/step in AGPR<->VGPR copies.
+typedef __bf16 bf16x4_t __attribute__((ext_vector_type(4)));
+__device__ __forceinline__ bf16x4_t ds_tr16p(const unsigned short* p) {
+    typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 v4;
+    auto lp = (__attribute__((address_space(3))) v4*)(p);
+    return __builtin_amdgcn_ds_read_tr16_b64_v4bf16(lp);
 }
 
 // One block: TK k x 64 c output tile for ONE (r,s), summing the pixel range
@@ -567,7 +575,7 @@ __device__ __forceinline__ unsigned long long ds_tr16(unsigned off) {
 // TK=128 for K>=128 layers (halves the scatter cost per MFMA and the
 // dout re-reads); TK=64 otherwise.
 // POW2: P*Q and Q are powers of two (shift decode); else runtime div.
-template <int STRIDE, bool POW2, int TK>
+template <int STRIDE, bool POW2, int TK, bool AL = true>
 __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     const unsigned short* __restrict__ dout,  // [Nb,P,Q,K]
     const unsigned short* __restrict__ in,    // [Nb,H,W,C]
@@ -638,10 +646,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
                 }
 #pragma unroll
                 for (int h = 0; h < KH; ++h)
-                    areg[rr][h] = load16(aptr[rr] + h * 64, k0 + h * 64 + cc8, K);
+                    areg[rr][h] = load16<AL>(aptr[rr] + h * 64, k0 + h * 64 + cc8, K);
                 int hh = p * STRIDE - pad + r, ww = q * STRIDE - pad + s;
                 if (hh >= 0 && hh < H && ww >= 0 && ww < W)
-                    bv = load16(in + ((long)(n * H + hh) * W + ww) * C + c0 + cc8,
+                    bv = load16<AL>(in + ((long)(n * H + hh) * W + ww) * C + c0 + cc8,
                                 c0 + cc8, C);
             }
             breg[rr] = bv;
@@ -701,37 +709,26 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
         }
     };
 
-    union U64x8 { unsigned long long q[2]; bf16x8_t v; };
+    union U64x8 { bf16x4_t h[2]; bf16x8_t v; };
+    const char* lb = (const char*)&lds;
     auto mfma_step = [&](int buf) {     // literal buf only
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
             U64x8 a[MI], bfr[2];
 #pragma unroll
             for (int mi = 0; mi < MI; ++mi) {
-                a[mi].q[0] = ds_tr16(roA[kk][mi][0] + (buf ? A1 : 0));
-                a[mi].q[1] = ds_tr16(roA[kk][mi][1] + (buf ? A1 : 0));
+                a[mi].h[0] = ds_tr16p((const unsigned short*)
+                    (lb + roA[kk][mi][0] + (buf ? A1 : 0)));
+                a[mi].h[1] = ds_tr16p((const unsigned short*)
+                    (lb + roA[kk][mi][1] + (buf ? A1 : 0)));
             }
 #pragma unroll
             for (int nj = 0; nj < 2; ++nj) {
-                bfr[nj].q[0] = ds_tr16(roB[kk][nj][0] + (buf ? BB1 : 0));
-                bfr[nj].q[1] = ds_tr16(roB[kk][nj][1] + (buf ? BB1 : 0));
+                bfr[nj].h[0] = ds_tr16p((const unsigned short*)
+                    (lb + roB[kk][nj][0] + (buf ? BB1 : 0)));
+                bfr[nj].h[1] = ds_tr16p((const unsigned short*)
+                    (lb + roB[kk][nj][1] + (buf ? BB1 : 0)));
             }
-            // one wait for the whole batch; the value operands pin the
-            // mfma uses behind it
-            if constexpr (MI == 2)
-                asm volatile("s_waitcnt lgkmcnt(0)"
-                             : "+v"(a[0].q[0]), "+v"(a[0].q[1]),
-                               "+v"(a[1].q[0]), "+v"(a[1].q[1]),
-                               "+v"(bfr[0].q[0]), "+v"(bfr[0].q[1]),
-                               "+v"(bfr[1].q[0]), "+v"(bfr[1].q[1]));
-            else
-                asm volatile("s_waitcnt lgkmcnt(0)"
-                             : "+v"(a[0].q[0]), "+v"(a[0].q[1]),
-                               "+v"(a[1].q[0]), "+v"(a[1].q[1]),
-                               "+v"(a[2].q[0]), "+v"(a[2].q[1]),
-                               "+v"(a[3].q[0]), "+v"(a[3].q[1]),
-                               "+v"(bfr[0].q[0]), "+v"(bfr[0].q[1]),
-                               "+v"(bfr[1].q[0]), "+v"(bfr[1].q[1]));
 #pragma unroll
             for (int mi = 0; mi < MI; ++mi)
 #pragma unroll
@@ -778,7 +775,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
 // flattened (r,s,c) axis fits one 64-column tile, so one block covers every
 // tap in a single pass over its pixel chunk — vs the generic kernel's
 // R*S separate 64-c tiles at C/64 utilization each.
-template <int STRIDE, bool POW2>
+template <int STRIDE, bool POW2, bool AL = true>
 __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
     const unsigned short* __restrict__ dout,  // [Nb,P,Q,K]
     const unsigned short* __restrict__ in,    // [Nb,H,W,C]
@@ -840,7 +837,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
                     int rem = (int)(m % ((long)P * Q));
                     p = rem / Q; q = rem % Q;
                 }
-                av = load16(dout + m * K + k0 + cc8, k0 + cc8, K);
+                av = load16<AL>(dout + m * K + k0 + cc8, k0 + cc8, K);
                 int h0 = p * STRIDE - pad, w0 = q * STRIDE - pad;
                 const unsigned short* pix =
                     in + (long)n * H * W * C + ((long)h0 * W + w0) * C;
@@ -893,26 +890,26 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
         }
     };
 
-    union U64x8 { unsigned long long q[2]; bf16x8_t v; };
+    union U64x8 { bf16x4_t h[2]; bf16x8_t v; };
+    const char* lb = (const char*)&lds;
     auto mfma_step = [&](int buf) {     // literal buf only
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
             U64x8 a[2], bfr[2];
 #pragma unroll
             for (int mi = 0; mi < 2; ++mi) {
-                a[mi].q[0] = ds_tr16(roA[kk][mi][0] + (buf ? A1 : 0));
-                a[mi].q[1] = ds_tr16(roA[kk][mi][1] + (buf ? A1 : 0));
+                a[mi].h[0] = ds_tr16p((const unsigned short*)
+                    (lb + roA[kk][mi][0] + (buf ? A1 : 0)));
+                a[mi].h[1] = ds_tr16p((const unsigned short*)
+                    (lb + roA[kk][mi][1] + (buf ? A1 : 0)));
             }
 #pragma unroll
             for (int nj = 0; nj < 2; ++nj) {
-                bfr[nj].q[0] = ds_tr16(roB[kk][nj][0] + (buf ? BB1 : 0));
-                bfr[nj].q[1] = ds_tr16(roB[kk][nj][1] + (buf ? BB1 : 0));
+                bfr[nj].h[0] = ds_tr16p((const unsigned short*)
+                    (lb + roB[kk][nj][0] + (buf ? BB1 : 0)));
+                bfr[nj].h[1] = ds_tr16p((const unsigned short*)
+                    (lb + roB[kk][nj][1] + (buf ? BB1 : 0)));
             }
-            asm volatile("s_waitcnt lgkmcnt(0)"
-                         : "+v"(a[0].q[0]), "+v"(a[0].q[1]),
-                           "+v"(a[1].q[0]), "+v"(a[1].q[1]),
-                           "+v"(bfr[0].q[0]), "+v"(bfr[0].q[1]),
-                           "+v"(bfr[1].q[0]), "+v"(bfr[1].q[1]));
 #pragma unroll
             for (int mi = 0; mi < 2; ++mi)
 #pragma unroll
@@ -1024,7 +1021,7 @@ __global__ __launch_bounds__(256) void colsum_fold_kernel(
         long M_ = DG ? (long)Nb * H * W : (long)Nb * P * Q;                   \
         int Nout_ = DG ? C : K;                                               \
         long grid = ((M_ + TM - 1) / TM) * ((Nout_ + TN - 1) / TN);           \
-        hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, ST, DG, SM>),            \
+        hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, ST, DG, SM, ALV>),       \
             dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,            \
             (const unsigned short*)src, (const unsigned short*)wgt,           \
             (const unsigned short*)bias, (unsigned short*)dst,                \
@@ -1036,16 +1033,23 @@ extern "C" void ps_conv_fwd(
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int stride, int pad, void* strm)
 {
-    if (R * S > 1 && R * S * C <= 64) {   // flattened-contraction stem path
-        if (stride == 1) LAUNCH_GEMM(128, 64, 1, false, true);
-        else             LAUNCH_GEMM(128, 64, 2, false, true);
-    } else if (K >= 128) {
-        if (stride == 1) LAUNCH_GEMM(128, 128, 1, false, false);
-        else             LAUNCH_GEMM(128, 128, 2, false, false);
-    } else {
-        if (stride == 1) LAUNCH_GEMM(256, 64, 1, false, false);
-        else             LAUNCH_GEMM(256, 64, 2, false, false);
-    }
+    bool al = ((C | K) & 7) == 0;
+#define FWD_BODY()                                                            \
+    do {                                                                      \
+        if (R * S > 1 && R * S * C <= 64) {                                   \
+            if (stride == 1) LAUNCH_GEMM(128, 64, 1, false, true);            \
+            else             LAUNCH_GEMM(128, 64, 2, false, true);            \
+        } else if (K >= 128) {                                                \
+            if (stride == 1) LAUNCH_GEMM(128, 128, 1, false, false);          \
+            else             LAUNCH_GEMM(128, 128, 2, false, false);          \
+        } else {                                                              \
+            if (stride == 1) LAUNCH_GEMM(256, 64, 1, false, false);           \
+            else             LAUNCH_GEMM(256, 64, 2, false, false);           \
+        }                                                                     \
+    } while (0)
+    if (al) { constexpr bool ALV = true; FWD_BODY(); }
+    else    { constexpr bool ALV = false; FWD_BODY(); }
+#undef FWD_BODY
 }
 
 // wgt here is the TRANSPOSED weight wT[R,S,C,K] (host permutes once per
@@ -1056,9 +1060,16 @@ extern "C" void ps_conv_dgrad(
     int R, int S, int stride, int pad, void* strm)
 {
     const void* bias = nullptr;
+    bool al = ((C | K) & 7) == 0;
     if (stride == 1) {
-        if (C >= 128) LAUNCH_GEMM(128, 128, 1, true, false);
-        else          LAUNCH_GEMM(128, 64, 1, true, false);
+#define DG_BODY()                                                             \
+        do {                                                                  \
+            if (C >= 128) LAUNCH_GEMM(128, 128, 1, true, false);              \
+            else          LAUNCH_GEMM(128, 64, 1, true, false);               \
+        } while (0)
+        if (al) { constexpr bool ALV = true; DG_BODY(); }
+        else    { constexpr bool ALV = false; DG_BODY(); }
+#undef DG_BODY
     } else {
         // 4 parity-class launches (see conv_dgrad2_kernel); classes with no
         // contributing (r,s) — e.g. 3 of 4 for a 1x1 stride-2 conv — are
@@ -1079,14 +1090,16 @@ extern "C" void ps_conv_dgrad(
             if (R <= r0 || S <= s0) continue;
             int Hc = (H - hp + 1) >> 1, Wc = (W - wp + 1) >> 1;
             long M_ = (long)Nb * Hc * Wc;
-#define DG2(TM, TN)                                                           \
-            hipLaunchKernelGGL((conv_dgrad2_kernel<TM, TN>),                  \
+#define DG2(TM, TN, ALV)                                                      \
+            hipLaunchKernelGGL((conv_dgrad2_kernel<TM, TN, ALV>),             \
                 dim3((unsigned)(((M_ + TM - 1) / TM) * ((C + TN - 1) / TN))), \
                 dim3(256), 0, (hipStream_t)strm, dout,                        \
                 (const unsigned short*)wgt, (unsigned short*)dst,             \
                 Nb, H, W, C, K, P, Q, R, S, pad, hp, wp, Hc, Wc)
-            if (C >= 128) DG2(128, 128);
-            else          DG2(128, 64);
+            if (al) { if (C >= 128) DG2(128, 128, true);
+                      else          DG2(128, 64, true); }
+            else    { if (C >= 128) DG2(128, 128, false);
+                      else          DG2(128, 64, false); }
 #undef DG2
         }
     }
@@ -1111,7 +1124,7 @@ extern "C" void ps_conv_wgrad(
         int l2pq_ = ilog2_exact((long)P * Q), l2q_ = ilog2_exact(Q);
         bool pw = l2pq_ >= 0 && l2q_ >= 0;
 #define WGS(ST, PW)                                                           \
-        hipLaunchKernelGGL((conv_wgrad_small_kernel<ST, PW>),                 \
+        hipLaunchKernelGGL((conv_wgrad_small_kernel<ST, PW, false>),          \
             dim3((unsigned)grid_s), dim3(256), 0, (hipStream_t)strm,          \
             (const unsigned short*)dout, (const unsigned short*)in,           \
             (float*)partial_f32, Nb, H, W, C, K, P, Q, R, S, pad, split,      \
@@ -1134,15 +1147,19 @@ extern "C" void ps_conv_wgrad(
     int l2pq = ilog2_exact((long)P * Q), l2q = ilog2_exact(Q);
     bool pow2 = l2pq >= 0 && l2q >= 0;
 #define WG_LAUNCH(ST, PW, TKV)                                                \
-    hipLaunchKernelGGL((conv_wgrad_kernel<ST, PW, TKV>),                      \
+    hipLaunchKernelGGL((conv_wgrad_kernel<ST, PW, TKV, ALV>),                 \
         dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,                \
         (const unsigned short*)dout, (const unsigned short*)in,               \
         (float*)partial_f32, Nb, H, W, C, K, P, Q, R, S, pad, split, chunk,   \
         l2pq, l2q, per_xcd)
 #define WG_TK(ST, PW) do { if (TK == 128) WG_LAUNCH(ST, PW, 128);             \
                            else WG_LAUNCH(ST, PW, 64); } while (0)
-    if (stride == 1) { if (pow2) WG_TK(1, true); else WG_TK(1, false); }
-    else             { if (pow2) WG_TK(2, true); else WG_TK(2, false); }
+#define WG_AL(ST, PW) do { if (((C | K) & 7) == 0) {                          \
+        constexpr bool ALV = true; WG_TK(ST, PW);                             \
+    } else { constexpr bool ALV = false; WG_TK(ST, PW); } } while (0)
+    if (stride == 1) { if (pow2) WG_AL(1, true); else WG_AL(1, false); }
+    else             { if (pow2) WG_AL(2, true); else WG_AL(2, false); }
+#undef WG_AL
 #undef WG_TK
 #undef WG_LAUNCH
     long n = (long)K * R * S * C;
