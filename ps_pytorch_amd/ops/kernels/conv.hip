@@ -1033,7 +1033,7 @@ extern "C" void ps_conv_fwd(
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int stride, int pad, void* strm)
 {
-    bool al = ((C | K) & 7) == 0;
+    bool al = (C & 63) == 0;           // full 64-chunk contraction coverage
 #define FWD_BODY()                                                            \
     do {                                                                      \
         if (R * S > 1 && R * S * C <= 64) {                                   \
@@ -1060,7 +1060,7 @@ extern "C" void ps_conv_dgrad(
     int R, int S, int stride, int pad, void* strm)
 {
     const void* bias = nullptr;
-    bool al = ((C | K) & 7) == 0;
+    bool al = (K & 63) == 0;           // contraction runs over K
     if (stride == 1) {
 #define DG_BODY()                                                             \
         do {                                                                  \
@@ -1154,7 +1154,7 @@ extern "C" void ps_conv_wgrad(
         l2pq, l2q, per_xcd)
 #define WG_TK(ST, PW) do { if (TK == 128) WG_LAUNCH(ST, PW, 128);             \
                            else WG_LAUNCH(ST, PW, 64); } while (0)
-#define WG_AL(ST, PW) do { if (((C | K) & 7) == 0) {                          \
+#define WG_AL(ST, PW) do { if ((C & 63) == 0 && K % TK == 0) {                \
         constexpr bool ALV = true; WG_TK(ST, PW);                             \
     } else { constexpr bool ALV = false; WG_TK(ST, PW); } } while (0)
     if (stride == 1) { if (pow2) WG_AL(1, true); else WG_AL(1, false); }
