@@ -49,7 +49,7 @@ def _wgrad_split(M: int, K: int, C: int, R: int, S: int) -> int:
     else:
         tk = 128 if K >= 128 else 64
         tiles = ((K + tk - 1) // tk) * R * S * ((C + 63) // 64)
-    want = max(1, 512 // max(tiles, 1))
+    want = max(1, 768 // max(tiles, 1))
     max_split = max(1, M // 64)
     return max(1, min(want, max_split, 256))
 
