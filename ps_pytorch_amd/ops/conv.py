@@ -49,7 +49,10 @@ def _wgrad_split(M: int, K: int, C: int, R: int, S: int) -> int:
     else:
         tk = 128 if K >= 128 else 64
         tiles = ((K + tk - 1) // tk) * R * S * ((C + 63) // 64)
-    want = max(1, 768 // max(tiles, 1))
+    # target ~2 blocks/CU of fill; more split also means more partial-slab
+    # traffic in the reduce, so keep a floor of 2 rather than over-splitting
+    target = int(os.environ.get('PS_WG_WANT', '512'))
+    want = max(2, target // max(tiles, 1))
     max_split = max(1, M // 64)
     return max(1, min(want, max_split, 256))
 
