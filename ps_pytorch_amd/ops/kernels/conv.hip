@@ -360,22 +360,35 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
 // s == (wp+pad) mod 2 contribute — so instead of predicating 3/4 of the MFMA
 // work to zero (what a direct stride-2 GEMM does), launch 4 kernels, one per
 // class, each contracting only its valid (r,s) subset at full tile density.
+// All parity classes ride ONE launch: `ends` = inclusive-exclusive prefix
+// of per-class block counts, `codes` = packed (hp<<1|wp) per slot — the
+// four separate launches were short-pipeline/launch-overhead bound.
 template <int TM, int TN, bool AL = true>
 __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
     const unsigned short* __restrict__ dout, // [Nb,P,Q,K]
     const unsigned short* __restrict__ wgt,  // wT [R,S,C,K]
     unsigned short* __restrict__ dx,         // [Nb,H,W,C]
     int Nb, int H, int W, int C, int K, int P, int Q,
-    int R, int S, int pad, int hp, int wp, int Hc, int Wc)
+    int R, int S, int pad, int4 ends, int4 codes)
 {
+    int bx = blockIdx.x;
+    int cls;
+    if      (bx < ends.x) cls = 0;
+    else if (bx < ends.y) { cls = 1; bx -= ends.x; }
+    else if (bx < ends.z) { cls = 2; bx -= ends.y; }
+    else                  { cls = 3; bx -= ends.z; }
+    const int code = (cls == 0) ? codes.x : (cls == 1) ? codes.y
+                     : (cls == 2) ? codes.z : codes.w;
+    const int hp = code >> 1, wp = code & 1;
+    const int Hc = (H - hp + 1) >> 1, Wc = (W - wp + 1) >> 1;
     __shared__ __attribute__((aligned(16))) FwdLds<TM, TN> lds;
     constexpr bool SWZ = (TN == 64);
     constexpr int AR = TM / 32;
     constexpr int BR = TN / 32;
     const long M = (long)Nb * Hc * Wc;
     const int tiles_n = (C + TN - 1) / TN;
-    const long m0 = (blockIdx.x / tiles_n) * (long)TM;
-    const int n0 = (blockIdx.x % tiles_n) * TN;
+    const long m0 = (bx / tiles_n) * (long)TM;
+    const int n0 = (bx % tiles_n) * TN;
     const int t = threadIdx.x;
     const int trow = t >> 3;
     const int cc8 = (t & 7) * 8;
@@ -1084,24 +1097,35 @@ extern "C" void ps_conv_dgrad(
         if (any_empty)
             (void)hipMemsetAsync(dst, 0, (long)Nb * H * W * C * 2,
                                  (hipStream_t)strm);
+        int TM_ = 128, TN_ = (C >= 128) ? 128 : 64;
+        int tiles_n_ = (C + TN_ - 1) / TN_;
+        int ends_[4] = {0, 0, 0, 0}, codes_[4] = {0, 0, 0, 0};
+        int slot = 0, total = 0;
         for (int hp = 0; hp < 2 && hp < H; ++hp)
         for (int wp = 0; wp < 2 && wp < W; ++wp) {
             int r0 = (hp + pad) & 1, s0 = (wp + pad) & 1;
             if (R <= r0 || S <= s0) continue;
             int Hc = (H - hp + 1) >> 1, Wc = (W - wp + 1) >> 1;
             long M_ = (long)Nb * Hc * Wc;
-#define DG2(TM, TN, ALV)                                                      \
-            hipLaunchKernelGGL((conv_dgrad2_kernel<TM, TN, ALV>),             \
-                dim3((unsigned)(((M_ + TM - 1) / TM) * ((C + TN - 1) / TN))), \
-                dim3(256), 0, (hipStream_t)strm, dout,                        \
-                (const unsigned short*)wgt, (unsigned short*)dst,             \
-                Nb, H, W, C, K, P, Q, R, S, pad, hp, wp, Hc, Wc)
-            if (al) { if (C >= 128) DG2(128, 128, true);
-                      else          DG2(128, 64, true); }
-            else    { if (C >= 128) DG2(128, 128, false);
-                      else          DG2(128, 64, false); }
-#undef DG2
+            total += (int)(((M_ + TM_ - 1) / TM_) * tiles_n_);
+            ends_[slot] = total;
+            codes_[slot] = (hp << 1) | wp;
+            ++slot;
         }
+        for (int i = slot; i < 4; ++i) { ends_[i] = total; codes_[i] = 0; }
+        if (total == 0) return;
+        int4 ends = make_int4(ends_[0], ends_[1], ends_[2], ends_[3]);
+        int4 codes = make_int4(codes_[0], codes_[1], codes_[2], codes_[3]);
+#define DG2(TM, TN, ALV)                                                      \
+        hipLaunchKernelGGL((conv_dgrad2_kernel<TM, TN, ALV>),                 \
+            dim3((unsigned)total), dim3(256), 0, (hipStream_t)strm, dout,     \
+            (const unsigned short*)wgt, (unsigned short*)dst,                 \
+            Nb, H, W, C, K, P, Q, R, S, pad, ends, codes)
+        if (al) { if (C >= 128) DG2(128, 128, true);
+                  else          DG2(128, 64, true); }
+        else    { if (C >= 128) DG2(128, 128, false);
+                  else          DG2(128, 64, false); }
+#undef DG2
     }
 }
 
