@@ -87,11 +87,11 @@ __device__ __forceinline__ V16 load16(const unsigned short* src, int cc, int Cn)
 // TN==64 tiles drop the 72-pad and XOR-swizzle 8-element granules instead
 // (same bank spreading, 11% less LDS) — that fits 2 blocks/CU at 256x64 and
 // 3 at 128x64, which the padded layout cannot.
-template <int TM, int TN>
+template <int TM, int TN, int NBUF = 2>
 struct FwdLds {
-    static constexpr int PITCH = (TN == 64) ? 64 : LDSP;
-    unsigned short A[2][TM][PITCH];
-    unsigned short B[2][TN][PITCH];
+    static constexpr int PITCH = 64;   // swizzled granules, no pad (all TN)
+    unsigned short A[NBUF][TM][PITCH];
+    unsigned short B[NBUF][TN][PITCH];
 };
 
 template <bool SWZ>
@@ -112,8 +112,8 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int pad)
 {
-    __shared__ __attribute__((aligned(16))) FwdLds<TM, TN> lds;
-    constexpr bool SWZ = (TN == 64);
+    __shared__ __attribute__((aligned(16))) FwdLds<TM, TN, SMALL ? 1 : 2> lds;
+    constexpr bool SWZ = true;
     constexpr int AR = TM / 32;        // A rows staged per thread
     constexpr int BR = TN / 32;        // B rows staged per thread
     const int Cin = DGRAD ? K : C;     // contraction channel count
@@ -382,7 +382,7 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
     const int hp = code >> 1, wp = code & 1;
     const int Hc = (H - hp + 1) >> 1, Wc = (W - wp + 1) >> 1;
     __shared__ __attribute__((aligned(16))) FwdLds<TM, TN> lds;
-    constexpr bool SWZ = (TN == 64);
+    constexpr bool SWZ = true;
     constexpr int AR = TM / 32;
     constexpr int BR = TN / 32;
     const long M = (long)Nb * Hc * Wc;
