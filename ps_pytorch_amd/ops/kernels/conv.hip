@@ -1056,8 +1056,8 @@ extern "C" void ps_conv_fwd(
             if (stride == 1) LAUNCH_GEMM(128, 128, 1, false, false);          \
             else             LAUNCH_GEMM(128, 128, 2, false, false);          \
         } else {                                                              \
-            if (stride == 1) LAUNCH_GEMM(256, 64, 1, false, false);           \
-            else             LAUNCH_GEMM(256, 64, 2, false, false);           \
+            if (stride == 1) LAUNCH_GEMM(128, 64, 1, false, false);           \
+            else             LAUNCH_GEMM(128, 64, 2, false, false);           \
         }                                                                     \
     } while (0)
     if (al) { constexpr bool ALV = true; FWD_BODY(); }
