@@ -41,13 +41,18 @@ def _supported(x: torch.Tensor, w: torch.Tensor, stride, padding,
     return R <= 7 and S <= 7
 
 
-def _wgrad_split(M: int, K: int, C: int, R: int, S: int) -> int:
-    """Pick the split-K factor: enough blocks to fill 256 CUs (~8 blocks/CU)
-    without exploding the f32 partial buffer."""
+def _wgrad_split(M: int, K: int, C: int, R: int, S: int,
+                 stride: int = 0, pad: int = -1, P: int = 0, Q: int = 0) -> int:
+    """Pick the split-K factor: enough blocks to fill 256 CUs (~2 blocks/CU)
+    without exploding the f32 partial buffer. Mirrors the kernel dispatch in
+    conv.hip (generic / small-RSC / row-halo)."""
+    tk = 128 if K >= 128 else 64
     if R * S > 1 and R * S * C <= 64:       # flattened small-RSC kernel
         tiles = (K + 63) // 64
+    elif (stride == 1 and R == 3 and S == 3 and pad == 1 and Q == P
+          and 0 < Q <= 32 and (Q & (Q - 1)) == 0 and (P & (P - 1)) == 0):
+        tiles = ((K + tk - 1) // tk) * 3 * ((C + 63) // 64)   # row-halo
     else:
-        tk = 128 if K >= 128 else 64
         tiles = ((K + tk - 1) // tk) * R * S * ((C + 63) // 64)
     # target ~2 blocks/CU of fill; more split also means more partial-slab
     # traffic in the reduce, so keep a floor of 2 rather than over-splitting
@@ -102,7 +107,7 @@ class _ConvFn(torch.autograd.Function):
                               current_stream_ptr())
         if ctx.needs_input_grad[1]:
             M = Nb * P * Q
-            split = _wgrad_split(M, K, C, R, S)
+            split = _wgrad_split(M, K, C, R, S, stride, pad, P, Q)
             partial = torch.empty(split * K * R * S * C,
                                   dtype=torch.float32, device=x.device)
             dw = torch.empty_like(w).contiguous(memory_format=_CL)

@@ -784,6 +784,229 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     }
 }
 
+// Row-halo wgrad for the dominant 3x3 stride-1 pad-1 family: one block owns
+// (k-tile, c-tile, ONE r) and walks whole output rows. Per 32-pixel step it
+// stages the dout rows once plus the matching input rows with a 2-pixel
+// halo — all S=3 s-taps then read the SAME staged rows at shifted pixel
+// positions (a row shift in the [pixel][chan] subtile image keeps the
+// transpose-read 8-B alignment; a channel shift would not). 3x the MACs per
+// staged byte of the generic kernel, and dout/in leave L2 R times instead
+// of R*S times. Requires: stride 1, R=S=3, pad 1 (so H=P, W=Q and the
+// (n*H+h) axis is globally linear in the output row index), Q a power of
+// two <= 32, P a power of two.
+//
+// Contraction axis = 32 consecutive output pixels (exactly one
+// mfma_f32_16x16x32 depth): ROWS = 32/Q output rows per step; each lane's
+// (row, q) split of its contraction index is a compile-run constant.
+template <int TK, bool AL>
+__global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
+    const unsigned short* __restrict__ dout,  // [Nb,P,Q,K]
+    const unsigned short* __restrict__ in,    // [Nb,H,W,C]
+    float* __restrict__ partial,              // [SPLIT][K][9*C]
+    int Nb, int H, int W, int C, int K, int P, int Q,
+    int split, int ipr, int l2q, int l2p, int per_xcd)
+{
+    // ipr: output rows per split chunk (multiple of ROWS)
+    constexpr int PAD = 1;
+    const int ROWS = 32 >> l2q;                 // output rows per step
+    const int HW2 = W + 2;                      // halo row width
+    __shared__ __attribute__((aligned(16))) struct {
+        unsigned short A[2][TK / 16][32 * 16 + 16];     // dout [k-sub][m*16+o]
+        unsigned short B[2][4][48 * 16 + 16];           // in   [c-sub][hp*16+o]
+    } lds;                                               // hp = rib*HW2 + j
+    constexpr int MI = TK / 32;
+    const int tiles_k = (K + TK - 1) / TK;
+    const int tiles_c = (C + 63) >> 6;
+    long l = (long)(blockIdx.x & 7) * per_xcd + (blockIdx.x >> 3);
+    const long nlog = (long)tiles_k * tiles_c * 3 * split;
+    if (l >= nlog) return;
+    int b = (int)l;
+    const int k0 = (b % tiles_k) * TK; b /= tiles_k;
+    const int c0 = (b % tiles_c) * 64; b /= tiles_c;
+    const int r = b % 3; b /= 3;
+    const int sid = b;
+    const long rows_total = (long)Nb * P;
+    const long row0 = (long)sid * ipr;
+    const long row1 = (row0 + ipr < rows_total) ? row0 + ipr : rows_total;
+
+    const int t = threadIdx.x;
+    const int cc8 = (t & 7) * 8;
+    const int lane = t & 63, wid = t >> 6;
+    const int wm = wid >> 1, wn = wid & 1;
+    const int fr = lane & 15, fq = lane >> 4;
+    f32x4_t acc[3][MI][2];
+#pragma unroll
+    for (int s = 0; s < 3; ++s)
+#pragma unroll
+        for (int i = 0; i < MI; ++i)
+#pragma unroll
+            for (int j = 0; j < 2; ++j) acc[s][i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    const int nsteps = (int)((row1 - row0 + (1 << (5 - l2q)) - 1)
+                             >> (5 - l2q));   // 32 pixels/step
+
+    // ---- running pointers ----
+    // dout: linear. in rows: (n*H + h) = rowidx + (r - PAD)  (H == P), so
+    // one linear pointer too; validity needs h = p + r - PAD in [0, H).
+    const unsigned short* pdout = dout + (row0 << l2q) * K + k0 + cc8;
+    const long inrow0 = row0 + r - PAD;          // global in-row of rib 0
+    const unsigned short* pin = in + inrow0 * W * C + c0 + cc8;
+    long prow = row0;                            // current step's first row
+
+    // per-thread static staging table for B: quantum u covers halo pixel
+    // (rib, j) channel-octet cc8 (8 threads per pixel as usual)
+    const int nbq = ROWS * HW2;                  // halo pixels per step
+    int tb_rib[5], tb_j[5];
+    int nb_t = 0;
+#pragma unroll
+    for (int u = 0; u < 5; ++u) {
+        int idx = (t >> 3) + 32 * u;
+        if (idx < nbq) { tb_rib[nb_t] = idx / HW2; tb_j[nb_t] = idx % HW2; ++nb_t; }
+    }
+    // A staging: m = (t>>3) + 32*u covers 32 m-positions... 32 rows of TK:
+    // each thread stages TK/64 quanta per m-row group (like generic wgrad)
+    V16 aregs[TK / 64], bregs[5];
+
+    auto load_step = [&]() {
+        // A: dout[m][k0 + h*64 + cc8], m = t>>3 (32 rows x 8 thr)
+        bool mrow_ok = (prow + ((t >> 3) >> l2q)) < rows_total;
+#pragma unroll
+        for (int h = 0; h < TK / 64; ++h)
+            aregs[h] = mrow_ok
+                ? load16<AL>(pdout + (long)(t >> 3) * K + h * 64,
+                             k0 + h * 64 + cc8, K)
+                : zero16();
+        // B: halo rows
+        for (int u = 0; u < nb_t; ++u) {
+            int rib = tb_rib[u], j = tb_j[u];
+            // validity: p = (prow+rib) % P ; h = p + r - PAD in [0,H)
+            int p = (int)((prow + rib) & (P - 1));
+            int h = p + r - PAD;
+            int wpx = j - 1;
+            bool v = h >= 0 && h < H && wpx >= 0 && wpx < W
+                     && (prow + rib) < rows_total;
+            bregs[u] = v ? load16<AL>(pin + ((long)rib * W + wpx) * C,
+                                      c0 + cc8, C)
+                         : zero16();
+        }
+        pdout += (long)32 * K;
+        pin += (long)ROWS * W * C;
+        prow += ROWS;
+    };
+
+    // LDS write/read offsets (subtile layout as the generic tr kernels)
+    constexpr unsigned A1 = sizeof(lds.A[0]);
+    constexpr unsigned BB1 = sizeof(lds.B[0]);
+    unsigned short* wrA[TK / 64];
+#pragma unroll
+    for (int h = 0; h < TK / 64; ++h) {
+        int kk8 = h * 64 + cc8;
+        wrA[h] = &lds.A[0][kk8 >> 4][(t >> 3) * 16 + (kk8 & 15)];
+    }
+    unsigned short* wrB[5];
+    for (int u = 0; u < nb_t; ++u) {
+        int hp = tb_rib[u] * HW2 + tb_j[u];
+        wrB[u] = &lds.B[0][cc8 >> 4][hp * 16 + (cc8 & 15)];
+    }
+
+    auto write_lds = [&](int buf) {     // literal buf only
+#pragma unroll
+        for (int h = 0; h < TK / 64; ++h)
+            *(uint4*)((char*)wrA[h] + (buf ? A1 : 0)) = aregs[h].u4;
+        for (int u = 0; u < nb_t; ++u)
+            *(uint4*)((char*)wrB[u] + (buf ? BB1 : 0)) = bregs[u].u4;
+    };
+
+    // read offsets: A-frag lane (fr, fq) element e = fq*8 + i*4 + (fr>>2);
+    // B-frag for tap s: halo position = rib(e)*HW2 + q(e) + s
+    const char* lb = (const char*)&lds;
+    unsigned roA[MI][2];
+    unsigned roB[3][2][2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+        int e = fq * 8 + i * 4 + (fr >> 2);
+        int qo = (fr & 3) * 4;
+#pragma unroll
+        for (int mi = 0; mi < MI; ++mi)
+            roA[mi][i] = (unsigned)((char*)&lds.A[0][wm * MI + mi]
+                                    [e * 16 + qo] - lb);
+        int rib = e >> l2q, q = e & (Q - 1);
+#pragma unroll
+        for (int s = 0; s < 3; ++s) {
+            int hp = rib * HW2 + q + s;
+#pragma unroll
+            for (int nj = 0; nj < 2; ++nj)
+                roB[s][nj][i] = (unsigned)((char*)&lds.B[0][wn * 2 + nj]
+                                           [hp * 16 + qo] - lb);
+        }
+    }
+
+    union U64x8 { bf16x4_t h[2]; bf16x8_t v; };
+    auto mfma_step = [&](int buf) {     // literal buf only
+        U64x8 a[MI];
+#pragma unroll
+        for (int mi = 0; mi < MI; ++mi) {
+            a[mi].h[0] = ds_tr16p((const unsigned short*)
+                (lb + roA[mi][0] + (buf ? A1 : 0)));
+            a[mi].h[1] = ds_tr16p((const unsigned short*)
+                (lb + roA[mi][1] + (buf ? A1 : 0)));
+        }
+#pragma unroll
+        for (int s = 0; s < 3; ++s) {
+            U64x8 bf[2];
+#pragma unroll
+            for (int nj = 0; nj < 2; ++nj) {
+                bf[nj].h[0] = ds_tr16p((const unsigned short*)
+                    (lb + roB[s][nj][0] + (buf ? BB1 : 0)));
+                bf[nj].h[1] = ds_tr16p((const unsigned short*)
+                    (lb + roB[s][nj][1] + (buf ? BB1 : 0)));
+            }
+#pragma unroll
+            for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+                for (int nj = 0; nj < 2; ++nj)
+                    acc[s][mi][nj] = MFMA_BF16(a[mi].v, bf[nj].v,
+                                               acc[s][mi][nj]);
+        }
+    };
+
+    load_step();
+    write_lds(0);
+    if (nsteps > 1) load_step();
+    __syncthreads();
+    for (int it = 0; it < nsteps; it += 2) {
+        if (it + 1 < nsteps) {
+            write_lds(1);
+            if (it + 2 < nsteps) load_step();
+        }
+        mfma_step(0);
+        __syncthreads();
+        if (it + 1 >= nsteps) break;
+        if (it + 2 < nsteps) {
+            write_lds(0);
+            if (it + 3 < nsteps) load_step();
+        }
+        mfma_step(1);
+        __syncthreads();
+    }
+
+    const int RSC = 9 * C;
+    float* dstp = partial + (long)sid * K * RSC;
+#pragma unroll
+    for (int s = 0; s < 3; ++s)
+#pragma unroll
+    for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int nj = 0; nj < 2; ++nj)
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+        int k = k0 + wm * (MI * 16) + mi * 16 + fq * 4 + e;
+        int c = c0 + wn * 32 + nj * 16 + fr;
+        if (k < K && c < C)
+            dstp[(long)k * RSC + (r * 3 + s) * C + c] = acc[s][mi][nj][e];
+    }
+}
+
 // Small-RSC wgrad (ResNet stem 3x3x3=27, LeNet conv1 5x5x1=25): the whole
 // flattened (r,s,c) axis fits one 64-column tile, so one block covers every
 // tap in a single pass over its pixel chunk — vs the generic kernel's
@@ -1162,6 +1385,38 @@ extern "C" void ps_conv_wgrad(
                            (hipStream_t)strm, (unsigned short*)dw,
                            (const float*)partial_f32, n_, split);
         return;
+    }
+    // 3x3 s1 pad1 family -> row-halo kernel (see conv_wgrad_row_kernel)
+    {
+        int l2q_ = ilog2_exact(Q), l2pq_ = ilog2_exact((long)P * Q);
+        if (stride == 1 && R == 3 && S == 3 && pad == 1 && P == H && Q == W
+            && l2q_ >= 0 && l2pq_ >= 0 && Q <= 32) {
+            int TKr = (K >= 128) ? 128 : 64;
+            int tkr = (K + TKr - 1) / TKr, tcr = (C + 63) / 64;
+            long rows_total = (long)Nb * P;
+            int ROWS = 32 >> l2q_;
+            long ipr_l = (rows_total + split - 1) / split;
+            ipr_l = ((ipr_l + ROWS - 1) / ROWS) * ROWS;
+            long nlog = (long)tkr * tcr * 3 * split;
+            int pxc = (int)((nlog + 7) / 8);
+            long grid_r = (long)pxc * 8;
+            bool alr = ((C & 63) == 0) && (K % TKr == 0);
+#define WGR(TKV, ALV)                                                         \
+            hipLaunchKernelGGL((conv_wgrad_row_kernel<TKV, ALV>),             \
+                dim3((unsigned)grid_r), dim3(256), 0, (hipStream_t)strm,      \
+                (const unsigned short*)dout, (const unsigned short*)in,       \
+                (float*)partial_f32, Nb, H, W, C, K, P, Q,                    \
+                split, (int)ipr_l, l2q_, 0, pxc)
+            if (TKr == 128) { if (alr) WGR(128, true); else WGR(128, false); }
+            else            { if (alr) WGR(64, true); else WGR(64, false); }
+#undef WGR
+            long n_ = (long)K * 9 * C;
+            int blocks_; ew_grid(n_ / 4, 256, &blocks_);
+            hipLaunchKernelGGL(reduce_slabs_kernel, dim3(blocks_), dim3(256),
+                               0, (hipStream_t)strm, (unsigned short*)dw,
+                               (const float*)partial_f32, n_, split);
+            return;
+        }
     }
     int TK = (K >= 128) ? 128 : 64;
     int tiles_k = (K + TK - 1) / TK, tiles_c = (C + 63) / 64;

@@ -67,7 +67,7 @@ def main():
         dw = torch.empty_like(w)
         wt = w.permute(2, 3, 1, 0).contiguous()
         M = Nb * P * P
-        split = _wgrad_split(M, K, C, R, R)
+        split = _wgrad_split(M, K, C, R, R, stride, pad, P, P)
         partial = torch.empty(split * K * R * R * C, dtype=torch.float32,
                               device='cuda')
         flops = 2.0 * M * K * R * R * C
