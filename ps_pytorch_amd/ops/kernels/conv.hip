@@ -855,13 +855,18 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
 
     // per-thread static staging table for B: quantum u covers halo pixel
     // (rib, j) channel-octet cc8 (8 threads per pixel as usual)
+    // per-thread static staging assignment: quantum u covers halo pixel
+    // (rib, j); indices are compile-time per u so everything stays in
+    // registers (a compacted runtime-count loop forced these to scratch)
     const int nbq = ROWS * HW2;                  // halo pixels per step
     int tb_rib[5], tb_j[5];
-    int nb_t = 0;
+    bool tb_v[5];
 #pragma unroll
     for (int u = 0; u < 5; ++u) {
         int idx = (t >> 3) + 32 * u;
-        if (idx < nbq) { tb_rib[nb_t] = idx / HW2; tb_j[nb_t] = idx % HW2; ++nb_t; }
+        tb_v[u] = idx < nbq;
+        int ix = tb_v[u] ? idx : 0;
+        tb_rib[u] = ix / HW2; tb_j[u] = ix % HW2;
     }
     // A staging: m = (t>>3) + 32*u covers 32 m-positions... 32 rows of TK:
     // each thread stages TK/64 quanta per m-row group (like generic wgrad)
@@ -877,7 +882,9 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
                              k0 + h * 64 + cc8, K)
                 : zero16();
         // B: halo rows
-        for (int u = 0; u < nb_t; ++u) {
+#pragma unroll
+        for (int u = 0; u < 5; ++u) {
+            if (!tb_v[u]) continue;
             int rib = tb_rib[u], j = tb_j[u];
             // validity: p = (prow+rib) % P ; h = p + r - PAD in [0,H)
             int p = (int)((prow + rib) & (P - 1));
@@ -904,7 +911,8 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
         wrA[h] = &lds.A[0][kk8 >> 4][(t >> 3) * 16 + (kk8 & 15)];
     }
     unsigned short* wrB[5];
-    for (int u = 0; u < nb_t; ++u) {
+#pragma unroll
+    for (int u = 0; u < 5; ++u) {
         int hp = tb_rib[u] * HW2 + tb_j[u];
         wrB[u] = &lds.B[0][cc8 >> 4][hp * 16 + (cc8 & 15)];
     }
@@ -913,8 +921,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
 #pragma unroll
         for (int h = 0; h < TK / 64; ++h)
             *(uint4*)((char*)wrA[h] + (buf ? A1 : 0)) = aregs[h].u4;
-        for (int u = 0; u < nb_t; ++u)
-            *(uint4*)((char*)wrB[u] + (buf ? BB1 : 0)) = bregs[u].u4;
+#pragma unroll
+        for (int u = 0; u < 5; ++u)
+            if (tb_v[u])
+                *(uint4*)((char*)wrB[u] + (buf ? BB1 : 0)) = bregs[u].u4;
     };
 
     // read offsets: A-frag lane (fr, fq) element e = fq*8 + i*4 + (fr>>2);
@@ -1391,7 +1401,7 @@ extern "C" void ps_conv_wgrad(
         int l2q_ = ilog2_exact(Q), l2pq_ = ilog2_exact((long)P * Q);
         if (stride == 1 && R == 3 && S == 3 && pad == 1 && P == H && Q == W
             && l2q_ >= 0 && l2pq_ >= 0 && Q <= 32) {
-            int TKr = (K >= 128) ? 128 : 64;
+            const int TKr = 64;     // TK=128 acc pressure costs a wave/SIMD
             int tkr = (K + TKr - 1) / TKr, tcr = (C + 63) / 64;
             long rows_total = (long)Nb * P;
             int ROWS = 32 >> l2q_;
@@ -1407,8 +1417,7 @@ extern "C" void ps_conv_wgrad(
                 (const unsigned short*)dout, (const unsigned short*)in,       \
                 (float*)partial_f32, Nb, H, W, C, K, P, Q,                    \
                 split, (int)ipr_l, l2q_, 0, pxc)
-            if (TKr == 128) { if (alr) WGR(128, true); else WGR(128, false); }
-            else            { if (alr) WGR(64, true); else WGR(64, false); }
+            if (alr) WGR(64, true); else WGR(64, false);
 #undef WGR
             long n_ = (long)K * 9 * C;
             int blocks_; ew_grid(n_ / 4, 256, &blocks_);
