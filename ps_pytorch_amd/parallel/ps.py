@@ -69,7 +69,8 @@ class ParameterServer:
         self.transport = PSTransport(self.flat, self.wire_dtype, self.device,
                                      self.rank, self.world,
                                      mode=cfg.aggregation,
-                                     compress=cfg.compress)
+                                     compress=cfg.compress,
+                                     comm_type=cfg.comm_type)
         self.ctrl = (ControlPlane(self.rank, self.world)
                      if cfg.mode == 'kill' else None)
         # f32 master copy + optimizer state in HBM

@@ -81,9 +81,11 @@ class PSTransport:
     def __init__(self, flat: FlatSpace, wire_dtype: torch.dtype,
                  device: torch.device, rank: int, world: int,
                  group: Optional[dist.ProcessGroup] = None,
-                 mode: str = 'collective', compress: bool = False):
+                 mode: str = 'collective', compress: bool = False,
+                 comm_type: str = 'Bcast'):
         if mode not in ('collective', 'gather'):
             raise ValueError(f"unknown aggregation mode {mode!r}")
+        self.comm_type = comm_type
         self.flat = flat
         self.rank = rank
         self.world = world
@@ -148,6 +150,18 @@ class PSTransport:
     # ---- weights: PS -> all ----
 
     def broadcast_weights(self) -> None:
+        if self.comm_type == 'Async':
+            # reference's deprecated P2P weight distribution
+            # (distributed_worker.py:201-219): explicit PS->worker sends
+            # instead of the one-collective broadcast.
+            if self.rank == PS_RANK:
+                works = [dist.isend(self.wire_w, dst=w, group=self.group)
+                         for w in range(1, self.world)]
+                for wk in works:
+                    wk.wait()
+            else:
+                dist.recv(self.wire_w, src=PS_RANK, group=self.group)
+            return
         dist.broadcast(self.wire_w, src=PS_RANK, group=self.group)
 
     def unpack_weights_into(self, dst: torch.Tensor) -> None:

@@ -61,7 +61,8 @@ class DistributedWorker:
         self.transport = PSTransport(self.flat, self.wire_dtype, self.device,
                                      self.rank, self.world,
                                      mode=cfg.aggregation,
-                                     compress=cfg.compress)
+                                     compress=cfg.compress,
+                                     comm_type=cfg.comm_type)
         # straggler handling (ref resnet_split.py:503-728): 'kill' polls the
         # PS's gloo signal from backward hooks; 'timeout' aborts locally past
         # --kill-threshold seconds. Both raise StepKilled mid-backward and

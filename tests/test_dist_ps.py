@@ -18,11 +18,12 @@ SEED = 5
 LR, MOM = 0.1, 0.9
 
 
-def _cfg(overlap: bool) -> JobConfig:
+def _cfg(overlap: bool, comm_type: str = 'Bcast') -> JobConfig:
     return JobConfig(network='LeNet', dataset='MNIST', batch_size=BS,
                      lr=LR, momentum=MOM, seed=SEED, max_steps=STEPS,
                      compress_grad='None', wire_dtype='fp32',
                      compute_dtype='fp32', overlap=overlap,
+                     comm_type=comm_type,
                      bucket_mb=0.25,    # several buckets even for LeNet
                      log_interval=10 ** 9, eval_freq=10 ** 9)
 
@@ -34,11 +35,12 @@ def _worker_batches(rank: int):
     return xs, ys
 
 
-def _role(rank: int, world: int, port: int, overlap: bool):
+def _role(rank: int, world: int, port: int, overlap: bool,
+          comm_type: str = 'Bcast'):
     from ps_pytorch_amd.parallel.transport import init_distributed
     from ps_pytorch_amd.parallel.ps import ParameterServer
     from ps_pytorch_amd.parallel.worker import DistributedWorker
-    cfg = _cfg(overlap)
+    cfg = _cfg(overlap, comm_type)
     env = init_distributed(backend='gloo')
     if rank == 0:
         ps = ParameterServer(cfg, rank, world, env['device'])
@@ -85,6 +87,16 @@ def test_golden_step_no_overlap():
 
 def test_golden_step_with_overlap():
     results = run_dist(_role, world=3, args=(True,))
+    got = torch.from_numpy(results[0])
+    ref = _serial_reference()
+    assert torch.allclose(got, ref, atol=1e-5, rtol=1e-5), \
+        (got - ref).abs().max()
+
+
+def test_golden_step_async_comm_type():
+    """--comm-type Async: P2P weight fan-out instead of broadcast
+    (ref distributed_worker.py:201-219) — same golden result."""
+    results = run_dist(_role, world=3, args=(False, 'Async'))
     got = torch.from_numpy(results[0])
     ref = _serial_reference()
     assert torch.allclose(got, ref, atol=1e-5, rtol=1e-5), \
