@@ -32,7 +32,7 @@ def parse():
     p.add_argument('--network', type=str, default='ResNet18')
     p.add_argument('--dataset', type=str, default='Cifar10')
     p.add_argument('--compress-grad', type=str, default='compress')
-    p.add_argument('--bucket-mb', type=float, default=25.0)
+    p.add_argument('--bucket-mb', type=float, default=4.0)
     p.add_argument('--no-overlap', action='store_true')
     p.add_argument('--aggregation', type=str, default='collective',
                    help='collective | gather (PS engine fan-in mode)')

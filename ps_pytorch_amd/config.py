@@ -66,7 +66,7 @@ def add_fit_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
                              'dtype end-to-end; SURVEY.md §2.4 dtype-quirk note)')
     parser.add_argument('--compute-dtype', type=str, default='bf16',
                         help='bf16 | fp32 : worker compute dtype on GPU')
-    parser.add_argument('--bucket-mb', type=float, default=25.0,
+    parser.add_argument('--bucket-mb', type=float, default=4.0,
                         help='gradient bucket size (MB) for overlapped RCCL ops')
     parser.add_argument('--overlap', type=str2bool, nargs='?', const=True, default=True,
                         help='overlap gradient push with backward (side HIP stream)')
@@ -110,7 +110,7 @@ class JobConfig:
     enable_gpu: bool = False
     wire_dtype: str = 'fp32'
     compute_dtype: str = 'bf16'
-    bucket_mb: float = 25.0
+    bucket_mb: float = 4.0
     overlap: bool = True
     aggregation: str = 'collective'
     engine: str = 'ps'
