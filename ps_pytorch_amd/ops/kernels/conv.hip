@@ -44,8 +44,6 @@ typedef float f32x4_t __attribute__((ext_vector_type(4)));
 
 #define MFMA_BF16(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0)
 
-#define LDSP 72   // padded LDS row length (bf16 elems): 144 B, 16B-aligned
-
 union V16 {                       // one 16-byte staging quantum (8 bf16)
     uint4 u4;
     unsigned short us[8];
@@ -548,13 +546,6 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
 }
 
 // ---------------------------------------------------------------- wgrad
-
-// XOR-swizzled transposed LDS image: element (row, mcol) of a 64x64 tile
-// lives at [row][ swz(row, mcol>>3)*8 + (mcol&7) ], pitch 64 (no pad — the
-// swizzle does the bank spreading).
-__device__ __forceinline__ int wg_swz(int row, int g) {
-    return g ^ (row & 7) ^ ((row >> 3) & 7);
-}
 
 // Natural-layout wgrad LDS: both operands stored [pixel][channel] exactly as
 // loaded (vector ds_write_b128 staging — no transpose scatter), reorganized
