@@ -18,25 +18,32 @@
 //          pixel chunks into f32 partials + deterministic slab reduce (fixed
 //          order, no atomics — replicas must stay bit-identical).
 //
-// Performance structure (guide §5/§6: T14 register-staged pipeline):
-//   * fwd/dgrad: 4-wave blocks, wave tile 64x64 (4x4 mfma_f32_16x16x32_bf16),
-//     block tile 256x64 (N<=64 output channels) or 128x128; BK=64; ONE
-//     register staging set, 2 LDS buffers, one barrier per k-step: write
-//     tile t+1 to LDS after the barrier, immediately issue loads of t+2,
-//     then MFMA tile t while the loads fly.
-//   * stride is a template parameter — no integer div/mod in inner loops
-//     (stride-2 dgrad parity checks become bit ops).
-//   * wgrad: contraction runs over pixels; both operands need an
-//     [out-idx][pixel] LDS image, i.e. a transpose on store. The store
-//     scatter is XOR-swizzled on 8-element granules (g ^= (row&7)^(row>>3))
-//     so the 8 lanes sharing one pixel column hit 8 different banks; reads
-//     unswizzle the same way (2-way conflict worst case on ds_read_b128).
-//     Pixel->(n,p,q) decode is shifts when P*Q and Q are powers of two
-//     (every CIFAR shape), runtime div otherwise.
-//   * wgrad grid is XCD-swizzled: all blocks covering the SAME pixel chunk
-//     (same split id, all (r,s)/k/c tiles) are placed on the SAME XCD
-//     (block b runs on XCD b%8, microarch guide) so the 9x re-read of
-//     dout/in per (r,s) comes from that XCD's L2, not HBM.
+// Performance structure (guide §5/§6; PMC/disassembly evidence: profiles/):
+//   * fwd/dgrad: 4-wave blocks, 128x128 / 128x64 tiles (wave tile 64x64 or
+//     64x32 of mfma_f32_16x16x32_bf16), BK=64; T14 pipeline — ONE register
+//     staging set, 2 LDS buffers, one barrier per k-step: write tile t+1 to
+//     LDS after the barrier, immediately issue loads of t+2, MFMA tile t
+//     while the loads fly. Staging addresses are affine-walked (per-row
+//     running pointers + wave-uniform step deltas + precomputed tap-validity
+//     bitmasks) — a per-step address rebuild made the kernels VALU-bound.
+//   * stride is a template parameter; stride-2 dgrad runs by PARITY CLASS
+//     (conv_dgrad2_kernel, one launch for all four classes) at full tile
+//     density instead of predicating 3/4 of the MFMAs to zero.
+//   * AL template: channel counts with full 64-chunk coverage take
+//     unconditional 16-B vector loads (the runtime tail guard was ~1/3 of
+//     scalar-pipe traffic); LeNet-style ragged shapes take the guarded path.
+//   * wgrad: contraction runs over pixels. Operands stage NATURALLY
+//     ([pixel][chan] 16-chan subtiles, 32-B row stride, vector writes) and
+//     fragments come back through gfx950's hardware transpose read
+//     (__builtin_amdgcn_ds_read_tr16_b64_v4bf16). The 3x3 s1 pad1 family
+//     uses conv_wgrad_row_kernel: one r-tap per block, all three s-taps
+//     served from shared W+2-wide halo rows at shifted (still 8-B-aligned)
+//     LDS positions — 3x the MACs per staged byte, R instead of R*S L2
+//     re-reads. Pixel->(n,p,q) decode is shifts when the dims are powers
+//     of two (every CIFAR shape), runtime div otherwise.
+//   * wgrad grids are XCD-swizzled: all blocks covering the SAME pixel
+//     chunk land on the SAME XCD (block b runs on XCD b%8) so re-reads of
+//     dout/in come from that XCD's L2, not HBM.
 #include "common.h"
 
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
