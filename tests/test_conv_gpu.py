@@ -27,7 +27,7 @@ SHAPES = [
 
 
 def _rel_err(a: torch.Tensor, b: torch.Tensor) -> float:
-    d = (a.float() - b.float()).abs().max()
+    d = (a.detach().float() - b.detach().float()).abs().max()
     return float(d / b.float().abs().max().clamp_min(1e-6))
 
 
