@@ -121,52 +121,86 @@ __global__ __launch_bounds__(256) void bn_stats_kernel(
     }
 }
 
-// ---------------- k1b: fold per-block partials into ws[0:2C] ----------------
-// one 256-thread block per output WORD: threads stride the nblocks rows of
-// that word's column, LDS tree-reduce to one value. (A thread-per-word
-// version ran 2 waves on the whole chip summing 512 values serially —
-// 118 us; this one is a few us.)
-__global__ __launch_bounds__(256) void bn_fold_partials_kernel(
-    const float* __restrict__ partial, float* __restrict__ ws, int nblocks, int C)
+
+// ---------------- fused fold + finalize ----------------
+// one block per channel: threads stride the per-block partial rows for the
+// channel's (sum, sq) columns, LDS tree-reduce, thread 0 runs the finalize
+// math — replaces the fold(2C blocks)+finalize(launch) pair.
+template <typename PT>
+__global__ __launch_bounds__(256) void bn_fwd_foldfin_kernel(
+    const float* __restrict__ partial, float* __restrict__ ws,
+    const PT* __restrict__ gamma, const PT* __restrict__ beta,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    float* __restrict__ save_mean, float* __restrict__ save_invstd,
+    int nblocks, long M, int C, float momentum, float eps)
 {
-    __shared__ float s[256];
-    const int w = blockIdx.x;                 // word in [0, 2C)
-    float acc = 0.f;
-    for (int b = threadIdx.x; b < nblocks; b += 256)
-        acc += partial[(long)b * 2 * C + w];
-    s[threadIdx.x] = acc;
+    __shared__ float s1[256], s2[256];
+    const int c = blockIdx.x;
+    float a1 = 0.f, a2 = 0.f;
+    for (int b = threadIdx.x; b < nblocks; b += 256) {
+        a1 += partial[(long)b * 2 * C + c];
+        a2 += partial[(long)b * 2 * C + C + c];
+    }
+    s1[threadIdx.x] = a1; s2[threadIdx.x] = a2;
     __syncthreads();
     for (int k = 128; k > 0; k >>= 1) {
-        if ((int)threadIdx.x < k) s[threadIdx.x] += s[threadIdx.x + k];
+        if ((int)threadIdx.x < k) {
+            s1[threadIdx.x] += s1[threadIdx.x + k];
+            s2[threadIdx.x] += s2[threadIdx.x + k];
+        }
         __syncthreads();
     }
-    if (threadIdx.x == 0) ws[w] = s[0];
+    if (threadIdx.x == 0) {
+        float mean = s1[0] / (float)M;
+        float var = fmaxf(s2[0] / (float)M - mean * mean, 0.f);
+        float invstd = rsqrtf(var + eps);
+        save_mean[c] = mean;
+        save_invstd[c] = invstd;
+        if (running_mean) {
+            float unbiased = var * (float)M / (float)(M > 1 ? M - 1 : 1);
+            running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+            running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+        }
+        float g = pt_to_f32<PT>(gamma[c]), bb = pt_to_f32<PT>(beta[c]);
+        float scale = g * invstd;
+        ws[2 * C + c] = scale;
+        ws[3 * C + c] = bb - mean * scale;
+    }
 }
 
-// ---------------- k2: forward finalize ----------------
-template <typename PT>   // param dtype (gamma/beta)
-__global__ __launch_bounds__(256) void bn_fwd_finalize_kernel(
-    float* __restrict__ ws, const PT* __restrict__ gamma,
-    const PT* __restrict__ beta, float* __restrict__ running_mean,
-    float* __restrict__ running_var, float* __restrict__ save_mean,
-    float* __restrict__ save_invstd, long M, int C, float momentum, float eps)
+template <typename PT>
+__global__ __launch_bounds__(256) void bn_bwd_foldfin_kernel(
+    const float* __restrict__ partial, float* __restrict__ ws,
+    const PT* __restrict__ gamma, const float* __restrict__ save_mean,
+    const float* __restrict__ save_invstd, PT* __restrict__ dgamma,
+    PT* __restrict__ dbeta, int nblocks, long M, int C)
 {
-    int c = blockIdx.x * blockDim.x + threadIdx.x;
-    if (c >= C) return;
-    float mean = ws[c] / (float)M;
-    float var = fmaxf(ws[C + c] / (float)M - mean * mean, 0.f);
-    float invstd = rsqrtf(var + eps);
-    save_mean[c] = mean;
-    save_invstd[c] = invstd;
-    if (running_mean) {
-        float unbiased = var * (float)M / (float)(M > 1 ? M - 1 : 1);
-        running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
-        running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    __shared__ float s1[256], s2[256];
+    const int c = blockIdx.x;
+    float a1 = 0.f, a2 = 0.f;
+    for (int b = threadIdx.x; b < nblocks; b += 256) {
+        a1 += partial[(long)b * 2 * C + c];
+        a2 += partial[(long)b * 2 * C + C + c];
     }
-    float g = pt_to_f32<PT>(gamma[c]), b = pt_to_f32<PT>(beta[c]);
-    float scale = g * invstd;
-    ws[2 * C + c] = scale;
-    ws[3 * C + c] = b - mean * scale;
+    s1[threadIdx.x] = a1; s2[threadIdx.x] = a2;
+    __syncthreads();
+    for (int k = 128; k > 0; k >>= 1) {
+        if ((int)threadIdx.x < k) {
+            s1[threadIdx.x] += s1[threadIdx.x + k];
+            s2[threadIdx.x] += s2[threadIdx.x + k];
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        float sum_dy = s1[0], sum_dyx = s2[0];
+        float invstd = save_invstd[c];
+        dgamma[c] = pt_from_f32<PT>(sum_dyx);
+        dbeta[c] = pt_from_f32<PT>(sum_dy);
+        float a = pt_to_f32<PT>(gamma[c]) * invstd;
+        ws[2 * C + c] = a;
+        ws[3 * C + c] = -a * invstd * sum_dyx / (float)M;
+        ws[4 * C + c] = -a * sum_dy / (float)M;
+    }
 }
 
 template <typename PT>
@@ -276,25 +310,6 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
     }
 }
 
-// ---------------- k5: backward finalize ----------------
-template <typename PT>
-__global__ __launch_bounds__(256) void bn_bwd_finalize_kernel(
-    float* __restrict__ ws, const PT* __restrict__ gamma,
-    const float* __restrict__ save_mean, const float* __restrict__ save_invstd,
-    PT* __restrict__ dgamma, PT* __restrict__ dbeta, long M, int C)
-{
-    int c = blockIdx.x * blockDim.x + threadIdx.x;
-    if (c >= C) return;
-    float sum_dy = ws[c], sum_dyx = ws[C + c];
-    float invstd = save_invstd[c];
-    dgamma[c] = pt_from_f32<PT>(sum_dyx);
-    dbeta[c] = pt_from_f32<PT>(sum_dy);
-    float a = pt_to_f32<PT>(gamma[c]) * invstd;
-    ws[2 * C + c] = a;
-    ws[3 * C + c] = -a * invstd * sum_dyx / (float)M;   // b: coeff of (x-mean)
-    ws[4 * C + c] = -a * sum_dy / (float)M;             // c: constant
-}
-
 // ---------------- k6: backward dx (+dres) ----------------
 template <typename T, bool RELU, bool DRES>
 __global__ __launch_bounds__(256) void bn_bwd_dx_kernel(
@@ -358,11 +373,10 @@ static void bn_fwd_t(const void* x, void* y, const void* gamma, const void* beta
         int nb = stats_blocks(M, Ci, 4);
         hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(nb), b256, 0, s,
                            (const T*)x, (float*)partial, M, Ci);
-        hipLaunchKernelGGL(bn_fold_partials_kernel, dim3(2 * Ci), b256, 0, s,
-                           (const float*)partial, wsf, nb, Ci);
-        hipLaunchKernelGGL((bn_fwd_finalize_kernel<PT>), dim3((Ci + 255) / 256), b256, 0, s,
-                           wsf, (const PT*)gamma, (const PT*)beta, (float*)rmean,
-                           (float*)rvar, (float*)smean, (float*)sinvstd, M, Ci,
+        hipLaunchKernelGGL((bn_fwd_foldfin_kernel<PT>), dim3(Ci), b256, 0, s,
+                           (const float*)partial, wsf, (const PT*)gamma,
+                           (const PT*)beta, (float*)rmean, (float*)rvar,
+                           (float*)smean, (float*)sinvstd, nb, M, Ci,
                            momentum, eps);
     } else {
         hipLaunchKernelGGL((bn_eval_finalize_kernel<PT>), dim3((Ci + 255) / 256), b256, 0, s,
@@ -419,11 +433,10 @@ static void bn_bwd_t(const void* x, const void* y, const void* dy,
         hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), dim3(nb), b256, 0, s,
                            (const T*)x, (const T*)y, (const T*)dy,
                            (const float*)smean, (const float*)sinvstd, (float*)partial, M, Ci);
-    hipLaunchKernelGGL(bn_fold_partials_kernel, dim3(2 * Ci), b256, 0, s,
-                       (const float*)partial, wsf, nb, Ci);
-    hipLaunchKernelGGL((bn_bwd_finalize_kernel<PT>), dim3((Ci + 255) / 256), b256, 0, s,
-                       wsf, (const PT*)gamma, (const float*)smean,
-                       (const float*)sinvstd, (PT*)dgamma, (PT*)dbeta, M, Ci);
+    hipLaunchKernelGGL((bn_bwd_foldfin_kernel<PT>), dim3(Ci), b256, 0, s,
+                       (const float*)partial, wsf, (const PT*)gamma,
+                       (const float*)smean, (const float*)sinvstd,
+                       (PT*)dgamma, (PT*)dbeta, nb, M, Ci);
     int blocks = stats_blocks(M, Ci, 4);
     if (blocks > 2048) blocks = 2048;
     if (relu) {
