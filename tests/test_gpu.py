@@ -120,3 +120,22 @@ def test_lenet_gpu_loss_decreases():
     losses = [tr.train_step(x, y) for _ in range(40)]
     torch.cuda.synchronize()
     assert min(losses[-5:]) < losses[0], losses  # memorizes one batch
+
+
+def test_resnet18_gpu_loss_decreases():
+    """End-to-end ResNet-18 on the full hand-written kernel path (PsConv2d +
+    fused BN + fused SGD): memorizing one batch must drive the loss down
+    (SURVEY.md §4 convergence-as-test)."""
+    from ps_pytorch_amd.config import JobConfig
+    from ps_pytorch_amd.trainer import NNTrainer
+    cfg = JobConfig(network='ResNet18', dataset='Cifar10', batch_size=64,
+                    lr=0.05, momentum=0.9, enable_gpu=True)
+    tr = NNTrainer(cfg, device=torch.device('cuda', 0))
+    tr.build_model()
+    torch.manual_seed(7)
+    x = torch.randn(64, 3, 32, 32, device='cuda', dtype=tr.compute_dtype) \
+        .contiguous(memory_format=torch.channels_last)
+    y = torch.randint(0, 10, (64,), device='cuda')
+    losses = [tr.train_step(x, y) for _ in range(40)]
+    torch.cuda.synchronize()
+    assert min(losses[-5:]) < 0.5 * losses[0], losses
