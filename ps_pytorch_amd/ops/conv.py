@@ -47,8 +47,8 @@ def _wgrad_split(M: int, K: int, C: int, R: int, S: int,
     without exploding the f32 partial buffer. Mirrors the kernel dispatch in
     conv.hip (generic / small-RSC / row-halo)."""
     tk = 128 if K >= 128 else 64
-    if R * S > 1 and R * S * C <= 64:       # flattened small-RSC kernel
-        tiles = (K + 63) // 64
+    if R * S > 1 and R * S * C <= 192:      # flattened small-RSC kernel
+        tiles = ((K + 63) // 64) * ((R * S * C + 63) // 64)
     elif (stride == 1 and R == 3 and S == 3 and pad == 1 and Q == P
           and 0 < Q <= 32 and (Q & (Q - 1)) == 0 and (P & (P - 1)) == 0):
         tiles = ((K + 63) // 64) * 3 * ((C + 63) // 64)       # row-halo (TK=64)

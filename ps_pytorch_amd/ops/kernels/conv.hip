@@ -199,22 +199,9 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         for (int j = 0; j < NJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
     const int KC = (Cin + 63) >> 6;          // contraction chunks per (r,s)
-    const int nsteps = SMALL ? 1 : R * S * KC;
+    const int nsteps = SMALL ? (R * S * C + 63) >> 6 : R * S * KC;
     int lr = 0, ls = 0, lc = 0;              // load-pointer step state
 
-    // SMALL: per-lane flattened (r,s,c) gather table for the 8 contraction
-    // elements this thread stages (element e = cc8+u -> (e/(S*C), (e/C)%S,
-    // e%C)); a handful of integer divides once per kernel.
-    int tre[SMALL ? 8 : 1], tse[SMALL ? 8 : 1], tce[SMALL ? 8 : 1];
-    if constexpr (SMALL) {
-#pragma unroll
-        for (int u = 0; u < 8; ++u) {
-            int e = cc8 + u;
-            tre[u] = e / (S * C);
-            tse[u] = (e / C) % S;
-            tce[u] = e % C;
-        }
-    }
 
     V16 areg[AR], breg[BR];
 
@@ -239,6 +226,19 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
     // then advance every pointer by the wave-uniform step delta
     auto load_step = [&]() {
         const int cbase = (lc << 6) + cc8;     // contraction offset (tail guard)
+        // SMALL: per-lane flattened (r,s,c) gather table for this chunk's 8
+        // contraction elements (e = lc*64 + cc8+u); <=3 chunks, a handful
+        // of divides each.
+        int tre[SMALL ? 8 : 1], tse[SMALL ? 8 : 1], tce[SMALL ? 8 : 1];
+        if constexpr (SMALL) {
+#pragma unroll
+            for (int u = 0; u < 8; ++u) {
+                int e = (lc << 6) + cc8 + u;
+                tre[u] = e / (S * C);
+                tse[u] = (e / C) % S;
+                tce[u] = e % C;
+            }
+        }
 #pragma unroll
         for (int rr = 0; rr < AR; ++rr) {
             if constexpr (SMALL) {
@@ -248,7 +248,7 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
 #pragma unroll
                 for (int u = 0; u < 8; ++u) {
                     int hh = ax[rr] + tre[u], ww = ay[rr] + tse[u];
-                    if (rowv && cc8 + u < RSC
+                    if (rowv && (lc << 6) + cc8 + u < RSC
                         && hh >= 0 && hh < H && ww >= 0 && ww < W)
                         v.us[u] = src[abase[rr]
                                       + ((long)tre[u] * W + tse[u]) * C + tce[u]];
@@ -262,11 +262,15 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
 #pragma unroll
         for (int rr = 0; rr < BR; ++rr) {
             if constexpr (SMALL)
-                breg[rr] = colv[rr] ? load16<AL>(pB[rr], cc8, R * S * C) : zero16();
+                breg[rr] = colv[rr]
+                    ? load16<AL>(pB[rr] + (lc << 6), (lc << 6) + cc8, R * S * C)
+                    : zero16();
             else
                 breg[rr] = colv[rr] ? load16<AL>(pB[rr], cbase, Cin) : zero16();
         }
-        if constexpr (!SMALL) {
+        if constexpr (SMALL) {
+            ++lc;
+        } else {
             long dA, dB;
             bool rs_adv = (++lc == KC);
             if (rs_adv) {
@@ -1032,8 +1036,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
     const long M = (long)Nb * P * Q;
     const int RSC = R * S * C;
     const int tiles_k = (K + 63) >> 6;
+    const int nc = (RSC + 63) >> 6;           // flattened-rsc 64-chunks
     int b = blockIdx.x;
     const int k0 = (b % tiles_k) * 64; b /= tiles_k;
+    const int e0 = (b % nc) * 64; b /= nc;    // this block's rsc chunk
     const int sid = b;
     const long mbeg = (long)sid * chunk;
     const long mend = (mbeg + chunk < M) ? mbeg + chunk : M;
@@ -1055,7 +1061,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
     int tre[8], tse[8], toff[8];
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
-        int e = cc8 + u;
+        int e = e0 + cc8 + u;
         tre[u] = e / (S * C);
         tse[u] = (e / C) % S;
         toff[u] = (tre[u] * W + tse[u]) * C + e % C;
@@ -1088,7 +1094,8 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
 #pragma unroll
                 for (int u = 0; u < 8; ++u) {
                     int hh = h0 + tre[u], ww = w0 + tse[u];
-                    if (cc8 + u < RSC && hh >= 0 && hh < H && ww >= 0 && ww < W)
+                    if (e0 + cc8 + u < RSC
+                        && hh >= 0 && hh < H && ww >= 0 && ww < W)
                         bv.us[u] = pix[toff[u]];
                 }
             }
@@ -1190,7 +1197,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
 #pragma unroll
     for (int e = 0; e < 4; ++e) {
         int k = k0 + wm * 32 + mi * 16 + fq * 4 + e;
-        int col = wn * 32 + nj * 16 + fr;
+        int col = e0 + wn * 32 + nj * 16 + fr;
         if (k < K && col < RSC)
             dstp[(long)k * RSC + col] = acc[mi][nj][e];
     }
@@ -1280,7 +1287,7 @@ extern "C" void ps_conv_fwd(
     bool al = (C & 63) == 0;           // full 64-chunk contraction coverage
 #define FWD_BODY()                                                            \
     do {                                                                      \
-        if (R * S > 1 && R * S * C <= 64) {                                   \
+        if (R * S > 1 && R * S * C <= 192) {                                  \
             if (stride == 1) LAUNCH_GEMM(128, 64, 1, false, true);            \
             else             LAUNCH_GEMM(128, 64, 2, false, true);            \
         } else if (K >= 128) {                                                \
@@ -1373,9 +1380,10 @@ extern "C" void ps_conv_wgrad(
     long M = (long)Nb * P * Q;
     long chunk64 = (M + (long)split * 64 - 1) / ((long)split * 64);
     int chunk = (int)(chunk64 * 64);
-    if (R * S > 1 && R * S * C <= 64) {     // flattened stem/LeNet path
+    if (R * S > 1 && R * S * C <= 192) {    // flattened stem/LeNet path
         int tiles_k = (K + 63) / 64;
-        long grid_s = (long)tiles_k * split;
+        int nc_s = (R * S * C + 63) / 64;
+        long grid_s = (long)tiles_k * nc_s * split;
         int l2pq_ = ilog2_exact((long)P * Q), l2q_ = ilog2_exact(Q);
         bool pw = l2pq_ >= 0 && l2q_ >= 0;
 #define WGS(ST, PW)                                                           \
