@@ -108,7 +108,8 @@ __device__ __forceinline__ int gswz(int row, int g) {
 // 3*3*3=27, LeNet conv1 5*5*1=25) — flatten (r,s,c) into the contraction
 // axis via a per-lane gather table and run ONE k-step instead of R*S, so
 // the MFMA utilization is RSC/64 of a full tile instead of C/64 per step.
-template <int TM, int TN, int STRIDE, bool DGRAD, bool SMALL = false, bool AL = true>
+template <int TM, int TN, int STRIDE, bool DGRAD, bool SMALL = false, bool AL = true,
+          int NBUF = 2>   // NBUF=1 only for the one-step (RSC<=64) SMALL path
 __global__ __launch_bounds__(256) void conv_gemm_kernel(
     const unsigned short* __restrict__ src,  // fwd: in [Nb,H,W,C]; dgrad: dout [Nb,P,Q,K]
     const unsigned short* __restrict__ wgt,  // fwd: w [K,R,S,C]; dgrad: wT [R,S,C,K]
@@ -117,7 +118,7 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int pad)
 {
-    __shared__ __attribute__((aligned(16))) FwdLds<TM, TN, SMALL ? 1 : 2> lds;
+    __shared__ __attribute__((aligned(16))) FwdLds<TM, TN, NBUF> lds;
     constexpr bool SWZ = true;
     constexpr int AR = TM / 32;        // A rows staged per thread
     constexpr int BR = TN / 32;        // B rows staged per thread
@@ -1267,12 +1268,13 @@ __global__ __launch_bounds__(256) void colsum_fold_kernel(
 
 // ---------------------------------------------------------------- C API
 
-#define LAUNCH_GEMM(TM, TN, ST, DG, SM)                                       \
+#define LAUNCH_GEMM(TM, TN, ST, DG, SM) LAUNCH_GEMM_NB(TM, TN, ST, DG, SM, 2)
+#define LAUNCH_GEMM_NB(TM, TN, ST, DG, SM, NBV)                               \
     do {                                                                      \
         long M_ = DG ? (long)Nb * H * W : (long)Nb * P * Q;                   \
         int Nout_ = DG ? C : K;                                               \
         long grid = ((M_ + TM - 1) / TM) * ((Nout_ + TN - 1) / TN);           \
-        hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, ST, DG, SM, ALV>),       \
+        hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, ST, DG, SM, ALV, NBV>), \
             dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,            \
             (const unsigned short*)src, (const unsigned short*)wgt,           \
             (const unsigned short*)bias, (unsigned short*)dst,                \
@@ -1287,7 +1289,10 @@ extern "C" void ps_conv_fwd(
     bool al = (C & 63) == 0;           // full 64-chunk contraction coverage
 #define FWD_BODY()                                                            \
     do {                                                                      \
-        if (R * S > 1 && R * S * C <= 192) {                                  \
+        if (R * S > 1 && R * S * C <= 64) {     /* one-step: single buffer */\
+            if (stride == 1) LAUNCH_GEMM_NB(128, 64, 1, false, true, 1);      \
+            else             LAUNCH_GEMM_NB(128, 64, 2, false, true, 1);      \
+        } else if (R * S > 1 && R * S * C <= 192) {                           \
             if (stride == 1) LAUNCH_GEMM(128, 64, 1, false, true);            \
             else             LAUNCH_GEMM(128, 64, 2, false, true);            \
         } else if (K >= 128) {                                                \
