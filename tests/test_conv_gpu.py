@@ -23,6 +23,8 @@ SHAPES = [
     (4, 1, 28, 28, 20, 5, 1, 0, True),       # LeNet conv1
     (4, 20, 12, 12, 50, 5, 1, 0, True),      # LeNet conv2 (C,K % 8 != 0)
     (3, 16, 9, 7, 24, 3, 1, 1, False),       # ragged M (tile tails)
+    (2, 3, 64, 64, 64, 7, 2, 3, False),      # R50-style stem 7x7 s2
+                                             # (RSC=147 multi-chunk flattened)
 ]
 
 
