@@ -1295,6 +1295,9 @@ extern "C" void ps_conv_fwd(
         } else if (R * S > 1 && R * S * C <= 192) {                           \
             if (stride == 1) LAUNCH_GEMM(128, 64, 1, false, true);            \
             else             LAUNCH_GEMM(128, 64, 2, false, true);            \
+        } else if (R * S == 1 && C <= 64) {     /* 1x1: one k-step */         \
+            if (stride == 1) LAUNCH_GEMM_NB(128, 64, 1, false, false, 1);     \
+            else             LAUNCH_GEMM_NB(128, 64, 2, false, false, 1);     \
         } else if (K >= 128) {                                                \
             if (stride == 1) LAUNCH_GEMM(128, 128, 1, false, false);          \
             else             LAUNCH_GEMM(128, 128, 2, false, false);          \
