@@ -78,6 +78,23 @@ class _FusedBNFn(torch.autograd.Function):
         return (dx, dgamma, dbeta, dres, None, None, None, None, None)
 
 
+def _bn_eval_fused(x, gamma, beta, residual, rmean, rvar, eps, relu):
+    lib = require_lib()
+    x = x.contiguous(memory_format=_CL)
+    if residual is not None:
+        residual = residual.contiguous(memory_format=_CL)
+    N, C, H, W = x.shape
+    y = torch.empty_like(x)
+    ws = torch.empty(5 * C, dtype=torch.float32, device=x.device)
+    lib.ps_bn_fwd(x.data_ptr(), y.data_ptr(), gamma.data_ptr(),
+                  beta.data_ptr(), rmean.data_ptr(), rvar.data_ptr(),
+                  0, 0, ws.data_ptr(), 0,
+                  residual.data_ptr() if residual is not None else 0,
+                  N * H * W, C, 0.0, float(eps), 0, int(relu),
+                  dtype_tag(x.dtype), current_stream_ptr())
+    return y
+
+
 class PsBatchNorm2d(nn.Module):
     """BatchNorm2d with optional fused residual-add + ReLU epilogue.
 
@@ -114,6 +131,13 @@ class PsBatchNorm2d(nn.Module):
             return _FusedBNFn.apply(x, self.weight, self.bias, residual,
                                     self.running_mean, self.running_var,
                                     self.momentum, self.eps, self.relu)
+        if (not self.training and _kernel_ok(x)
+                and not torch.is_grad_enabled()):
+            # eval-mode fused path (running-stats normalize + res + relu)
+            self._ensure_f32_stats()
+            return _bn_eval_fused(x, self.weight, self.bias, residual,
+                                  self.running_mean, self.running_var,
+                                  self.eps, self.relu)
         # torch fallback (CPU, eval mode, unsupported shapes)
         rm = self.running_mean
         rv = self.running_var

@@ -151,3 +151,24 @@ def test_gpu_kernel_f32_exactish():
     assert torch.allclose(y, yr, atol=1e-4), (y - yr).abs().max()
     assert torch.allclose(x.grad, x2.grad, atol=1e-4)
     assert torch.allclose(m.weight.grad, ref.weight.grad, atol=1e-3)
+
+
+@pytest.mark.gpu
+def test_bn_eval_fused_matches_torch():
+    """Eval-mode fused BN (running-stats normalize + residual + relu) vs
+    torch eval batch_norm."""
+    torch.manual_seed(11)
+    m = PsBatchNorm2d(64, relu=True).cuda().to(torch.bfloat16)
+    m.running_mean.data = torch.randn(64, device='cuda') * 0.1
+    m.running_var.data = torch.rand(64, device='cuda') + 0.5
+    m.eval()
+    x = torch.randn(8, 64, 16, 16, device='cuda', dtype=torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    res = torch.randn_like(x)
+    with torch.no_grad():
+        y = m(x, residual=res)
+        ref = F.relu(F.batch_norm(
+            x.float(), m.running_mean, m.running_var, m.weight.float(),
+            m.bias.float(), False, 0.1, m.eps) + res.float())
+    err = (y.float() - ref).abs().max() / ref.abs().max().clamp_min(1e-6)
+    assert float(err) < 0.03, float(err)
