@@ -864,12 +864,16 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
     const int nbq = ROWS * HW2;                  // halo pixels per step
     int tb_rib[5], tb_j[5];
     bool tb_v[5];
+    long tb_off[5];                     // static in-row element offset
 #pragma unroll
     for (int u = 0; u < 5; ++u) {
         int idx = (t >> 3) + 32 * u;
         tb_v[u] = idx < nbq;
         int ix = tb_v[u] ? idx : 0;
         tb_rib[u] = ix / HW2; tb_j[u] = ix % HW2;
+        int wpx = tb_j[u] - 1;
+        tb_v[u] = tb_v[u] && wpx >= 0 && wpx < W;   // fold w-validity in
+        tb_off[u] = ((long)tb_rib[u] * W + (wpx < 0 ? 0 : wpx)) * C;
     }
     // A staging: m = (t>>3) + 32*u covers 32 m-positions... 32 rows of TK:
     // each thread stages TK/64 quanta per m-row group (like generic wgrad)
@@ -884,19 +888,16 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
                 ? load16<AL>(pdout + (long)(t >> 3) * K + h * 64,
                              k0 + h * 64 + cc8, K)
                 : zero16();
-        // B: halo rows
+        // B: halo rows (w-validity and in-row offsets are precomputed;
+        // only the h / image-range part depends on the step)
 #pragma unroll
         for (int u = 0; u < 5; ++u) {
-            if (!tb_v[u]) continue;
-            int rib = tb_rib[u], j = tb_j[u];
-            // validity: p = (prow+rib) % P ; h = p + r - PAD in [0,H)
+            int rib = tb_rib[u];
             int p = (int)((prow + rib) & (P - 1));
             int h = p + r - PAD;
-            int wpx = j - 1;
-            bool v = h >= 0 && h < H && wpx >= 0 && wpx < W
+            bool v = tb_v[u] && h >= 0 && h < H
                      && (prow + rib) < rows_total;
-            bregs[u] = v ? load16<AL>(pin + ((long)rib * W + wpx) * C,
-                                      c0 + cc8, C)
+            bregs[u] = v ? load16<AL>(pin + tb_off[u], c0 + cc8, C)
                          : zero16();
         }
         pdout += (long)32 * K;
