@@ -863,16 +863,17 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
     // registers (a compacted runtime-count loop forced these to scratch)
     const int nbq = ROWS * HW2;                  // halo pixels per step
     int tb_rib[5], tb_j[5];
-    bool tb_v[5];
+    bool tb_ex[5];                      // quantum exists (gates the write)
+    bool tb_v[5];                       // exists AND w in range (gates load)
     long tb_off[5];                     // static in-row element offset
 #pragma unroll
     for (int u = 0; u < 5; ++u) {
         int idx = (t >> 3) + 32 * u;
-        tb_v[u] = idx < nbq;
-        int ix = tb_v[u] ? idx : 0;
+        tb_ex[u] = idx < nbq;
+        int ix = tb_ex[u] ? idx : 0;
         tb_rib[u] = ix / HW2; tb_j[u] = ix % HW2;
         int wpx = tb_j[u] - 1;
-        tb_v[u] = tb_v[u] && wpx >= 0 && wpx < W;   // fold w-validity in
+        tb_v[u] = tb_ex[u] && wpx >= 0 && wpx < W;
         tb_off[u] = ((long)tb_rib[u] * W + (wpx < 0 ? 0 : wpx)) * C;
     }
     // A staging: m = (t>>3) + 32*u covers 32 m-positions... 32 rows of TK:
@@ -927,7 +928,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
             *(uint4*)((char*)wrA[h] + (buf ? A1 : 0)) = aregs[h].u4;
 #pragma unroll
         for (int u = 0; u < 5; ++u)
-            if (tb_v[u])
+            if (tb_ex[u])
                 *(uint4*)((char*)wrB[u] + (buf ? BB1 : 0)) = bregs[u].u4;
     };
 
