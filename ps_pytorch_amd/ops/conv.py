@@ -50,7 +50,7 @@ def _wgrad_split(M: int, K: int, C: int, R: int, S: int,
     if R * S > 1 and R * S * C <= 192:      # flattened small-RSC kernel
         tiles = ((K + 63) // 64) * ((R * S * C + 63) // 64)
     elif (stride == 1 and R == 3 and S == 3 and pad == 1 and Q == P
-          and 0 < Q <= 32 and (Q & (Q - 1)) == 0 and (P & (P - 1)) == 0):
+          and Q >= 4):
         tiles = ((K + 63) // 64) * 3 * ((C + 63) // 64)       # row-halo (TK=64)
     else:
         tiles = ((K + tk - 1) // tk) * R * S * ((C + 63) // 64)
