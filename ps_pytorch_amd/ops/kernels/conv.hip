@@ -45,6 +45,7 @@
 //     chunk land on the SAME XCD (block b runs on XCD b%8) so re-reads of
 //     dout/in come from that XCD's L2, not HBM.
 #include "common.h"
+#include <stdlib.h>
 
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
 typedef float f32x4_t __attribute__((ext_vector_type(4)));
@@ -1411,10 +1412,13 @@ extern "C" void ps_conv_wgrad(
                            (const float*)partial_f32, n_, split);
         return;
     }
-    // 3x3 s1 pad1 family -> row-halo kernel (see conv_wgrad_row_kernel)
+    // 3x3 s1 pad1 family -> row-halo kernel (see conv_wgrad_row_kernel);
+    // PS_WG_ROW_OFF=1 falls back to the generic kernel (A/B)
     {
-        if (stride == 1 && R == 3 && S == 3 && pad == 1 && P == H && Q == W
-            && Q >= 4) {
+        static int row_en = -1;
+        if (row_en < 0) row_en = getenv("PS_WG_ROW_OFF") ? 0 : 1;
+        if (row_en && stride == 1 && R == 3 && S == 3 && pad == 1
+            && P == H && Q == W && Q >= 4) {
             const int TKr = 64;     // TK=128 acc pressure costs a wave/SIMD
             int tkr = (K + TKr - 1) / TKr, tcr = (C + 63) / 64;
             int ipc = (Nb + split - 1) / split;
