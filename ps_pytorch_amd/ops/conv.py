@@ -50,7 +50,8 @@ def _wgrad_split(M: int, K: int, C: int, R: int, S: int,
     if R * S > 1 and R * S * C <= 192:      # flattened small-RSC kernel
         tiles = ((K + 63) // 64) * ((R * S * C + 63) // 64)
     elif (stride == 1 and R == 3 and S == 3 and pad == 1 and Q == P
-          and Q >= 4 and not os.environ.get('PS_WG_ROW_OFF')):
+          and 0 < Q <= 32 and (Q & (Q - 1)) == 0 and (P & (P - 1)) == 0
+          and not os.environ.get('PS_WG_ROW_OFF')):
         tiles = ((K + 63) // 64) * 3 * ((C + 63) // 64)       # row-halo (TK=64)
     else:
         tiles = ((K + tk - 1) // tk) * R * S * ((C + 63) // 64)

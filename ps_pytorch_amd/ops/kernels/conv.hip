@@ -788,28 +788,32 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     }
 }
 
-// Row-halo wgrad for the 3x3 stride-1 pad-1 family (any square-ish shape
-// with P==H, Q==W): one block owns (k-tile, c-tile, ONE r) and walks the
-// output in 32-pixel steps — ROWS=floor(32/Q) whole rows when Q<=32, else
-// 32-wide strips of one row. Each step stages the dout pixels once plus the
-// matching input rows with a 2-pixel halo; all S=3 s-taps read the SAME
-// staged rows at shifted positions (row shifts in the [pixel][chan] subtile
-// image keep the transpose-read 8-B alignment). Per-lane pixel
-// decompositions are computed ONCE (a handful of divides at init); the walk
-// itself advances two pointers by scalar step deltas — no pow2 requirement.
+// Row-halo wgrad for the dominant 3x3 stride-1 pad-1 family: one block owns
+// (k-tile, c-tile, ONE r) and walks whole output rows. Per 32-pixel step it
+// stages the dout rows once plus the matching input rows with a 2-pixel
+// halo — all S=3 s-taps then read the SAME staged rows at shifted pixel
+// positions (a row shift in the [pixel][chan] subtile image keeps the
+// transpose-read 8-B alignment; a channel shift would not). 3x the MACs per
+// staged byte of the generic kernel, and dout/in leave L2 R times instead
+// of R*S times. Requires: stride 1, R=S=3, pad 1 (so H=P, W=Q and the
+// (n*H+h) axis is globally linear in the output row index), Q a power of
+// two <= 32, P a power of two.
+//
+// Contraction axis = 32 consecutive output pixels (exactly one
+// mfma_f32_16x16x32 depth): ROWS = 32/Q output rows per step; each lane's
+// (row, q) split of its contraction index is a compile-run constant.
 template <int TK, bool AL>
 __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
     const unsigned short* __restrict__ dout,  // [Nb,P,Q,K]
     const unsigned short* __restrict__ in,    // [Nb,H,W,C]
     float* __restrict__ partial,              // [SPLIT][K][9*C]
     int Nb, int H, int W, int C, int K, int P, int Q,
-    int split, int ipc, int per_xcd)
+    int split, int ipr, int l2q, int l2p, int per_xcd)
 {
+    // ipr: output rows per split chunk (multiple of ROWS)
     constexpr int PAD = 1;
-    const int QS = (Q <= 32) ? Q : 32;          // strip width
-    const int ROWS = (Q <= 32) ? (32 / Q) : 1;  // output rows per step
-    const int STRIPS = (Q + 31) >> 5;           // strips per row
-    const int HW2 = QS + 2;                     // halo row width
+    const int ROWS = 32 >> l2q;                 // output rows per step
+    const int HW2 = W + 2;                      // halo row width
     __shared__ __attribute__((aligned(16))) struct {
         unsigned short A[2][TK / 16][32 * 16 + 16];     // dout [k-sub][m*16+o]
         unsigned short B[2][4][48 * 16 + 16];           // in   [c-sub][hp*16+o]
@@ -825,8 +829,9 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
     const int c0 = (b % tiles_c) * 64; b /= tiles_c;
     const int r = b % 3; b /= 3;
     const int sid = b;
-    const int n0 = sid * ipc;
-    const int n1 = (n0 + ipc < Nb) ? n0 + ipc : Nb;
+    const long rows_total = (long)Nb * P;
+    const long row0 = (long)sid * ipr;
+    const long row1 = (row0 + ipr < rows_total) ? row0 + ipr : rows_total;
 
     const int t = threadIdx.x;
     const int cc8 = (t & 7) * 8;
@@ -841,77 +846,75 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
 #pragma unroll
             for (int j = 0; j < 2; ++j) acc[s][i][j] = {0.f, 0.f, 0.f, 0.f};
 
-    const int rgroups = (P + ROWS - 1) / ROWS;  // row-groups per image
-    const int spi = rgroups * STRIPS;           // steps per image
-    const int nsteps = (n1 > n0) ? (n1 - n0) * spi : 0;
+    const int nsteps = (int)((row1 - row0 + (1 << (5 - l2q)) - 1)
+                             >> (5 - l2q));   // 32 pixels/step
 
-    // ---- scalar walk state + step deltas (in pixels) ----
-    // pdout tracks pixel (n, p, sbase); pin tracks in element
-    // (n*H + p + r - PAD) * W + sbase  (h == p + r - PAD since H == P).
-    int pcur = 0, scur = 0;                     // row base, strip index
-    const unsigned short* pdout = dout + ((long)n0 * P * Q) * K + k0 + cc8;
-    const unsigned short* pin = in
-        + ((long)n0 * H + (r - PAD)) * W * C + c0 + cc8;
-    const long d_strip = 32;
-    const long d_row = (long)ROWS * Q - 32 * (STRIPS - 1);
-    const long d_img = d_row - (long)(rgroups * ROWS - P) * Q;
+    // ---- running pointers ----
+    // dout: linear. in rows: (n*H + h) = rowidx + (r - PAD)  (H == P), so
+    // one linear pointer too; validity needs h = p + r - PAD in [0, H).
+    const unsigned short* pdout = dout + (row0 << l2q) * K + k0 + cc8;
+    const long inrow0 = row0 + r - PAD;          // global in-row of rib 0
+    const unsigned short* pin = in + inrow0 * W * C + c0 + cc8;
+    long prow = row0;                            // current step's first row
 
-    // ---- per-thread static tables (init-time divides only) ----
-    // A: thread stages pixel midx = t>>3 of the step
-    const int midx = t >> 3;
-    const int a_dr = midx / QS, a_dq = midx % QS;
-    const long aoff = ((long)a_dr * Q + a_dq) * K;
-    // B: quantum u covers halo position (rib, j)
-    const int nbq = ROWS * HW2;
+    // per-thread static staging table for B: quantum u covers halo pixel
+    // (rib, j) channel-octet cc8 (8 threads per pixel as usual)
+    // per-thread static staging assignment: quantum u covers halo pixel
+    // (rib, j); indices are compile-time per u so everything stays in
+    // registers (a compacted runtime-count loop forced these to scratch)
+    const int nbq = ROWS * HW2;                  // halo pixels per step
     int tb_rib[5], tb_j[5];
-    bool tb_ex[5];
-    long tb_off[5];
+    bool tb_ex[5];                      // quantum exists (gates the write)
+    bool tb_v[5];                       // exists AND w in range (gates load)
+    long tb_off[5];                     // static in-row element offset
 #pragma unroll
     for (int u = 0; u < 5; ++u) {
         int idx = (t >> 3) + 32 * u;
         tb_ex[u] = idx < nbq;
         int ix = tb_ex[u] ? idx : 0;
         tb_rib[u] = ix / HW2; tb_j[u] = ix % HW2;
-        tb_off[u] = ((long)tb_rib[u] * W + tb_j[u] - 1) * C;
+        int wpx = tb_j[u] - 1;
+        tb_v[u] = tb_ex[u] && wpx >= 0 && wpx < W;
+        tb_off[u] = ((long)tb_rib[u] * W + (wpx < 0 ? 0 : wpx)) * C;
     }
-
-    const int nsub = TK / 64;
+    // A staging: m = (t>>3) + 32*u covers 32 m-positions... 32 rows of TK:
+    // each thread stages TK/64 quanta per m-row group (like generic wgrad)
     V16 aregs[TK / 64], bregs[5];
 
     auto load_step = [&]() {
-        const int sbase = scur * 32;
-        bool av = (pcur + a_dr < P) && (sbase + a_dq < Q);
+        // A: dout[m][k0 + h*64 + cc8], m = t>>3 (32 rows x 8 thr)
+        bool mrow_ok = (prow + ((t >> 3) >> l2q)) < rows_total;
 #pragma unroll
-        for (int h = 0; h < nsub; ++h)
-            aregs[h] = av ? load16<AL>(pdout + aoff + h * 64,
-                                       k0 + h * 64 + cc8, K)
-                          : zero16();
+        for (int h = 0; h < TK / 64; ++h)
+            aregs[h] = mrow_ok
+                ? load16<AL>(pdout + (long)(t >> 3) * K + h * 64,
+                             k0 + h * 64 + cc8, K)
+                : zero16();
+        // B: halo rows (w-validity and in-row offsets are precomputed;
+        // only the h / image-range part depends on the step)
 #pragma unroll
         for (int u = 0; u < 5; ++u) {
-            int hh = pcur + tb_rib[u] + r - PAD;
-            int ww = sbase + tb_j[u] - 1;
-            bool v = tb_ex[u] && hh >= 0 && hh < H && ww >= 0 && ww < W;
+            int rib = tb_rib[u];
+            int p = (int)((prow + rib) & (P - 1));
+            int h = p + r - PAD;
+            bool v = tb_v[u] && h >= 0 && h < H
+                     && (prow + rib) < rows_total;
             bregs[u] = v ? load16<AL>(pin + tb_off[u], c0 + cc8, C)
                          : zero16();
         }
-        long d;
-        if (++scur == STRIPS) {
-            scur = 0;
-            pcur += ROWS;
-            if (pcur >= P) { pcur = 0; d = d_img; }
-            else d = d_row;
-        } else d = d_strip;
-        pdout += d * K;
-        pin += d * C;
+        pdout += (long)32 * K;
+        pin += (long)ROWS * W * C;
+        prow += ROWS;
     };
 
+    // LDS write/read offsets (subtile layout as the generic tr kernels)
     constexpr unsigned A1 = sizeof(lds.A[0]);
     constexpr unsigned BB1 = sizeof(lds.B[0]);
     unsigned short* wrA[TK / 64];
 #pragma unroll
     for (int h = 0; h < TK / 64; ++h) {
         int kk8 = h * 64 + cc8;
-        wrA[h] = &lds.A[0][kk8 >> 4][midx * 16 + (kk8 & 15)];
+        wrA[h] = &lds.A[0][kk8 >> 4][(t >> 3) * 16 + (kk8 & 15)];
     }
     unsigned short* wrB[5];
 #pragma unroll
@@ -930,8 +933,8 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
                 *(uint4*)((char*)wrB[u] + (buf ? BB1 : 0)) = bregs[u].u4;
     };
 
-    // read offsets: frag element e -> (drow, dq); A image row = e; B halo
-    // position = drow*HW2 + dq + s
+    // read offsets: A-frag lane (fr, fq) element e = fq*8 + i*4 + (fr>>2);
+    // B-frag for tap s: halo position = rib(e)*HW2 + q(e) + s
     const char* lb = (const char*)&lds;
     unsigned roA[MI][2];
     unsigned roB[3][2][2];
@@ -943,10 +946,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
         for (int mi = 0; mi < MI; ++mi)
             roA[mi][i] = (unsigned)((char*)&lds.A[0][wm * MI + mi]
                                     [e * 16 + qo] - lb);
-        int edr = e / QS, edq = e % QS;
+        int rib = e >> l2q, q = e & (Q - 1);
 #pragma unroll
         for (int s = 0; s < 3; ++s) {
-            int hp = edr * HW2 + edq + s;
+            int hp = rib * HW2 + q + s;
 #pragma unroll
             for (int nj = 0; nj < 2; ++nj)
                 roB[s][nj][i] = (unsigned)((char*)&lds.B[0][wn * 2 + nj]
@@ -983,26 +986,24 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
         }
     };
 
-    if (nsteps > 0) {
-        load_step();
-        write_lds(0);
-        if (nsteps > 1) load_step();
-        __syncthreads();
-        for (int it = 0; it < nsteps; it += 2) {
-            if (it + 1 < nsteps) {
-                write_lds(1);
-                if (it + 2 < nsteps) load_step();
-            }
-            mfma_step(0);
-            __syncthreads();
-            if (it + 1 >= nsteps) break;
-            if (it + 2 < nsteps) {
-                write_lds(0);
-                if (it + 3 < nsteps) load_step();
-            }
-            mfma_step(1);
-            __syncthreads();
+    load_step();
+    write_lds(0);
+    if (nsteps > 1) load_step();
+    __syncthreads();
+    for (int it = 0; it < nsteps; it += 2) {
+        if (it + 1 < nsteps) {
+            write_lds(1);
+            if (it + 2 < nsteps) load_step();
         }
+        mfma_step(0);
+        __syncthreads();
+        if (it + 1 >= nsteps) break;
+        if (it + 2 < nsteps) {
+            write_lds(0);
+            if (it + 3 < nsteps) load_step();
+        }
+        mfma_step(1);
+        __syncthreads();
     }
 
     const int RSC = 9 * C;
@@ -1417,11 +1418,15 @@ extern "C" void ps_conv_wgrad(
     {
         static int row_en = -1;
         if (row_en < 0) row_en = getenv("PS_WG_ROW_OFF") ? 0 : 1;
-        if (row_en && stride == 1 && R == 3 && S == 3 && pad == 1
-            && P == H && Q == W && Q >= 4) {
+        int l2q_ = ilog2_exact(Q), l2pq_ = ilog2_exact((long)P * Q);
+        if (row_en && stride == 1 && R == 3 && S == 3 && pad == 1 && P == H
+            && Q == W && l2q_ >= 0 && l2pq_ >= 0 && Q <= 32) {
             const int TKr = 64;     // TK=128 acc pressure costs a wave/SIMD
             int tkr = (K + TKr - 1) / TKr, tcr = (C + 63) / 64;
-            int ipc = (Nb + split - 1) / split;
+            long rows_total = (long)Nb * P;
+            int ROWS = 32 >> l2q_;
+            long ipr_l = (rows_total + split - 1) / split;
+            ipr_l = ((ipr_l + ROWS - 1) / ROWS) * ROWS;
             long nlog = (long)tkr * tcr * 3 * split;
             int pxc = (int)((nlog + 7) / 8);
             long grid_r = (long)pxc * 8;
@@ -1431,7 +1436,7 @@ extern "C" void ps_conv_wgrad(
                 dim3((unsigned)grid_r), dim3(256), 0, (hipStream_t)strm,      \
                 (const unsigned short*)dout, (const unsigned short*)in,       \
                 (float*)partial_f32, Nb, H, W, C, K, P, Q,                    \
-                split, ipc, pxc)
+                split, (int)ipr_l, l2q_, 0, pxc)
             if (alr) WGR(64, true); else WGR(64, false);
 #undef WGR
             long n_ = (long)K * 9 * C;
