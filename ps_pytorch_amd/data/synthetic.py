@@ -26,7 +26,7 @@ from ..config import input_shape_of, num_classes_of
 # Dataset sizes mirror the real datasets (CIFAR: 50k/10k etc.) but are
 # capped so CPU tests stay fast; synthetic => size is a free parameter.
 _TRAIN_SIZE = {'mnist': 60000, 'cifar10': 50000, 'cifar100': 50000,
-               'svhn': 73257, 'imagenet-syn': 20000}
+               'svhn': 73257, 'imagenet-syn': 8192}   # 224^2 f32: keep resident size sane
 _TEST_SIZE = {'mnist': 10000, 'cifar10': 10000, 'cifar100': 10000,
               'svhn': 26032, 'imagenet-syn': 2000}
 
