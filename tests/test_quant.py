@@ -81,3 +81,30 @@ def test_acc_into():
     acc = a.clone()
     F.acc_into(acc, bb)
     assert torch.allclose(acc, a + bb.float())
+
+
+# property test: q8 roundtrip error bound holds for arbitrary sizes/scales
+try:
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=30, deadline=None)
+    @given(n=st.integers(min_value=1, max_value=5000),
+           scale=st.floats(min_value=1e-6, max_value=1e3),
+           seed=st.integers(min_value=0, max_value=2 ** 31 - 1))
+    def test_q8_roundtrip_error_bound(n, scale, seed):
+        from ps_pytorch_amd.ops import functional as F
+        g = torch.Generator().manual_seed(seed)
+        x = torch.randn(n, generator=g) * scale
+        _, tot = F.q8_layout(n)
+        p = torch.zeros(tot, dtype=torch.uint8)
+        F.pack_q8(p, x)
+        y = torch.empty(n)
+        F.unpack_q8(y, p)
+        # per-256-block error <= half a quantization step of that block
+        xb = torch.zeros(((n + 255) // 256) * 256)
+        xb[:n] = x
+        step = xb.view(-1, 256).abs().amax(dim=1) / 127.0
+        bound = step.repeat_interleave(256)[:n] * 0.5 + 1e-12
+        assert torch.all((y - x).abs() <= bound + 1e-6 * x.abs())
+except ImportError:       # pragma: no cover
+    pass
