@@ -91,7 +91,7 @@ try:
     @given(n=st.integers(min_value=1, max_value=5000),
            scale=st.floats(min_value=1e-6, max_value=1e3),
            seed=st.integers(min_value=0, max_value=2 ** 31 - 1))
-    def test_q8_roundtrip_error_bound(n, scale, seed):
+    def test_q8_roundtrip_bound_hypothesis(n, scale, seed):
         from ps_pytorch_amd.ops import functional as F
         g = torch.Generator().manual_seed(seed)
         x = torch.randn(n, generator=g) * scale
