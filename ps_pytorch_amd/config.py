@@ -115,6 +115,24 @@ class JobConfig:
     aggregation: str = 'collective'
     engine: str = 'ps'
 
+    def __post_init__(self) -> None:
+        if self.mode not in ('normal', 'kill', 'timeout'):
+            raise ValueError(f"unknown --mode {self.mode!r}")
+        if self.aggregation not in ('collective', 'gather'):
+            raise ValueError(f"unknown --aggregation {self.aggregation!r}")
+        # --mode kill needs arrival-order fan-in: the PS's kill verdict fires
+        # from the gather drain's first-k quota (parallel/ps.py); a collective
+        # reduce has no arrival order and would never send a verdict, hanging
+        # every worker at its end-of-step verdict recv. Normalize rather than
+        # reject: the reference's --mode kill ran on its (gather-like)
+        # per-layer P2P protocol without a separate aggregation flag.
+        if self.mode == 'kill' and self.aggregation != 'gather':
+            import warnings
+            warnings.warn("--mode kill requires --aggregation gather "
+                          "(arrival-order first-k drives the kill verdict); "
+                          "switching aggregation to 'gather'")
+            self.aggregation = 'gather'
+
     @property
     def compress(self) -> bool:
         return str(self.compress_grad).lower() in ('compress', 'true', '1')

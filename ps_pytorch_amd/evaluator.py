@@ -56,8 +56,25 @@ class DistributedEvaluator:
                 time.sleep(self.poll_interval)
 
     def _load_model(self, path: str) -> None:
+        """Load a model_step_<k> checkpoint. PS checkpoints carry parameters
+        only (no BN buffers — the reference never syncs them and its master
+        saves params, sync_replicas_master_nn.py:264-270); worker-saved
+        ResNet/VGG checkpoints carry the full state_dict. Strictness adapts:
+        missing BUFFER keys are tolerated iff the checkpoint has no buffers,
+        while a missing parameter or an unexpected key is always an error
+        (a silent strict=False here masked renamed keys — ADVICE r1)."""
         sd = torch.load(path, map_location='cpu', weights_only=True)
-        self.network.load_state_dict(sd, strict=False)
+        param_keys = {k for k, _ in self.network.named_parameters()}
+        buffer_keys = {k for k, _ in self.network.named_buffers()}
+        has_buffers = any(k in buffer_keys for k in sd)
+        missing, unexpected = self.network.load_state_dict(sd, strict=False)
+        bad_missing = [k for k in missing
+                       if k in param_keys or has_buffers]
+        if bad_missing or unexpected:
+            raise RuntimeError(
+                f"checkpoint {path} does not match model "
+                f"{self.cfg.network}: missing={bad_missing} "
+                f"unexpected={list(unexpected)}")
         self.network.to(self.device)
 
     @torch.no_grad()

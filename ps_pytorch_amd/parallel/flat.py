@@ -75,15 +75,25 @@ class FlatSpace:
         p0 = self.params[0]
         self.dtype = dtype or p0.dtype
         self.device = device or p0.device
+        # Parameter offsets are aligned up to ALIGN elements so every bucket
+        # boundary (always a param boundary, see _partition) is 16-B aligned
+        # even for bf16 storage: the gather-mode HIP kernels (ps_acc,
+        # ps_pack_q8, ps_unpack_q8) issue 8/16-byte vector accesses relative
+        # to the bucket start. Pad gaps live inside buckets; their grads stay
+        # zero and their weights are broadcast as-is (dead bytes on the wire,
+        # < ALIGN elems per param).
+        ALIGN = 8
         self.offsets: List[int] = []
         total = 0
         for p in self.params:
+            total = (total + ALIGN - 1) & ~(ALIGN - 1)
             self.offsets.append(total)
             total += p.numel()
         self.total = total
-        # pad storage to a multiple of 4 elements: full-buffer HIP kernels
+        self.num_params = sum(p.numel() for p in self.params)
+        # pad storage to a multiple of ALIGN: full-buffer HIP kernels
         # (fused_sgd) run purely 16B-vectorized; pad grads stay zero.
-        self.padded = (total + 3) & ~3
+        self.padded = (total + ALIGN - 1) & ~(ALIGN - 1)
 
         # Per-param storage layout: 4D channels_last params are stored in
         # their (O,H,W,I) storage order and re-exposed as strided views, so
@@ -107,20 +117,25 @@ class FlatSpace:
         self.buckets = self._partition(bucket_bytes)
 
     def _partition(self, bucket_bytes: int) -> List[Bucket]:
+        """Contiguous buckets at ALIGNED boundaries: a bucket ends at the
+        next param's aligned offset (the alignment gap travels inside the
+        bucket), and the last bucket ends at `padded` — so every bucket
+        slice start/size is 16-B aligned for the vectorized HIP kernels."""
         elem = self.flat_w.element_size()
         max_elems = max(1, bucket_bytes // elem)
+        nparams = len(self.params)
         buckets: List[Bucket] = []
         cur_ids: List[int] = []
         cur_start = 0
         for i, (p, off) in enumerate(zip(self.params, self.offsets)):
-            end = off + p.numel()
+            end = self.offsets[i + 1] if i + 1 < nparams else self.padded
             cur_ids.append(i)
             if end - cur_start >= max_elems:
                 buckets.append(Bucket(len(buckets), cur_start, end, cur_ids))
                 cur_ids = []
                 cur_start = end
         if cur_ids:
-            buckets.append(Bucket(len(buckets), cur_start, self.total, cur_ids))
+            buckets.append(Bucket(len(buckets), cur_start, self.padded, cur_ids))
         return buckets
 
     def _view(self, flat: torch.Tensor, p: nn.Parameter, off: int,

@@ -70,7 +70,9 @@ class ParameterServer:
                                      self.rank, self.world,
                                      mode=cfg.aggregation,
                                      compress=cfg.compress,
-                                     comm_type=cfg.comm_type)
+                                     comm_type=cfg.comm_type,
+                                     track_killed=(cfg.mode == 'timeout'
+                                                   and cfg.aggregation == 'gather'))
         self.ctrl = (ControlPlane(self.rank, self.world)
                      if cfg.mode == 'kill' else None)
         # f32 master copy + optimizer state in HBM
@@ -96,8 +98,15 @@ class ParameterServer:
         if t.mode == 'gather':
             t.post_gather_recvs()
             on_quota = self.ctrl.signal if self.ctrl is not None else None
-            t.drain_arrivals(self.cfg.num_aggregate, on_quota=on_quota)
+            counts = t.drain_arrivals(self.cfg.num_aggregate, on_quota=on_quota)
             grad = t.acc_g
+            # under-quota buckets (timeout-killed workers excluded from the
+            # count): pre-scale so the optimizer's global 1/k yields the
+            # exact average over the real arrivals
+            k = max(1, min(self.cfg.num_aggregate, self.world - 1))
+            for b, cnt in zip(self.flat.buckets, counts):
+                if 0 < cnt < k:
+                    grad[b.start:b.end].mul_(k / cnt)
         else:
             t.recv_buckets(self.flat.buckets)
             t.wait_all()

@@ -82,7 +82,7 @@ class PSTransport:
                  device: torch.device, rank: int, world: int,
                  group: Optional[dist.ProcessGroup] = None,
                  mode: str = 'collective', compress: bool = False,
-                 comm_type: str = 'Bcast'):
+                 comm_type: str = 'Bcast', track_killed: bool = False):
         if mode not in ('collective', 'gather'):
             raise ValueError(f"unknown aggregation mode {mode!r}")
         self.comm_type = comm_type
@@ -94,10 +94,35 @@ class PSTransport:
         self.wire_dtype = wire_dtype
         self.mode = mode
         self.compress = bool(compress) and mode == 'gather'
+        # liveness bound for the PS fan-in (a crashed worker otherwise hangs
+        # the drain forever — the reference shares this property, but an
+        # 8-GPU production run deserves a loud error over a silent hang)
+        self.drain_timeout = float(os.environ.get('PS_DRAIN_TIMEOUT', '600'))
         n = flat.padded
         self.wire_w = torch.zeros(n, dtype=wire_dtype, device=device)
         self.wire_g = torch.zeros(n, dtype=wire_dtype, device=device)
         self._works: List[dist.Work] = []
+        # program-order matching guard: NCCL IGNORES P2P tags — matching of
+        # the gather-mode isend/irecv pairs rests entirely on per-pair FIFO
+        # order, so every rank MUST touch buckets in index order. This
+        # counter turns a violation into an assert instead of a silent
+        # payload mismatch. (The tag= arguments below still matter for the
+        # gloo test backend, which does honor them.)
+        self._next_push = 0
+        # timeout-mode arrival marking: a locally-aborted worker's zero
+        # payloads must not occupy --num-aggregate quota slots (the
+        # reference's first-k only counts real arrivals). Workers send a
+        # 1-byte killed flag per step on a gloo side-group; the PS excludes
+        # flagged workers from the quota count.
+        self.track_killed = bool(track_killed) and mode == 'gather'
+        if self.track_killed and dist.is_initialized():
+            self._flag_group = dist.new_group(backend='gloo')
+            self._flag_buf = torch.zeros(1, dtype=torch.uint8)
+            if rank == PS_RANK:
+                self._flag_bufs = {w: torch.zeros(1, dtype=torch.uint8)
+                                   for w in range(1, world)}
+        else:
+            self.track_killed = False
 
         if mode == 'gather':
             if self.compress:
@@ -186,6 +211,12 @@ class PSTransport:
         gather:     pack (q8 or wire dtype), ncclSend to rank 0.
         killed=True sends a zero payload without touching the grads (the
         straggler-abort path: matching stays intact, compute is skipped)."""
+        # NCCL ignores tags: matching is per-pair FIFO, so bucket order is
+        # the protocol. Enforce it (reset in wait_all at step end).
+        assert b.index == self._next_push, (
+            f"bucket push out of order: got {b.index}, expected "
+            f"{self._next_push} (per-pair FIFO matching would corrupt)")
+        self._next_push += 1
         if self.mode == 'collective':
             src = self.flat.grad_slice(b)
             wire = self.grad_wire_slice(b)
@@ -260,17 +291,25 @@ class PSTransport:
         its --num-aggregate first-k selection (:179-207). Later arrivals are
         drained (matching) but discarded. `on_quota()` fires once, the
         moment every bucket has its quota while some recvs are still in
-        flight (the straggler-kill trigger). Returns per-bucket counts."""
+        flight (the straggler-kill trigger). Returns per-bucket counts.
+
+        With track_killed (timeout mode), a worker that aborted locally sent
+        a killed=1 flag on the gloo side-group; its zero payloads are drained
+        for matching but excluded from the quota count — the reference's
+        first-k only ever counts real gradients."""
         assert self.rank == PS_RANK and self.mode == 'gather'
         buckets = self.flat.buckets
         k = max(1, min(num_aggregate, self.world - 1))
         self.acc_g.zero_()
         counts = [0] * len(buckets)
         left = {w: len(buckets) for w in range(1, self.world)}
+        killed_of: dict = {}
         quota_fired = False
         for bi, w in self._arrivals():
             left[w] -= 1
-            if counts[bi] < k:
+            if self.track_killed and w not in killed_of:
+                killed_of[w] = self._recv_killed_flag(w)
+            if counts[bi] < k and not killed_of.get(w, False):
                 b = buckets[bi]
                 if self.compress:
                     ops_f.unpack_q8(self.acc_g[b.start:b.end],
@@ -287,6 +326,26 @@ class PSTransport:
             on_quota([])
         return counts
 
+    # -- killed-flag side channel (timeout mode, gather) --
+
+    def send_killed_flag(self, killed: bool) -> None:
+        """Worker: one flag byte per step (non-blocking — the PS only recvs
+        it lazily at this worker's first arrival, so a blocking send here
+        before the bucket pushes would deadlock; completion joins wait_all)."""
+        if not self.track_killed or self.rank == PS_RANK:
+            return
+        self._flag_buf.fill_(1 if killed else 0)
+        self._works.append(dist.isend(self._flag_buf, dst=PS_RANK,
+                                      group=self._flag_group, tag=78))
+
+    def _recv_killed_flag(self, w: int) -> bool:
+        """PS: consume worker w's flag for this step (exactly one per step;
+        blocks at most until w finishes its pushes — it sends the flag right
+        after launching them, or immediately upon aborting)."""
+        buf = self._flag_bufs[w]
+        dist.recv(buf, src=w, group=self._flag_group, tag=78)
+        return bool(int(buf))
+
     def _arrivals(self):
         """Yield (bucket_idx, worker) in completion order, draining
         self._pending. NCCL work objects expose real event-backed
@@ -294,13 +353,23 @@ class PSTransport:
         inside wait(), so one waiter thread per worker blocks through that
         worker's buckets in FIFO send order and feeds a completion queue
         (cross-worker arrival order is what the queue observes)."""
+        deadline = time.monotonic() + self.drain_timeout
         if dist.get_backend(self.group) == 'nccl':
+            idle_spins = 0
             while self._pending:
                 done = [key for key, wk in self._pending.items()
                         if wk.is_completed()]
                 if not done:
-                    time.sleep(0)
+                    # back off after a few hot spins: pinning a host core for
+                    # the whole fan-in window buys nothing once the first
+                    # poll came back empty (ADVICE r1); 50 us keeps drain
+                    # latency noise well under a bucket's wire time.
+                    idle_spins += 1
+                    time.sleep(0 if idle_spins < 64 else 50e-6)
+                    if time.monotonic() > deadline:
+                        self._drain_timeout_error()
                     continue
+                idle_spins = 0
                 for key in done:
                     self._pending.pop(key)
                     yield key
@@ -321,15 +390,30 @@ class PSTransport:
             t.start()
         total = len(self._pending)
         self._pending.clear()
-        for _ in range(total):
-            yield q.get()
+        for i in range(total):
+            try:
+                yield q.get(timeout=max(0.001, deadline - time.monotonic()))
+            except _queue.Empty:
+                self._pending = {(i, -1): None}   # non-empty for the error msg
+                self._drain_timeout_error(total - i)
         for t in threads:
             t.join()
+
+    def _drain_timeout_error(self, missing: Optional[int] = None):
+        who = (sorted(self._pending.keys()) if missing is None
+               else f"{missing} recvs")
+        raise RuntimeError(
+            f"PS gradient fan-in stalled > {self.drain_timeout:.0f}s "
+            f"(PS_DRAIN_TIMEOUT); outstanding (bucket, worker) recvs: {who}. "
+            f"A worker likely crashed — check its log; partial aggregation "
+            f"(--num-aggregate) still requires every worker to SEND (zeros "
+            f"when aborted), matching the reference's Waitany drain.")
 
     def wait_all(self) -> None:
         for w in self._works:
             w.wait()
         self._works.clear()
+        self._next_push = 0
 
     def barrier(self) -> None:
         if dist.is_initialized():

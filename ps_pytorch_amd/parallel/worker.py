@@ -62,7 +62,9 @@ class DistributedWorker:
                                      self.rank, self.world,
                                      mode=cfg.aggregation,
                                      compress=cfg.compress,
-                                     comm_type=cfg.comm_type)
+                                     comm_type=cfg.comm_type,
+                                     track_killed=(cfg.mode == 'timeout'
+                                                   and cfg.aggregation == 'gather'))
         # straggler handling (ref resnet_split.py:503-728): 'kill' polls the
         # PS's gloo signal from backward hooks; 'timeout' aborts locally past
         # --kill-threshold seconds. Both raise StepKilled mid-backward and
@@ -128,6 +130,9 @@ class DistributedWorker:
         self.transport.unpack_weights_into(self.flat.flat_w)
 
     def push_gradients(self, killed: bool = False) -> None:
+        # timeout-mode marking: tell the PS whether this step's payloads are
+        # real before it counts them toward the --num-aggregate quota
+        self.transport.send_killed_flag(killed)
         if self.cfg.overlap:
             if not killed:
                 self._flush_ready()

@@ -13,8 +13,13 @@ def test_views_alias_storage():
     fs.flat_w.fill_(0.5)
     for p in m.parameters():
         assert torch.all(p.data == 0.5)
-    assert fs.total == 431080
-    assert fs.padded % 4 == 0
+    assert fs.num_params == 431080      # LeNet parameter count
+    assert fs.total >= fs.num_params    # alignment gaps between params
+    assert fs.total - fs.num_params < 8 * len(fs.params)
+    assert fs.padded % 8 == 0
+    # every param offset 16-B aligned (vectorized kernels)
+    for off in fs.offsets:
+        assert off % 8 == 0
 
 
 def test_reverse_order_layout():
@@ -30,9 +35,12 @@ def test_bucket_partition_covers_everything():
     m = build_model('ResNet18')
     fs = FlatSpace(m, bucket_bytes=4 * 1024 * 1024)
     assert fs.buckets[0].start == 0
-    assert fs.buckets[-1].end == fs.total
+    assert fs.buckets[-1].end == fs.padded
     for a, b in zip(fs.buckets, fs.buckets[1:]):
         assert a.end == b.start
+    # bucket boundaries 16-B aligned for the gather-mode vector kernels
+    for b in fs.buckets:
+        assert b.start % 8 == 0 and b.end % 8 == 0
     covered = set()
     for b in fs.buckets:
         covered.update(b.param_ids)

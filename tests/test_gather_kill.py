@@ -153,3 +153,62 @@ def test_timeout_mode_aborts_locally():
                    args=({'mode': 'timeout', 'kill_threshold': 0.0}, False))
     assert res[1] == STEPS and res[2] == STEPS
     assert res[0] is not None
+
+
+def test_kill_with_collective_normalizes_to_gather():
+    """--mode kill with --aggregation collective would deadlock (the PS's
+    kill verdict only fires from the gather drain); JobConfig normalizes the
+    combination to gather with a warning."""
+    import warnings
+    with warnings.catch_warnings(record=True) as rec:
+        warnings.simplefilter('always')
+        cfg = _cfg(mode='kill', aggregation='collective')
+    assert cfg.aggregation == 'gather'
+    assert any('gather' in str(w.message) for w in rec)
+    # the valid combination stays untouched, no warning
+    with warnings.catch_warnings(record=True) as rec:
+        warnings.simplefilter('always')
+        cfg = _cfg(mode='kill', aggregation='gather')
+    assert cfg.aggregation == 'gather' and not rec
+
+
+def _role_one_timeout(rank: int, world: int, port: int, cfg_kw: dict):
+    """Like _role, but ONLY rank 2 gets an instant timeout threshold."""
+    from ps_pytorch_amd.parallel.transport import init_distributed
+    from ps_pytorch_amd.parallel.ps import ParameterServer
+    from ps_pytorch_amd.parallel.worker import DistributedWorker
+    kw = dict(cfg_kw)
+    if rank == 2:
+        kw['kill_threshold'] = 0.0
+    cfg = _cfg(**kw)
+    env = init_distributed(backend='gloo')
+    if rank == 0:
+        ps = ParameterServer(cfg, rank, world, env['device'])
+        ps.build_model(10)
+        for _ in range(STEPS):
+            ps.step()
+        return ps.master_w[:ps.flat.total].clone()
+    w = DistributedWorker(cfg, rank, world, env['device'])
+    w.build_model(10)
+    xs, ys = _batches(rank, same=True)
+    kills = 0
+    for i in range(STEPS):
+        if w.train_step(xs[i], ys[i]) is None:
+            kills += 1
+    return kills
+
+
+def test_timeout_killed_worker_excluded_from_quota():
+    """A timeout-aborted worker's zero payloads must NOT occupy
+    --num-aggregate quota slots or dilute the average: with
+    rank 2 instantly aborting and num_aggregate=2, the update must equal
+    the single-real-worker serial reference exactly (count-corrected
+    scaling), not a half-diluted one."""
+    res = run_dist(_role_one_timeout, world=3,
+                   args=({'mode': 'timeout', 'kill_threshold': 30.0,
+                          'num_aggregate': 2},))
+    assert res[2] == STEPS            # rank 2 aborted every step
+    assert res[1] == 0                # rank 1 never aborted
+    got = torch.from_numpy(res[0])
+    ref = _serial(_identity, workers=(1,), same_data=True, scale=1.0)
+    assert torch.allclose(got, ref, atol=1e-5, rtol=1e-5), (got - ref).abs().max()
