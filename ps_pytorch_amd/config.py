@@ -58,6 +58,10 @@ def add_fit_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
                         help='shared directory for model_step_<k> checkpoints')
     parser.add_argument('--compress-grad', type=str, default='compress', metavar='N',
                         help='compress | None : wire compression for gradients')
+    parser.add_argument('--data-dir', type=str, default=None, metavar='DIR',
+                        help='directory holding real dataset files (MNIST idx / '
+                             'CIFAR pickle batches / SVHN .mat); falls back to '
+                             './<name>_data, then synthetic data')
     parser.add_argument('--enable-gpu', type=str2bool, nargs='?', const=True,
                         default=False, help='run compute on GPUs (one rank per GPU)')
     # --- MI355X-native knobs (new; not in the reference) ---
@@ -107,6 +111,7 @@ class JobConfig:
     eval_freq: int = 50
     train_dir: str = 'output/models/'
     compress_grad: str = 'compress'
+    data_dir: Optional[str] = None
     enable_gpu: bool = False
     wire_dtype: str = 'fp32'
     compute_dtype: str = 'bf16'

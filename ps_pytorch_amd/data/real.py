@@ -1,0 +1,293 @@
+"""Real-dataset pipeline: self-contained readers + device-resident batching
+with on-GPU augmentation.
+
+Reference parity: src/util.py:21-106 `prepare_data` loads MNIST / CIFAR-10 /
+CIFAR-100 / SVHN through torchvision with per-dataset transforms, and
+src/data/data_prepare.py pre-downloads the files on every node (data
+locality: nothing moves over the wire, README.md:24). This image has no
+network and no torchvision, so the readers below parse the standard on-disk
+formats directly (MNIST idx, CIFAR pickle batches, SVHN .mat via scipy) —
+drop the files the reference's downloader would have fetched into
+`--data-dir` (or ./<name>_data, the reference's paths) and they are used.
+
+MI355X-native loader design: the full uint8 dataset is tiny next to 288 GB
+HBM3E (CIFAR-10 train = 150 MB), so it lives ON DEVICE and augmentation runs
+as vectorized tensor ops on the GPU per batch — reflect-pad-4 + random
+32-crop + horizontal flip + normalize, matching the reference's torchvision
+transform stack (util.py:36-47) — instead of per-sample PIL work on host
+CPUs feeding a multiprocessing loader (the reference's vendored
+my_data_loader.py). Zero H2D traffic per step; the only per-batch work is
+one gather and a fused normalize/cast.
+"""
+from __future__ import annotations
+
+import gzip
+import os
+import pickle
+import struct
+from typing import Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..config import input_shape_of, num_classes_of
+from ..utils.logging import get_logger
+
+logger = get_logger('ps_pytorch_amd.data')
+
+# normalization constants, verbatim from the reference transform stacks
+_NORM = {
+    'mnist':    ((0.1307,), (0.3081,)),                       # util.py:27
+    'cifar10':  (tuple(x / 255.0 for x in (125.3, 123.0, 113.9)),
+                 tuple(x / 255.0 for x in (63.0, 62.1, 66.7))),  # util.py:35
+    'cifar100': (tuple(x / 255.0 for x in (125.3, 123.0, 113.9)),
+                 tuple(x / 255.0 for x in (63.0, 62.1, 66.7))),
+    'svhn':     ((0.4914, 0.4822, 0.4465), (0.2023, 0.1994, 0.2010)),
+}
+# train-time augmentation per dataset: (pad, pad_mode, hflip)
+_AUG = {
+    'mnist':    (0, None, False),
+    'cifar10':  (4, 'reflect', True),     # util.py:38-44
+    'cifar100': (4, 'reflect', True),
+    'svhn':     (4, 'constant', True),    # RandomCrop(32, padding=4) default
+}
+
+
+def _open_maybe_gz(path: str):
+    return gzip.open(path, 'rb') if path.endswith('.gz') else open(path, 'rb')
+
+
+def _find(root: str, names) -> Optional[str]:
+    """First existing file among `names` (each tried plain and .gz) under
+    root or root's immediate subdirectories."""
+    dirs = [root]
+    if os.path.isdir(root):
+        dirs += [os.path.join(root, d) for d in sorted(os.listdir(root))
+                 if os.path.isdir(os.path.join(root, d))]
+    for d in dirs:
+        for n in names:
+            for cand in (os.path.join(d, n), os.path.join(d, n + '.gz')):
+                if os.path.isfile(cand):
+                    return cand
+    return None
+
+
+# ---- format readers (numpy uint8 [N,C,H,W] + int64 labels) ----
+
+def read_idx_images(path: str) -> np.ndarray:
+    with _open_maybe_gz(path) as f:
+        magic, n, rows, cols = struct.unpack('>IIII', f.read(16))
+        if magic != 2051:
+            raise ValueError(f"{path}: bad idx3 magic {magic}")
+        buf = f.read(n * rows * cols)
+    return np.frombuffer(buf, dtype=np.uint8).reshape(n, 1, rows, cols)
+
+
+def read_idx_labels(path: str) -> np.ndarray:
+    with _open_maybe_gz(path) as f:
+        magic, n = struct.unpack('>II', f.read(8))
+        if magic != 2049:
+            raise ValueError(f"{path}: bad idx1 magic {magic}")
+        buf = f.read(n)
+    return np.frombuffer(buf, dtype=np.uint8).astype(np.int64)
+
+
+def load_mnist(root: str, split: str) -> Tuple[np.ndarray, np.ndarray]:
+    prefix = 'train' if split == 'train' else 't10k'
+    img = _find(root, [f'{prefix}-images-idx3-ubyte', f'{prefix}-images.idx3-ubyte'])
+    lab = _find(root, [f'{prefix}-labels-idx1-ubyte', f'{prefix}-labels.idx1-ubyte'])
+    if img is None or lab is None:
+        raise FileNotFoundError(f"MNIST idx files not under {root}")
+    return read_idx_images(img), read_idx_labels(lab)
+
+
+def load_cifar(root: str, split: str, fine100: bool = False):
+    if fine100:
+        name = 'train' if split == 'train' else 'test'
+        path = _find(root, [name])
+        if path is None:
+            raise FileNotFoundError(f"CIFAR-100 pickle '{name}' not under {root}")
+        with _open_maybe_gz(path) as f:
+            d = pickle.load(f, encoding='latin1')
+        x = np.asarray(d['data'], dtype=np.uint8).reshape(-1, 3, 32, 32)
+        y = np.asarray(d['fine_labels'], dtype=np.int64)
+        return x, y
+    names = ([f'data_batch_{i}' for i in range(1, 6)] if split == 'train'
+             else ['test_batch'])
+    xs, ys = [], []
+    for n in names:
+        path = _find(root, [n])
+        if path is None:
+            raise FileNotFoundError(f"CIFAR-10 pickle '{n}' not under {root}")
+        with _open_maybe_gz(path) as f:
+            d = pickle.load(f, encoding='latin1')
+        xs.append(np.asarray(d['data'], dtype=np.uint8).reshape(-1, 3, 32, 32))
+        ys.append(np.asarray(d['labels'], dtype=np.int64))
+    return np.concatenate(xs), np.concatenate(ys)
+
+
+def load_svhn(root: str, split: str):
+    from scipy.io import loadmat
+    path = _find(root, [f'{split}_32x32.mat'])
+    if path is None:
+        raise FileNotFoundError(f"SVHN {split}_32x32.mat not under {root}")
+    m = loadmat(path)
+    x = np.transpose(m['X'], (3, 2, 0, 1)).astype(np.uint8)   # HWCN -> NCHW
+    y = m['y'].astype(np.int64).reshape(-1)
+    y[y == 10] = 0            # torchvision SVHN label convention
+    return x, y
+
+
+_READERS = {
+    'mnist': lambda r, s: load_mnist(r, s),
+    'cifar10': lambda r, s: load_cifar(r, s, fine100=False),
+    'cifar100': lambda r, s: load_cifar(r, s, fine100=True),
+    'svhn': lambda r, s: load_svhn(r, s),
+}
+
+
+def dataset_root(dataset: str, data_dir: Optional[str] = None) -> Optional[str]:
+    """Resolve where this dataset's files live; None if nothing present.
+    Checked in order: explicit --data-dir, $PS_DATA_ROOT[/<name>*], and the
+    reference's conventional ./<name>_data paths (data_prepare.py)."""
+    d = dataset.lower()
+    cands = []
+    for base in (data_dir, os.environ.get('PS_DATA_ROOT')):
+        if base:
+            cands += [base, os.path.join(base, f'{d}_data'), os.path.join(base, d)]
+    cands += [f'./{d}_data', d and f'./{d}']
+    probe = {'mnist': ['train-images-idx3-ubyte'],
+             'cifar10': ['data_batch_1'],
+             'cifar100': ['train'],
+             'svhn': ['train_32x32.mat']}[d] if d in _READERS else None
+    if probe is None:
+        return None
+    for c in cands:
+        if c and os.path.isdir(c) and _find(c, probe) is not None:
+            return c
+    return None
+
+
+class RealDataset:
+    """Device-resident real dataset: uint8 images + labels on `device`."""
+
+    def __init__(self, dataset: str, split: str, root: str,
+                 device: torch.device = torch.device('cpu'),
+                 dtype: torch.dtype = torch.float32):
+        d = dataset.lower()
+        if d not in _READERS:
+            raise ValueError(f"no real-data reader for {dataset!r}")
+        self.name = d
+        self.split = split
+        self.shape = input_shape_of(d)
+        self.num_classes = num_classes_of(d)
+        x, y = _READERS[d](root, split)
+        if tuple(x.shape[1:]) != tuple(self.shape):
+            raise ValueError(f"{dataset} files under {root} have shape "
+                             f"{x.shape[1:]}, expected {self.shape}")
+        self.dtype = dtype
+        # raw uint8 stays resident; normalization happens per batch (fused
+        # with the augmentation gather, output in compute dtype)
+        self.x_u8 = torch.from_numpy(np.ascontiguousarray(x)).to(device)
+        self.y = torch.from_numpy(np.ascontiguousarray(y)).to(device)
+        mean, std = _NORM[d]
+        C = self.shape[0]
+        self.mean = (torch.tensor(mean, dtype=torch.float32, device=device)
+                     .view(1, C, 1, 1) * 255.0)
+        self.inv_std = 1.0 / (torch.tensor(std, dtype=torch.float32,
+                                           device=device).view(1, C, 1, 1) * 255.0)
+        self.pad, self.pad_mode, self.hflip = _AUG[d]
+
+    def __len__(self) -> int:
+        return self.x_u8.shape[0]
+
+    # -- per-batch pipeline (all on self.x_u8's device) --
+
+    def normalize(self, xb_u8: torch.Tensor) -> torch.Tensor:
+        return ((xb_u8.float() - self.mean) * self.inv_std).to(self.dtype)
+
+    def augment(self, xb_u8: torch.Tensor, gen: torch.Generator) -> torch.Tensor:
+        """Train transform stack (vectorized over the batch): pad + random
+        crop + random hflip on uint8, then normalize+cast. Matches the
+        reference's torchvision stack (util.py:36-47) sample-for-sample in
+        distribution; randomness comes from `gen` (CPU generator for
+        deterministic multi-rank tests)."""
+        B, C, H, W = xb_u8.shape
+        p = self.pad
+        if p:
+            xf = xb_u8.float()
+            mode = self.pad_mode if self.pad_mode != 'constant' else 'constant'
+            xp = torch.nn.functional.pad(xf, (p, p, p, p), mode=mode)
+            oy = torch.randint(0, 2 * p + 1, (B,), generator=gen)
+            ox = torch.randint(0, 2 * p + 1, (B,), generator=gen)
+            oy = oy.to(xp.device)
+            ox = ox.to(xp.device)
+            # gather the BxCxHxW crops: index grids [B,1,H,W]
+            ar_h = torch.arange(H, device=xp.device)
+            ar_w = torch.arange(W, device=xp.device)
+            iy = (oy.view(B, 1) + ar_h.view(1, H))            # [B,H]
+            ix = (ox.view(B, 1) + ar_w.view(1, W))            # [B,W]
+            xp = xp.gather(2, iy.view(B, 1, H, 1).expand(B, C, H, W + 2 * p))
+            xb = xp.gather(3, ix.view(B, 1, 1, W).expand(B, C, H, W))
+        else:
+            xb = xb_u8.float()
+        if self.hflip:
+            fmask = (torch.rand(B, generator=gen) < 0.5).to(xb.device)
+            xb = torch.where(fmask.view(B, 1, 1, 1), xb.flip(-1), xb)
+        return ((xb - self.mean) * self.inv_std).to(self.dtype)
+
+
+class RealResidentLoader:
+    """ResidentLoader-compatible surface over a RealDataset: iteration yields
+    (data, target); next_batch() wraps epochs; train split augments."""
+
+    def __init__(self, ds: RealDataset, batch_size: int, shuffle: bool = True,
+                 seed: int = 1, drop_last: bool = True, augment: bool = True):
+        self.ds = ds
+        self.batch_size = batch_size
+        self.shuffle = shuffle
+        self.drop_last = drop_last
+        self.augment = augment and ds.split == 'train'
+        self._epoch = 0
+        self._pos = 0
+        self._seed = seed
+        self._gen = torch.Generator().manual_seed(seed * 40503 + 7)
+        self._perm = self._make_perm()
+
+    @property
+    def dataset(self):
+        return self.ds
+
+    def _make_perm(self) -> torch.Tensor:
+        if not self.shuffle:
+            return torch.arange(len(self.ds), device=self.ds.x_u8.device)
+        g = torch.Generator().manual_seed(self._seed * 7919 + self._epoch)
+        return torch.randperm(len(self.ds), generator=g).to(self.ds.x_u8.device)
+
+    def __len__(self) -> int:
+        n = len(self.ds)
+        return (n // self.batch_size if self.drop_last
+                else -(-n // self.batch_size))
+
+    def _batch(self, i: int):
+        idx = self._perm[i * self.batch_size:(i + 1) * self.batch_size]
+        xb = self.ds.x_u8.index_select(0, idx)
+        yb = self.ds.y.index_select(0, idx)
+        if self.augment:
+            return self.ds.augment(xb, self._gen), yb
+        return self.ds.normalize(xb), yb
+
+    def __iter__(self):
+        for i in range(len(self)):
+            yield self._batch(i)
+        self._epoch += 1
+        self._perm = self._make_perm()
+
+    def next_batch(self):
+        if self._pos >= len(self):
+            self._pos = 0
+            self._epoch += 1
+            self._perm = self._make_perm()
+        i = self._pos
+        self._pos += 1
+        return self._batch(i)
