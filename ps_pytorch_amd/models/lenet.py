@@ -17,6 +17,7 @@ import torch
 import torch.nn as nn
 
 from ..ops.conv import PsConv2d
+from ..ops.linear import PsLinear
 import torch.nn.functional as F
 
 
@@ -26,8 +27,8 @@ class LeNet(nn.Module):
         self.conv1 = PsConv2d(in_channels, 20, kernel_size=5)
         self.conv2 = PsConv2d(20, 50, kernel_size=5)
         # 28x28 -> conv5 -> 24 -> pool -> 12 -> conv5 -> 8 -> pool -> 4
-        self.fc1 = nn.Linear(50 * 4 * 4, 500)
-        self.fc2 = nn.Linear(500, num_classes)
+        self.fc1 = PsLinear(50 * 4 * 4, 500)
+        self.fc2 = PsLinear(500, num_classes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         x = F.max_pool2d(F.relu(self.conv1(x)), 2)

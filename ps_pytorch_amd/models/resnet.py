@@ -18,6 +18,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..ops.conv import PsConv2d
+from ..ops.linear import PsLinear
 from ..ops.modules import PsBatchNorm2d
 
 
@@ -81,7 +82,7 @@ class ResNet(nn.Module):
         self.layer2 = self._make_layer(block, 128, num_blocks[1], stride=2)
         self.layer3 = self._make_layer(block, 256, num_blocks[2], stride=2)
         self.layer4 = self._make_layer(block, 512, num_blocks[3], stride=2)
-        self.linear = nn.Linear(512 * block.expansion, num_classes)
+        self.linear = PsLinear(512 * block.expansion, num_classes)
 
     def _make_layer(self, block, planes, n, stride):
         strides = [stride] + [1] * (n - 1)

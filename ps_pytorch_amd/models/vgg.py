@@ -5,6 +5,7 @@ from __future__ import annotations
 import torch.nn as nn
 
 from ..ops.conv import PsConv2d
+from ..ops.linear import PsLinear
 from ..ops.modules import PsBatchNorm2d
 
 _CFG = {
@@ -39,7 +40,7 @@ class VGG(nn.Module):
                  batch_norm: bool = False):
         super().__init__()
         self.features = _make_layers(_CFG[name], in_channels, batch_norm)
-        self.classifier = nn.Linear(512, num_classes)
+        self.classifier = PsLinear(512, num_classes)
 
     def forward(self, x):
         x = self.features(x).flatten(1)
