@@ -18,6 +18,7 @@ import torch.nn as nn
 
 from ..ops.conv import PsConv2d
 from ..ops.linear import PsLinear
+from ..ops.pool import max_pool2d as ps_max_pool2d
 import torch.nn.functional as F
 
 
@@ -31,8 +32,8 @@ class LeNet(nn.Module):
         self.fc2 = PsLinear(500, num_classes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = F.max_pool2d(F.relu(self.conv1(x)), 2)
-        x = F.max_pool2d(F.relu(self.conv2(x)), 2)
+        x = ps_max_pool2d(F.relu(self.conv1(x)), 2)
+        x = ps_max_pool2d(F.relu(self.conv2(x)), 2)
         x = x.flatten(1)
         x = F.relu(self.fc1(x))
         return self.fc2(x)

@@ -19,6 +19,7 @@ import torch.nn.functional as F
 
 from ..ops.conv import PsConv2d
 from ..ops.linear import PsLinear
+from ..ops.pool import max_pool2d as ps_max_pool2d, global_avg_pool
 from ..ops.modules import PsBatchNorm2d
 
 
@@ -95,12 +96,12 @@ class ResNet(nn.Module):
     def forward(self, x):
         out = self.bn1(self.conv1(x))
         if self.imagenet_stem:
-            out = F.max_pool2d(out, 3, stride=2, padding=1)
+            out = ps_max_pool2d(out, 3, stride=2, padding=1)
         out = self.layer1(out)
         out = self.layer2(out)
         out = self.layer3(out)
         out = self.layer4(out)
-        out = F.adaptive_avg_pool2d(out, 1).flatten(1)
+        out = global_avg_pool(out)
         return self.linear(out)
 
 

@@ -68,6 +68,18 @@ def _try_load() -> None:
         lib.ps_bn_bwd.argtypes = [ctypes.c_void_p] * 12 + [
             ctypes.c_long, ctypes.c_long, ctypes.c_int, ctypes.c_int,
             ctypes.c_void_p]
+        lib.ps_softmax_ce_fwd.argtypes = [ctypes.c_void_p] * 5 + [
+            ctypes.c_long, ctypes.c_int, ctypes.c_void_p]
+        lib.ps_softmax_ce_bwd.argtypes = [ctypes.c_void_p] * 5 + [
+            ctypes.c_long, ctypes.c_int, ctypes.c_void_p]
+        lib.ps_maxpool_fwd.argtypes = [ctypes.c_void_p] * 3 + [
+            ctypes.c_int] * 10 + [ctypes.c_void_p]
+        lib.ps_maxpool_bwd.argtypes = [ctypes.c_void_p] * 3 + [
+            ctypes.c_int] * 10 + [ctypes.c_void_p]
+        lib.ps_gavgpool_fwd.argtypes = [ctypes.c_void_p] * 2 + [
+            ctypes.c_int] * 3 + [ctypes.c_void_p]
+        lib.ps_gavgpool_bwd.argtypes = [ctypes.c_void_p] * 2 + [
+            ctypes.c_int] * 3 + [ctypes.c_void_p]
         _LIB = lib
     except OSError as e:  # pragma: no cover
         _LIB_ERR = str(e)

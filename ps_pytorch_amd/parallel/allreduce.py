@@ -25,6 +25,8 @@ import torch
 import torch.distributed as dist
 import torch.nn.functional as F
 
+from ..ops.loss import cross_entropy as ps_cross_entropy
+
 from ..config import JobConfig, input_shape_of, num_classes_of
 from ..models import build_model
 from ..optim import FlatSGD
@@ -117,7 +119,7 @@ class AllReduceTrainer:
         self.flat.zero_grads()
         self._reset()
         out = self.network(data)
-        loss = F.cross_entropy(out.float(), target)
+        loss = ps_cross_entropy(out, target)
         loss.backward()
         if self.cfg.overlap:
             self._flush_ready()

@@ -18,6 +18,8 @@ from typing import List, Optional
 import torch
 import torch.nn.functional as F
 
+from ..ops.loss import cross_entropy as ps_cross_entropy
+
 from ..config import JobConfig, input_shape_of, num_classes_of
 from ..models import build_model
 from ..parallel.flat import FlatSpace, prep_model
@@ -158,7 +160,7 @@ class DistributedWorker:
         try:
             t0 = time.time()
             out = self.network(data)
-            loss = F.cross_entropy(out.float(), target)
+            loss = ps_cross_entropy(out, target)
             self.f_dur = time.time() - t0
             t0 = time.time()
             loss.backward()

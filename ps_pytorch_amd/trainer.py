@@ -16,6 +16,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from .ops.loss import cross_entropy as ps_cross_entropy
+
 from .config import JobConfig, num_classes_of, input_shape_of
 from .models import build_model
 from .optim import FlatSGD, FlatAdam
@@ -57,7 +59,7 @@ class NNTrainer:
 
     def _loss(self, data, target):
         out = self.network(data)
-        return F.cross_entropy(out.float(), target), out
+        return ps_cross_entropy(out, target), out
 
     def _step_body(self, data, target) -> torch.Tensor:
         self.flat.zero_grads()
