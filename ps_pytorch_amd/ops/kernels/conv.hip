@@ -100,9 +100,24 @@ struct FwdLds {
     unsigned short B[NBUF][TN][PITCH];
 };
 
-template <bool SWZ>
+// Granule swizzle σ(row, g): position of logical granule g (16 B) within
+// row's 128-B span. Must be a permutation of g for each row (writes).
+//
+// Mode 1 (r1): g ^ (row&7) ^ ((row>>3)&7). PMC showed ~8 pairwise bank
+//   collisions per ds_read_b128 lane group (94M conflict cycles on fwd l3):
+//   each b128 group mixes two fq half-sets {rows E1, granule g0} and
+//   {rows E2, g0^1}, and this σ maps both onto the SAME 8 bank classes.
+// Mode 2 (r2 fix): g ^ (((row>>1)&3)<<1). Bank class = (8*(row&1) + σ)
+//   mod 16; per group the even-row lanes give σ(E1,g0) = {0,2,4,6} and
+//   σ(E2,g0^1) = {5,7,1,3} (disjoint), same for odd rows — a full 16-class
+//   bijection per lane group, i.e. conflict-free by the documented
+//   (a/4) mod 64 b128 banking. Verified on hardware via
+//   SQ_LDS_BANK_CONFLICT; PS_SWZ=1 selects the old layout for A/B.
+template <int SWZM>
 __device__ __forceinline__ int gswz(int row, int g) {
-    return SWZ ? (g ^ (row & 7) ^ ((row >> 3) & 7)) : g;
+    if constexpr (SWZM == 2)
+        return g ^ (((row >> 1) & 3) << 1);
+    return g ^ (row & 7) ^ ((row >> 3) & 7);
 }
 
 // SMALL=true: the whole R*S*C contraction fits one 64-chunk (ResNet stem
@@ -110,7 +125,8 @@ __device__ __forceinline__ int gswz(int row, int g) {
 // axis via a per-lane gather table and run ONE k-step instead of R*S, so
 // the MFMA utilization is RSC/64 of a full tile instead of C/64 per step.
 template <int TM, int TN, int STRIDE, bool DGRAD, bool SMALL = false, bool AL = true,
-          int NBUF = 2>   // NBUF=1 only for the one-step (RSC<=64) SMALL path
+          int NBUF = 2,   // NBUF=1 only for the one-step (RSC<=64) SMALL path
+          int SWZM = 2>   // LDS granule swizzle mode (see gswz)
 __global__ __launch_bounds__(256) void conv_gemm_kernel(
     const unsigned short* __restrict__ src,  // fwd: in [Nb,H,W,C]; dgrad: dout [Nb,P,Q,K]
     const unsigned short* __restrict__ wgt,  // fwd: w [K,R,S,C]; dgrad: wT [R,S,C,K]
@@ -120,7 +136,7 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
     int R, int S, int pad)
 {
     __shared__ __attribute__((aligned(16))) FwdLds<TM, TN, NBUF> lds;
-    constexpr bool SWZ = true;
+    constexpr int SWZ = SWZM;
     constexpr int AR = TM / 32;        // A rows staged per thread
     constexpr int BR = TN / 32;        // B rows staged per thread
     const int Cin = DGRAD ? K : C;     // contraction channel count
@@ -374,7 +390,7 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
 // All parity classes ride ONE launch: `ends` = inclusive-exclusive prefix
 // of per-class block counts, `codes` = packed (hp<<1|wp) per slot — the
 // four separate launches were short-pipeline/launch-overhead bound.
-template <int TM, int TN, bool AL = true>
+template <int TM, int TN, bool AL = true, int SWZM = 2>
 __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
     const unsigned short* __restrict__ dout, // [Nb,P,Q,K]
     const unsigned short* __restrict__ wgt,  // wT [R,S,C,K]
@@ -393,7 +409,7 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
     const int hp = code >> 1, wp = code & 1;
     const int Hc = (H - hp + 1) >> 1, Wc = (W - wp + 1) >> 1;
     __shared__ __attribute__((aligned(16))) FwdLds<TM, TN> lds;
-    constexpr bool SWZ = true;
+    constexpr int SWZ = SWZM;
     constexpr int AR = TM / 32;
     constexpr int BR = TN / 32;
     const long M = (long)Nb * Hc * Wc;
@@ -1271,13 +1287,25 @@ __global__ __launch_bounds__(256) void colsum_fold_kernel(
 
 // ---------------------------------------------------------------- C API
 
+// PS_SWZ=1 selects the round-1 LDS swizzle for on-hardware A/B (gswz).
+static inline int swz_mode() {
+    static int m = -1;
+    if (m < 0) { const char* e = getenv("PS_SWZ"); m = e ? atoi(e) : 2; }
+    return m;
+}
+
 #define LAUNCH_GEMM(TM, TN, ST, DG, SM) LAUNCH_GEMM_NB(TM, TN, ST, DG, SM, 2)
 #define LAUNCH_GEMM_NB(TM, TN, ST, DG, SM, NBV)                               \
+    do {                                                                      \
+        if (swz_mode() == 1) LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, NBV, 1);      \
+        else                 LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, NBV, 2);      \
+    } while (0)
+#define LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, NBV, SWV)                          \
     do {                                                                      \
         long M_ = DG ? (long)Nb * H * W : (long)Nb * P * Q;                   \
         int Nout_ = DG ? C : K;                                               \
         long grid = ((M_ + TM - 1) / TM) * ((Nout_ + TN - 1) / TN);           \
-        hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, ST, DG, SM, ALV, NBV>), \
+        hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, ST, DG, SM, ALV, NBV, SWV>), \
             dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,            \
             (const unsigned short*)src, (const unsigned short*)wgt,           \
             (const unsigned short*)bias, (unsigned short*)dst,                \
@@ -1365,11 +1393,14 @@ extern "C" void ps_conv_dgrad(
         if (total == 0) return;
         int4 ends = make_int4(ends_[0], ends_[1], ends_[2], ends_[3]);
         int4 codes = make_int4(codes_[0], codes_[1], codes_[2], codes_[3]);
-#define DG2(TM, TN, ALV)                                                      \
-        hipLaunchKernelGGL((conv_dgrad2_kernel<TM, TN, ALV>),                 \
+#define DG2_SW(TM, TN, ALV, SWV)                                              \
+        hipLaunchKernelGGL((conv_dgrad2_kernel<TM, TN, ALV, SWV>),            \
             dim3((unsigned)total), dim3(256), 0, (hipStream_t)strm, dout,     \
             (const unsigned short*)wgt, (unsigned short*)dst,                 \
             Nb, H, W, C, K, P, Q, R, S, pad, ends, codes)
+#define DG2(TM, TN, ALV) do {                                                 \
+        if (swz_mode() == 1) DG2_SW(TM, TN, ALV, 1);                          \
+        else                 DG2_SW(TM, TN, ALV, 2); } while (0)
         if (al) { if (C >= 128) DG2(128, 128, true);
                   else          DG2(128, 64, true); }
         else    { if (C >= 128) DG2(128, 128, false);
