@@ -1260,18 +1260,24 @@ __global__ __launch_bounds__(256) void colsum_part_kernel(
     long M, int K)
 {
     __shared__ float red[256];
-    int k = threadIdx.x & 63;
-    int sub = threadIdx.x >> 6;           // 4 m-partitions per block
-    float acc = 0.f;
-    if (k < K)
-        for (long m = (long)blockIdx.x * 4 + sub; m < M;
-             m += (long)gridDim.x * 4)
-            acc += bf16_to_f32(dout[m * K + k]);
-    red[threadIdx.x] = acc;
-    __syncthreads();
-    if (sub == 0 && k < K)
-        partial[(long)blockIdx.x * K + k] =
-            red[k] + red[64 + k] + red[128 + k] + red[192 + k];
+    const int lane = threadIdx.x & 63;
+    const int sub = threadIdx.x >> 6;     // 4 m-partitions per block
+    // outer loop covers K > 64 (Linear bias: K = 500/1000; conv biases are
+    // small but this path is shared with PsLinear's db)
+    for (int kb = 0; kb < K; kb += 64) {
+        int k = kb + lane;
+        float acc = 0.f;
+        if (k < K)
+            for (long m = (long)blockIdx.x * 4 + sub; m < M;
+                 m += (long)gridDim.x * 4)
+                acc += bf16_to_f32(dout[m * K + k]);
+        red[threadIdx.x] = acc;
+        __syncthreads();
+        if (sub == 0 && k < K)
+            partial[(long)blockIdx.x * K + k] =
+                red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane];
+        __syncthreads();
+    }
 }
 
 __global__ __launch_bounds__(256) void colsum_fold_kernel(
