@@ -83,6 +83,11 @@ class _ConvFn(torch.autograd.Function):
                         current_stream_ptr())
         ctx.save_for_backward(x, wc)
         ctx.conf = (stride, pad, b is not None)
+        # steal-mode targets: write dw/db straight into the flat_g slices
+        # (parallel/flat.py attach_grads) — AccumulateGrad then steals the
+        # returned view instead of launching an add per param.
+        ctx.gtgt = (getattr(w, '_ps_flat_grad', None),
+                    getattr(b, '_ps_flat_grad', None) if b is not None else None)
         return out
 
     @staticmethod
@@ -111,13 +116,19 @@ class _ConvFn(torch.autograd.Function):
             split = _wgrad_split(M, K, C, R, S, stride, pad, P, Q)
             partial = torch.empty(split * K * R * S * C,
                                   dtype=torch.float32, device=x.device)
-            dw = torch.empty_like(w).contiguous(memory_format=_CL)
+            wt_tgt = ctx.gtgt[0]
+            dw = (wt_tgt if wt_tgt is not None and wt_tgt.dtype == w.dtype
+                  and wt_tgt.is_cuda
+                  else torch.empty_like(w).contiguous(memory_format=_CL))
             lib.ps_conv_wgrad(dout.data_ptr(), x.data_ptr(),
                               partial.data_ptr(), dw.data_ptr(),
                               Nb, H, W, C, K, P, Q, R, S, stride, pad,
                               split, current_stream_ptr())
         if has_bias and ctx.needs_input_grad[2]:
-            db = torch.empty(K, dtype=dout.dtype, device=dout.device)
+            b_tgt = ctx.gtgt[1]
+            db = (b_tgt if b_tgt is not None and b_tgt.dtype == dout.dtype
+                  and b_tgt.is_cuda
+                  else torch.empty(K, dtype=dout.dtype, device=dout.device))
             bpart = torch.empty(512 * K, dtype=torch.float32,
                                 device=dout.device)
             lib.ps_conv_bias_grad(db.data_ptr(), dout.data_ptr(),

@@ -596,6 +596,19 @@ struct WgradLds {
 // address; the 16-lane group's lanes get their column's 4 values. The
 // compiler-modeled builtin (ck_tile idiom) — an inline-asm version with a
 // tied-operand waitcnt barrier cost ~150 VALU/step in AGPR<->VGPR copies.
+// Pixel-row placement permutation for the [pixel][16-chan] subtiles:
+// row stride is 32 B, so rows 8 apart land 256 B apart = the same 64-dword
+// bank window, and every tr16 read's 32-lane group spans rows {C+j} and
+// {C+8+j} -> a guaranteed 2-way conflict on all 16 pairs (measured: 33% of
+// wgrad LDS cycles). pswz(r) = r ^ ((r>>3)&1)<<2 satisfies
+// phi(r+8) = phi(r)^4 (mod-8 bank class), so the two 4-row windows of any
+// read group land on disjoint bank classes, for the shifted s-tap B rows of
+// the row-halo kernel too. Applied identically at write and read setup
+// (all precomputed — zero inner-loop cost).
+__device__ __forceinline__ int pswz(int r) {
+    return r ^ (((r >> 3) & 1) << 2);
+}
+
 typedef __bf16 bf16x4_t __attribute__((ext_vector_type(4)));
 __device__ __forceinline__ bf16x4_t ds_tr16p(const unsigned short* p) {
     typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 v4;
@@ -706,9 +719,9 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
 #pragma unroll
         for (int h = 0; h < KH; ++h) {
             int kk8 = h * 64 + cc8;
-            wrA[rr][h] = &lds.A[0][kk8 >> 4][midx * 16 + (kk8 & 15)];
+            wrA[rr][h] = &lds.A[0][kk8 >> 4][pswz(midx) * 16 + (kk8 & 15)];
         }
-        wrB[rr] = &lds.B[0][cc8 >> 4][midx * 16 + (cc8 & 15)];
+        wrB[rr] = &lds.B[0][cc8 >> 4][pswz(midx) * 16 + (cc8 & 15)];
     }
     // transpose-read offsets: frag (mi,kk,half i) = block rows
     // m = kk*32 + fq*8 + i*4 .. +4, cols = 16-chan subtile; lane fr reads
@@ -720,7 +733,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
         for (int i = 0; i < 2; ++i) {
-            int m = kk * 32 + fq * 8 + i * 4 + (fr >> 2);
+            int m = pswz(kk * 32 + fq * 8 + i * 4 + (fr >> 2));
             int qo = (fr & 3) * 4;
 #pragma unroll
             for (int mi = 0; mi < MI; ++mi)
@@ -930,12 +943,12 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
 #pragma unroll
     for (int h = 0; h < TK / 64; ++h) {
         int kk8 = h * 64 + cc8;
-        wrA[h] = &lds.A[0][kk8 >> 4][(t >> 3) * 16 + (kk8 & 15)];
+        wrA[h] = &lds.A[0][kk8 >> 4][pswz(t >> 3) * 16 + (kk8 & 15)];
     }
     unsigned short* wrB[5];
 #pragma unroll
     for (int u = 0; u < 5; ++u) {
-        int hp = tb_rib[u] * HW2 + tb_j[u];
+        int hp = pswz(tb_rib[u] * HW2 + tb_j[u]);
         wrB[u] = &lds.B[0][cc8 >> 4][hp * 16 + (cc8 & 15)];
     }
 
@@ -961,11 +974,11 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
 #pragma unroll
         for (int mi = 0; mi < MI; ++mi)
             roA[mi][i] = (unsigned)((char*)&lds.A[0][wm * MI + mi]
-                                    [e * 16 + qo] - lb);
+                                    [pswz(e) * 16 + qo] - lb);
         int rib = e >> l2q, q = e & (Q - 1);
 #pragma unroll
         for (int s = 0; s < 3; ++s) {
-            int hp = rib * HW2 + q + s;
+            int hp = pswz(rib * HW2 + q + s);
 #pragma unroll
             for (int nj = 0; nj < 2; ++nj)
                 roB[s][nj][i] = (unsigned)((char*)&lds.B[0][wn * 2 + nj]
@@ -1131,7 +1144,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
     unsigned short* wrB[2];
 #pragma unroll
     for (int rr = 0; rr < 2; ++rr) {
-        int midx = trow + 32 * rr;
+        int midx = pswz(trow + 32 * rr);
         wrA[rr] = &lds.A[0][cc8 >> 4][midx * 16 + (cc8 & 15)];
         wrB[rr] = &lds.B[0][cc8 >> 4][midx * 16 + (cc8 & 15)];
     }
@@ -1141,7 +1154,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
         for (int i = 0; i < 2; ++i) {
-            int m = kk * 32 + fq * 8 + i * 4 + (fr >> 2);
+            int m = pswz(kk * 32 + fq * 8 + i * 4 + (fr >> 2));
             int qo = (fr & 3) * 4;
 #pragma unroll
             for (int mi = 0; mi < 2; ++mi)

@@ -53,6 +53,9 @@ class _FusedBNFn(torch.autograd.Function):
         ctx.save_for_backward(x, y, gamma, save_mean, save_invstd)
         ctx.relu = relu
         ctx.has_res = residual is not None
+        # steal-mode flat_g targets (see parallel/flat.py attach_grads)
+        ctx.gtgt = (getattr(gamma, '_ps_flat_grad', None),
+                    getattr(beta, '_ps_flat_grad', None))
         return y
 
     @staticmethod
@@ -64,8 +67,11 @@ class _FusedBNFn(torch.autograd.Function):
         M = N * H * W
         dx = torch.empty_like(x)
         dres = torch.empty_like(x) if ctx.has_res else None
-        dgamma = torch.empty_like(gamma)
-        dbeta = torch.empty_like(gamma)
+        g_tgt, b_tgt = ctx.gtgt
+        dgamma = (g_tgt if g_tgt is not None and g_tgt.dtype == gamma.dtype
+                  and g_tgt.is_cuda else torch.empty_like(gamma))
+        dbeta = (b_tgt if b_tgt is not None and b_tgt.dtype == gamma.dtype
+                 and b_tgt.is_cuda else torch.empty_like(gamma))
         ws = torch.empty(5 * C, dtype=torch.float32, device=x.device)
         partial = torch.empty(1024 * 2 * C, dtype=torch.float32, device=x.device)
         lib.ps_bn_bwd(

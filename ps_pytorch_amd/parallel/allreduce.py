@@ -60,7 +60,7 @@ class AllReduceTrainer:
         net = prep_model(net, self.device, self.compute_dtype)
         self.network = net
         self.flat = FlatSpace(net, bucket_bytes=int(cfg.bucket_mb * 2 ** 20))
-        self.flat.attach_grads()
+        self.flat.attach_grads(steal=(self.device.type == 'cuda'))
         # init-time param sync (ref data_parallel_dist.py:45-46) — one flat
         # broadcast (ranks are identically seeded; this guards non-determinism)
         if dist.is_initialized() and self.world > 1:
@@ -89,6 +89,7 @@ class AllReduceTrainer:
 
     def _make_hook(self, pid: int):
         def hook(_param):
+            self.flat.ensure_grad_in_flat(pid)
             bi = self._param_bucket[pid]
             self._pending[bi] -= 1
             if self._pending[bi] == 0:
@@ -123,6 +124,8 @@ class AllReduceTrainer:
         loss.backward()
         if self.cfg.overlap:
             self._flush_ready()
+        else:
+            self.flat.harvest_grads()   # steal mode: catch fallback grads
         for b in self.flat.buckets[self._next_launch:]:
             self._push(b)
         self._next_launch = len(self.flat.buckets)

@@ -149,13 +149,57 @@ class FlatSpace:
 
     # ---- gradient plumbing ----
 
-    def attach_grads(self) -> None:
-        """Point every p.grad at its flat_g slice (autograd accumulates +=)."""
+    def attach_grads(self, steal: bool = False) -> None:
+        """Wire autograd output into flat_g.
+
+        view mode (default, CPU/torch ops): p.grad = flat_g view; autograd
+        ACCUMULATES into it (one at::add kernel per param per step).
+
+        steal mode (GPU kernel path): p.grad starts None and every Ps op's
+        backward WRITES its weight/bias grad directly into the param's
+        flat_g slice (published here as p._ps_flat_grad) and returns that
+        view — AccumulateGrad then steals instead of adding, removing ~60
+        elementwise add kernels per ResNet-18 step. Ops that fell back to
+        torch produce a foreign tensor; ensure_grad_in_flat() detects that
+        by pointer and copies, so kill-switches (PS_CONV=0 …) stay correct.
+        """
+        self._steal = steal
+        self._gviews = []
+        self._gptrs = []
         for pid, (p, off) in enumerate(zip(self.params, self.offsets)):
-            p.grad = self._view(self.flat_g, p, off, pid)
+            v = self._view(self.flat_g, p, off, pid)
+            self._gviews.append(v)
+            self._gptrs.append(v.data_ptr())
+            p._ps_flat_grad = v
+            p.grad = None if steal else v
 
     def zero_grads(self) -> None:
         self.flat_g.zero_()
+        if getattr(self, '_steal', False):
+            for p in self.params:
+                p.grad = None
+
+    def ensure_grad_in_flat(self, pid: int) -> None:
+        """Steal mode: if this param's grad landed outside flat_g (a torch
+        fallback op), copy it into the slice. No-op when the Ps kernels
+        produced it in place (pointer match)."""
+        if not getattr(self, '_steal', False):
+            return
+        p = self.params[pid]
+        g = p.grad
+        if g is None or g.data_ptr() == self._gptrs[pid]:
+            return
+        if not getattr(self, '_warned_foreign', False):
+            self._warned_foreign = True
+            import warnings
+            warnings.warn("grad for a parameter landed outside flat_g "
+                          "(torch-fallback op?); copying per step")
+        self._gviews[pid].copy_(g)
+
+    def harvest_grads(self) -> None:
+        """Steal mode: make sure every param's grad is in flat_g."""
+        for pid in range(len(self.params)):
+            self.ensure_grad_in_flat(pid)
 
     def grad_slice(self, b: Bucket) -> torch.Tensor:
         return self.flat_g[b.start:b.end]

@@ -59,7 +59,7 @@ class DistributedWorker:
         net = prep_model(net, self.device, self.compute_dtype)
         self.network = net
         self.flat = FlatSpace(net, bucket_bytes=int(cfg.bucket_mb * 2 ** 20))
-        self.flat.attach_grads()
+        self.flat.attach_grads(steal=(self.device.type == 'cuda'))
         self.transport = PSTransport(self.flat, self.wire_dtype, self.device,
                                      self.rank, self.world,
                                      mode=cfg.aggregation,
@@ -96,6 +96,7 @@ class DistributedWorker:
     def _make_hook(self, pid: int):
         def hook(_param):
             self._check_abort()
+            self.flat.ensure_grad_in_flat(pid)
             bi = self._param_bucket[pid]
             self._pending[bi] -= 1
             if self._pending[bi] == 0:
@@ -135,6 +136,8 @@ class DistributedWorker:
         # timeout-mode marking: tell the PS whether this step's payloads are
         # real before it counts them toward the --num-aggregate quota
         self.transport.send_killed_flag(killed)
+        if not self.cfg.overlap and not killed:
+            self.flat.harvest_grads()   # steal mode: catch fallback grads
         if self.cfg.overlap:
             if not killed:
                 self._flush_ready()

@@ -55,7 +55,9 @@ class NNTrainer:
         # f32 master copy + fused update; flat_g is the "wire" (local = trivially summed)
         self.master_w = self.flat.flat_w.detach().to(torch.float32).clone()
         self.optimizer = FlatSGD(self.master_w, lr=cfg.lr, momentum=cfg.momentum)
-        self.flat.attach_grads()
+        # steal mode on GPU: Ps ops write grads into flat_g directly (no
+        # per-param AccumulateGrad add kernels); view mode on CPU
+        self.flat.attach_grads(steal=(self.device.type == 'cuda'))
 
     def _loss(self, data, target):
         out = self.network(data)
@@ -65,6 +67,7 @@ class NNTrainer:
         self.flat.zero_grads()
         loss, _ = self._loss(data, target)
         loss.backward()
+        self.flat.harvest_grads()   # steal mode: catch torch-fallback grads
         # fused: master update + re-pack into the live (possibly bf16) params
         self.optimizer.step(self.flat.flat_g, grad_scale=1.0,
                             wire_out=self.flat.flat_w)
