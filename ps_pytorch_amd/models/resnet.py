@@ -17,7 +17,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops.conv import PsConv2d
+from ..ops.conv import PsConv2d, conv_with_passthrough
 from ..ops.linear import PsLinear
 from ..ops.pool import max_pool2d as ps_max_pool2d, global_avg_pool
 from ..ops.modules import PsBatchNorm2d
@@ -40,8 +40,12 @@ class BasicBlock(nn.Module):
             )
 
     def forward(self, x):
-        out = self.bn1(self.conv1(x))
-        return self.bn2(self.conv2(out), residual=self.shortcut(x))
+        # conv1 passes x through so the residual branch's gradient fuses
+        # into conv1's dgrad epilogue (ops/conv.py _ConvCarryFn) instead of
+        # an autograd elementwise add at the fork
+        h, x_sc = conv_with_passthrough(self.conv1, x)
+        out = self.bn1(h)
+        return self.bn2(self.conv2(out), residual=self.shortcut(x_sc))
 
 
 class Bottleneck(nn.Module):
@@ -63,9 +67,10 @@ class Bottleneck(nn.Module):
             )
 
     def forward(self, x):
-        out = self.bn1(self.conv1(x))
+        h, x_sc = conv_with_passthrough(self.conv1, x)
+        out = self.bn1(h)
         out = self.bn2(self.conv2(out))
-        return self.bn3(self.conv3(out), residual=self.shortcut(x))
+        return self.bn3(self.conv3(out), residual=self.shortcut(x_sc))
 
 
 class ResNet(nn.Module):

@@ -53,7 +53,7 @@ def _try_load() -> None:
         lib.ps_acc.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                ctypes.c_long, ctypes.c_int, ctypes.c_void_p]
         lib.ps_conv_fwd.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int] * 11 + [ctypes.c_void_p]
-        lib.ps_conv_dgrad.argtypes = [ctypes.c_void_p] * 3 + [ctypes.c_int] * 11 + [ctypes.c_void_p]
+        lib.ps_conv_dgrad.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int] * 11 + [ctypes.c_void_p]
         lib.ps_conv_wgrad.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int] * 12 + [ctypes.c_void_p]
         lib.ps_conv_bias_grad.argtypes = [ctypes.c_void_p] * 3 + [
             ctypes.c_long, ctypes.c_int, ctypes.c_void_p]

@@ -126,12 +126,16 @@ __device__ __forceinline__ int gswz(int row, int g) {
 // the MFMA utilization is RSC/64 of a full tile instead of C/64 per step.
 template <int TM, int TN, int STRIDE, bool DGRAD, bool SMALL = false, bool AL = true,
           int NBUF = 2,   // NBUF=1 only for the one-step (RSC<=64) SMALL path
-          int SWZM = 2>   // LDS granule swizzle mode (see gswz)
+          int SWZM = 2,   // LDS granule swizzle mode (see gswz)
+          bool ACCF = false>  // dgrad: epilogue adds `carry` (residual-fork
+                              // grad accumulation fused in — kills the
+                              // autograd at::add at every block input fork)
 __global__ __launch_bounds__(256) void conv_gemm_kernel(
     const unsigned short* __restrict__ src,  // fwd: in [Nb,H,W,C]; dgrad: dout [Nb,P,Q,K]
     const unsigned short* __restrict__ wgt,  // fwd: w [K,R,S,C]; dgrad: wT [R,S,C,K]
     const unsigned short* __restrict__ bias, // [K] or null (fwd only)
     unsigned short* __restrict__ dst,        // fwd: out [Nb,P,Q,K]; dgrad: dx [Nb,H,W,C]
+    const unsigned short* __restrict__ carry,// [dst shape] or null (ACCF)
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int pad)
 {
@@ -374,8 +378,12 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
 #pragma unroll
         for (int e = 0; e < 4; ++e) {
             long om = m0 + wm * 64 + mi * 16 + fq * 4 + e;
-            if (om < M && col < Nout)
-                dst[om * Nout + col] = f32_to_bf16(acc[mi][nj][e] + bv);
+            if (om < M && col < Nout) {
+                float v = acc[mi][nj][e] + bv;
+                if constexpr (ACCF)
+                    v += bf16_to_f32(carry[om * Nout + col]);
+                dst[om * Nout + col] = f32_to_bf16(v);
+            }
         }
     }
 }
@@ -390,11 +398,12 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
 // All parity classes ride ONE launch: `ends` = inclusive-exclusive prefix
 // of per-class block counts, `codes` = packed (hp<<1|wp) per slot — the
 // four separate launches were short-pipeline/launch-overhead bound.
-template <int TM, int TN, bool AL = true, int SWZM = 2>
+template <int TM, int TN, bool AL = true, int SWZM = 2, bool ACCF = false>
 __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
     const unsigned short* __restrict__ dout, // [Nb,P,Q,K]
     const unsigned short* __restrict__ wgt,  // wT [R,S,C,K]
     unsigned short* __restrict__ dx,         // [Nb,H,W,C]
+    const unsigned short* __restrict__ carry,// [Nb,H,W,C] or null (ACCF)
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int pad, int4 ends, int4 codes)
 {
@@ -568,7 +577,10 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
                 int n = (int)(om / ((long)Hc * Wc));
                 int rem = (int)(om % ((long)Hc * Wc));
                 int h = hp + 2 * (rem / Wc), w = wp + 2 * (rem % Wc);
-                dx[((long)(n * H + h) * W + w) * C + col] = f32_to_bf16(acc[mi][nj][e]);
+                long oi = ((long)(n * H + h) * W + w) * C + col;
+                float v = acc[mi][nj][e];
+                if constexpr (ACCF) v += bf16_to_f32(carry[oi]);
+                dx[oi] = f32_to_bf16(v);
             }
         }
     }
@@ -1316,18 +1328,24 @@ static inline int swz_mode() {
 #define LAUNCH_GEMM(TM, TN, ST, DG, SM) LAUNCH_GEMM_NB(TM, TN, ST, DG, SM, 2)
 #define LAUNCH_GEMM_NB(TM, TN, ST, DG, SM, NBV)                               \
     do {                                                                      \
-        if (swz_mode() == 1) LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, NBV, 1);      \
-        else                 LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, NBV, 2);      \
+        if (swz_mode() == 1) LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, NBV, 1, false); \
+        else                 LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, NBV, 2, false); \
     } while (0)
-#define LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, NBV, SWV)                          \
+#define LAUNCH_GEMM_ACC(TM, TN, ST, DG, SM, ACV)                              \
+    do {                                                                      \
+        if (swz_mode() == 1) LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, 2, 1, ACV);   \
+        else                 LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, 2, 2, ACV);   \
+    } while (0)
+#define LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, NBV, SWV, ACV)                     \
     do {                                                                      \
         long M_ = DG ? (long)Nb * H * W : (long)Nb * P * Q;                   \
         int Nout_ = DG ? C : K;                                               \
         long grid = ((M_ + TM - 1) / TM) * ((Nout_ + TN - 1) / TN);           \
-        hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, ST, DG, SM, ALV, NBV, SWV>), \
+        hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, ST, DG, SM, ALV, NBV, SWV, ACV>), \
             dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,            \
             (const unsigned short*)src, (const unsigned short*)wgt,           \
             (const unsigned short*)bias, (unsigned short*)dst,                \
+            (const unsigned short*)carry,                                     \
             Nb, H, W, C, K, P, Q, R, S, pad);                                 \
     } while (0)
 
@@ -1336,6 +1354,7 @@ extern "C" void ps_conv_fwd(
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int stride, int pad, void* strm)
 {
+    const void* carry = nullptr;       // fwd never accumulates
     bool al = (C & 63) == 0;           // full 64-chunk contraction coverage
 #define FWD_BODY()                                                            \
     do {                                                                      \
@@ -1363,21 +1382,26 @@ extern "C" void ps_conv_fwd(
 
 // wgt here is the TRANSPOSED weight wT[R,S,C,K] (host permutes once per
 // backward; ~us for the largest ResNet tensor).
+// carry != null fuses a same-shaped bf16 tensor into the dx epilogue
+// (dx = dgrad + carry): the residual-fork gradient accumulation that
+// autograd otherwise runs as a separate elementwise add.
 extern "C" void ps_conv_dgrad(
-    const void* src, const void* wgt, void* dst,
+    const void* src, const void* wgt, void* dst, const void* carry,
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int stride, int pad, void* strm)
 {
     const void* bias = nullptr;
     bool al = (K & 63) == 0;           // contraction runs over K
     if (stride == 1) {
-#define DG_BODY()                                                             \
+#define DG_BODY(ACV)                                                          \
         do {                                                                  \
-            if (C >= 128) LAUNCH_GEMM(128, 128, 1, true, false);              \
-            else          LAUNCH_GEMM(128, 64, 1, true, false);               \
+            if (C >= 128) LAUNCH_GEMM_ACC(128, 128, 1, true, false, ACV);     \
+            else          LAUNCH_GEMM_ACC(128, 64, 1, true, false, ACV);      \
         } while (0)
-        if (al) { constexpr bool ALV = true; DG_BODY(); }
-        else    { constexpr bool ALV = false; DG_BODY(); }
+        if (al) { constexpr bool ALV = true;
+                  if (carry) DG_BODY(true); else DG_BODY(false); }
+        else    { constexpr bool ALV = false;
+                  if (carry) DG_BODY(true); else DG_BODY(false); }
 #undef DG_BODY
     } else {
         // 4 parity-class launches (see conv_dgrad2_kernel); classes with no
@@ -1390,9 +1414,16 @@ extern "C" void ps_conv_dgrad(
             int r0 = (hp + pad) & 1, s0 = (wp + pad) & 1;
             if (R <= r0 || S <= s0) any_empty = true;
         }
-        if (any_empty)
-            (void)hipMemsetAsync(dst, 0, (long)Nb * H * W * C * 2,
-                                 (hipStream_t)strm);
+        if (any_empty) {
+            // parity classes with no contributing taps: dx = carry (or 0)
+            if (carry)
+                (void)hipMemcpyAsync(dst, carry, (long)Nb * H * W * C * 2,
+                                     hipMemcpyDeviceToDevice,
+                                     (hipStream_t)strm);
+            else
+                (void)hipMemsetAsync(dst, 0, (long)Nb * H * W * C * 2,
+                                     (hipStream_t)strm);
+        }
         int TM_ = 128, TN_ = (C >= 128) ? 128 : 64;
         int tiles_n_ = (C + TN_ - 1) / TN_;
         int ends_[4] = {0, 0, 0, 0}, codes_[4] = {0, 0, 0, 0};
@@ -1412,14 +1443,18 @@ extern "C" void ps_conv_dgrad(
         if (total == 0) return;
         int4 ends = make_int4(ends_[0], ends_[1], ends_[2], ends_[3]);
         int4 codes = make_int4(codes_[0], codes_[1], codes_[2], codes_[3]);
-#define DG2_SW(TM, TN, ALV, SWV)                                              \
-        hipLaunchKernelGGL((conv_dgrad2_kernel<TM, TN, ALV, SWV>),            \
+#define DG2_SW(TM, TN, ALV, SWV, ACV)                                         \
+        hipLaunchKernelGGL((conv_dgrad2_kernel<TM, TN, ALV, SWV, ACV>),       \
             dim3((unsigned)total), dim3(256), 0, (hipStream_t)strm, dout,     \
             (const unsigned short*)wgt, (unsigned short*)dst,                 \
+            (const unsigned short*)carry,                                     \
             Nb, H, W, C, K, P, Q, R, S, pad, ends, codes)
+#define DG2_AC(TM, TN, ALV, SWV) do {                                         \
+        if (carry) DG2_SW(TM, TN, ALV, SWV, true);                            \
+        else       DG2_SW(TM, TN, ALV, SWV, false); } while (0)
 #define DG2(TM, TN, ALV) do {                                                 \
-        if (swz_mode() == 1) DG2_SW(TM, TN, ALV, 1);                          \
-        else                 DG2_SW(TM, TN, ALV, 2); } while (0)
+        if (swz_mode() == 1) DG2_AC(TM, TN, ALV, 1);                          \
+        else                 DG2_AC(TM, TN, ALV, 2); } while (0)
         if (al) { if (C >= 128) DG2(128, 128, true);
                   else          DG2(128, 64, true); }
         else    { if (C >= 128) DG2(128, 128, false);

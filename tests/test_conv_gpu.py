@@ -82,3 +82,42 @@ def test_conv_wgrad_deterministic():
         out.backward(dout)
         grads.append(wv.grad.clone())
     assert torch.equal(grads[0], grads[1])
+
+
+CARRY_SHAPES = [
+    (8, 64, 32, 32, 64, 3, 1, 1),     # BasicBlock conv1 s1 (gemm ACCF)
+    (8, 64, 32, 32, 128, 3, 2, 1),    # downsample conv1 s2 (dgrad2 ACCF)
+    (8, 64, 32, 32, 128, 1, 2, 0),    # 1x1 s2 (empty parity classes: carry copy)
+]
+
+
+@pytest.mark.parametrize("shape", CARRY_SHAPES)
+def test_conv_carry_fuses_residual_grad(shape):
+    """_ConvCarryFn: grad w.r.t. x must equal dgrad(dy) + dcarry — the
+    residual-fork accumulation fused into the dx epilogue (ACCF kernels)."""
+    from ps_pytorch_amd.ops.conv import _ConvCarryFn
+    Nb, C, H, W, K, R, stride, pad = shape
+    g = torch.Generator().manual_seed(hash(shape) % (2 ** 31))
+    x = torch.randn(Nb, C, H, W, generator=g).to('cuda', torch.bfloat16) \
+        .contiguous(memory_format=_CL).requires_grad_(True)
+    w = (torch.randn(K, C, R, R, generator=g) / (R * R * C) ** 0.5) \
+        .to('cuda', torch.bfloat16).contiguous(memory_format=_CL) \
+        .requires_grad_(True)
+    scale = torch.randn(Nb, C, H, W, generator=g).to('cuda', torch.bfloat16) \
+        .contiguous(memory_format=_CL)
+
+    out, xp = _ConvCarryFn.apply(x, w, None, stride, pad)
+    dout = torch.randn(out.shape, generator=g).to('cuda', torch.bfloat16) \
+        .contiguous(memory_format=_CL)
+    # residual branch consumes xp; its grad (scale) rides dcarry
+    loss = (out.float() * dout.float()).sum() + (xp.float() * scale.float()).sum()
+    loss.backward()
+
+    xr = x.detach().float().requires_grad_(True)
+    wr = w.detach().float().requires_grad_(True)
+    outr = F.conv2d(xr, wr, None, stride=stride, padding=pad)
+    lossr = (outr * dout.float()).sum() + (xr * scale.float()).sum()
+    lossr.backward()
+
+    assert _rel_err(x.grad, xr.grad) < 0.03, f"dx+carry {_rel_err(x.grad, xr.grad)}"
+    assert _rel_err(w.grad, wr.grad) < 0.03
