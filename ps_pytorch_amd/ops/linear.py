@@ -73,7 +73,7 @@ class _LinearFn(torch.autograd.Function):
             lib.ps_wt_transpose(wt.data_ptr(), w.data_ptr(), N, K,
                                 current_stream_ptr())
             dx = torch.empty_like(x)
-            lib.ps_conv_dgrad(dy.data_ptr(), wt.data_ptr(), dx.data_ptr(),
+            lib.ps_conv_dgrad(dy.data_ptr(), wt.data_ptr(), dx.data_ptr(), 0,
                               M, 1, 1, K, N, 1, 1, 1, 1, 1, 0,
                               current_stream_ptr())
             dx = dx.reshape(shape)
