@@ -86,8 +86,8 @@ class _ConvFn(torch.autograd.Function):
         # steal-mode targets: write dw/db straight into the flat_g slices
         # (parallel/flat.py attach_grads) — AccumulateGrad then steals the
         # returned view instead of launching an add per param.
-        ctx.gtgt = (getattr(w, '_ps_flat_grad', None),
-                    getattr(b, '_ps_flat_grad', None) if b is not None else None)
+        ctx.gtgt = (getattr(w, '_ps_flat_grad_fn', None),
+                    getattr(b, '_ps_flat_grad_fn', None) if b is not None else None)
         return out
 
     @staticmethod
@@ -121,7 +121,7 @@ class _ConvFn(torch.autograd.Function):
             split = _wgrad_split(M, K, C, R, S, stride, pad, P, Q)
             partial = torch.empty(split * K * R * S * C,
                                   dtype=torch.float32, device=x.device)
-            wt_tgt = ctx.gtgt[0]
+            wt_tgt = ctx.gtgt[0]() if ctx.gtgt[0] is not None else None
             dw = (wt_tgt if wt_tgt is not None and wt_tgt.dtype == w.dtype
                   and wt_tgt.is_cuda
                   else torch.empty_like(w).contiguous(memory_format=_CL))
@@ -130,7 +130,7 @@ class _ConvFn(torch.autograd.Function):
                               Nb, H, W, C, K, P, Q, R, S, stride, pad,
                               split, current_stream_ptr())
         if has_bias and ctx.needs_input_grad[2]:
-            b_tgt = ctx.gtgt[1]
+            b_tgt = ctx.gtgt[1]() if ctx.gtgt[1] is not None else None
             db = (b_tgt if b_tgt is not None and b_tgt.dtype == dout.dtype
                   and b_tgt.is_cuda
                   else torch.empty(K, dtype=dout.dtype, device=dout.device))
@@ -164,11 +164,14 @@ class _ConvCarryFn(torch.autograd.Function):
         return dx, dw, db, None, None
 
 
+_CARRY = os.environ.get('PS_CARRY', '1') != '0'    # A/B: autograd add vs fused
+
+
 def conv_with_passthrough(mod: nn.Conv2d, x: torch.Tensor):
     """(conv(x), x) — on the kernel path the pass-through output carries the
     residual branch so its gradient fuses into dgrad; otherwise plain."""
-    if _supported(x, mod.weight, mod.stride, mod.padding,
-                  mod.dilation, mod.groups) and x.requires_grad:
+    if _CARRY and _supported(x, mod.weight, mod.stride, mod.padding,
+                             mod.dilation, mod.groups) and x.requires_grad:
         return _ConvCarryFn.apply(x, mod.weight, mod.bias,
                                   mod.stride[0], mod.padding[0])
     return mod(x), x

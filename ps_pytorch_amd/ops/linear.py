@@ -55,8 +55,8 @@ class _LinearFn(torch.autograd.Function):
         ctx.save_for_backward(x2, wc)
         ctx.conf = (shape, b is not None)
         # steal-mode flat_g targets (see parallel/flat.py attach_grads)
-        ctx.gtgt = (getattr(w, '_ps_flat_grad', None),
-                    getattr(b, '_ps_flat_grad', None) if b is not None else None)
+        ctx.gtgt = (getattr(w, '_ps_flat_grad_fn', None),
+                    getattr(b, '_ps_flat_grad_fn', None) if b is not None else None)
         return y.reshape(*shape[:-1], N)
 
     @staticmethod
@@ -81,14 +81,14 @@ class _LinearFn(torch.autograd.Function):
             split = _wgrad_split(M, N, K, 1, 1, stride=1, pad=0, P=1, Q=1)
             partial = torch.empty(split * N * K, dtype=torch.float32,
                                   device=x.device)
-            wt_tgt = ctx.gtgt[0]
+            wt_tgt = ctx.gtgt[0]() if ctx.gtgt[0] is not None else None
             dw = (wt_tgt if wt_tgt is not None and wt_tgt.dtype == w.dtype
                   and wt_tgt.is_cuda else torch.empty_like(w))
             lib.ps_conv_wgrad(dy.data_ptr(), x.data_ptr(), partial.data_ptr(),
                               dw.data_ptr(), M, 1, 1, K, N, 1, 1, 1, 1, 1, 0,
                               split, current_stream_ptr())
         if has_bias and ctx.needs_input_grad[2]:
-            b_tgt = ctx.gtgt[1]
+            b_tgt = ctx.gtgt[1]() if ctx.gtgt[1] is not None else None
             db = (b_tgt if b_tgt is not None and b_tgt.dtype == dy.dtype
                   and b_tgt.is_cuda
                   else torch.empty(N, dtype=dy.dtype, device=dy.device))

@@ -54,8 +54,8 @@ class _FusedBNFn(torch.autograd.Function):
         ctx.relu = relu
         ctx.has_res = residual is not None
         # steal-mode flat_g targets (see parallel/flat.py attach_grads)
-        ctx.gtgt = (getattr(gamma, '_ps_flat_grad', None),
-                    getattr(beta, '_ps_flat_grad', None))
+        ctx.gtgt = (getattr(gamma, '_ps_flat_grad_fn', None),
+                    getattr(beta, '_ps_flat_grad_fn', None))
         return y
 
     @staticmethod
@@ -67,7 +67,8 @@ class _FusedBNFn(torch.autograd.Function):
         M = N * H * W
         dx = torch.empty_like(x)
         dres = torch.empty_like(x) if ctx.has_res else None
-        g_tgt, b_tgt = ctx.gtgt
+        g_tgt = ctx.gtgt[0]() if ctx.gtgt[0] is not None else None
+        b_tgt = ctx.gtgt[1]() if ctx.gtgt[1] is not None else None
         dgamma = (g_tgt if g_tgt is not None and g_tgt.dtype == gamma.dtype
                   and g_tgt.is_cuda else torch.empty_like(gamma))
         dbeta = (b_tgt if b_tgt is not None and b_tgt.dtype == gamma.dtype

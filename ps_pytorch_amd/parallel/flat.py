@@ -164,20 +164,36 @@ class FlatSpace:
         by pointer and copies, so kill-switches (PS_CONV=0 …) stay correct.
         """
         self._steal = steal
-        self._gviews = []
         self._gptrs = []
+        elem = self.flat_g.element_size()
         for pid, (p, off) in enumerate(zip(self.params, self.offsets)):
-            v = self._view(self.flat_g, p, off, pid)
-            self._gviews.append(v)
-            self._gptrs.append(v.data_ptr())
-            p._ps_flat_grad = v
-            p.grad = None if steal else v
+            self._gptrs.append(self.flat_g.data_ptr() + off * elem)
+            # a FACTORY, not a held view: AccumulateGrad only steals a grad
+            # whose use_count is 1 at accumulation time — any live Python
+            # reference to the view forces a clone (one copyBuffer per param
+            # per step, measured). Ops call this per backward and let the
+            # fresh view die with the autograd edge.
+            p._ps_flat_grad_fn = self._view_factory(pid)
+            p.grad = None if steal else self._view(self.flat_g, p, off, pid)
 
     def zero_grads(self) -> None:
         self.flat_g.zero_()
         if getattr(self, '_steal', False):
             for p in self.params:
                 p.grad = None
+
+    def _view_factory(self, pid: int):
+        flat_g, off, p = self.flat_g, self.offsets[pid], self.params[pid]
+        shape, cl = tuple(p.shape), self._cl[pid]
+        numel = p.numel()
+
+        def make():
+            sl = flat_g[off:off + numel]
+            if cl:
+                O, I, H, W = shape
+                return sl.view(O, H, W, I).permute(0, 3, 1, 2)
+            return sl.view(shape)
+        return make
 
     def ensure_grad_in_flat(self, pid: int) -> None:
         """Steal mode: if this param's grad landed outside flat_g (a torch
@@ -194,7 +210,7 @@ class FlatSpace:
             import warnings
             warnings.warn("grad for a parameter landed outside flat_g "
                           "(torch-fallback op?); copying per step")
-        self._gviews[pid].copy_(g)
+        self._view(self.flat_g, p, self.offsets[pid], pid).copy_(g)
 
     def harvest_grads(self) -> None:
         """Steal mode: make sure every param's grad is in flat_g."""

@@ -33,7 +33,7 @@ def test_steal_harvest_cpu_fallback_matches_view_mode():
     fs = FlatSpace(net)
     fs.attach_grads(steal=True)
     for p in fs.params:
-        assert p.grad is None and hasattr(p, '_ps_flat_grad')
+        assert p.grad is None and hasattr(p, '_ps_flat_grad_fn')
     loss = F.cross_entropy(net(x), y)
     loss.backward()
     # grads landed in autograd-owned tensors, flat_g still zero
