@@ -62,7 +62,7 @@ def _try_load() -> None:
         lib.ps_fused_adam.argtypes = [ctypes.c_void_p] * 6 + [
             ctypes.c_long] + [ctypes.c_float] * 7 + [
             ctypes.c_int, ctypes.c_int, ctypes.c_void_p]
-        lib.ps_bn_fwd.argtypes = [ctypes.c_void_p] * 11 + [
+        lib.ps_bn_fwd.argtypes = [ctypes.c_void_p] * 12 + [
             ctypes.c_long, ctypes.c_long, ctypes.c_float, ctypes.c_float,
             ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p]
         lib.ps_bn_bwd.argtypes = [ctypes.c_void_p] * 12 + [

@@ -83,8 +83,11 @@ __global__ __launch_bounds__(256) void bn_stats_kernel(
     const int R = 256 / G;                  // rows per block sub-iteration
     const int cg = threadIdx.x % G;
     const int row_in_blk = threadIdx.x / G;
-    __shared__ float s_sum[256][8];
-    __shared__ float s_sq[256][8];
+    // [256][9]: the +1 pad breaks the tid-stride-8-dword pattern that put
+    // every 32-lane store group on 4 banks (8-way; SQ_LDS_BANK_CONFLICT
+    // measured 45% of this kernel's LDS cycles)
+    __shared__ float s_sum[256][9];
+    __shared__ float s_sq[256][9];
     f32x8 sum = {0, 0, 0, 0, 0, 0, 0, 0};
     f32x8 sq = {0, 0, 0, 0, 0, 0, 0, 0};
     const long chunk = 4L * R;
@@ -218,10 +221,14 @@ __global__ __launch_bounds__(256) void bn_eval_finalize_kernel(
 }
 
 // ---------------- k3: normalize (+residual)(+relu) ----------------
+// mask: bit-packed relu activity (bit k of byte [r*G+cg] = channel cg*8+k
+// pre-clamp value > 0), written in training so the backward reads 1/16th of
+// a tensor instead of re-reading y for the mask (two full passes saved).
 template <typename T, bool RELU, bool RES>
 __global__ __launch_bounds__(256) void bn_normalize_kernel(
     const T* __restrict__ x, T* __restrict__ y, const T* __restrict__ res,
-    const float* __restrict__ ws, long M, int C)
+    const float* __restrict__ ws, unsigned char* __restrict__ mask,
+    long M, int C)
 {
     // fixed channel-octet per thread: scale/shift loaded ONCE, rows strided
     // (the i%G form re-read 64 B of coefficients per 16 B of payload)
@@ -242,9 +249,17 @@ __global__ __launch_bounds__(256) void bn_normalize_kernel(
                 f32x8 v = VecIO<T>::load(x + o);
                 v = v * scale + shift;
                 if constexpr (RES) v += VecIO<T>::load(res + o);
-                if constexpr (RELU)
+                if constexpr (RELU) {
+                    if (mask) {
+                        unsigned char mb = 0;
+#pragma unroll
+                        for (int k = 0; k < 8; ++k)
+                            mb |= (v[k] > 0.f) << k;
+                        mask[r * G + cg] = mb;
+                    }
 #pragma unroll
                     for (int k = 0; k < 8; ++k) v[k] = fmaxf(v[k], 0.f);
+                }
                 VecIO<T>::store(y + o, v);
             }
         }
@@ -254,7 +269,8 @@ __global__ __launch_bounds__(256) void bn_normalize_kernel(
 // ---------------- k4: backward reduce ----------------
 template <typename T, bool RELU>
 __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
-    const T* __restrict__ x, const T* __restrict__ y, const T* __restrict__ dy,
+    const T* __restrict__ x, const unsigned char* __restrict__ mask,
+    const T* __restrict__ dy,
     const float* __restrict__ save_mean, const float* __restrict__ save_invstd,
     float* __restrict__ partial, long M, int C)
 {
@@ -262,8 +278,8 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
     const int R = 256 / G;
     const int cg = threadIdx.x % G;
     const int row_in_blk = threadIdx.x / G;
-    __shared__ float s_dy[256][8];
-    __shared__ float s_dyx[256][8];
+    __shared__ float s_dy[256][9];   // +1 pad: see bn_stats_kernel
+    __shared__ float s_dyx[256][9];
     f32x8 mean = *(const f32x8*)&save_mean[cg * 8];
     f32x8 invstd = *(const f32x8*)&save_invstd[cg * 8];
     f32x8 sum_dy = {0, 0, 0, 0, 0, 0, 0, 0};
@@ -280,9 +296,10 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
                 f32x8 d = VecIO<T>::load(dy + o);
                 f32x8 xv = VecIO<T>::load(x + o);
                 if constexpr (RELU) {
-                    f32x8 yo = VecIO<T>::load(y + o);
+                    unsigned char mb = mask[r * G + cg];
 #pragma unroll
-                    for (int k = 0; k < 8; ++k) d[k] = yo[k] > 0.f ? d[k] : 0.f;
+                    for (int k = 0; k < 8; ++k)
+                        d[k] = (mb >> k) & 1 ? d[k] : 0.f;
                 }
                 sum_dy += d;
                 sum_dyx += d * (xv - mean) * invstd;
@@ -313,7 +330,8 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
 // ---------------- k6: backward dx (+dres) ----------------
 template <typename T, bool RELU, bool DRES>
 __global__ __launch_bounds__(256) void bn_bwd_dx_kernel(
-    const T* __restrict__ x, const T* __restrict__ y, const T* __restrict__ dy,
+    const T* __restrict__ x, const unsigned char* __restrict__ mask,
+    const T* __restrict__ dy,
     T* __restrict__ dx, T* __restrict__ dres,
     const float* __restrict__ save_mean, const float* __restrict__ ws,
     long M, int C)
@@ -337,9 +355,10 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_kernel(
                 long o = r * C + cg * 8;
                 f32x8 d = VecIO<T>::load(dy + o);
                 if constexpr (RELU) {
-                    f32x8 yo = VecIO<T>::load(y + o);
+                    unsigned char mb = mask[r * G + cg];
 #pragma unroll
-                    for (int k = 0; k < 8; ++k) d[k] = yo[k] > 0.f ? d[k] : 0.f;
+                    for (int k = 0; k < 8; ++k)
+                        d[k] = (mb >> k) & 1 ? d[k] : 0.f;
                 }
                 if constexpr (DRES) VecIO<T>::store(dres + o, d);
                 f32x8 xv = VecIO<T>::load(x + o);
@@ -362,7 +381,8 @@ static inline int stats_blocks(long M, int C, int unroll) {
 template <typename T, typename PT>
 static void bn_fwd_t(const void* x, void* y, const void* gamma, const void* beta,
                      void* rmean, void* rvar, void* smean, void* sinvstd,
-                     void* ws, void* partial, const void* res, long M, long C,
+                     void* ws, void* partial, const void* res, void* mask,
+                     long M, long C,
                      float momentum, float eps, int training, int relu,
                      hipStream_t s)
 {
@@ -387,36 +407,36 @@ static void bn_fwd_t(const void* x, void* y, const void* gamma, const void* beta
     if (blocks > 2048) blocks = 2048;
     if (relu) {
         if (res) hipLaunchKernelGGL((bn_normalize_kernel<T, true, true>), dim3(blocks), b256, 0, s,
-                                    (const T*)x, (T*)y, (const T*)res, wsf, M, Ci);
+                                    (const T*)x, (T*)y, (const T*)res, wsf, (unsigned char*)mask, M, Ci);
         else     hipLaunchKernelGGL((bn_normalize_kernel<T, true, false>), dim3(blocks), b256, 0, s,
-                                    (const T*)x, (T*)y, nullptr, wsf, M, Ci);
+                                    (const T*)x, (T*)y, nullptr, wsf, (unsigned char*)mask, M, Ci);
     } else {
         if (res) hipLaunchKernelGGL((bn_normalize_kernel<T, false, true>), dim3(blocks), b256, 0, s,
-                                    (const T*)x, (T*)y, (const T*)res, wsf, M, Ci);
+                                    (const T*)x, (T*)y, (const T*)res, wsf, nullptr, M, Ci);
         else     hipLaunchKernelGGL((bn_normalize_kernel<T, false, false>), dim3(blocks), b256, 0, s,
-                                    (const T*)x, (T*)y, nullptr, wsf, M, Ci);
+                                    (const T*)x, (T*)y, nullptr, wsf, nullptr, M, Ci);
     }
 }
 
 extern "C" void ps_bn_fwd(const void* x, void* y, const void* gamma,
                           const void* beta, void* rmean, void* rvar,
                           void* smean, void* sinvstd, void* ws, void* partial,
-                          const void* res,
+                          const void* res, void* mask,
                           long M, long C, float momentum, float eps,
                           int training, int relu, int dtype, void* stream)
 {
     hipStream_t s = (hipStream_t)stream;
     if (dtype == PS_BF16)
         bn_fwd_t<unsigned short, unsigned short>(x, y, gamma, beta, rmean, rvar,
-                                                 smean, sinvstd, ws, partial, res, M, C,
+                                                 smean, sinvstd, ws, partial, res, mask, M, C,
                                                  momentum, eps, training, relu, s);
     else
         bn_fwd_t<float, float>(x, y, gamma, beta, rmean, rvar, smean, sinvstd,
-                               ws, partial, res, M, C, momentum, eps, training, relu, s);
+                               ws, partial, res, mask, M, C, momentum, eps, training, relu, s);
 }
 
 template <typename T, typename PT>
-static void bn_bwd_t(const void* x, const void* y, const void* dy,
+static void bn_bwd_t(const void* x, const void* mask, const void* dy,
                      const void* gamma, const void* smean, const void* sinvstd,
                      void* dx, void* dgamma, void* dbeta, void* dres, void* ws,
                      void* partial, long M, long C, int relu, hipStream_t s)
@@ -427,11 +447,11 @@ static void bn_bwd_t(const void* x, const void* y, const void* dy,
     int nb = stats_blocks(M, Ci, 4);
     if (relu)
         hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, true>), dim3(nb), b256, 0, s,
-                           (const T*)x, (const T*)y, (const T*)dy,
+                           (const T*)x, (const unsigned char*)mask, (const T*)dy,
                            (const float*)smean, (const float*)sinvstd, (float*)partial, M, Ci);
     else
         hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, false>), dim3(nb), b256, 0, s,
-                           (const T*)x, (const T*)y, (const T*)dy,
+                           (const T*)x, (const unsigned char*)mask, (const T*)dy,
                            (const float*)smean, (const float*)sinvstd, (float*)partial, M, Ci);
     hipLaunchKernelGGL((bn_bwd_foldfin_kernel<PT>), dim3(Ci), b256, 0, s,
                        (const float*)partial, wsf, (const PT*)gamma,
@@ -441,22 +461,24 @@ static void bn_bwd_t(const void* x, const void* y, const void* dy,
     if (blocks > 2048) blocks = 2048;
     if (relu) {
         if (dres) hipLaunchKernelGGL((bn_bwd_dx_kernel<T, true, true>), dim3(blocks), b256, 0, s,
-                                     (const T*)x, (const T*)y, (const T*)dy, (T*)dx, (T*)dres,
+                                     (const T*)x, (const unsigned char*)mask, (const T*)dy, (T*)dx, (T*)dres,
                                      (const float*)smean, wsf, M, Ci);
         else      hipLaunchKernelGGL((bn_bwd_dx_kernel<T, true, false>), dim3(blocks), b256, 0, s,
-                                     (const T*)x, (const T*)y, (const T*)dy, (T*)dx, nullptr,
+                                     (const T*)x, (const unsigned char*)mask, (const T*)dy, (T*)dx, nullptr,
                                      (const float*)smean, wsf, M, Ci);
     } else {
         if (dres) hipLaunchKernelGGL((bn_bwd_dx_kernel<T, false, true>), dim3(blocks), b256, 0, s,
-                                     (const T*)x, (const T*)y, (const T*)dy, (T*)dx, (T*)dres,
+                                     (const T*)x, (const unsigned char*)mask, (const T*)dy, (T*)dx, (T*)dres,
                                      (const float*)smean, wsf, M, Ci);
         else      hipLaunchKernelGGL((bn_bwd_dx_kernel<T, false, false>), dim3(blocks), b256, 0, s,
-                                     (const T*)x, (const T*)y, (const T*)dy, (T*)dx, nullptr,
+                                     (const T*)x, (const unsigned char*)mask, (const T*)dy, (T*)dx, nullptr,
                                      (const float*)smean, wsf, M, Ci);
     }
 }
 
-extern "C" void ps_bn_bwd(const void* x, const void* y, const void* dy,
+// mask: the bit-packed relu activity written by ps_bn_fwd (training+relu);
+// replaces the full y re-read the backward previously did for the mask.
+extern "C" void ps_bn_bwd(const void* x, const void* mask, const void* dy,
                           const void* gamma, const void* smean,
                           const void* sinvstd, void* dx, void* dgamma,
                           void* dbeta, void* dres, void* ws, void* partial,
@@ -465,9 +487,9 @@ extern "C" void ps_bn_bwd(const void* x, const void* y, const void* dy,
 {
     hipStream_t s = (hipStream_t)stream;
     if (dtype == PS_BF16)
-        bn_bwd_t<unsigned short, unsigned short>(x, y, dy, gamma, smean, sinvstd,
+        bn_bwd_t<unsigned short, unsigned short>(x, mask, dy, gamma, smean, sinvstd,
                                                  dx, dgamma, dbeta, dres, ws, partial, M, C, relu, s);
     else
-        bn_bwd_t<float, float>(x, y, dy, gamma, smean, sinvstd, dx, dgamma,
+        bn_bwd_t<float, float>(x, mask, dy, gamma, smean, sinvstd, dx, dgamma,
                                dbeta, dres, ws, partial, M, C, relu, s);
 }

@@ -369,6 +369,65 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
     }
 
     // ---- epilogue ----
+    // V2 (NBUF==2 variants): the MFMA C-layout leaves each lane holding
+    // column fragments, so the direct write is 32-64 scalar ushort stores
+    // per lane — a store-ISSUE-bound tail (guide T21). Stage the f32 tile
+    // through the (now idle) LDS buffers, read back row-major and emit one
+    // 16-B store per 8 columns; the ACCF carry add becomes a 16-B vector
+    // load here instead of scalar gathers.
+    if constexpr (NBUF == 2) {
+        static_assert(sizeof(lds) >= TM * TN * 4, "f32 staging fits LDS");
+        float* sf = reinterpret_cast<float*>(&lds);
+#pragma unroll
+        for (int nj = 0; nj < NJ; ++nj) {
+            int cl = wn * (NJ * 16) + nj * 16 + fr;
+            float bv = (!DGRAD && bias && n0 + cl < Nout)
+                       ? bf16_to_f32(bias[n0 + cl]) : 0.f;
+#pragma unroll
+            for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+            for (int e = 0; e < 4; ++e)
+                sf[(wm * 64 + mi * 16 + fq * 4 + e) * TN + cl]
+                    = acc[mi][nj][e] + bv;
+        }
+        __syncthreads();
+        constexpr int SEG = TN / 8;        // 8-col segments per tile row
+        constexpr int RPP = 256 / SEG;     // tile rows per pass
+        const int segc = (t % SEG) * 8;
+        const int rl0 = t / SEG;
+        const bool vec_ok = (Nout & 7) == 0;   // 16-B-aligned row segments
+#pragma unroll
+        for (int ps = 0; ps < TM / RPP; ++ps) {
+            int rl = ps * RPP + rl0;
+            long om = m0 + rl;
+            if (om >= M) continue;
+            int col = n0 + segc;
+            f32x4_t a0 = *(const f32x4_t*)&sf[rl * TN + segc];
+            f32x4_t a1 = *(const f32x4_t*)&sf[rl * TN + segc + 4];
+            float v[8];
+#pragma unroll
+            for (int k2 = 0; k2 < 4; ++k2) { v[k2] = a0[k2]; v[4 + k2] = a1[k2]; }
+            if (vec_ok && col + 8 <= Nout) {
+                V16 o;
+                if constexpr (ACCF) {
+                    V16 cv = *(const V16*)&carry[om * Nout + col];
+#pragma unroll
+                    for (int k2 = 0; k2 < 8; ++k2)
+                        v[k2] += bf16_to_f32(cv.us[k2]);
+                }
+#pragma unroll
+                for (int k2 = 0; k2 < 8; ++k2) o.us[k2] = f32_to_bf16(v[k2]);
+                *(uint4*)&dst[om * Nout + col] = o.u4;
+            } else {
+                for (int k2 = 0; k2 < 8 && col + k2 < Nout; ++k2) {
+                    float vv = v[k2];
+                    if constexpr (ACCF)
+                        vv += bf16_to_f32(carry[om * Nout + col + k2]);
+                    dst[om * Nout + col + k2] = f32_to_bf16(vv);
+                }
+            }
+        }
+    } else {
 #pragma unroll
     for (int nj = 0; nj < NJ; ++nj) {
         int col = n0 + wn * (NJ * 16) + nj * 16 + fr;
@@ -385,6 +444,7 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
                 dst[om * Nout + col] = f32_to_bf16(v);
             }
         }
+    }
     }
 }
 
@@ -565,22 +625,58 @@ __global__ __launch_bounds__(256) void conv_dgrad2_kernel(
         }
     }
 
+    // V2 epilogue: LDS-staged row-major read-back + 16-B stores (see the
+    // conv_gemm_kernel epilogue comment); rows scatter to parity positions
+    // but columns stay contiguous, so the segment stores are unchanged.
+    {
+        static_assert(sizeof(lds) >= TM * TN * 4, "f32 staging fits LDS");
+        float* sf = reinterpret_cast<float*>(&lds);
 #pragma unroll
-    for (int nj = 0; nj < NJ; ++nj) {
-        int col = n0 + wn * (NJ * 16) + nj * 16 + fr;
+        for (int nj = 0; nj < NJ; ++nj) {
+            int cl = wn * (NJ * 16) + nj * 16 + fr;
 #pragma unroll
-        for (int mi = 0; mi < 4; ++mi)
+            for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-        for (int e = 0; e < 4; ++e) {
-            long om = m0 + wm * 64 + mi * 16 + fq * 4 + e;
-            if (om < M && col < C) {
-                int n = (int)(om / ((long)Hc * Wc));
-                int rem = (int)(om % ((long)Hc * Wc));
-                int h = hp + 2 * (rem / Wc), w = wp + 2 * (rem % Wc);
-                long oi = ((long)(n * H + h) * W + w) * C + col;
-                float v = acc[mi][nj][e];
-                if constexpr (ACCF) v += bf16_to_f32(carry[oi]);
-                dx[oi] = f32_to_bf16(v);
+            for (int e = 0; e < 4; ++e)
+                sf[(wm * 64 + mi * 16 + fq * 4 + e) * TN + cl] = acc[mi][nj][e];
+        }
+        __syncthreads();
+        constexpr int SEG = TN / 8;
+        constexpr int RPP = 256 / SEG;
+        const int segc = (t % SEG) * 8;
+        const int rl0 = t / SEG;
+        const bool vec_ok = (C & 7) == 0;
+#pragma unroll
+        for (int ps = 0; ps < TM / RPP; ++ps) {
+            int rl = ps * RPP + rl0;
+            long om = m0 + rl;
+            if (om >= M) continue;
+            int col = n0 + segc;
+            int n = (int)(om / ((long)Hc * Wc));
+            int rem = (int)(om % ((long)Hc * Wc));
+            int h = hp + 2 * (rem / Wc), w = wp + 2 * (rem % Wc);
+            long oi = ((long)(n * H + h) * W + w) * C + col;
+            f32x4_t a0 = *(const f32x4_t*)&sf[rl * TN + segc];
+            f32x4_t a1 = *(const f32x4_t*)&sf[rl * TN + segc + 4];
+            float v[8];
+#pragma unroll
+            for (int k2 = 0; k2 < 4; ++k2) { v[k2] = a0[k2]; v[4 + k2] = a1[k2]; }
+            if (vec_ok && col + 8 <= C) {
+                if constexpr (ACCF) {
+                    V16 cv = *(const V16*)&carry[oi];
+#pragma unroll
+                    for (int k2 = 0; k2 < 8; ++k2) v[k2] += bf16_to_f32(cv.us[k2]);
+                }
+                V16 o;
+#pragma unroll
+                for (int k2 = 0; k2 < 8; ++k2) o.us[k2] = f32_to_bf16(v[k2]);
+                *(uint4*)&dx[oi] = o.u4;
+            } else {
+                for (int k2 = 0; k2 < 8 && col + k2 < C; ++k2) {
+                    float vv = v[k2];
+                    if constexpr (ACCF) vv += bf16_to_f32(carry[oi + k2]);
+                    dx[oi + k2] = f32_to_bf16(vv);
+                }
             }
         }
     }

@@ -42,15 +42,21 @@ class _FusedBNFn(torch.autograd.Function):
         save_invstd = torch.empty(C, dtype=torch.float32, device=x.device)
         ws = torch.empty(5 * C, dtype=torch.float32, device=x.device)
         partial = torch.empty(1024 * 2 * C, dtype=torch.float32, device=x.device)
+        # bit-packed relu mask (1 byte per 8 channels): the backward reads
+        # this instead of re-reading the full y tensor for the mask
+        mask = (torch.empty(M * (C // 8), dtype=torch.uint8, device=x.device)
+                if relu else None)
         lib.ps_bn_fwd(
             x.data_ptr(), y.data_ptr(), gamma.data_ptr(), beta.data_ptr(),
             running_mean.data_ptr(), running_var.data_ptr(),
             save_mean.data_ptr(), save_invstd.data_ptr(), ws.data_ptr(),
             partial.data_ptr(),
             residual.data_ptr() if residual is not None else 0,
+            mask.data_ptr() if mask is not None else 0,
             M, C, float(momentum), float(eps), 1, int(relu),
             dtype_tag(x.dtype), current_stream_ptr())
-        ctx.save_for_backward(x, y, gamma, save_mean, save_invstd)
+        ctx.save_for_backward(x, gamma, save_mean, save_invstd)
+        ctx.mask = mask
         ctx.relu = relu
         ctx.has_res = residual is not None
         # steal-mode flat_g targets (see parallel/flat.py attach_grads)
@@ -61,7 +67,8 @@ class _FusedBNFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         lib = require_lib()
-        x, y, gamma, save_mean, save_invstd = ctx.saved_tensors
+        x, gamma, save_mean, save_invstd = ctx.saved_tensors
+        mask = ctx.mask
         dy = dy.contiguous(memory_format=_CL)
         N, C, H, W = x.shape
         M = N * H * W
@@ -76,7 +83,8 @@ class _FusedBNFn(torch.autograd.Function):
         ws = torch.empty(5 * C, dtype=torch.float32, device=x.device)
         partial = torch.empty(1024 * 2 * C, dtype=torch.float32, device=x.device)
         lib.ps_bn_bwd(
-            x.data_ptr(), y.data_ptr(), dy.data_ptr(), gamma.data_ptr(),
+            x.data_ptr(), mask.data_ptr() if mask is not None else 0,
+            dy.data_ptr(), gamma.data_ptr(),
             save_mean.data_ptr(), save_invstd.data_ptr(), dx.data_ptr(),
             dgamma.data_ptr(), dbeta.data_ptr(),
             dres.data_ptr() if dres is not None else 0, ws.data_ptr(),
@@ -96,7 +104,7 @@ def _bn_eval_fused(x, gamma, beta, residual, rmean, rvar, eps, relu):
     lib.ps_bn_fwd(x.data_ptr(), y.data_ptr(), gamma.data_ptr(),
                   beta.data_ptr(), rmean.data_ptr(), rvar.data_ptr(),
                   0, 0, ws.data_ptr(), 0,
-                  residual.data_ptr() if residual is not None else 0,
+                  residual.data_ptr() if residual is not None else 0, 0,
                   N * H * W, C, 0.0, float(eps), 0, int(relu),
                   dtype_tag(x.dtype), current_stream_ptr())
     return y

@@ -173,8 +173,15 @@ class FlatSpace:
             # reference to the view forces a clone (one copyBuffer per param
             # per step, measured). Ops call this per backward and let the
             # fresh view die with the autograd edge.
-            p._ps_flat_grad_fn = self._view_factory(pid)
-            p.grad = None if steal else self._view(self.flat_g, p, off, pid)
+            if steal:
+                p._ps_flat_grad_fn = self._view_factory(pid)
+                p.grad = None
+            else:
+                # view mode: autograd accumulates INTO the view, so ops must
+                # NOT also write the slice (p.grad += view-of-itself = 2x)
+                if hasattr(p, '_ps_flat_grad_fn'):
+                    del p._ps_flat_grad_fn
+                p.grad = self._view(self.flat_g, p, off, pid)
 
     def zero_grads(self) -> None:
         self.flat_g.zero_()
