@@ -1344,18 +1344,52 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
     }
 }
 
-// deterministic slab reduce: dw[e] = sum_s partial[s][e] (fixed order);
-// float4 lanes — the scalar version ran ~18x off stream rate.
+// Stage-1 slab reduce for large split counts: the final fold kernel's
+// parallelism is capped by OUTPUT elements (K*RSC/4 lanes — 36 blocks for
+// ResNet-18 l1, 0.14 blocks/CU) while its work is split*n reads, which
+// left it ~7x off stream rate at split~170. Each (block.x, group=block.y)
+// sums its group's slabs in registers and stores into the group's first
+// slab (disjoint read/write sets per group: deterministic, in place).
+__global__ __launch_bounds__(256) void reduce_slabs_part_kernel(
+    float* __restrict__ partial, long n, int nslab, int per)
+{
+    EW_IDX
+    const int g = blockIdx.y;
+    const int s0 = g * per;
+    const int s1 = (s0 + per < nslab) ? s0 + per : nslab;
+    if (s0 >= nslab) return;
+    long nv = n >> 2;
+    for (long i = gid; i < nv; i += stride) {
+        f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
+        for (int sI = s0; sI < s1; ++sI) {
+            f32x4_t v = *(const f32x4_t*)(partial + (long)sI * n + i * 4);
+#pragma unroll
+            for (int k = 0; k < 4; ++k) acc[k] += v[k];
+        }
+        *(f32x4_t*)(partial + (long)s0 * n + i * 4) = acc;
+    }
+    if (gid == 0 && g == 0)
+        for (long i = nv << 2; i < n; ++i) {
+            // scalar tail: fold EVERY slab here (groups would race on the
+            // same tail words otherwise); the fold stage reads slab 0 only
+            float acc = 0.f;
+            for (int sI = 0; sI < nslab; ++sI) acc += partial[(long)sI * n + i];
+            partial[i] = acc;
+        }
+}
+
+// deterministic slab reduce: dw[e] = sum_s partial[s*stride_][e] (fixed
+// order); float4 lanes — the scalar version ran ~18x off stream rate.
 __global__ __launch_bounds__(256) void reduce_slabs_kernel(
     unsigned short* __restrict__ dw, const float* __restrict__ partial,
-    long n, int nslab)
+    long n, int nslab, int sstride, int tail_ready)
 {
     EW_IDX
     long nv = n >> 2;
     for (long i = gid; i < nv; i += stride) {
         f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
         for (int s = 0; s < nslab; ++s) {
-            f32x4_t v = *(const f32x4_t*)(partial + (long)s * n + i * 4);
+            f32x4_t v = *(const f32x4_t*)(partial + (long)s * sstride * n + i * 4);
 #pragma unroll
             for (int k = 0; k < 4; ++k) acc[k] += v[k];
         }
@@ -1366,10 +1400,34 @@ __global__ __launch_bounds__(256) void reduce_slabs_kernel(
     }
     if (gid == 0)
         for (long i = nv << 2; i < n; ++i) {
+            if (tail_ready) { dw[i] = f32_to_bf16(partial[i]); continue; }
             float acc = 0.f;
-            for (int s = 0; s < nslab; ++s) acc += partial[(long)s * n + i];
+            for (int s = 0; s < nslab; ++s)
+                acc += partial[(long)s * sstride * n + i];
             dw[i] = f32_to_bf16(acc);
         }
+}
+
+// two-stage launch helper: split > 8 goes through the filled stage-1 pass
+static void launch_reduce_slabs(unsigned short* dw, float* partial, long n,
+                                int split, hipStream_t strm)
+{
+    int blocks;
+    if (split > 8) {
+        const int groups = 8;
+        const int per = (split + groups - 1) / groups;
+        const int ngroups = (split + per - 1) / per;
+        ew_grid(n >> 2, 256, &blocks);
+        hipLaunchKernelGGL(reduce_slabs_part_kernel,
+                           dim3(blocks, ngroups), dim3(256), 0, strm,
+                           partial, n, split, per);
+        hipLaunchKernelGGL(reduce_slabs_kernel, dim3(blocks), dim3(256), 0,
+                           strm, dw, partial, n, ngroups, per, 1);
+        return;
+    }
+    ew_grid(n >> 2, 256, &blocks);
+    hipLaunchKernelGGL(reduce_slabs_kernel, dim3(blocks), dim3(256), 0, strm,
+                       dw, partial, n, split, 1, 0);
 }
 
 // column sum for bias grad: db[k] = sum_m dout[m][k], two-stage —
@@ -1588,10 +1646,8 @@ extern "C" void ps_conv_wgrad(
         else             { if (pw) WGS(2, true); else WGS(2, false); }
 #undef WGS
         long n_ = (long)K * R * S * C;
-        int blocks_; ew_grid(n_, 256, &blocks_);
-        hipLaunchKernelGGL(reduce_slabs_kernel, dim3(blocks_), dim3(256), 0,
-                           (hipStream_t)strm, (unsigned short*)dw,
-                           (const float*)partial_f32, n_, split);
+        launch_reduce_slabs((unsigned short*)dw, (float*)partial_f32, n_,
+                            split, (hipStream_t)strm);
         return;
     }
     // 3x3 s1 pad1 family -> row-halo kernel (see conv_wgrad_row_kernel);
@@ -1621,10 +1677,8 @@ extern "C" void ps_conv_wgrad(
             if (alr) WGR(64, true); else WGR(64, false);
 #undef WGR
             long n_ = (long)K * 9 * C;
-            int blocks_; ew_grid(n_ / 4, 256, &blocks_);
-            hipLaunchKernelGGL(reduce_slabs_kernel, dim3(blocks_), dim3(256),
-                               0, (hipStream_t)strm, (unsigned short*)dw,
-                               (const float*)partial_f32, n_, split);
+            launch_reduce_slabs((unsigned short*)dw, (float*)partial_f32, n_,
+                                split, (hipStream_t)strm);
             return;
         }
     }
@@ -1652,10 +1706,8 @@ extern "C" void ps_conv_wgrad(
 #undef WG_TK
 #undef WG_LAUNCH
     long n = (long)K * R * S * C;
-    int blocks; ew_grid(n, 256, &blocks);
-    hipLaunchKernelGGL(reduce_slabs_kernel, dim3(blocks), dim3(256), 0,
-                       (hipStream_t)strm, (unsigned short*)dw,
-                       (const float*)partial_f32, n, split);
+    launch_reduce_slabs((unsigned short*)dw, (float*)partial_f32, n, split,
+                        (hipStream_t)strm);
 }
 
 // partial_f32 must hold 512*K floats.
