@@ -717,6 +717,18 @@ __device__ __forceinline__ int pswz(int r) {
     return r ^ (((r >> 3) & 1) << 2);
 }
 
+// Lemire fast division for the non-pow2 pixel decode (ResNet-50's
+// Q=56/28/14/7): q = floor(n/d) = umul64hi(n, floor(2^64/d)+1), exact for
+// all 32-bit n and d > 1. The POW2=false wgrad kernels ran ~2 full divide
+// sequences per thread per 64-pixel step (~50 VALU against 16 MFMA).
+__device__ __forceinline__ unsigned fdiv_u32(unsigned n,
+                                             unsigned long long magic) {
+    return magic ? (unsigned)__umul64hi((unsigned long long)n, magic) : n;
+}
+static inline unsigned long long fdiv_magic(long d) {
+    return d > 1 ? (~0ULL) / (unsigned long long)d + 1 : 0ULL;
+}
+
 typedef __bf16 bf16x4_t __attribute__((ext_vector_type(4)));
 __device__ __forceinline__ bf16x4_t ds_tr16p(const unsigned short* p) {
     typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 v4;
@@ -736,7 +748,8 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
     float* __restrict__ partial,              // [SPLIT][K][R*S*C]
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int pad, int split, int chunk,
-    int l2pq, int l2q, int per_xcd)
+    int l2pq, int l2q, int per_xcd,
+    unsigned long long mpq, unsigned long long mq)
 {
     __shared__ __attribute__((aligned(16))) WgradLds<TK> lds;
     constexpr int KH = TK / 64;        // 64-wide k sub-chunks per tile
@@ -794,9 +807,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_kernel(
                     int rem = (int)m & ((1 << l2pq) - 1);
                     p = rem >> l2q; q = rem & ((1 << l2q) - 1);
                 } else {
-                    n = (int)(m / ((long)P * Q));
-                    int rem = (int)(m % ((long)P * Q));
-                    p = rem / Q; q = rem % Q;
+                    n = (int)fdiv_u32((unsigned)m, mpq);
+                    unsigned rem = (unsigned)m - (unsigned)n * (unsigned)(P * Q);
+                    p = (int)fdiv_u32(rem, mq);
+                    q = (int)(rem - (unsigned)p * (unsigned)Q);
                 }
 #pragma unroll
                 for (int h = 0; h < KH; ++h)
@@ -1171,7 +1185,8 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
     float* __restrict__ partial,              // [SPLIT][K][R*S*C]
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int pad, int split, int chunk,
-    int l2pq, int l2q)
+    int l2pq, int l2q,
+    unsigned long long mpq, unsigned long long mq)
 {
     __shared__ __attribute__((aligned(16))) WgradLds<64> lds;
     const long M = (long)Nb * P * Q;
@@ -1224,9 +1239,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
                     int rem = (int)m & ((1 << l2pq) - 1);
                     p = rem >> l2q; q = rem & ((1 << l2q) - 1);
                 } else {
-                    n = (int)(m / ((long)P * Q));
-                    int rem = (int)(m % ((long)P * Q));
-                    p = rem / Q; q = rem % Q;
+                    n = (int)fdiv_u32((unsigned)m, mpq);
+                    unsigned rem = (unsigned)m - (unsigned)n * (unsigned)(P * Q);
+                    p = (int)fdiv_u32(rem, mq);
+                    q = (int)(rem - (unsigned)p * (unsigned)Q);
                 }
                 av = load16<AL>(dout + m * K + k0 + cc8, k0 + cc8, K);
                 int h0 = p * STRIDE - pad, w0 = q * STRIDE - pad;
@@ -1636,12 +1652,14 @@ extern "C" void ps_conv_wgrad(
         long grid_s = (long)tiles_k * nc_s * split;
         int l2pq_ = ilog2_exact((long)P * Q), l2q_ = ilog2_exact(Q);
         bool pw = l2pq_ >= 0 && l2q_ >= 0;
+        unsigned long long mpq_ = fdiv_magic((long)P * Q);
+        unsigned long long mq_ = fdiv_magic(Q);
 #define WGS(ST, PW)                                                           \
         hipLaunchKernelGGL((conv_wgrad_small_kernel<ST, PW, false>),          \
             dim3((unsigned)grid_s), dim3(256), 0, (hipStream_t)strm,          \
             (const unsigned short*)dout, (const unsigned short*)in,           \
             (float*)partial_f32, Nb, H, W, C, K, P, Q, R, S, pad, split,      \
-            chunk, l2pq_, l2q_)
+            chunk, l2pq_, l2q_, mpq_, mq_)
         if (stride == 1) { if (pw) WGS(1, true); else WGS(1, false); }
         else             { if (pw) WGS(2, true); else WGS(2, false); }
 #undef WGS
@@ -1689,12 +1707,14 @@ extern "C" void ps_conv_wgrad(
     long grid = (long)per_xcd * 8;
     int l2pq = ilog2_exact((long)P * Q), l2q = ilog2_exact(Q);
     bool pow2 = l2pq >= 0 && l2q >= 0;
+    unsigned long long mpq = fdiv_magic((long)P * Q);
+    unsigned long long mq = fdiv_magic(Q);
 #define WG_LAUNCH(ST, PW, TKV)                                                \
     hipLaunchKernelGGL((conv_wgrad_kernel<ST, PW, TKV, ALV>),                 \
         dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,                \
         (const unsigned short*)dout, (const unsigned short*)in,               \
         (float*)partial_f32, Nb, H, W, C, K, P, Q, R, S, pad, split, chunk,   \
-        l2pq, l2q, per_xcd)
+        l2pq, l2q, per_xcd, mpq, mq)
 #define WG_TK(ST, PW) do { if (TK == 128) WG_LAUNCH(ST, PW, 128);             \
                            else WG_LAUNCH(ST, PW, 64); } while (0)
 #define WG_AL(ST, PW) do { if ((C & 63) == 0 && K % TK == 0) {                \

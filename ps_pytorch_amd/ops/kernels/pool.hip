@@ -63,8 +63,11 @@ __global__ __launch_bounds__(256) void maxpool_fwd_kernel(
 #pragma unroll
             for (int u = 0; u < 8; ++u) o[u] = f32_to_bf16(best[u]);
             *(ushort8_t*)yp = o;
+            unsigned long long a8 = 0;
 #pragma unroll
-            for (int u = 0; u < 8; ++u) ap[u] = (unsigned char)bidx[u];
+            for (int u = 0; u < 8; ++u)
+                a8 |= (unsigned long long)(bidx[u] & 0xff) << (8 * u);
+            *(unsigned long long*)ap = a8;
         } else {
             for (int u = 0; u < cw; ++u) {
                 yp[u] = f32_to_bf16(best[u]);
@@ -109,8 +112,19 @@ __global__ __launch_bounds__(256) void maxpool_bwd_kernel(
                 const unsigned char slot = (unsigned char)(r * S + s);
                 const unsigned char* ap = arg + opix * C + cb;
                 const unsigned short* gp = dy + opix * C + cb;
-                for (int u = 0; u < cw; ++u)
-                    if (ap[u] == slot) acc[u] += bf16_to_f32(gp[u]);
+                if (cw == 8) {
+                    // one 8-B load instead of 8 scalar byte loads (this
+                    // kernel was 16x off stream rate on the stem pool)
+                    unsigned long long a8 = *(const unsigned long long*)ap;
+                    ushort8_t g8 = *(const ushort8_t*)gp;
+#pragma unroll
+                    for (int u = 0; u < 8; ++u)
+                        if (((a8 >> (8 * u)) & 0xff) == slot)
+                            acc[u] += bf16_to_f32(g8[u]);
+                } else {
+                    for (int u = 0; u < cw; ++u)
+                        if (ap[u] == slot) acc[u] += bf16_to_f32(gp[u]);
+                }
             }
         }
         unsigned short* dp = dx + pix * C + cb;
