@@ -73,6 +73,24 @@ class _ConvFn(torch.autograd.Function):
         K, _, R, S = w.shape
         P = (H + 2 * pad - R) // stride + 1
         Q = (W + 2 * pad - S) // stride + 1
+        # C<=3 stems (CIFAR 3x3, R50 7x7, MNIST 5x5): pad channels to 4 so
+        # the SMALL gather runs as aligned 8-B tap loads (kernels' C4 path).
+        # The input is the dataset tensor (no dx needed), so only fwd/wgrad
+        # see the padded operands; dw is computed at C=4 and sliced back.
+        c4 = (C <= 3 and R * S > 1 and R * S * 4 <= 256
+              and not x.requires_grad)
+        ctx.c4 = c4
+        ctx.c_orig = C
+        if c4:
+            x4 = torch.empty((Nb, 4, H, W), dtype=x.dtype, device=x.device,
+                             memory_format=_CL)
+            lib.ps_pad4(x4.data_ptr(), x.data_ptr(), Nb * H * W, C,
+                        current_stream_ptr())
+            w4 = torch.empty((K, 4, R, S), dtype=w.dtype, device=w.device,
+                             memory_format=_CL)
+            lib.ps_pad4(w4.data_ptr(), wc.data_ptr(), K * R * S, C,
+                        current_stream_ptr())
+            x, wc, C = x4, w4, 4
         # NB: allocate channels_last DIRECTLY — empty().contiguous(CL) runs a
         # full transposing copy of uninitialized memory (~190us at l1 size)
         out = torch.empty((Nb, K, P, Q), dtype=x.dtype, device=x.device,
@@ -122,13 +140,26 @@ class _ConvFn(torch.autograd.Function):
             partial = torch.empty(split * K * R * S * C,
                                   dtype=torch.float32, device=x.device)
             wt_tgt = ctx.gtgt[0]() if ctx.gtgt[0] is not None else None
-            dw = (wt_tgt if wt_tgt is not None and wt_tgt.dtype == w.dtype
-                  and wt_tgt.is_cuda
-                  else torch.empty_like(w).contiguous(memory_format=_CL))
-            lib.ps_conv_wgrad(dout.data_ptr(), x.data_ptr(),
-                              partial.data_ptr(), dw.data_ptr(),
-                              Nb, H, W, C, K, P, Q, R, S, stride, pad,
-                              split, current_stream_ptr())
+            if ctx.c4:
+                # padded-channel stem: wgrad at C=4, then slice the real
+                # C_orig channels back into the (steal-target) dw
+                dw4 = torch.empty_like(w).contiguous(memory_format=_CL)
+                lib.ps_conv_wgrad(dout.data_ptr(), x.data_ptr(),
+                                  partial.data_ptr(), dw4.data_ptr(),
+                                  Nb, H, W, 4, K, P, Q, R, S, stride, pad,
+                                  split, current_stream_ptr())
+                dw = (wt_tgt if wt_tgt is not None and wt_tgt.is_cuda
+                      else torch.empty((K, ctx.c_orig, R, S), dtype=w.dtype,
+                                       device=w.device, memory_format=_CL))
+                dw.copy_(dw4[:, :ctx.c_orig])
+            else:
+                dw = (wt_tgt if wt_tgt is not None and wt_tgt.dtype == w.dtype
+                      and wt_tgt.is_cuda
+                      else torch.empty_like(w).contiguous(memory_format=_CL))
+                lib.ps_conv_wgrad(dout.data_ptr(), x.data_ptr(),
+                                  partial.data_ptr(), dw.data_ptr(),
+                                  Nb, H, W, C, K, P, Q, R, S, stride, pad,
+                                  split, current_stream_ptr())
         if has_bias and ctx.needs_input_grad[2]:
             b_tgt = ctx.gtgt[1]() if ctx.gtgt[1] is not None else None
             db = (b_tgt if b_tgt is not None and b_tgt.dtype == dout.dtype

@@ -127,9 +127,11 @@ __device__ __forceinline__ int gswz(int row, int g) {
 template <int TM, int TN, int STRIDE, bool DGRAD, bool SMALL = false, bool AL = true,
           int NBUF = 2,   // NBUF=1 only for the one-step (RSC<=64) SMALL path
           int SWZM = 2,   // LDS granule swizzle mode (see gswz)
-          bool ACCF = false>  // dgrad: epilogue adds `carry` (residual-fork
+          bool ACCF = false,  // dgrad: epilogue adds `carry` (residual-fork
                               // grad accumulation fused in — kills the
                               // autograd at::add at every block input fork)
+          bool C4 = false>    // SMALL with C==4 (padded stem): tap-aligned
+                              // 8-B gather loads instead of scalar
 __global__ __launch_bounds__(256) void conv_gemm_kernel(
     const unsigned short* __restrict__ src,  // fwd: in [Nb,H,W,C]; dgrad: dout [Nb,P,Q,K]
     const unsigned short* __restrict__ wgt,  // fwd: w [K,R,S,C]; dgrad: wT [R,S,C,K]
@@ -252,7 +254,7 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         // contraction elements (e = lc*64 + cc8+u); <=3 chunks, a handful
         // of divides each.
         int tre[SMALL ? 8 : 1], tse[SMALL ? 8 : 1], tce[SMALL ? 8 : 1];
-        if constexpr (SMALL) {
+        if constexpr (SMALL && !C4) {
 #pragma unroll
             for (int u = 0; u < 8; ++u) {
                 int e = (lc << 6) + cc8 + u;
@@ -263,7 +265,27 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         }
 #pragma unroll
         for (int rr = 0; rr < AR; ++rr) {
-            if constexpr (SMALL) {
+            if constexpr (SMALL && C4) {
+                // padded-channel stem: the 8-elem quantum is exactly two
+                // (r,s) taps of 4 channels each — two aligned 8-B loads
+                const int RSC = R * S * 4;
+                const int e0 = (lc << 6) + cc8;
+                const int t0 = e0 >> 2;
+                V16 v = zero16();
+                bool rowv = ax[rr] > INT_MIN / 4;
+#pragma unroll
+                for (int half = 0; half < 2; ++half) {
+                    int t = t0 + half;
+                    int tr = t / S, ts = t - tr * S;
+                    int hh = ax[rr] + tr, ww = ay[rr] + ts;
+                    if (rowv && e0 + 4 * half < RSC
+                        && hh >= 0 && hh < H && ww >= 0 && ww < W)
+                        *(unsigned long long*)&v.us[4 * half] =
+                            *(const unsigned long long*)
+                                (src + abase[rr] + ((long)tr * W + ts) * 4);
+                }
+                areg[rr] = v;
+            } else if constexpr (SMALL) {
                 const int RSC = R * S * C;
                 V16 v = zero16();
                 bool rowv = ax[rr] > INT_MIN / 4;
@@ -283,7 +305,18 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         }
 #pragma unroll
         for (int rr = 0; rr < BR; ++rr) {
-            if constexpr (SMALL)
+            if constexpr (SMALL && C4) {
+                const int RSC = R * S * 4;
+                const int e0 = (lc << 6) + cc8;
+                V16 v = zero16();
+#pragma unroll
+                for (int half = 0; half < 2; ++half)
+                    if (colv[rr] && e0 + 4 * half < RSC)
+                        *(unsigned long long*)&v.us[4 * half] =
+                            *(const unsigned long long*)
+                                (pB[rr] + (lc << 6) + 4 * half);
+                breg[rr] = v;
+            } else if constexpr (SMALL)
                 breg[rr] = colv[rr]
                     ? load16<AL>(pB[rr] + (lc << 6), (lc << 6) + cc8, R * S * C)
                     : zero16();
@@ -1178,7 +1211,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
 // flattened (r,s,c) axis fits one 64-column tile, so one block covers every
 // tap in a single pass over its pixel chunk — vs the generic kernel's
 // R*S separate 64-c tiles at C/64 utilization each.
-template <int STRIDE, bool POW2, bool AL = true>
+template <int STRIDE, bool POW2, bool AL = true, bool C4 = false>
 __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
     const unsigned short* __restrict__ dout,  // [Nb,P,Q,K]
     const unsigned short* __restrict__ in,    // [Nb,H,W,C]
@@ -1213,14 +1246,22 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
         for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
     // per-lane flattened (r,s,c) gather table (as the SMALL fwd path);
-    // toff = the lane's constant element offset from the pixel origin
+    // toff = the lane's constant element offset from the pixel origin.
+    // C4: two aligned 8-B tap loads per quantum instead (padded stem).
     int tre[8], tse[8], toff[8];
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
         int e = e0 + cc8 + u;
-        tre[u] = e / (S * C);
-        tse[u] = (e / C) % S;
-        toff[u] = (tre[u] * W + tse[u]) * C + e % C;
+        if constexpr (C4) {
+            int tp = e >> 2;
+            tre[u] = tp / S;
+            tse[u] = tp - tre[u] * S;
+            toff[u] = (tre[u] * W + tse[u]) * 4;
+        } else {
+            tre[u] = e / (S * C);
+            tse[u] = (e / C) % S;
+            toff[u] = (tre[u] * W + tse[u]) * C + e % C;
+        }
     }
 
     const int nsteps = (int)((mend - mbeg + 63) >> 6);
@@ -1248,12 +1289,24 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
                 int h0 = p * STRIDE - pad, w0 = q * STRIDE - pad;
                 const unsigned short* pix =
                     in + (long)n * H * W * C + ((long)h0 * W + w0) * C;
+                if constexpr (C4) {
+#pragma unroll
+                    for (int half = 0; half < 2; ++half) {
+                        int u = 4 * half;
+                        int hh = h0 + tre[u], ww = w0 + tse[u];
+                        if (e0 + cc8 + u < RSC
+                            && hh >= 0 && hh < H && ww >= 0 && ww < W)
+                            *(unsigned long long*)&bv.us[u] =
+                                *(const unsigned long long*)(pix + toff[u]);
+                    }
+                } else {
 #pragma unroll
                 for (int u = 0; u < 8; ++u) {
                     int hh = h0 + tre[u], ww = w0 + tse[u];
                     if (e0 + cc8 + u < RSC
                         && hh >= 0 && hh < H && ww >= 0 && ww < W)
                         bv.us[u] = pix[toff[u]];
+                }
                 }
             }
             areg[rr] = av; breg[rr] = bv;
@@ -1506,6 +1559,18 @@ static inline int swz_mode() {
         if (swz_mode() == 1) LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, 2, 1, ACV);   \
         else                 LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, 2, 2, ACV);   \
     } while (0)
+#define LAUNCH_GEMM_CV(TM, TN, ST, SM, NBV)                                   \
+    do {                                                                      \
+        long M_ = (long)Nb * P * Q;                                           \
+        long grid = ((M_ + TM - 1) / TM) * ((K + TN - 1) / TN);               \
+        hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, ST, false, SM, ALV, NBV, \
+                                             2, false, true>),               \
+            dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,            \
+            (const unsigned short*)src, (const unsigned short*)wgt,           \
+            (const unsigned short*)bias, (unsigned short*)dst,                \
+            (const unsigned short*)carry,                                     \
+            Nb, H, W, C, K, P, Q, R, S, pad);                                 \
+    } while (0)
 #define LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, NBV, SWV, ACV)                     \
     do {                                                                      \
         long M_ = DG ? (long)Nb * H * W : (long)Nb * P * Q;                   \
@@ -1528,7 +1593,13 @@ extern "C" void ps_conv_fwd(
     bool al = (C & 63) == 0;           // full 64-chunk contraction coverage
 #define FWD_BODY()                                                            \
     do {                                                                      \
-        if (R * S > 1 && R * S * C <= 64) {     /* one-step: single buffer */\
+        if (R * S > 1 && C == 4 && R * S * 4 <= 64) {   /* padded stem */     \
+            if (stride == 1) LAUNCH_GEMM_CV(128, 64, 1, true, 1);             \
+            else             LAUNCH_GEMM_CV(128, 64, 2, true, 1);             \
+        } else if (R * S > 1 && C == 4 && R * S * 4 <= 256) {                 \
+            if (stride == 1) LAUNCH_GEMM_CV(128, 64, 1, true, 2);             \
+            else             LAUNCH_GEMM_CV(128, 64, 2, true, 2);             \
+        } else if (R * S > 1 && R * S * C <= 64) {  /* one-step: 1 buffer */  \
             if (stride == 1) LAUNCH_GEMM_NB(128, 64, 1, false, true, 1);      \
             else             LAUNCH_GEMM_NB(128, 64, 2, false, true, 1);      \
         } else if (R * S > 1 && R * S * C <= 192) {                           \
@@ -1646,7 +1717,8 @@ extern "C" void ps_conv_wgrad(
     long M = (long)Nb * P * Q;
     long chunk64 = (M + (long)split * 64 - 1) / ((long)split * 64);
     int chunk = (int)(chunk64 * 64);
-    if (R * S > 1 && R * S * C <= 192) {    // flattened stem/LeNet path
+    if (R * S > 1 && ((C == 4 && R * S * 4 <= 256) || R * S * C <= 192)) {
+        // flattened stem/LeNet path (C==4 = padded-channel stem)
         int tiles_k = (K + 63) / 64;
         int nc_s = (R * S * C + 63) / 64;
         long grid_s = (long)tiles_k * nc_s * split;
@@ -1654,14 +1726,19 @@ extern "C" void ps_conv_wgrad(
         bool pw = l2pq_ >= 0 && l2q_ >= 0;
         unsigned long long mpq_ = fdiv_magic((long)P * Q);
         unsigned long long mq_ = fdiv_magic(Q);
-#define WGS(ST, PW)                                                           \
-        hipLaunchKernelGGL((conv_wgrad_small_kernel<ST, PW, false>),          \
+#define WGS(ST, PW, CV)                                                       \
+        hipLaunchKernelGGL((conv_wgrad_small_kernel<ST, PW, false, CV>),      \
             dim3((unsigned)grid_s), dim3(256), 0, (hipStream_t)strm,          \
             (const unsigned short*)dout, (const unsigned short*)in,           \
             (float*)partial_f32, Nb, H, W, C, K, P, Q, R, S, pad, split,      \
             chunk, l2pq_, l2q_, mpq_, mq_)
-        if (stride == 1) { if (pw) WGS(1, true); else WGS(1, false); }
-        else             { if (pw) WGS(2, true); else WGS(2, false); }
+        if (C == 4) {
+            if (stride == 1) { if (pw) WGS(1, true, true); else WGS(1, false, true); }
+            else             { if (pw) WGS(2, true, true); else WGS(2, false, true); }
+        } else {
+            if (stride == 1) { if (pw) WGS(1, true, false); else WGS(1, false, false); }
+            else             { if (pw) WGS(2, true, false); else WGS(2, false, false); }
+        }
 #undef WGS
         long n_ = (long)K * R * S * C;
         launch_reduce_slabs((unsigned short*)dw, (float*)partial_f32, n_,

@@ -121,3 +121,31 @@ def test_conv_carry_fuses_residual_grad(shape):
 
     assert _rel_err(x.grad, xr.grad) < 0.03, f"dx+carry {_rel_err(x.grad, xr.grad)}"
     assert _rel_err(w.grad, wr.grad) < 0.03
+
+
+@pytest.mark.parametrize("shape4", [
+    (8, 3, 32, 32, 64, 3, 1, 1),     # CIFAR stem (RSC4=36, one-step)
+    (4, 3, 64, 64, 64, 7, 2, 3),     # R50-style stem 7x7 s2 (RSC4=196)
+    (4, 1, 28, 28, 20, 5, 1, 0),     # LeNet conv1 (C=1 -> pad 4)
+])
+def test_conv_c4_padded_stem_path(shape4):
+    """C<=3 stems with a non-grad input take the padded-channel (C4) fast
+    gather; fwd output and dw (computed at C=4, sliced) must match torch."""
+    Nb, C, H, W, K, R, stride, pad = shape4
+    g = torch.Generator().manual_seed(hash(shape4) % (2 ** 31))
+    x = torch.randn(Nb, C, H, W, generator=g).to('cuda', torch.bfloat16) \
+        .contiguous(memory_format=_CL)            # NO requires_grad: C4 path
+    w = (torch.randn(K, C, R, R, generator=g) / (R * R * C) ** 0.5) \
+        .to('cuda', torch.bfloat16).contiguous(memory_format=_CL) \
+        .requires_grad_(True)
+    out = _ConvFn.apply(x, w, None, stride, pad)
+    dout = torch.randn(out.shape, generator=g).to('cuda', torch.bfloat16) \
+        .contiguous(memory_format=_CL)
+    out.backward(dout)
+
+    xr = x.detach().float()
+    wr = w.detach().float().requires_grad_(True)
+    outr = F.conv2d(xr, wr, None, stride=stride, padding=pad)
+    outr.backward(dout.float())
+    assert _rel_err(out, outr) < 0.03, f"c4 fwd {_rel_err(out, outr)}"
+    assert _rel_err(w.grad, wr.grad) < 0.03, f"c4 wgrad {_rel_err(w.grad, wr.grad)}"
