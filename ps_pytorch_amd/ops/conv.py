@@ -47,7 +47,9 @@ def _wgrad_split(M: int, K: int, C: int, R: int, S: int,
     without exploding the f32 partial buffer. Mirrors the kernel dispatch in
     conv.hip (generic / small-RSC / row-halo)."""
     tk = 128 if K >= 128 else 64
-    if R * S > 1 and R * S * C <= 192:      # flattened small-RSC kernel
+    if R * S > 1 and ((C == 4 and R * S * 4 <= 256) or R * S * C <= 192):
+        # flattened small-RSC kernel (C==4 = padded-channel stem) — MUST
+        # mirror conv.hip's dispatch or the split target misjudges the grid
         tiles = ((K + 63) // 64) * ((R * S * C + 63) // 64)
     elif (stride == 1 and R == 3 and S == 3 and pad == 1 and Q == P
           and 0 < Q <= 32 and (Q & (Q - 1)) == 0 and (P & (P - 1)) == 0
