@@ -39,6 +39,17 @@ static inline void ew_grid(long n_vec, int threads, int* blocks) {
     *blocks = (int)(want < cap ? (want > 0 ? want : 1) : cap);
 }
 
+// Lemire fast division: q = floor(n/d) = umul64hi(n, floor(2^64/d)+1),
+// exact for all 32-bit n and d > 1 (d == 1 encoded as magic 0). Replaces
+// per-element integer division in pixel/channel decode loops.
+__device__ __forceinline__ unsigned fdiv_u32(unsigned n,
+                                             unsigned long long magic) {
+    return magic ? (unsigned)__umul64hi((unsigned long long)n, magic) : n;
+}
+static inline unsigned long long fdiv_magic(long d) {
+    return d > 1 ? (~0ULL) / (unsigned long long)d + 1 : 0ULL;
+}
+
 // Tail convention: kernels vectorize 4 elements/lane; the final n%4 scalar
 // elements are handled by lane (gid == 0) of each kernel with scalar code.
 #define EW_IDX long gid = (long)blockIdx.x * blockDim.x + threadIdx.x; \

@@ -48,6 +48,7 @@
 #include <stdlib.h>
 
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
+typedef unsigned short ushort8_t __attribute__((ext_vector_type(8)));
 typedef float f32x4_t __attribute__((ext_vector_type(4)));
 
 #define MFMA_BF16(a, b, c) __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0)
@@ -71,8 +72,26 @@ __device__ __forceinline__ V16 load16(const unsigned short* src, int cc, int Cn)
     V16 v;
     if constexpr (AL) {
         v.u4 = *(const uint4*)src;
-    } else if (cc + 8 <= Cn && ((Cn | cc) & 7) == 0) {
-        v.u4 = *(const uint4*)src;
+    } else if (cc + 8 <= Cn) {
+        // in-bounds but possibly misaligned (rows of odd-length tensors,
+        // e.g. LeNet's K=500/50): pick the widest aligned access. The
+        // branches diverge by row parity, but two masked passes still beat
+        // the 8-scalar path this used to take for EVERY chunk whenever
+        // Cn % 8 != 0.
+        size_t a = (size_t)src;
+        if ((a & 15) == 0) {
+            v.u4 = *(const uint4*)src;
+        } else if ((a & 7) == 0) {
+            ((unsigned long long*)&v)[0] = ((const unsigned long long*)src)[0];
+            ((unsigned long long*)&v)[1] = ((const unsigned long long*)src)[1];
+        } else if ((a & 3) == 0) {
+#pragma unroll
+            for (int u = 0; u < 4; ++u)
+                ((unsigned int*)&v)[u] = ((const unsigned int*)src)[u];
+        } else {
+#pragma unroll
+            for (int u = 0; u < 8; ++u) v.us[u] = src[u];
+        }
     } else {
         v = zero16();
         for (int u = 0; u < 8; ++u)
@@ -748,18 +767,6 @@ struct WgradLds {
 // (all precomputed — zero inner-loop cost).
 __device__ __forceinline__ int pswz(int r) {
     return r ^ (((r >> 3) & 1) << 2);
-}
-
-// Lemire fast division for the non-pow2 pixel decode (ResNet-50's
-// Q=56/28/14/7): q = floor(n/d) = umul64hi(n, floor(2^64/d)+1), exact for
-// all 32-bit n and d > 1. The POW2=false wgrad kernels ran ~2 full divide
-// sequences per thread per 64-pixel step (~50 VALU against 16 MFMA).
-__device__ __forceinline__ unsigned fdiv_u32(unsigned n,
-                                             unsigned long long magic) {
-    return magic ? (unsigned)__umul64hi((unsigned long long)n, magic) : n;
-}
-static inline unsigned long long fdiv_magic(long d) {
-    return d > 1 ? (~0ULL) / (unsigned long long)d + 1 : 0ULL;
 }
 
 typedef __bf16 bf16x4_t __attribute__((ext_vector_type(4)));
@@ -1507,24 +1514,59 @@ __global__ __launch_bounds__(256) void colsum_part_kernel(
     float* __restrict__ partial, const unsigned short* __restrict__ dout,
     long M, int K)
 {
-    __shared__ float red[256];
-    const int lane = threadIdx.x & 63;
-    const int sub = threadIdx.x >> 6;     // 4 m-partitions per block
-    // outer loop covers K > 64 (Linear bias: K = 500/1000; conv biases are
-    // small but this path is shared with PsLinear's db)
-    for (int kb = 0; kb < K; kb += 64) {
-        int k = kb + lane;
-        float acc = 0.f;
-        if (k < K)
-            for (long m = (long)blockIdx.x * 4 + sub; m < M;
-                 m += (long)gridDim.x * 4)
-                acc += bf16_to_f32(dout[m * K + k]);
-        red[threadIdx.x] = acc;
+    // row-major layout (bn_stats-style): thread t covers channel octet
+    // kg = t % G of row r = t / G, reading 16-B vectors — the previous
+    // column-per-lane form read K-strided scalars (fully uncoalesced,
+    // ~20x off stream rate on LeNet's [524288, 50] bias grad).
+    const int G = (K + 7) >> 3;
+    const int Rr = 256 / G;               // rows per block iteration (>=1)
+    const int kg = threadIdx.x % G;
+    const int rib = threadIdx.x / G;
+    const int kb = kg * 8;
+    const int cw = (kb + 8 <= K) ? 8 : (K - kb);
+    const bool active = rib < Rr && Rr > 0;
+    __shared__ float red[256][9];         // +1 pad (bank stride)
+    float acc[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) acc[u] = 0.f;
+    if (active && Rr > 0) {
+        const long chunk = 4L * Rr;
+        const bool aligned = ((K & 7) == 0);
+        for (long base = (long)blockIdx.x * chunk + rib; base < M;
+             base += (long)gridDim.x * chunk) {
+#pragma unroll
+            for (int u4 = 0; u4 < 4; ++u4) {
+                long r = base + (long)u4 * Rr;
+                if (r >= M) continue;
+                const unsigned short* p = dout + r * K + kb;
+                if (aligned && cw == 8) {
+                    ushort8_t v = *(const ushort8_t*)p;
+#pragma unroll
+                    for (int u = 0; u < 8; ++u) acc[u] += bf16_to_f32(v[u]);
+                } else {
+                    for (int u = 0; u < cw; ++u) acc[u] += bf16_to_f32(p[u]);
+                }
+            }
+        }
+    }
+#pragma unroll
+    for (int u = 0; u < 8; ++u) red[threadIdx.x][u] = acc[u];
+    __syncthreads();
+    // deterministic tree fold over rib (log2(Rr) steps, non-pow2 safe)
+    int live = Rr > 0 ? Rr : 1;
+    while (live > 1) {
+        int half = (live + 1) >> 1;
+        if (active && rib < live - half) {
+#pragma unroll
+            for (int u = 0; u < 8; ++u)
+                red[threadIdx.x][u] += red[(rib + half) * G + kg][u];
+        }
         __syncthreads();
-        if (sub == 0 && k < K)
-            partial[(long)blockIdx.x * K + k] =
-                red[lane] + red[64 + lane] + red[128 + lane] + red[192 + lane];
-        __syncthreads();
+        live = half;
+    }
+    if (rib == 0) {
+        float* dst = partial + (long)blockIdx.x * K + kb;
+        for (int u = 0; u < cw; ++u) dst[u] = red[kg][u];
     }
 }
 

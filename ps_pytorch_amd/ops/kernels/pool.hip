@@ -17,17 +17,23 @@ __global__ __launch_bounds__(256) void maxpool_fwd_kernel(
     unsigned short* __restrict__ y, unsigned char* __restrict__ arg,
     const unsigned short* __restrict__ x,
     int Nb, int H, int W, int C, int P, int Q,
-    int R, int S, int st, int pd)
+    int R, int S, int st, int pd,
+    unsigned long long mcn, unsigned long long mq, unsigned long long mpq)
 {
     EW_IDX
     const int c8n = (C + 7) >> 3;
     const long total = (long)Nb * P * Q * c8n;
     for (long i = gid; i < total; i += stride) {
-        int cb = (int)(i % c8n) * 8;
-        long pix = i / c8n;
-        int q = (int)(pix % Q);
-        int p = (int)((pix / Q) % P);
-        int n = (int)(pix / ((long)P * Q));
+        // magic-division decode (per-element / and % were ~30 VALU each)
+        unsigned pixu = fdiv_u32((unsigned)i, mcn);
+        int cb = (int)((unsigned)i - pixu * (unsigned)c8n) * 8;
+        long pix = pixu;
+        unsigned nu = fdiv_u32(pixu, mpq);
+        unsigned rem = pixu - nu * (unsigned)(P * Q);
+        unsigned pu = fdiv_u32(rem, mq);
+        int q = (int)(rem - pu * (unsigned)Q);
+        int p = (int)pu;
+        int n = (int)nu;
         const int cw = (cb + 8 <= C) ? 8 : (C - cb);
         float best[8];
         int bidx[8];
@@ -83,17 +89,22 @@ __global__ __launch_bounds__(256) void maxpool_bwd_kernel(
     unsigned short* __restrict__ dx, const unsigned short* __restrict__ dy,
     const unsigned char* __restrict__ arg,
     int Nb, int H, int W, int C, int P, int Q,
-    int R, int S, int st, int pd)
+    int R, int S, int st, int pd,
+    unsigned long long mcn, unsigned long long mw, unsigned long long mhw)
 {
     EW_IDX
     const int c8n = (C + 7) >> 3;
     const long total = (long)Nb * H * W * c8n;
     for (long i = gid; i < total; i += stride) {
-        int cb = (int)(i % c8n) * 8;
-        long pix = i / c8n;
-        int w = (int)(pix % W);
-        int h = (int)((pix / W) % H);
-        int n = (int)(pix / ((long)H * W));
+        unsigned pixu = fdiv_u32((unsigned)i, mcn);
+        int cb = (int)((unsigned)i - pixu * (unsigned)c8n) * 8;
+        long pix = pixu;
+        unsigned nu = fdiv_u32(pixu, mhw);
+        unsigned rem = pixu - nu * (unsigned)(H * W);
+        unsigned hu = fdiv_u32(rem, mw);
+        int w = (int)(rem - hu * (unsigned)W);
+        int h = (int)hu;
+        int n = (int)nu;
         const int cw = (cb + 8 <= C) ? 8 : (C - cb);
         float acc[8];
 #pragma unroll
@@ -208,7 +219,9 @@ extern "C" void ps_maxpool_fwd(
     hipLaunchKernelGGL(maxpool_fwd_kernel, dim3(blocks), dim3(256), 0,
                        (hipStream_t)strm, (unsigned short*)y,
                        (unsigned char*)arg, (const unsigned short*)x,
-                       Nb, H, W, C, P, Q, R, S, st, pd);
+                       Nb, H, W, C, P, Q, R, S, st, pd,
+                       fdiv_magic((C + 7) >> 3), fdiv_magic(Q),
+                       fdiv_magic((long)P * Q));
 }
 
 extern "C" void ps_maxpool_bwd(
@@ -220,7 +233,9 @@ extern "C" void ps_maxpool_bwd(
     hipLaunchKernelGGL(maxpool_bwd_kernel, dim3(blocks), dim3(256), 0,
                        (hipStream_t)strm, (unsigned short*)dx,
                        (const unsigned short*)dy, (const unsigned char*)arg,
-                       Nb, H, W, C, P, Q, R, S, st, pd);
+                       Nb, H, W, C, P, Q, R, S, st, pd,
+                       fdiv_magic((C + 7) >> 3), fdiv_magic(W),
+                       fdiv_magic((long)H * W));
 }
 
 extern "C" void ps_gavgpool_fwd(void* y, const void* x, int Nb, int HW,
