@@ -33,9 +33,24 @@ class FlatAdam:
 
     @torch.no_grad()
     def step(self, grad_sum: torch.Tensor, grad_scale: float = 1.0,
-             wire_out: Optional[torch.Tensor] = None) -> None:
+             wire_out: Optional[torch.Tensor] = None,
+             region=None, advance: bool = True) -> None:
+        """region=(start, end): slice update for the pipelined per-bucket
+        path; pass advance=True exactly once per logical step (first
+        bucket) so the bias-correction t stays per-step."""
         from ..ops.functional import fused_adam_step
-        self.t += 1
+        if advance:
+            self.t += 1
+        if region is not None:
+            st, en = region
+            fused_adam_step(self.w[st:en], grad_sum[st:en],
+                            self.exp_avg[st:en], self.exp_avg_sq[st:en],
+                            self.t, self.lr, self.beta1, self.beta2, self.eps,
+                            self.weight_decay, grad_scale,
+                            self.max_exp_avg_sq[st:en] if self.max_exp_avg_sq
+                            is not None else None,
+                            wire_out[st:en] if wire_out is not None else None)
+            return
         fused_adam_step(self.w, grad_sum, self.exp_avg, self.exp_avg_sq,
                         self.t, self.lr, self.beta1, self.beta2, self.eps,
                         self.weight_decay, grad_scale,

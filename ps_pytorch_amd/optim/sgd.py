@@ -30,7 +30,17 @@ class FlatSGD:
 
     @torch.no_grad()
     def step(self, grad_sum: torch.Tensor, grad_scale: float = 1.0,
-             wire_out: Optional[torch.Tensor] = None) -> None:
+             wire_out: Optional[torch.Tensor] = None,
+             region=None, advance: bool = True) -> None:   # advance: parity with FlatAdam
+        """region=(start, end): update only that flat slice — the per-bucket
+        pipelined update/broadcast path (elementwise math: slicing exact)."""
+        if region is not None:
+            st, en = region
+            fused_sgd_step(self.w[st:en], grad_sum[st:en], self.m[st:en],
+                           self.lr, self.momentum, self.weight_decay,
+                           grad_scale, self.nesterov,
+                           wire_out[st:en] if wire_out is not None else None)
+            return
         fused_sgd_step(self.w, grad_sum, self.m, self.lr, self.momentum,
                        self.weight_decay, grad_scale, self.nesterov, wire_out)
 

@@ -22,6 +22,7 @@ import time
 from typing import Optional
 
 import torch
+import torch.distributed as dist
 
 from ..config import JobConfig, input_shape_of, num_classes_of
 from ..models import build_model
@@ -84,6 +85,10 @@ class ParameterServer:
                                      momentum=cfg.momentum)
         # first broadcast payload
         self.transport.pack_weights_from(self.master_w)
+        # pipelined-broadcast mode: step k's weights go out at the tail of
+        # step k-1; step 0's go out here (matching the workers' first fetch)
+        if self.transport.bcast_bucketed and dist.is_initialized():
+            self.transport.bcast_all_buckets()
 
     @property
     def grad_scale(self) -> float:
@@ -94,6 +99,8 @@ class ParameterServer:
     def step(self) -> None:
         """One synchronous step (mirrors DistributedWorker.train_step order)."""
         t = self.transport
+        if t.bcast_bucketed:
+            return self._step_pipelined()
         t.broadcast_weights()
         if t.mode == 'gather':
             t.post_gather_recvs()
@@ -114,6 +121,29 @@ class ParameterServer:
         # fused: average-scale + momentum + update + re-pack next payload
         self.optimizer.step(grad, grad_scale=self.grad_scale,
                             wire_out=t.wire_w)
+        self.cur_step += 1
+
+    def _step_pipelined(self) -> None:
+        """Collective-mode pipeline: per bucket, wait fan-in -> fused slice
+        update -> broadcast that slice immediately. The fan-in tail of later
+        buckets and the whole optimizer step hide under wire time; the
+        per-rank collective order matches the workers' (transport.py
+        bcast_bucketed note). The step's own weights went out at the tail of
+        the previous step (step 0: at build_model)."""
+        t = self.transport
+        t.wait_bcasts()                      # previous tail fully on the wire
+        t.recv_buckets(self.flat.buckets)    # post async per-bucket reduces
+        last = self.cur_step + 1 >= self.cfg.max_steps
+        for i, b in enumerate(self.flat.buckets):
+            t.wait_reduce(i)
+            self.optimizer.step(t.wire_g, grad_scale=self.grad_scale,
+                                wire_out=t.wire_w, region=(b.start, b.end),
+                                advance=(i == 0))
+            if not last:
+                # workers exit after their final step: no dangling
+                # collective for them to match
+                t.bcast_bucket(b)
+        t._works.clear()
         self.cur_step += 1
 
     def start(self) -> None:

@@ -102,6 +102,7 @@ class PSTransport:
         self.wire_w = torch.zeros(n, dtype=wire_dtype, device=device)
         self.wire_g = torch.zeros(n, dtype=wire_dtype, device=device)
         self._works: List[dist.Work] = []
+        self._bcast_works: List[dist.Work] = []
         # program-order matching guard: NCCL IGNORES P2P tags — matching of
         # the gather-mode isend/irecv pairs rests entirely on per-pair FIFO
         # order, so every rank MUST touch buckets in index order. This
@@ -174,7 +175,43 @@ class PSTransport:
 
     # ---- weights: PS -> all ----
 
+    # Pipelined per-bucket update/broadcast (collective mode, Bcast
+    # comm_type): the PS waits each bucket's fan-in, runs the fused update
+    # on that slice, and broadcasts it immediately — the fan-in tail of
+    # later buckets and the whole optimizer step hide under wire time.
+    # Per-rank collective order stays identical on every rank:
+    #   worker: [bcast(k) x B, reduce(k) x B, bcast(k+1) x B, ...]
+    #   PS:     [bcast(0,init) x B, reduce(0) x B, bcast(0,tail) x B, ...]
+    # (the PS skips the tail broadcast of the final step — no dangling
+    # collective for workers that have exited their loop).
+    @property
+    def bcast_bucketed(self) -> bool:
+        return (self.mode == 'collective' and self.comm_type == 'Bcast'
+                and os.environ.get('PS_BCAST_PIPE', '1') != '0')
+
+    def bcast_bucket(self, b: Bucket) -> None:
+        self._bcast_works.append(
+            dist.broadcast(self.wire_w[b.start:b.end], src=PS_RANK,
+                           group=self.group, async_op=True))
+
+    def bcast_all_buckets(self) -> None:
+        for b in self.flat.buckets:
+            self.bcast_bucket(b)
+
+    def wait_bcasts(self) -> None:
+        for w in self._bcast_works:
+            w.wait()
+        self._bcast_works.clear()
+
+    def wait_reduce(self, i: int) -> None:
+        """Wait the i-th posted reduce (collective-mode PS pipeline)."""
+        self._works[i].wait()
+
     def broadcast_weights(self) -> None:
+        if self.rank != PS_RANK and self.bcast_bucketed:
+            self.bcast_all_buckets()
+            self.wait_bcasts()
+            return
         if self.comm_type == 'Async':
             # reference's deprecated P2P weight distribution
             # (distributed_worker.py:201-219): explicit PS->worker sends

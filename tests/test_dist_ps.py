@@ -101,3 +101,24 @@ def test_golden_step_async_comm_type():
     ref = _serial_reference()
     assert torch.allclose(got, ref, atol=1e-5, rtol=1e-5), \
         (got - ref).abs().max()
+
+
+def test_bcast_pipeline_mode_matrix(monkeypatch):
+    """The pipelined per-bucket update/broadcast engages exactly for
+    collective + Bcast (and can be killed with PS_BCAST_PIPE=0); gather and
+    Async keep the single-collective broadcast."""
+    import torch
+    from ps_pytorch_amd.models import build_model
+    from ps_pytorch_amd.parallel.flat import FlatSpace
+    from ps_pytorch_amd.parallel.transport import PSTransport
+    fs = FlatSpace(build_model('LeNet', in_channels=1))
+    dev = torch.device('cpu')
+
+    def mk(**kw):
+        return PSTransport(fs, torch.float32, dev, 0, 1, **kw)
+
+    assert mk(mode='collective', comm_type='Bcast').bcast_bucketed
+    assert not mk(mode='gather', comm_type='Bcast').bcast_bucketed
+    assert not mk(mode='collective', comm_type='Async').bcast_bucketed
+    monkeypatch.setenv('PS_BCAST_PIPE', '0')
+    assert not mk(mode='collective', comm_type='Bcast').bcast_bucketed
