@@ -137,10 +137,30 @@ class PsBatchNorm2d(nn.Module):
             self.running_mean.data = self.running_mean.data.float()
             self.running_var.data = self.running_var.data.float()
 
+    # num_batches_tracked: the canonical buffer update is a device-side
+    # long add — 20 kernel launches per ResNet-18 step for a counter nobody
+    # reads during training. Count on the host and materialize into the
+    # buffer only when a state_dict is taken (or synced in from a load).
+    def _sync_nbt_out(self, *a):
+        host = getattr(self, '_nbt_host', 0)
+        if host:
+            self.num_batches_tracked.fill_(int(self.num_batches_tracked)
+                                           + host)
+            self._nbt_host = 0
+
+    def _load_from_state_dict(self, *args, **kw):
+        self._nbt_host = 0
+        super()._load_from_state_dict(*args, **kw)
+
+    def _save_to_state_dict(self, *args, **kw):
+        # (root state_dict() recurses here, not into child .state_dict())
+        self._sync_nbt_out()
+        super()._save_to_state_dict(*args, **kw)
+
     def forward(self, x: torch.Tensor,
                 residual: Optional[torch.Tensor] = None) -> torch.Tensor:
         if self.training:
-            self.num_batches_tracked += 1
+            self._nbt_host = getattr(self, '_nbt_host', 0) + 1
         if self.training and _kernel_ok(x):
             self._ensure_f32_stats()
             return _FusedBNFn.apply(x, self.weight, self.bias, residual,
