@@ -118,6 +118,12 @@ class NNTrainer:
         self._gx.copy_(data, non_blocking=True)
         self._gy.copy_(target, non_blocking=True)
         self._graph.replay()
+        # host-side BN batch counters don't tick during replay (host code
+        # doesn't run); keep them canonical here
+        from .ops.modules import PsBatchNorm2d
+        for m in self.network.modules():
+            if isinstance(m, PsBatchNorm2d):
+                m._nbt_host = getattr(m, '_nbt_host', 0) + 1
         self.cur_step += 1
         return self._gloss
 
