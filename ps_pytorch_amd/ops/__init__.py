@@ -82,6 +82,9 @@ def _try_load() -> None:
             ctypes.c_int] * 3 + [ctypes.c_void_p]
         lib.ps_pad4.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                 ctypes.c_long, ctypes.c_int, ctypes.c_void_p]
+        lib.ps_padc.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                ctypes.c_long, ctypes.c_int, ctypes.c_int,
+                                ctypes.c_void_p]
         _LIB = lib
     except OSError as e:  # pragma: no cover
         _LIB_ERR = str(e)

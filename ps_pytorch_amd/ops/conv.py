@@ -47,9 +47,11 @@ def _wgrad_split(M: int, K: int, C: int, R: int, S: int,
     without exploding the f32 partial buffer. Mirrors the kernel dispatch in
     conv.hip (generic / small-RSC / row-halo)."""
     tk = 128 if K >= 128 else 64
-    if R * S > 1 and ((C == 4 and R * S * 4 <= 256) or R * S * C <= 192):
-        # flattened small-RSC kernel (C==4 = padded-channel stem) — MUST
-        # mirror conv.hip's dispatch or the split target misjudges the grid
+    if R * S > 1 and ((C == 4 and R * S * 4 <= 256)
+                      or (C % 8 == 0 and C % 64 and R * S * C <= 768)
+                      or R * S * C <= 192):
+        # flattened small-RSC kernel (padded-channel stems incl. C%8) —
+        # MUST mirror conv.hip's dispatch or the split target misjudges
         tiles = ((K + 63) // 64) * ((R * S * C + 63) // 64)
     elif (stride == 1 and R == 3 and S == 3 and pad == 1 and Q == P
           and 0 < Q <= 32 and (Q & (Q - 1)) == 0 and (P & (P - 1)) == 0
@@ -75,24 +77,32 @@ class _ConvFn(torch.autograd.Function):
         K, _, R, S = w.shape
         P = (H + 2 * pad - R) // stride + 1
         Q = (W + 2 * pad - S) // stride + 1
-        # C<=3 stems (CIFAR 3x3, R50 7x7, MNIST 5x5): pad channels to 4 so
-        # the SMALL gather runs as aligned 8-B tap loads (kernels' C4 path).
-        # The input is the dataset tensor (no dx needed), so only fwd/wgrad
-        # see the padded operands; dw is computed at C=4 and sliced back.
-        c4 = (C <= 3 and R * S > 1 and R * S * 4 <= 256
-              and not x.requires_grad)
-        ctx.c4 = c4
+        # Padded-channel fast paths (SMALL gather with aligned tap loads):
+        #  * C<=3 stems -> 4 (two 8-B halves per quantum; dataset inputs,
+        #    no dx needed)
+        #  * ragged C%8 (LeNet conv2's C=20 -> 24) -> next multiple of 8
+        #    (quantum within one tap: 16-B loads; dx computed padded and
+        #    sliced back)
+        cp = 0
+        if R * S > 1:
+            if C <= 3 and R * S * 4 <= 256 and not x.requires_grad:
+                cp = 4
+            elif C % 8 and C < 64:
+                c8 = (C + 7) & ~7
+                if c8 % 64 and R * S * c8 <= 768:
+                    cp = c8
+        ctx.cp = cp
         ctx.c_orig = C
-        if c4:
-            x4 = torch.empty((Nb, 4, H, W), dtype=x.dtype, device=x.device,
+        if cp:
+            xp = torch.empty((Nb, cp, H, W), dtype=x.dtype, device=x.device,
                              memory_format=_CL)
-            lib.ps_pad4(x4.data_ptr(), x.data_ptr(), Nb * H * W, C,
+            lib.ps_padc(xp.data_ptr(), x.data_ptr(), Nb * H * W, C, cp,
                         current_stream_ptr())
-            w4 = torch.empty((K, 4, R, S), dtype=w.dtype, device=w.device,
+            wp = torch.empty((K, cp, R, S), dtype=w.dtype, device=w.device,
                              memory_format=_CL)
-            lib.ps_pad4(w4.data_ptr(), wc.data_ptr(), K * R * S, C,
+            lib.ps_padc(wp.data_ptr(), wc.data_ptr(), K * R * S, C, cp,
                         current_stream_ptr())
-            x, wc, C = x4, w4, 4
+            x, wc, C = xp, wp, cp
         # NB: allocate channels_last DIRECTLY — empty().contiguous(CL) runs a
         # full transposing copy of uninitialized memory (~190us at l1 size)
         out = torch.empty((Nb, K, P, Q), dtype=x.dtype, device=x.device,
@@ -130,30 +140,44 @@ class _ConvFn(torch.autograd.Function):
             lib.ps_wt_transpose(wt.data_ptr(), w.data_ptr(), K, R * S * C,
                                 current_stream_ptr())
             dx = torch.empty_like(x).contiguous(memory_format=_CL)
+            if dcarry is not None and ctx.cp:
+                # padded path: fuse the carry AFTER the slice instead
+                dxp_carry = dcarry
+                dcarry = None
+            else:
+                dxp_carry = None
             if dcarry is not None:
                 dcarry = dcarry.contiguous(memory_format=_CL)
             lib.ps_conv_dgrad(dout.data_ptr(), wt.data_ptr(), dx.data_ptr(),
                               dcarry.data_ptr() if dcarry is not None else 0,
                               Nb, H, W, C, K, P, Q, R, S, stride, pad,
                               current_stream_ptr())
+            if ctx.cp:
+                # computed at padded C: slice the real channels back
+                dxs = torch.empty((Nb, ctx.c_orig, H, W), dtype=dx.dtype,
+                                  device=dx.device, memory_format=_CL)
+                dxs.copy_(dx[:, :ctx.c_orig])
+                if dxp_carry is not None:
+                    dxs.add_(dxp_carry)
+                dx = dxs
         if ctx.needs_input_grad[1]:
             M = Nb * P * Q
             split = _wgrad_split(M, K, C, R, S, stride, pad, P, Q)
             partial = torch.empty(split * K * R * S * C,
                                   dtype=torch.float32, device=x.device)
             wt_tgt = ctx.gtgt[0]() if ctx.gtgt[0] is not None else None
-            if ctx.c4:
-                # padded-channel stem: wgrad at C=4, then slice the real
-                # C_orig channels back into the (steal-target) dw
-                dw4 = torch.empty_like(w).contiguous(memory_format=_CL)
+            if ctx.cp:
+                # padded-channel stem: wgrad at the padded C, then slice
+                # the real C_orig channels back into the (steal-target) dw
+                dwp = torch.empty_like(w).contiguous(memory_format=_CL)
                 lib.ps_conv_wgrad(dout.data_ptr(), x.data_ptr(),
-                                  partial.data_ptr(), dw4.data_ptr(),
-                                  Nb, H, W, 4, K, P, Q, R, S, stride, pad,
-                                  split, current_stream_ptr())
+                                  partial.data_ptr(), dwp.data_ptr(),
+                                  Nb, H, W, ctx.cp, K, P, Q, R, S, stride,
+                                  pad, split, current_stream_ptr())
                 dw = (wt_tgt if wt_tgt is not None and wt_tgt.is_cuda
                       else torch.empty((K, ctx.c_orig, R, S), dtype=w.dtype,
                                        device=w.device, memory_format=_CL))
-                dw.copy_(dw4[:, :ctx.c_orig])
+                dw.copy_(dwp[:, :ctx.c_orig])
             else:
                 dw = (wt_tgt if wt_tgt is not None and wt_tgt.dtype == w.dtype
                       and wt_tgt.is_cuda

@@ -10,6 +10,8 @@
 // here are bucket slices at parameter boundaries, not always 16B-sized).
 #include "common.h"
 
+typedef unsigned short ushort8_t __attribute__((ext_vector_type(8)));
+
 __global__ __launch_bounds__(256) void pack_bf16_kernel(
     unsigned short* __restrict__ dst, const float* __restrict__ src, long n)
 {
@@ -73,4 +75,32 @@ extern "C" void ps_pad4(void* dst, const void* src, long npix, int C,
     hipLaunchKernelGGL(pad4_kernel, dim3(blocks), dim3(256), 0,
                        (hipStream_t)stream, (unsigned short*)dst,
                        (const unsigned short*)src, npix, C);
+}
+
+// general channel pad C -> CP (CP % 8 == 0, CP <= 64): out[p][0:CP] =
+// {in[p][0:C], 0...}; 16-B vector stores per octet.
+__global__ __launch_bounds__(256) void padc_kernel(
+    unsigned short* __restrict__ dst, const unsigned short* __restrict__ src,
+    long npix, int C, int CP)
+{
+    EW_IDX
+    for (long i = gid; i < npix; i += stride) {
+        const unsigned short* s = src + i * C;
+        unsigned short* d = dst + i * CP;
+        for (int cb = 0; cb < CP; cb += 8) {
+            ushort8_t o = {0, 0, 0, 0, 0, 0, 0, 0};
+#pragma unroll
+            for (int u = 0; u < 8; ++u)
+                if (cb + u < C) o[u] = s[cb + u];
+            *(ushort8_t*)(d + cb) = o;
+        }
+    }
+}
+
+extern "C" void ps_padc(void* dst, const void* src, long npix, int C,
+                        int CP, void* stream) {
+    int blocks; ew_grid(npix, 256, &blocks);
+    hipLaunchKernelGGL(padc_kernel, dim3(blocks), dim3(256), 0,
+                       (hipStream_t)stream, (unsigned short*)dst,
+                       (const unsigned short*)src, npix, C, CP);
 }

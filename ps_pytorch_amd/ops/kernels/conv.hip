@@ -149,8 +149,9 @@ template <int TM, int TN, int STRIDE, bool DGRAD, bool SMALL = false, bool AL = 
           bool ACCF = false,  // dgrad: epilogue adds `carry` (residual-fork
                               // grad accumulation fused in — kills the
                               // autograd at::add at every block input fork)
-          bool C4 = false>    // SMALL with C==4 (padded stem): tap-aligned
-                              // 8-B gather loads instead of scalar
+          int CPM = 0>    // SMALL padded-channel gather: 4 = C==4 stem
+                          // (two 8-B tap halves), 8 = C%8==0 (quantum
+                          // within one tap: single 16-B load), 0 = off
 __global__ __launch_bounds__(256) void conv_gemm_kernel(
     const unsigned short* __restrict__ src,  // fwd: in [Nb,H,W,C]; dgrad: dout [Nb,P,Q,K]
     const unsigned short* __restrict__ wgt,  // fwd: w [K,R,S,C]; dgrad: wT [R,S,C,K]
@@ -273,7 +274,7 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         // contraction elements (e = lc*64 + cc8+u); <=3 chunks, a handful
         // of divides each.
         int tre[SMALL ? 8 : 1], tse[SMALL ? 8 : 1], tce[SMALL ? 8 : 1];
-        if constexpr (SMALL && !C4) {
+        if constexpr (SMALL && CPM == 0) {
 #pragma unroll
             for (int u = 0; u < 8; ++u) {
                 int e = (lc << 6) + cc8 + u;
@@ -284,7 +285,23 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         }
 #pragma unroll
         for (int rr = 0; rr < AR; ++rr) {
-            if constexpr (SMALL && C4) {
+            if constexpr (SMALL && CPM == 8) {
+                // C%8==0: the 8-elem quantum lies within one (r,s) tap —
+                // one guarded 16-B aligned load (LeNet conv2: C 20->24)
+                const int RSC = R * S * C;
+                const int e0 = (lc << 6) + cc8;
+                const int tp = e0 / C;
+                const int inner = e0 - tp * C;
+                const int tr = tp / S, ts = tp - tr * S;
+                V16 v = zero16();
+                bool rowv = ax[rr] > INT_MIN / 4;
+                int hh = ax[rr] + tr, ww = ay[rr] + ts;
+                if (rowv && e0 < RSC && hh >= 0 && hh < H
+                    && ww >= 0 && ww < W)
+                    v.u4 = *(const uint4*)
+                        (src + abase[rr] + ((long)tr * W + ts) * C + inner);
+                areg[rr] = v;
+            } else if constexpr (SMALL && CPM == 4) {
                 // padded-channel stem: the 8-elem quantum is exactly two
                 // (r,s) taps of 4 channels each — two aligned 8-B loads
                 const int RSC = R * S * 4;
@@ -324,7 +341,16 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         }
 #pragma unroll
         for (int rr = 0; rr < BR; ++rr) {
-            if constexpr (SMALL && C4) {
+            if constexpr (SMALL && CPM == 8) {
+                // weight rows are RSC-contiguous and RSC%8==0: in-bounds
+                // quanta are full 16-B aligned loads
+                const int RSC = R * S * C;
+                const int e0 = (lc << 6) + cc8;
+                V16 v = zero16();
+                if (colv[rr] && e0 < RSC)
+                    v.u4 = *(const uint4*)(pB[rr] + (lc << 6));
+                breg[rr] = v;
+            } else if constexpr (SMALL && CPM == 4) {
                 const int RSC = R * S * 4;
                 const int e0 = (lc << 6) + cc8;
                 V16 v = zero16();
@@ -1218,7 +1244,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
 // flattened (r,s,c) axis fits one 64-column tile, so one block covers every
 // tap in a single pass over its pixel chunk — vs the generic kernel's
 // R*S separate 64-c tiles at C/64 utilization each.
-template <int STRIDE, bool POW2, bool AL = true, bool C4 = false>
+template <int STRIDE, bool POW2, bool AL = true, int CPM = 0>
 __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
     const unsigned short* __restrict__ dout,  // [Nb,P,Q,K]
     const unsigned short* __restrict__ in,    // [Nb,H,W,C]
@@ -1259,7 +1285,12 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
         int e = e0 + cc8 + u;
-        if constexpr (C4) {
+        if constexpr (CPM == 8) {
+            int tp = e / C;
+            tre[u] = tp / S;
+            tse[u] = tp - tre[u] * S;
+            toff[u] = (tre[u] * W + tse[u]) * C + (e - tp * C);
+        } else if constexpr (CPM == 4) {
             int tp = e >> 2;
             tre[u] = tp / S;
             tse[u] = tp - tre[u] * S;
@@ -1296,7 +1327,13 @@ __global__ __launch_bounds__(256) void conv_wgrad_small_kernel(
                 int h0 = p * STRIDE - pad, w0 = q * STRIDE - pad;
                 const unsigned short* pix =
                     in + (long)n * H * W * C + ((long)h0 * W + w0) * C;
-                if constexpr (C4) {
+                if constexpr (CPM == 8) {
+                    // quantum within one tap: single guarded 16-B load
+                    int hh = h0 + tre[0], ww = w0 + tse[0];
+                    if (e0 + cc8 < RSC
+                        && hh >= 0 && hh < H && ww >= 0 && ww < W)
+                        bv.u4 = *(const uint4*)(pix + toff[0]);
+                } else if constexpr (CPM == 4) {
 #pragma unroll
                     for (int half = 0; half < 2; ++half) {
                         int u = 4 * half;
@@ -1601,12 +1638,13 @@ static inline int swz_mode() {
         if (swz_mode() == 1) LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, 2, 1, ACV);   \
         else                 LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, 2, 2, ACV);   \
     } while (0)
-#define LAUNCH_GEMM_CV(TM, TN, ST, SM, NBV)                                   \
+#define LAUNCH_GEMM_CV(TM, TN, ST, SM, NBV) LAUNCH_GEMM_CP(TM, TN, ST, SM, NBV, 4)
+#define LAUNCH_GEMM_CP(TM, TN, ST, SM, NBV, CPV)                              \
     do {                                                                      \
         long M_ = (long)Nb * P * Q;                                           \
         long grid = ((M_ + TM - 1) / TM) * ((K + TN - 1) / TN);               \
         hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, ST, false, SM, ALV, NBV, \
-                                             2, false, true>),               \
+                                             2, false, CPV>),               \
             dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,            \
             (const unsigned short*)src, (const unsigned short*)wgt,           \
             (const unsigned short*)bias, (unsigned short*)dst,                \
@@ -1641,6 +1679,10 @@ extern "C" void ps_conv_fwd(
         } else if (R * S > 1 && C == 4 && R * S * 4 <= 256) {                 \
             if (stride == 1) LAUNCH_GEMM_CV(128, 64, 1, true, 2);             \
             else             LAUNCH_GEMM_CV(128, 64, 2, true, 2);             \
+        } else if (R * S > 1 && (C & 7) == 0 && (C & 63)                      \
+                   && R * S * C <= 768) {        /* padded/ragged C%8 */      \
+            if (stride == 1) LAUNCH_GEMM_CP(128, 64, 1, true, 2, 8);          \
+            else             LAUNCH_GEMM_CP(128, 64, 2, true, 2, 8);          \
         } else if (R * S > 1 && R * S * C <= 64) {  /* one-step: 1 buffer */  \
             if (stride == 1) LAUNCH_GEMM_NB(128, 64, 1, false, true, 1);      \
             else             LAUNCH_GEMM_NB(128, 64, 2, false, true, 1);      \
@@ -1759,8 +1801,11 @@ extern "C" void ps_conv_wgrad(
     long M = (long)Nb * P * Q;
     long chunk64 = (M + (long)split * 64 - 1) / ((long)split * 64);
     int chunk = (int)(chunk64 * 64);
-    if (R * S > 1 && ((C == 4 && R * S * 4 <= 256) || R * S * C <= 192)) {
-        // flattened stem/LeNet path (C==4 = padded-channel stem)
+    if (R * S > 1 && ((C == 4 && R * S * 4 <= 256)
+                      || ((C & 7) == 0 && (C & 63) && R * S * C <= 768)
+                      || R * S * C <= 192)) {
+        // flattened stem/LeNet path (C==4 = padded stem; C%8==0 = padded
+        // or naturally-aligned ragged channels, e.g. LeNet conv2 20->24)
         int tiles_k = (K + 63) / 64;
         int nc_s = (R * S * C + 63) / 64;
         long grid_s = (long)tiles_k * nc_s * split;
@@ -1775,11 +1820,14 @@ extern "C" void ps_conv_wgrad(
             (float*)partial_f32, Nb, H, W, C, K, P, Q, R, S, pad, split,      \
             chunk, l2pq_, l2q_, mpq_, mq_)
         if (C == 4) {
-            if (stride == 1) { if (pw) WGS(1, true, true); else WGS(1, false, true); }
-            else             { if (pw) WGS(2, true, true); else WGS(2, false, true); }
+            if (stride == 1) { if (pw) WGS(1, true, 4); else WGS(1, false, 4); }
+            else             { if (pw) WGS(2, true, 4); else WGS(2, false, 4); }
+        } else if ((C & 7) == 0 && (C & 63)) {
+            if (stride == 1) { if (pw) WGS(1, true, 8); else WGS(1, false, 8); }
+            else             { if (pw) WGS(2, true, 8); else WGS(2, false, 8); }
         } else {
-            if (stride == 1) { if (pw) WGS(1, true, false); else WGS(1, false, false); }
-            else             { if (pw) WGS(2, true, false); else WGS(2, false, false); }
+            if (stride == 1) { if (pw) WGS(1, true, 0); else WGS(1, false, 0); }
+            else             { if (pw) WGS(2, true, 0); else WGS(2, false, 0); }
         }
 #undef WGS
         long n_ = (long)K * R * S * C;
