@@ -87,12 +87,20 @@ __global__ __launch_bounds__(256) void padc_kernel(
     for (long i = gid; i < npix; i += stride) {
         const unsigned short* s = src + i * C;
         unsigned short* d = dst + i * CP;
-        for (int cb = 0; cb < CP; cb += 8) {
+        int cb = 0;
+        for (; cb + 8 <= CP; cb += 8) {
             ushort8_t o = {0, 0, 0, 0, 0, 0, 0, 0};
 #pragma unroll
             for (int u = 0; u < 8; ++u)
                 if (cb + u < C) o[u] = s[cb + u];
             *(ushort8_t*)(d + cb) = o;
+        }
+        if (cb < CP) {                 // CP % 8 == 4 tail (the C<=3 stems)
+            ushort4_t o = {0, 0, 0, 0};
+#pragma unroll
+            for (int u = 0; u < 4; ++u)
+                if (cb + u < C) o[u] = s[cb + u];
+            *(ushort4_t*)(d + cb) = o;
         }
     }
 }
