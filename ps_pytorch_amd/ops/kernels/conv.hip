@@ -149,15 +149,22 @@ template <int TM, int TN, int STRIDE, bool DGRAD, bool SMALL = false, bool AL = 
           bool ACCF = false,  // dgrad: epilogue adds `carry` (residual-fork
                               // grad accumulation fused in — kills the
                               // autograd at::add at every block input fork)
-          int CPM = 0>    // SMALL padded-channel gather: 4 = C==4 stem
+          int CPM = 0,    // SMALL padded-channel gather: 4 = C==4 stem
                           // (two 8-B tap halves), 8 = C%8==0 (quantum
                           // within one tap: single 16-B load), 0 = off
+          bool GLDS = false>  // stage via global_load_lds (AL && !SMALL):
+                              // drops the ds_write pass + staging VGPRs;
+                              // invalid taps load from the zero page and
+                              // the granule swizzle moves to the SOURCE
+                              // address (self-inverse XOR, guide rule 21)
 __global__ __launch_bounds__(256) void conv_gemm_kernel(
     const unsigned short* __restrict__ src,  // fwd: in [Nb,H,W,C]; dgrad: dout [Nb,P,Q,K]
     const unsigned short* __restrict__ wgt,  // fwd: w [K,R,S,C]; dgrad: wT [R,S,C,K]
     const unsigned short* __restrict__ bias, // [K] or null (fwd only)
     unsigned short* __restrict__ dst,        // fwd: out [Nb,P,Q,K]; dgrad: dx [Nb,H,W,C]
     const unsigned short* __restrict__ carry,// [dst shape] or null (ACCF)
+    const unsigned short* __restrict__ zpage,// 64 zero shorts (GLDS source
+                                             // for invalid taps/cols)
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int pad)
 {
@@ -176,6 +183,14 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
     const int t = threadIdx.x;
     const int trow = t >> 3;           // 0..31
     const int cc8 = (t & 7) * 8;       // this thread's 8-elem contraction chunk
+    // GLDS staging geometry: wave w fills rows [w*TM/4, (w+1)*TM/4) of A
+    // (and the TN analog of B) with AR (resp. BR) 1-KiB global_load_lds
+    // per wave; lane l covers (row = base + rr*8 + (l>>3), position l&7).
+    // The LDS image position p of row r holds granule gswz(r, p) — the
+    // swizzle is an involution, so the source granule IS gswz(r, l&7).
+    const int glane = t & 63, gwid = t >> 6;
+    const int ga_row0 = gwid * (TM / 4) + (glane >> 3);
+    const int gb_row0 = gwid * (TN / 4) + (glane >> 3);
 
     // ---- per-row state (computed once) ----
     // !SMALL: running global pointer pA (advanced by a wave-uniform delta
@@ -188,7 +203,9 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
     long abase[SMALL ? AR : 1]; int ax[SMALL ? AR : 1], ay[SMALL ? AR : 1];
 #pragma unroll
     for (int rr = 0; rr < AR; ++rr) {
-        long m = m0 + trow + 32 * rr;
+        const int lrowA = GLDS ? (ga_row0 + rr * 8) : (trow + 32 * rr);
+        const int ac8 = GLDS ? gswz<SWZ>(lrowA, t & 7) * 8 : cc8;
+        long m = m0 + lrowA;
         pA[rr] = src; vm[rr] = 0;
         if constexpr (SMALL) { ax[rr] = INT_MIN / 2; ay[rr] = 0; abase[rr] = 0; }
         if (m < M) {
@@ -200,7 +217,7 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
                     ax[rr] = h0; ay[rr] = w0;
                     abase[rr] = ((long)(n * H + h0) * W + w0) * C;
                 } else {
-                    pA[rr] = src + ((long)(n * H + h0) * W + w0) * C + cc8;
+                    pA[rr] = src + ((long)(n * H + h0) * W + w0) * C + ac8;
                     unsigned long long msk = 0;
                     for (int r = 0; r < R; ++r)
                         for (int s = 0; s < S; ++s)
@@ -214,7 +231,7 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
                 int h = rem / W, w = rem % W;
                 // stride-1 only (stride-2 dgrad runs conv_dgrad2_kernel)
                 int p0 = h + pad, q0 = w + pad;
-                pA[rr] = src + (long)n * P * Q * K + ((long)p0 * Q + q0) * K + cc8;
+                pA[rr] = src + (long)n * P * Q * K + ((long)p0 * Q + q0) * K + ac8;
                 unsigned long long msk = 0;
                 for (int r = 0; r < R; ++r)
                     for (int s = 0; s < S; ++s)
@@ -254,15 +271,17 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
     bool colv[BR];
 #pragma unroll
     for (int rr = 0; rr < BR; ++rr) {
-        int col = n0 + trow + 32 * rr;
+        const int lrowB = GLDS ? (gb_row0 + rr * 8) : (trow + 32 * rr);
+        const int bc8 = GLDS ? gswz<SWZ>(lrowB, t & 7) * 8 : cc8;
+        int col = n0 + lrowB;
         colv[rr] = col < Nout;
         if (!colv[rr]) col = 0;
         if constexpr (SMALL)
-            pB[rr] = wgt + (long)col * R * S * C + cc8;
+            pB[rr] = wgt + (long)col * R * S * C + bc8;
         else if constexpr (!DGRAD)
-            pB[rr] = wgt + (long)col * R * S * C + cc8;   // (r,s,c0)=(0,0,0)
+            pB[rr] = wgt + (long)col * R * S * C + bc8;   // (r,s,c0)=(0,0,0)
         else
-            pB[rr] = wgt + (long)col * K + cc8;
+            pB[rr] = wgt + (long)col * K + bc8;
     }
     const int KCm1 = (KC - 1) * 64;
 
@@ -408,6 +427,53 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         }
     };
 
+    // GLDS staging: issue the next tile's 1-KiB global_load_lds pieces
+    // (lane-linear dest = exactly our [row][granule] image with the
+    // swizzle folded into the source granule, see setup) then advance the
+    // affine pointers. No registers staged, no ds_write pass; the
+    // following __syncthreads drains the DMA (hipcc emits vmcnt(0) there).
+    auto glds_step = [&](int buf) {
+        if constexpr (GLDS && !SMALL) {
+#pragma unroll
+            for (int rr = 0; rr < AR; ++rr) {
+                const unsigned short* gsrc = (vm[rr] & 1) ? pA[rr] : zpage;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) void*)gsrc,
+                    (__attribute__((address_space(3))) void*)
+                        &lds.A[buf][gwid * (TM / 4) + rr * 8][0], 16, 0, 0);
+            }
+#pragma unroll
+            for (int rr = 0; rr < BR; ++rr) {
+                const unsigned short* gsrc = colv[rr] ? pB[rr] : zpage;
+                __builtin_amdgcn_global_load_lds(
+                    (const __attribute__((address_space(1))) void*)gsrc,
+                    (__attribute__((address_space(3))) void*)
+                        &lds.B[buf][gwid * (TN / 4) + rr * 8][0], 16, 0, 0);
+            }
+            long dA, dB;
+            bool rs_adv = (++lc == KC);
+            if (rs_adv) {
+                lc = 0;
+                bool rwrap = (++ls == S);
+                if (rwrap) { ls = 0; ++lr; }
+                if constexpr (!DGRAD) {
+                    dA = (long)(rwrap ? (W - S + 1) : 1) * C - KCm1;
+                    dB = (long)C - KCm1;
+                } else {
+                    dA = (long)(rwrap ? (S - 1 - Q) : -1) * K - KCm1;
+                    dB = (long)C * K - KCm1;
+                }
+            } else { dA = 64; dB = 64; }
+#pragma unroll
+            for (int rr = 0; rr < AR; ++rr) {
+                pA[rr] += dA;
+                if (rs_adv) vm[rr] >>= 1;
+            }
+#pragma unroll
+            for (int rr = 0; rr < BR; ++rr) pB[rr] += dB;
+        }
+    };
+
     auto mfma_step = [&](int buf) {
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
@@ -433,6 +499,16 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
     };
 
     // ---- pipeline: 2 LDS buffers, one reg set, one barrier per step ----
+    if constexpr (GLDS && !SMALL) {
+        glds_step(0);
+        __syncthreads();
+        for (int it = 0; it < nsteps; ++it) {
+            if (it + 1 < nsteps)
+                glds_step((it + 1) & 1);     // flight covers mfma(it)
+            mfma_step(it & 1);
+            __syncthreads();
+        }
+    } else {
     load_step();
     write_lds(0);
     if (nsteps > 1) load_step();
@@ -444,6 +520,7 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         }
         mfma_step(it & 1);
         __syncthreads();
+    }
     }
 
     // ---- epilogue ----
@@ -1627,6 +1704,25 @@ static inline int swz_mode() {
     return m;
 }
 
+// PS_GLDS=0 reverts the aligned conv GEMMs to register staging (A/B).
+static inline int glds_mode() {
+    static int m = -1;
+    if (m < 0) { const char* e = getenv("PS_GLDS"); m = e ? atoi(e) : 1; }
+    return m;
+}
+
+// 16-B zero source for GLDS invalid taps/columns (thread-safe one-time
+// device alloc; first use happens in warmup, before any graph capture)
+static const unsigned short* zpage_ptr() {
+    static const unsigned short* z = [] {
+        void* p = nullptr;
+        (void)hipMalloc(&p, 128);
+        (void)hipMemset(p, 0, 128);
+        return (const unsigned short*)p;
+    }();
+    return z;
+}
+
 #define LAUNCH_GEMM(TM, TN, ST, DG, SM) LAUNCH_GEMM_NB(TM, TN, ST, DG, SM, 2)
 #define LAUNCH_GEMM_NB(TM, TN, ST, DG, SM, NBV)                               \
     do {                                                                      \
@@ -1648,20 +1744,30 @@ static inline int swz_mode() {
             dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,            \
             (const unsigned short*)src, (const unsigned short*)wgt,           \
             (const unsigned short*)bias, (unsigned short*)dst,                \
-            (const unsigned short*)carry,                                     \
+            (const unsigned short*)carry, nullptr,                            \
             Nb, H, W, C, K, P, Q, R, S, pad);                                 \
     } while (0)
-#define LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, NBV, SWV, ACV)                     \
+#define LAUNCH_GEMM_GL(TM, TN, ST, DG, SM, NBV, SWV, ACV, GLV)                \
     do {                                                                      \
         long M_ = DG ? (long)Nb * H * W : (long)Nb * P * Q;                   \
         int Nout_ = DG ? C : K;                                               \
         long grid = ((M_ + TM - 1) / TM) * ((Nout_ + TN - 1) / TN);           \
-        hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, ST, DG, SM, ALV, NBV, SWV, ACV>), \
+        hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, ST, DG, SM, ALV, NBV,    \
+                                             SWV, ACV, 0, GLV>),              \
             dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,            \
             (const unsigned short*)src, (const unsigned short*)wgt,           \
             (const unsigned short*)bias, (unsigned short*)dst,                \
-            (const unsigned short*)carry,                                     \
+            (const unsigned short*)carry, zpage_ptr(),                        \
             Nb, H, W, C, K, P, Q, R, S, pad);                                 \
+    } while (0)
+#define LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, NBV, SWV, ACV)                     \
+    do {                                                                      \
+        if constexpr (ALV && !(SM)) {                                         \
+            if (glds_mode()) LAUNCH_GEMM_GL(TM, TN, ST, DG, SM, NBV, SWV, ACV, true); \
+            else             LAUNCH_GEMM_GL(TM, TN, ST, DG, SM, NBV, SWV, ACV, false); \
+        } else {                                                              \
+            LAUNCH_GEMM_GL(TM, TN, ST, DG, SM, NBV, SWV, ACV, false);         \
+        }                                                                     \
     } while (0)
 
 extern "C" void ps_conv_fwd(
