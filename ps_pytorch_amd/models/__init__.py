@@ -14,13 +14,19 @@ _FACTORY = {
     'resnet50': ResNet50,
     'resnet101': ResNet101,
     'resnet152': ResNet152,
-    'vgg11': VGG11,
+    # ref util.py:19 maps "VGG11" to vgg11_bn — keep that behavior;
+    # the plain variants stay reachable as VGG11_plain etc.
+    'vgg11': VGG11_BN,
+    'vgg11_plain': VGG11,
     'vgg11_bn': VGG11_BN,
-    'vgg13': VGG13,
+    'vgg13_plain': VGG13,
+    'vgg13': VGG13_BN,
     'vgg13_bn': VGG13_BN,
-    'vgg16': VGG16,
+    'vgg16_plain': VGG16,
+    'vgg16': VGG16_BN,
     'vgg16_bn': VGG16_BN,
-    'vgg19': VGG19,
+    'vgg19_plain': VGG19,
+    'vgg19': VGG19_BN,
     'vgg19_bn': VGG19_BN,
 }
 

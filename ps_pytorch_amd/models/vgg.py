@@ -7,6 +7,20 @@ import torch.nn as nn
 from ..ops.conv import PsConv2d
 from ..ops.linear import PsLinear
 from ..ops.modules import PsBatchNorm2d
+from ..ops.pool import max_pool2d as ps_max_pool2d
+
+
+class PsMaxPool2d(nn.Module):
+    """nn.MaxPool2d(2,2) on the in-tree NHWC kernels (torch fallback on
+    CPU); a module so make_layers stays an nn.Sequential."""
+
+    def __init__(self, kernel_size: int, stride: int):
+        super().__init__()
+        self.kernel_size = kernel_size
+        self.stride = stride
+
+    def forward(self, x):
+        return ps_max_pool2d(x, self.kernel_size, self.stride)
 
 _CFG = {
     'VGG11': [64, 'M', 128, 'M', 256, 256, 'M', 512, 512, 'M', 512, 512, 'M'],
@@ -23,7 +37,7 @@ def _make_layers(cfg, in_channels: int, batch_norm: bool) -> nn.Sequential:
     c = in_channels
     for v in cfg:
         if v == 'M':
-            layers.append(nn.MaxPool2d(2, 2))
+            layers.append(PsMaxPool2d(2, 2))
         else:
             layers.append(PsConv2d(c, v, 3, padding=1, bias=not batch_norm))
             if batch_norm:

@@ -284,13 +284,14 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
     f32x8 invstd = *(const f32x8*)&save_invstd[cg * 8];
     f32x8 sum_dy = {0, 0, 0, 0, 0, 0, 0, 0};
     f32x8 sum_dyx = {0, 0, 0, 0, 0, 0, 0, 0};
-    // contiguous 16-row chunks per block iteration: the 8-deep form still
-    // measured 82% SQ_WAIT_ANY (two 16-B streams + mask; latency-bound)
-    const long chunk = 16L * R;
+    // contiguous 8-row chunks per block iteration (16-deep REGRESSED
+    // 467 -> 716 us/step: the wider span thrashes DRAM pages — same
+    // failure mode as round 1's gridDim-strided unroll)
+    const long chunk = 8L * R;
     for (long base = (long)blockIdx.x * chunk + row_in_blk; base < M;
          base += (long)gridDim.x * chunk) {
 #pragma unroll
-        for (int u = 0; u < 16; ++u) {
+        for (int u = 0; u < 8; ++u) {
             long r = base + (long)u * R;
             if (r < M) {
                 long o = r * C + cg * 8;
@@ -445,7 +446,7 @@ static void bn_bwd_t(const void* x, const void* mask, const void* dy,
     int Ci = (int)C;
     float* wsf = (float*)ws;
     dim3 b256(256);
-    int nb = stats_blocks(M, Ci, 16);
+    int nb = stats_blocks(M, Ci, 8);
     if (relu)
         hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, true>), dim3(nb), b256, 0, s,
                            (const T*)x, (const unsigned char*)mask, (const T*)dy,

@@ -1688,11 +1688,22 @@ __global__ __launch_bounds__(256) void colsum_fold_kernel(
     unsigned short* __restrict__ db, const float* __restrict__ partial,
     int nblk, int K)
 {
-    int k = blockIdx.x * 256 + threadIdx.x;
-    if (k >= K) return;
+    // one block per output channel; threads stride the partial rows and
+    // tree-reduce in LDS. (The old 1-2-block serial 512-row scan was a
+    // dependent-latency chain — 117 us per VGG bias grad.)
+    __shared__ float red[256];
+    const int k = blockIdx.x;
     float acc = 0.f;
-    for (int b = 0; b < nblk; ++b) acc += partial[(long)b * K + k];
-    db[k] = f32_to_bf16(acc);
+    for (int b = threadIdx.x; b < nblk; b += 256)
+        acc += partial[(long)b * K + k];
+    red[threadIdx.x] = acc;
+    __syncthreads();
+#pragma unroll
+    for (int w = 128; w; w >>= 1) {
+        if ((int)threadIdx.x < w) red[threadIdx.x] += red[threadIdx.x + w];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) db[k] = f32_to_bf16(red[0]);
 }
 
 // ---------------------------------------------------------------- C API
@@ -2011,7 +2022,7 @@ extern "C" void ps_conv_bias_grad(
     hipLaunchKernelGGL(colsum_part_kernel, dim3(nblk), dim3(256), 0,
                        (hipStream_t)strm, (float*)partial_f32,
                        (const unsigned short*)dout, M, K);
-    hipLaunchKernelGGL(colsum_fold_kernel, dim3((K + 255) / 256), dim3(256), 0,
+    hipLaunchKernelGGL(colsum_fold_kernel, dim3(K), dim3(256), 0,
                        (hipStream_t)strm, (unsigned short*)db,
                        (const float*)partial_f32, nblk, K);
 }
