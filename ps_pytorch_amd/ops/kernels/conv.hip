@@ -1317,6 +1317,204 @@ __global__ __launch_bounds__(256) void conv_wgrad_row_kernel(
     }
 }
 
+
+// GLDS variant of the row-halo wgrad: both operands stage through
+// global_load_lds into LINEAR [row][128 B] images — no staging registers,
+// no guarded ds_write pass (the register/write form was issue-bound:
+// SQ_ACTIVE_INST_ANY 48%). The tr16 fragment reads move to
+// octet-swizzled positions; sigma(row, oct) = oct ^ rho(row) with
+// rho(e) = (((e>>1)&1) + ((e>>3)&1)*2) << 1 spreads every {C+j, C+8+j}
+// 4-row read window over disjoint 8-dword bank blocks (the involution is
+// folded into the SOURCE granule, so DMA stays lane-linear). Invalid
+// pixels source the zero page. TK = 64 only.
+__device__ __forceinline__ int wrho(int e) {
+    return (((e >> 1) & 1) + ((e >> 3) & 1) * 2) << 1;
+}
+
+template <bool AL>
+__global__ __launch_bounds__(256) void conv_wgrad_row_glds_kernel(
+    const unsigned short* __restrict__ dout,  // [Nb,P,Q,K]
+    const unsigned short* __restrict__ in,    // [Nb,H,W,C]
+    float* __restrict__ partial,              // [SPLIT][K][9*C]
+    const unsigned short* __restrict__ zpage,
+    int Nb, int H, int W, int C, int K, int P, int Q,
+    int split, int ipr, int l2q, int l2p, int per_xcd)
+{
+    constexpr int TK = 64;
+    constexpr int PAD = 1;
+    const int ROWS = 32 >> l2q;
+    const int HW2 = W + 2;
+    // linear images: A 32 pixel-rows, B 64 halo rows (>= worst nbq=40;
+    // over-staged rows are zero-page sourced and never read)
+    __shared__ __attribute__((aligned(16))) struct {
+        unsigned short A[2][32][64];
+        unsigned short B[2][64][64];
+    } lds;
+    constexpr unsigned A1 = sizeof(lds.A[0]);
+    constexpr unsigned BB1 = sizeof(lds.B[0]);
+    constexpr int MI = TK / 32;
+    const int tiles_k = (K + TK - 1) / TK;
+    const int tiles_c = (C + 63) >> 6;
+    long l = (long)(blockIdx.x & 7) * per_xcd + (blockIdx.x >> 3);
+    const long nlog = (long)tiles_k * tiles_c * 3 * split;
+    if (l >= nlog) return;
+    int b = (int)l;
+    const int k0 = (b % tiles_k) * TK; b /= tiles_k;
+    const int c0 = (b % tiles_c) * 64; b /= tiles_c;
+    const int r = b % 3; b /= 3;
+    const int sid = b;
+    const long rows_total = (long)Nb * P;
+    const long row0 = (long)sid * ipr;
+    const long row1 = (row0 + ipr < rows_total) ? row0 + ipr : rows_total;
+
+    const int t = threadIdx.x;
+    const int lane = t & 63, wid = t >> 6;
+    const int wm = wid >> 1, wn = wid & 1;
+    const int fr = lane & 15, fq = lane >> 4;
+    f32x4_t acc[3][MI][2];
+#pragma unroll
+    for (int si = 0; si < 3; ++si)
+#pragma unroll
+        for (int i = 0; i < MI; ++i)
+#pragma unroll
+            for (int j = 0; j < 2; ++j) acc[si][i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    const int nsteps = (int)((row1 - row0 + (1 << (5 - l2q)) - 1)
+                             >> (5 - l2q));
+
+    // ---- staging assignments (per-lane constants) ----
+    // A: wave w stages pixels w*8 + (lane>>3), octet (lane&7) ^ rho(pixel)
+    const int apix = wid * 8 + (lane >> 3);
+    const int aoct = ((lane & 7) ^ wrho(apix));
+    const unsigned short* pdout = dout + (row0 << l2q) * K + k0
+                                  + (long)apix * K + aoct * 8;
+    // B: wave w stages halo rows {w*16 + j*8 + (lane>>3) : j 0..1}
+    const long inrow0 = row0 + r - PAD;
+    const unsigned short* pin0 = in + inrow0 * W * C + c0;
+    int b_rib[2], b_w[2], b_oct[2];
+    bool b_wok[2];
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+        int hp = wid * 16 + j * 8 + (lane >> 3);
+        b_oct[j] = ((lane & 7) ^ wrho(hp));
+        b_rib[j] = hp / HW2;            // halo row -> (in-row, w-1)
+        int wpx = hp - b_rib[j] * HW2 - 1;
+        b_wok[j] = wpx >= 0 && wpx < W && hp < ROWS * HW2;
+        b_w[j] = wpx < 0 ? 0 : wpx;
+    }
+    long prow = row0;
+
+    auto glds_stage = [&](int buf) {
+        {   // A: one 1-KiB piece per wave
+            bool ok = (prow + (apix >> l2q)) < rows_total && apix < 32;
+            const unsigned short* gsrc = ok ? pdout : zpage;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gsrc,
+                (__attribute__((address_space(3))) void*)
+                    &lds.A[buf][wid * 8][0], 16, 0, 0);
+        }
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {   // B: two pieces per wave
+            int rib = b_rib[j];
+            int p = (int)((prow + rib) & (P - 1));
+            int h = p + r - PAD;
+            bool ok = b_wok[j] && h >= 0 && h < H
+                      && (prow + rib) < rows_total;
+            const unsigned short* gsrc = ok
+                ? pin0 + ((long)rib * W + b_w[j]) * C + b_oct[j] * 8
+                : zpage;
+            __builtin_amdgcn_global_load_lds(
+                (const __attribute__((address_space(1))) void*)gsrc,
+                (__attribute__((address_space(3))) void*)
+                    &lds.B[buf][wid * 16 + j * 8][0], 16, 0, 0);
+        }
+        pdout += (long)32 * K;
+        pin0 += (long)ROWS * W * C;
+        prow += ROWS;
+    };
+
+    // ---- tr16 read offsets (octet-swizzled linear image) ----
+    const char* lb0 = (const char*)&lds;
+    unsigned roA[MI][2];
+    unsigned roB[3][2][2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+        int e = fq * 8 + i * 4 + (fr >> 2);
+        int qd = fr & 3;                      // quad within 16-ch subtile
+#pragma unroll
+        for (int mi = 0; mi < MI; ++mi) {
+            int o = (wm * MI + mi) * 2 + (qd >> 1);
+            roA[mi][i] = (unsigned)((char*)&lds.A[0][e]
+                             [(o ^ wrho(e)) * 8 + (qd & 1) * 4] - lb0);
+        }
+        int rib = e >> l2q, q = e & (Q - 1);
+#pragma unroll
+        for (int si = 0; si < 3; ++si) {
+            int hp = rib * HW2 + q + si;
+#pragma unroll
+            for (int nj = 0; nj < 2; ++nj) {
+                int o = (wn * 2 + nj) * 2 + (qd >> 1);
+                roB[si][nj][i] = (unsigned)((char*)&lds.B[0][hp]
+                                 [(o ^ wrho(hp)) * 8 + (qd & 1) * 4] - lb0);
+            }
+        }
+    }
+
+    union U64x8 { bf16x4_t h[2]; bf16x8_t v; };
+    auto mfma_step = [&](int buf) {
+        U64x8 a[MI];
+#pragma unroll
+        for (int mi = 0; mi < MI; ++mi) {
+            a[mi].h[0] = ds_tr16p((const unsigned short*)
+                (lb0 + roA[mi][0] + (buf ? A1 : 0)));
+            a[mi].h[1] = ds_tr16p((const unsigned short*)
+                (lb0 + roA[mi][1] + (buf ? A1 : 0)));
+        }
+#pragma unroll
+        for (int si = 0; si < 3; ++si) {
+            U64x8 bf[2];
+#pragma unroll
+            for (int nj = 0; nj < 2; ++nj) {
+                bf[nj].h[0] = ds_tr16p((const unsigned short*)
+                    (lb0 + roB[si][nj][0] + (buf ? BB1 : 0)));
+                bf[nj].h[1] = ds_tr16p((const unsigned short*)
+                    (lb0 + roB[si][nj][1] + (buf ? BB1 : 0)));
+            }
+#pragma unroll
+            for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+                for (int nj = 0; nj < 2; ++nj)
+                    acc[si][mi][nj] = MFMA_BF16(a[mi].v, bf[nj].v,
+                                                acc[si][mi][nj]);
+        }
+    };
+
+    glds_stage(0);
+    __syncthreads();
+    for (int it = 0; it < nsteps; ++it) {
+        if (it + 1 < nsteps)
+            glds_stage((it + 1) & 1);
+        mfma_step(it & 1);
+        __syncthreads();
+    }
+
+    const int RSC = 9 * C;
+    float* dstp = partial + (long)sid * K * RSC;
+#pragma unroll
+    for (int si = 0; si < 3; ++si)
+#pragma unroll
+    for (int mi = 0; mi < MI; ++mi)
+#pragma unroll
+    for (int nj = 0; nj < 2; ++nj)
+#pragma unroll
+    for (int e = 0; e < 4; ++e) {
+        int k = k0 + wm * (MI * 16) + mi * 16 + fq * 4 + e;
+        int c = c0 + wn * 32 + nj * 16 + fr;
+        if (k < K && c < C)
+            dstp[(long)k * RSC + (r * 3 + si) * C + c] = acc[si][mi][nj][e];
+    }
+}
+
 // Small-RSC wgrad (ResNet stem 3x3x3=27, LeNet conv1 5x5x1=25): the whole
 // flattened (r,s,c) axis fits one 64-column tile, so one block covers every
 // tap in a single pass over its pixel chunk — vs the generic kernel's
@@ -1960,6 +2158,11 @@ extern "C" void ps_conv_wgrad(
         int l2q_ = ilog2_exact(Q), l2pq_ = ilog2_exact((long)P * Q);
         if (row_en && stride == 1 && R == 3 && S == 3 && pad == 1 && P == H
             && Q == W && l2q_ >= 0 && l2pq_ >= 0 && Q <= 32) {
+            static int wg_glds = -1;
+            if (wg_glds < 0) {
+                const char* e = getenv("PS_WG_GLDS");
+                wg_glds = e ? atoi(e) : 0;
+            }
             const int TKr = 64;     // TK=128 acc pressure costs a wave/SIMD
             int tkr = (K + TKr - 1) / TKr, tcr = (C + 63) / 64;
             long rows_total = (long)Nb * P;
@@ -1976,7 +2179,19 @@ extern "C" void ps_conv_wgrad(
                 (const unsigned short*)dout, (const unsigned short*)in,       \
                 (float*)partial_f32, Nb, H, W, C, K, P, Q,                    \
                 split, (int)ipr_l, l2q_, 0, pxc)
-            if (alr) WGR(64, true); else WGR(64, false);
+#define WGRG(ALV)                                                             \
+            hipLaunchKernelGGL((conv_wgrad_row_glds_kernel<ALV>),             \
+                dim3((unsigned)grid_r), dim3(256), 0, (hipStream_t)strm,      \
+                (const unsigned short*)dout, (const unsigned short*)in,       \
+                (float*)partial_f32, zpage_ptr(), Nb, H, W, C, K, P, Q,       \
+                split, (int)ipr_l, l2q_, 0, pxc)
+            // GLDS form stages unconditional 16-B pieces: aligned
+            // channel/K counts only (alr); ragged shapes keep the
+            // register-staged kernel
+            if (wg_glds && alr) WGRG(true);
+            else if (alr)       WGR(64, true);
+            else                WGR(64, false);
+#undef WGRG
 #undef WGR
             long n_ = (long)K * 9 * C;
             launch_reduce_slabs((unsigned short*)dw, (float*)partial_f32, n_,
