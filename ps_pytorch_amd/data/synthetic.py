@@ -40,7 +40,8 @@ class SyntheticDataset:
     def __init__(self, dataset: str = 'MNIST', split: str = 'train',
                  size: Optional[int] = None, seed: int = 1,
                  device: torch.device = torch.device('cpu'),
-                 dtype: torch.dtype = torch.float32):
+                 dtype: torch.dtype = torch.float32,
+                 template_seed: Optional[int] = None):
         d = dataset.lower()
         self.name = d
         self.shape = input_shape_of(d)
@@ -48,7 +49,11 @@ class SyntheticDataset:
         n = size if size is not None else (_TRAIN_SIZE if split == 'train'
                                            else _TEST_SIZE).get(d, 10000)
         g = torch.Generator().manual_seed(seed * 1000003 + (0 if split == 'train' else 1))
-        tg = torch.Generator().manual_seed(seed)  # split-independent templates
+        # class templates must be keyed by the JOB seed only: the per-rank
+        # sample seed must NOT leak in, or every shard (and the test split)
+        # would model a different classification problem
+        tg = torch.Generator().manual_seed(
+            seed if template_seed is None else template_seed)
         templates = 0.5 * torch.randn((self.num_classes,) + self.shape, generator=tg)
         y = torch.randint(0, self.num_classes, (n,), generator=g)
         x = torch.randn((n,) + self.shape, generator=g) * 0.5 + templates[y]
@@ -124,8 +129,9 @@ def prepare_data(args_or_cfg, rank: int = 0, num_shards: int = 1,
     bs = getattr(cfg, 'batch_size', 128)
     tbs = getattr(cfg, 'test_batch_size', 500)
     train = SyntheticDataset(dataset, 'train', size=train_size,
-                             seed=seed * 131 + rank, device=device, dtype=dtype)
+                             seed=seed * 131 + rank, device=device,
+                             dtype=dtype, template_seed=seed)
     test = SyntheticDataset(dataset, 'test', size=test_size, seed=seed,
-                            device=device, dtype=dtype)
+                            device=device, dtype=dtype, template_seed=seed)
     return (ResidentLoader(train, bs, shuffle=True, seed=seed + rank),
             ResidentLoader(test, tbs, shuffle=False, drop_last=False))
