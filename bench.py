@@ -82,6 +82,13 @@ def main():
         import torch.distributed as dist
         env = init_distributed()
         device = env['device']
+        # timing barriers run on a HOST-SIDE gloo group: a barrier on the
+        # main (RCCL) communicator would be a device collective that the PS
+        # enqueues AFTER its pipelined tail broadcasts while workers
+        # enqueue it BEFORE their matching receives — mismatched collective
+        # order deadlocks NCCL (gloo tolerates it, which is why CPU tests
+        # can't catch this one)
+        barrier_grp = dist.new_group(backend='gloo')
         from ps_pytorch_amd.parallel.ps import ParameterServer
         from ps_pytorch_amd.parallel.worker import DistributedWorker
         if args.engine == 'allreduce':
@@ -118,7 +125,7 @@ def main():
             step()
         if use_cuda:
             torch.cuda.synchronize()
-        dist.barrier()
+        dist.barrier(group=barrier_grp)
         if use_cuda:
             torch.cuda.synchronize()
         t0 = time.time()
@@ -126,7 +133,7 @@ def main():
             step()
         if use_cuda:
             torch.cuda.synchronize()
-        dist.barrier()
+        dist.barrier(group=barrier_grp)
         if use_cuda:
             torch.cuda.synchronize()
         elapsed = time.time() - t0
