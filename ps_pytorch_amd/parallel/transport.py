@@ -453,6 +453,10 @@ class PSTransport:
         self._next_push = 0
 
     def barrier(self) -> None:
+        """Device-collective barrier on the MAIN group. Do NOT call this
+        between a pipelined tail broadcast (PS) and the workers' next fetch
+        — per-rank collective order would diverge and deadlock NCCL. Use a
+        gloo side-group for host-side synchronization (bench.py does)."""
         if dist.is_initialized():
             dist.barrier(group=self.group)
 
