@@ -52,7 +52,7 @@ def _try_load() -> None:
                                      ctypes.c_long, ctypes.c_int, ctypes.c_void_p]
         lib.ps_acc.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
                                ctypes.c_long, ctypes.c_int, ctypes.c_void_p]
-        lib.ps_conv_fwd.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int] * 11 + [ctypes.c_void_p]
+        lib.ps_conv_fwd.argtypes = [ctypes.c_void_p] * 5 + [ctypes.c_int] * 11 + [ctypes.c_void_p]
         lib.ps_conv_dgrad.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int] * 11 + [ctypes.c_void_p]
         lib.ps_conv_wgrad.argtypes = [ctypes.c_void_p] * 4 + [ctypes.c_int] * 12 + [ctypes.c_void_p]
         lib.ps_conv_bias_grad.argtypes = [ctypes.c_void_p] * 3 + [
@@ -62,7 +62,8 @@ def _try_load() -> None:
         lib.ps_fused_adam.argtypes = [ctypes.c_void_p] * 6 + [
             ctypes.c_long] + [ctypes.c_float] * 7 + [
             ctypes.c_int, ctypes.c_int, ctypes.c_void_p]
-        lib.ps_bn_fwd.argtypes = [ctypes.c_void_p] * 12 + [
+        lib.ps_bn_fwd.argtypes = [ctypes.c_void_p] * 13 + [
+            ctypes.c_int,
             ctypes.c_long, ctypes.c_long, ctypes.c_float, ctypes.c_float,
             ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p]
         lib.ps_bn_bwd.argtypes = [ctypes.c_void_p] * 12 + [

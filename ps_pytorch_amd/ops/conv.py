@@ -107,10 +107,26 @@ class _ConvFn(torch.autograd.Function):
         # full transposing copy of uninitialized memory (~190us at l1 size)
         out = torch.empty((Nb, K, P, Q), dtype=x.dtype, device=x.device,
                           memory_format=_CL)
+        # generic (NBUF=2) fwd paths also emit per-channel sum/sumsq
+        # partials for a following BatchNorm (attribute handshake); MUST
+        # mirror conv.hip's FWD_BODY dispatch
+        rsc = R * S * C
+        generic_fwd = not ((R * S > 1 and (rsc <= 192 or
+                                           (C == 4 and R * S * 4 <= 256) or
+                                           (C % 8 == 0 and C % 64 and rsc <= 768)))
+                           or (R * S == 1 and C <= 64))
+        stats = None
+        if generic_fwd and K % 8 == 0 and torch.is_grad_enabled():
+            tiles_m = (Nb * P * Q + 127) // 128
+            stats = torch.empty(tiles_m * 2 * K, dtype=torch.float32,
+                                device=x.device)
         lib.ps_conv_fwd(x.data_ptr(), wc.data_ptr(),
                         b.data_ptr() if b is not None else 0, out.data_ptr(),
+                        stats.data_ptr() if stats is not None else 0,
                         Nb, H, W, C, K, P, Q, R, S, stride, pad,
                         current_stream_ptr())
+        if stats is not None:
+            out._ps_bn_stats = (stats, (Nb * P * Q + 127) // 128)
         ctx.save_for_backward(x, wc)
         ctx.conf = (stride, pad, b is not None)
         # steal-mode targets: write dw/db straight into the flat_g slices

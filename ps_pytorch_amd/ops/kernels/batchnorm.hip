@@ -384,6 +384,7 @@ template <typename T, typename PT>
 static void bn_fwd_t(const void* x, void* y, const void* gamma, const void* beta,
                      void* rmean, void* rvar, void* smean, void* sinvstd,
                      void* ws, void* partial, const void* res, void* mask,
+                     const void* ext_stats, int ext_nblk,
                      long M, long C,
                      float momentum, float eps, int training, int relu,
                      hipStream_t s)
@@ -392,11 +393,18 @@ static void bn_fwd_t(const void* x, void* y, const void* gamma, const void* beta
     float* wsf = (float*)ws;
     dim3 b256(256);
     if (training) {
-        int nb = stats_blocks(M, Ci, 8);
-        hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(nb), b256, 0, s,
-                           (const T*)x, (float*)partial, M, Ci);
+        // ext_stats: sum/sumsq partials already produced by the producing
+        // conv's epilogue ([ext_nblk][2C]) — skip the stats re-read pass
+        const float* part = ext_stats ? (const float*)ext_stats
+                                      : (const float*)partial;
+        int nb = ext_nblk;
+        if (!ext_stats) {
+            nb = stats_blocks(M, Ci, 8);
+            hipLaunchKernelGGL((bn_stats_kernel<T>), dim3(nb), b256, 0, s,
+                               (const T*)x, (float*)partial, M, Ci);
+        }
         hipLaunchKernelGGL((bn_fwd_foldfin_kernel<PT>), dim3(Ci), b256, 0, s,
-                           (const float*)partial, wsf, (const PT*)gamma,
+                           part, wsf, (const PT*)gamma,
                            (const PT*)beta, (float*)rmean, (float*)rvar,
                            (float*)smean, (float*)sinvstd, nb, M, Ci,
                            momentum, eps);
@@ -424,17 +432,20 @@ extern "C" void ps_bn_fwd(const void* x, void* y, const void* gamma,
                           const void* beta, void* rmean, void* rvar,
                           void* smean, void* sinvstd, void* ws, void* partial,
                           const void* res, void* mask,
+                          const void* ext_stats, int ext_nblk,
                           long M, long C, float momentum, float eps,
                           int training, int relu, int dtype, void* stream)
 {
     hipStream_t s = (hipStream_t)stream;
     if (dtype == PS_BF16)
         bn_fwd_t<unsigned short, unsigned short>(x, y, gamma, beta, rmean, rvar,
-                                                 smean, sinvstd, ws, partial, res, mask, M, C,
+                                                 smean, sinvstd, ws, partial, res, mask,
+                                                 ext_stats, ext_nblk, M, C,
                                                  momentum, eps, training, relu, s);
     else
         bn_fwd_t<float, float>(x, y, gamma, beta, rmean, rvar, smean, sinvstd,
-                               ws, partial, res, mask, M, C, momentum, eps, training, relu, s);
+                               ws, partial, res, mask, ext_stats, ext_nblk,
+                               M, C, momentum, eps, training, relu, s);
 }
 
 template <typename T, typename PT>

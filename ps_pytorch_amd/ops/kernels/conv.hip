@@ -152,11 +152,16 @@ template <int TM, int TN, int STRIDE, bool DGRAD, bool SMALL = false, bool AL = 
           int CPM = 0,    // SMALL padded-channel gather: 4 = C==4 stem
                           // (two 8-B tap halves), 8 = C%8==0 (quantum
                           // within one tap: single 16-B load), 0 = off
-          bool GLDS = false>  // stage via global_load_lds (AL && !SMALL):
+          bool GLDS = false,  // stage via global_load_lds (AL && !SMALL):
                               // drops the ds_write pass + staging VGPRs;
                               // invalid taps load from the zero page and
                               // the granule swizzle moves to the SOURCE
                               // address (self-inverse XOR, guide rule 21)
+          bool STATS = false> // fwd epilogue also accumulates per-channel
+                              // sum/sumsq of the (rounded) outputs into
+                              // stats[tile_m][2*Nout] — the following
+                              // BatchNorm's stats pass without re-reading
+                              // the activation (bn_fwd_foldfin consumes it)
 __global__ __launch_bounds__(256) void conv_gemm_kernel(
     const unsigned short* __restrict__ src,  // fwd: in [Nb,H,W,C]; dgrad: dout [Nb,P,Q,K]
     const unsigned short* __restrict__ wgt,  // fwd: w [K,R,S,C]; dgrad: wT [R,S,C,K]
@@ -165,6 +170,7 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
     const unsigned short* __restrict__ carry,// [dst shape] or null (ACCF)
     const unsigned short* __restrict__ zpage,// 64 zero shorts (GLDS source
                                              // for invalid taps/cols)
+    float* __restrict__ stats,               // [tiles_m][2*Nout] or null
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int pad)
 {
@@ -551,6 +557,10 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
         const int segc = (t % SEG) * 8;
         const int rl0 = t / SEG;
         const bool vec_ok = (Nout & 7) == 0;   // 16-B-aligned row segments
+        float ssum[STATS ? 8 : 1], ssq[STATS ? 8 : 1];
+        if constexpr (STATS)
+#pragma unroll
+            for (int k2 = 0; k2 < 8; ++k2) { ssum[k2] = 0.f; ssq[k2] = 0.f; }
 #pragma unroll
         for (int ps = 0; ps < TM / RPP; ++ps) {
             int rl = ps * RPP + rl0;
@@ -572,13 +582,52 @@ __global__ __launch_bounds__(256) void conv_gemm_kernel(
                 }
 #pragma unroll
                 for (int k2 = 0; k2 < 8; ++k2) o.us[k2] = f32_to_bf16(v[k2]);
+                if constexpr (STATS)
+#pragma unroll
+                    for (int k2 = 0; k2 < 8; ++k2) {
+                        float rv = bf16_to_f32(o.us[k2]);   // rounded value
+                        ssum[k2] += rv;
+                        ssq[k2] += rv * rv;
+                    }
                 *(uint4*)&dst[om * Nout + col] = o.u4;
             } else {
                 for (int k2 = 0; k2 < 8 && col + k2 < Nout; ++k2) {
                     float vv = v[k2];
                     if constexpr (ACCF)
                         vv += bf16_to_f32(carry[om * Nout + col + k2]);
-                    dst[om * Nout + col + k2] = f32_to_bf16(vv);
+                    unsigned short ov = f32_to_bf16(vv);
+                    if constexpr (STATS) {
+                        float rv = bf16_to_f32(ov);
+                        ssum[k2] += rv;
+                        ssq[k2] += rv * rv;
+                    }
+                    dst[om * Nout + col + k2] = ov;
+                }
+            }
+        }
+        if constexpr (STATS) {
+            // reduce the RPP per-thread partials of each channel octet
+            // through LDS (reuse sf; deterministic fixed tree) and write
+            // this block's [2*Nout] stats row (tile_n slices are disjoint)
+            __syncthreads();
+            float* sp = sf;                       // [RPP][SEG][8] (+ sq)
+#pragma unroll
+            for (int k2 = 0; k2 < 8; ++k2) {
+                sp[(rl0 * SEG + t % SEG) * 8 + k2] = ssum[k2];
+                sp[(RPP * SEG + rl0 * SEG + t % SEG) * 8 + k2] = ssq[k2];
+            }
+            __syncthreads();
+            if (rl0 == 0) {
+                for (int rr2 = 1; rr2 < RPP; ++rr2)
+#pragma unroll
+                    for (int k2 = 0; k2 < 8; ++k2) {
+                        ssum[k2] += sp[(rr2 * SEG + t % SEG) * 8 + k2];
+                        ssq[k2] += sp[(RPP * SEG + rr2 * SEG + t % SEG) * 8 + k2];
+                    }
+                float* dst2 = stats + tile_m * 2 * (long)Nout + n0 + segc;
+                for (int k2 = 0; k2 < 8 && n0 + segc + k2 < Nout; ++k2) {
+                    dst2[k2] = ssum[k2];
+                    dst2[Nout + k2] = ssq[k2];
                 }
             }
         }
@@ -1953,21 +2002,26 @@ static const unsigned short* zpage_ptr() {
             dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,            \
             (const unsigned short*)src, (const unsigned short*)wgt,           \
             (const unsigned short*)bias, (unsigned short*)dst,                \
-            (const unsigned short*)carry, nullptr,                            \
+            (const unsigned short*)carry, nullptr, nullptr,                   \
             Nb, H, W, C, K, P, Q, R, S, pad);                                 \
     } while (0)
-#define LAUNCH_GEMM_GL(TM, TN, ST, DG, SM, NBV, SWV, ACV, GLV)                \
+#define LAUNCH_GEMM_GLS(TM, TN, ST, DG, SM, NBV, SWV, ACV, GLV, STV)          \
     do {                                                                      \
         long M_ = DG ? (long)Nb * H * W : (long)Nb * P * Q;                   \
         int Nout_ = DG ? C : K;                                               \
         long grid = ((M_ + TM - 1) / TM) * ((Nout_ + TN - 1) / TN);           \
         hipLaunchKernelGGL((conv_gemm_kernel<TM, TN, ST, DG, SM, ALV, NBV,    \
-                                             SWV, ACV, 0, GLV>),              \
+                                             SWV, ACV, 0, GLV, STV>),         \
             dim3((unsigned)grid), dim3(256), 0, (hipStream_t)strm,            \
             (const unsigned short*)src, (const unsigned short*)wgt,           \
             (const unsigned short*)bias, (unsigned short*)dst,                \
-            (const unsigned short*)carry, zpage_ptr(),                        \
+            (const unsigned short*)carry, zpage_ptr(), (float*)stats,         \
             Nb, H, W, C, K, P, Q, R, S, pad);                                 \
+    } while (0)
+#define LAUNCH_GEMM_GL(TM, TN, ST, DG, SM, NBV, SWV, ACV, GLV)                \
+    do {                                                                      \
+        if (!(DG) && stats) LAUNCH_GEMM_GLS(TM, TN, ST, DG, SM, NBV, SWV, ACV, GLV, true); \
+        else                LAUNCH_GEMM_GLS(TM, TN, ST, DG, SM, NBV, SWV, ACV, GLV, false); \
     } while (0)
 #define LAUNCH_GEMM_SW(TM, TN, ST, DG, SM, NBV, SWV, ACV)                     \
     do {                                                                      \
@@ -1979,8 +2033,12 @@ static const unsigned short* zpage_ptr() {
         }                                                                     \
     } while (0)
 
+// stats != null: the epilogue also writes per-channel sum/sumsq partials
+// [ceil(M/128)][2*K] for the following BatchNorm (generic NBUF=2 fwd paths
+// only — the python side mirrors the dispatch to know when to allocate).
 extern "C" void ps_conv_fwd(
     const void* src, const void* wgt, const void* bias, void* dst,
+    void* stats,
     int Nb, int H, int W, int C, int K, int P, int Q,
     int R, int S, int stride, int pad, void* strm)
 {
@@ -2031,6 +2089,7 @@ extern "C" void ps_conv_dgrad(
     int R, int S, int stride, int pad, void* strm)
 {
     const void* bias = nullptr;
+    const void* stats = nullptr;       // fwd-only epilogue feature
     bool al = (K & 63) == 0;           // contraction runs over K
     if (stride == 1) {
 #define DG_BODY(ACV)                                                          \

@@ -49,7 +49,7 @@ class _LinearFn(torch.autograd.Function):
         N = wc.shape[0]
         y = torch.empty((M, N), dtype=x.dtype, device=x.device)
         lib.ps_conv_fwd(x2.data_ptr(), wc.data_ptr(),
-                        b.data_ptr() if b is not None else 0, y.data_ptr(),
+                        b.data_ptr() if b is not None else 0, y.data_ptr(), 0,
                         M, 1, 1, K, N, 1, 1, 1, 1, 1, 0,
                         current_stream_ptr())
         ctx.save_for_backward(x2, wc)

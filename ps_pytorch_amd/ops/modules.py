@@ -46,6 +46,9 @@ class _FusedBNFn(torch.autograd.Function):
         # this instead of re-reading the full y tensor for the mask
         mask = (torch.empty(M * (C // 8), dtype=torch.uint8, device=x.device)
                 if relu else None)
+        # stats handshake: the producing conv's epilogue may have already
+        # written this activation's per-channel sum/sumsq partials
+        ext = getattr(x, '_ps_bn_stats', None)
         lib.ps_bn_fwd(
             x.data_ptr(), y.data_ptr(), gamma.data_ptr(), beta.data_ptr(),
             running_mean.data_ptr(), running_var.data_ptr(),
@@ -53,6 +56,8 @@ class _FusedBNFn(torch.autograd.Function):
             partial.data_ptr(),
             residual.data_ptr() if residual is not None else 0,
             mask.data_ptr() if mask is not None else 0,
+            ext[0].data_ptr() if ext is not None else 0,
+            ext[1] if ext is not None else 0,
             M, C, float(momentum), float(eps), 1, int(relu),
             dtype_tag(x.dtype), current_stream_ptr())
         ctx.save_for_backward(x, gamma, save_mean, save_invstd)
@@ -105,6 +110,7 @@ def _bn_eval_fused(x, gamma, beta, residual, rmean, rvar, eps, relu):
                   beta.data_ptr(), rmean.data_ptr(), rvar.data_ptr(),
                   0, 0, ws.data_ptr(), 0,
                   residual.data_ptr() if residual is not None else 0, 0,
+                  0, 0,
                   N * H * W, C, 0.0, float(eps), 0, int(relu),
                   dtype_tag(x.dtype), current_stream_ptr())
     return y
