@@ -24,6 +24,18 @@ _CL = torch.channels_last
 # A/B kill-switch: PS_CONV=0 routes every conv through torch/MIOpen.
 _ENABLED = os.environ.get('PS_CONV', '1') != '0'
 
+# BN-stats handshake sidechannel: _ConvFn.forward's epilogue may produce
+# per-channel sum/sumsq partials; apply() re-wraps the output tensor, so
+# the Function stashes them here and the module wrapper attaches them to
+# the wrapped output (sequential per caller thread — no races).
+_PENDING_STATS = None
+
+
+def _pop_stats():
+    global _PENDING_STATS
+    st, _PENDING_STATS = _PENDING_STATS, None
+    return st
+
 
 def _supported(x: torch.Tensor, w: torch.Tensor, stride, padding,
                dilation, groups) -> bool:
@@ -126,7 +138,11 @@ class _ConvFn(torch.autograd.Function):
                         Nb, H, W, C, K, P, Q, R, S, stride, pad,
                         current_stream_ptr())
         if stats is not None:
-            out._ps_bn_stats = (stats, (Nb * P * Q + 127) // 128)
+            # NOTE: apply() re-wraps forward's return value, so a python
+            # attribute set HERE would be lost — stash it for the module
+            # wrapper to attach to the wrapped output
+            global _PENDING_STATS
+            _PENDING_STATS = (stats, (Nb * P * Q + 127) // 128)
         ctx.save_for_backward(x, wc)
         ctx.conf = (stride, pad, b is not None)
         # steal-mode targets: write dw/db straight into the flat_g slices
@@ -248,8 +264,12 @@ def conv_with_passthrough(mod: nn.Conv2d, x: torch.Tensor):
     residual branch so its gradient fuses into dgrad; otherwise plain."""
     if _CARRY and _supported(x, mod.weight, mod.stride, mod.padding,
                              mod.dilation, mod.groups) and x.requires_grad:
-        return _ConvCarryFn.apply(x, mod.weight, mod.bias,
-                                  mod.stride[0], mod.padding[0])
+        out, xp = _ConvCarryFn.apply(x, mod.weight, mod.bias,
+                                     mod.stride[0], mod.padding[0])
+        st = _pop_stats()
+        if st is not None:
+            out._ps_bn_stats = st
+        return out, xp
     return mod(x), x
 
 
@@ -259,6 +279,10 @@ class PsConv2d(nn.Conv2d):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if _supported(x, self.weight, self.stride, self.padding,
                       self.dilation, self.groups):
-            return _ConvFn.apply(x, self.weight, self.bias,
-                                 self.stride[0], self.padding[0])
+            out = _ConvFn.apply(x, self.weight, self.bias,
+                                self.stride[0], self.padding[0])
+            st = _pop_stats()
+            if st is not None:
+                out._ps_bn_stats = st
+            return out
         return super().forward(x)
