@@ -128,7 +128,10 @@ class _ConvFn(torch.autograd.Function):
                                            (C % 8 == 0 and C % 64 and rsc <= 768)))
                            or (R * S == 1 and C <= 64))
         stats = None
-        if generic_fwd and K % 8 == 0 and torch.is_grad_enabled():
+        # NB: grad mode is force-disabled inside Function.forward, so
+        # torch.is_grad_enabled() is useless here; the param's
+        # requires_grad is the training signal (eval pays a few % of fwd)
+        if generic_fwd and K % 8 == 0 and w.requires_grad:
             tiles_m = (Nb * P * Q + 127) // 128
             stats = torch.empty(tiles_m * 2 * K, dtype=torch.float32,
                                 device=x.device)
