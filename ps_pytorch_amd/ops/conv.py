@@ -29,6 +29,7 @@ _ENABLED = os.environ.get('PS_CONV', '1') != '0'
 # the Function stashes them here and the module wrapper attaches them to
 # the wrapped output (sequential per caller thread — no races).
 _PENDING_STATS = None
+_BN_FUSE = os.environ.get('PS_BN_FUSE', '0') == '1'
 
 
 def _pop_stats():
@@ -128,10 +129,14 @@ class _ConvFn(torch.autograd.Function):
                                            (C % 8 == 0 and C % 64 and rsc <= 768)))
                            or (R * S == 1 and C <= 64))
         stats = None
+        # Measured NET NEGATIVE at -2.4% e2e (the epilogue's per-block
+        # serial stats reduce costs more than the bn_stats pass it saves:
+        # fwd 128x64 138 -> 245 us/call while bn_stats dropped 272 -> 38
+        # us/step) — default off; PS_BN_FUSE=1 re-enables the experiment.
         # NB: grad mode is force-disabled inside Function.forward, so
         # torch.is_grad_enabled() is useless here; the param's
-        # requires_grad is the training signal (eval pays a few % of fwd)
-        if generic_fwd and K % 8 == 0 and w.requires_grad:
+        # requires_grad is the training signal.
+        if (_BN_FUSE and generic_fwd and K % 8 == 0 and w.requires_grad):
             tiles_m = (Nb * P * Q + 127) // 128
             stats = torch.empty(tiles_m * 2 * K, dtype=torch.float32,
                                 device=x.device)
