@@ -2068,6 +2068,9 @@ extern "C" void ps_conv_fwd(
         } else if (K >= 128) {                                                \
             if (stride == 1) LAUNCH_GEMM(128, 128, 1, false, false);          \
             else             LAUNCH_GEMM(128, 128, 2, false, false);          \
+        } else if (K <= 32) {   /* skinny outputs: 2x the column util */      \
+            if (stride == 1) LAUNCH_GEMM(128, 32, 1, false, false);           \
+            else             LAUNCH_GEMM(128, 32, 2, false, false);           \
         } else {                                                              \
             if (stride == 1) LAUNCH_GEMM(128, 64, 1, false, false);           \
             else             LAUNCH_GEMM(128, 64, 2, false, false);           \
@@ -2094,8 +2097,9 @@ extern "C" void ps_conv_dgrad(
     if (stride == 1) {
 #define DG_BODY(ACV)                                                          \
         do {                                                                  \
-            if (C >= 128) LAUNCH_GEMM_ACC(128, 128, 1, true, false, ACV);     \
-            else          LAUNCH_GEMM_ACC(128, 64, 1, true, false, ACV);      \
+            if (C >= 128)     LAUNCH_GEMM_ACC(128, 128, 1, true, false, ACV); \
+            else if (C <= 32) LAUNCH_GEMM_ACC(128, 32, 1, true, false, ACV);  \
+            else              LAUNCH_GEMM_ACC(128, 64, 1, true, false, ACV);  \
         } while (0)
         if (al) { constexpr bool ALV = true;
                   if (carry) DG_BODY(true); else DG_BODY(false); }
