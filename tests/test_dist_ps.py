@@ -122,3 +122,13 @@ def test_bcast_pipeline_mode_matrix(monkeypatch):
     assert not mk(mode='collective', comm_type='Async').bcast_bucketed
     monkeypatch.setenv('PS_BCAST_PIPE', '0')
     assert not mk(mode='collective', comm_type='Bcast').bcast_bucketed
+
+
+def test_golden_step_bcast_pipeline_off(monkeypatch):
+    """PS_BCAST_PIPE=0 (single-collective broadcast) must produce the same
+    golden master as the default pipelined per-bucket path."""
+    monkeypatch.setenv('PS_BCAST_PIPE', '0')
+    res = run_dist(_role, world=3, args=(True,))
+    got = torch.from_numpy(res[0])
+    ref = _serial_reference()
+    assert torch.allclose(got, ref, atol=1e-5, rtol=1e-5), (got - ref).abs().max()
