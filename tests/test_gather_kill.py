@@ -212,3 +212,45 @@ def test_timeout_killed_worker_excluded_from_quota():
     got = torch.from_numpy(res[0])
     ref = _serial(_identity, workers=(1,), same_data=True, scale=1.0)
     assert torch.allclose(got, ref, atol=1e-5, rtol=1e-5), (got - ref).abs().max()
+
+
+def _role_dead_worker(rank: int, world: int, port: int):
+    """Worker 2 'crashes' (never pushes); PS must raise the drain-liveness
+    error instead of hanging forever (PS_DRAIN_TIMEOUT)."""
+    import os
+    os.environ['PS_DRAIN_TIMEOUT'] = '3'
+    from ps_pytorch_amd.parallel.transport import init_distributed
+    from ps_pytorch_amd.parallel.ps import ParameterServer
+    from ps_pytorch_amd.parallel.worker import DistributedWorker
+    cfg = _cfg(max_steps=1)
+    env = init_distributed(backend='gloo')
+    if rank == 0:
+        ps = ParameterServer(cfg, rank, world, env['device'])
+        ps.build_model(10)
+        try:
+            ps.step()
+        except RuntimeError as e:
+            return 'timeout' if 'stalled' in str(e) else f'other: {e}'
+        return 'no-error'
+    w = DistributedWorker(cfg, rank, world, env['device'])
+    w.build_model(10)
+    if rank == 1:
+        xs, ys = _batches(rank)
+        try:
+            w.train_step(xs[0], ys[0])   # blocks in wait_all (PS died) — ok
+        except Exception:
+            pass
+    # rank 2: participate in the weight broadcast, then go silent
+    elif rank == 2:
+        try:
+            w.fetch_weights()
+        except Exception:
+            pass
+    return 'done'
+
+
+def test_ps_drain_liveness_timeout():
+    """A crashed worker must produce a diagnostic RuntimeError on the PS
+    within PS_DRAIN_TIMEOUT, not a silent hang (VERDICT r1 weak #8)."""
+    res = run_dist(_role_dead_worker, world=3, timeout=120.0)
+    assert res[0] == 'timeout', res
