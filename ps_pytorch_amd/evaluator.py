@@ -80,9 +80,10 @@ class DistributedEvaluator:
     @torch.no_grad()
     def _evaluate_model(self, test_loader):
         self.network.eval()
+        pdtype = next(self.network.parameters()).dtype
         tot, loss_sum, p1_sum, p5_sum = 0, 0.0, 0.0, 0.0
         for data, target in test_loader:
-            data = data.to(self.device)
+            data = data.to(self.device, pdtype)   # loader dtype may differ
             target = target.to(self.device)
             out = self.network(data)
             loss_sum += float(F.cross_entropy(out, target, reduction='sum'))
