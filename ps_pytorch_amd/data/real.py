@@ -188,8 +188,9 @@ class RealDataset:
         self.dtype = dtype
         # raw uint8 stays resident; normalization happens per batch (fused
         # with the augmentation gather, output in compute dtype)
-        self.x_u8 = torch.from_numpy(np.ascontiguousarray(x)).to(device)
-        self.y = torch.from_numpy(np.ascontiguousarray(y)).to(device)
+        # copy: idx/pickle readers may hand back read-only mmap-backed views
+        self.x_u8 = torch.from_numpy(np.ascontiguousarray(x).copy()).to(device)
+        self.y = torch.from_numpy(np.ascontiguousarray(y).copy()).to(device)
         mean, std = _NORM[d]
         C = self.shape[0]
         self.mean = (torch.tensor(mean, dtype=torch.float32, device=device)

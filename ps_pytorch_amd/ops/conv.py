@@ -80,6 +80,23 @@ def _wgrad_split(M: int, K: int, C: int, R: int, S: int,
     return max(1, min(want, max_split, 256))
 
 
+# Ragged-K backward padding (LeNet K=20/50, fc N=500/10): the dgrad/wgrad
+# contraction runs over K, and K % 8 != 0 rows defeat load16's b128 path
+# (mostly 4-B misaligned loads plus a masked-scalar tail octet per row —
+# conv.hip:70-101). Padding dout rows and wT rows to a 64-multiple with
+# zeros is bitwise-neutral (fp32 accum of +0.0 terms) and makes the whole
+# reduction take the AL=true path. PS_PADK=0 disables (A/B).
+_PADK = os.environ.get('PS_PADK', '1') != '0'
+
+
+def _pad_rows(lib, src: torch.Tensor, nrows: int, K: int, Kp: int):
+    """[nrows, K] -> [nrows, Kp] with zero-filled tails (flat alloc)."""
+    dst = torch.empty(nrows * Kp, dtype=src.dtype, device=src.device)
+    lib.ps_padc(dst.data_ptr(), src.data_ptr(), nrows, K, Kp,
+                current_stream_ptr())
+    return dst
+
+
 class _ConvFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, b, stride, pad):
@@ -172,6 +189,14 @@ class _ConvFn(torch.autograd.Function):
         K, _, R, S = w.shape
         P, Q = dout.shape[2], dout.shape[3]
         dx = dw = db = None
+        # ragged-K: pad dout rows once, shared by dgrad and wgrad below
+        padk = (_PADK and K % 8 != 0
+                and (ctx.needs_input_grad[0] or ctx.needs_input_grad[1]))
+        if padk:
+            Kp = (K + 63) & ~63
+            doutp = _pad_rows(lib, dout, Nb * P * Q, K, Kp)
+        else:
+            Kp, doutp = K, dout
         if ctx.needs_input_grad[0]:
             # dgrad wants wT[R,S,C,K] in memory so its B stage is the same
             # contiguous-in-contraction load as fwd (no in-kernel transpose);
@@ -179,6 +204,8 @@ class _ConvFn(torch.autograd.Function):
             wt = torch.empty(R * S * C * K, dtype=w.dtype, device=w.device)
             lib.ps_wt_transpose(wt.data_ptr(), w.data_ptr(), K, R * S * C,
                                 current_stream_ptr())
+            if padk:   # pad the wT rows to match the padded contraction
+                wt = _pad_rows(lib, wt, R * S * C, K, Kp)
             dx = torch.empty_like(x).contiguous(memory_format=_CL)
             if dcarry is not None and ctx.cp:
                 # padded path: fuse the carry AFTER the slice instead
@@ -188,9 +215,9 @@ class _ConvFn(torch.autograd.Function):
                 dxp_carry = None
             if dcarry is not None:
                 dcarry = dcarry.contiguous(memory_format=_CL)
-            lib.ps_conv_dgrad(dout.data_ptr(), wt.data_ptr(), dx.data_ptr(),
+            lib.ps_conv_dgrad(doutp.data_ptr(), wt.data_ptr(), dx.data_ptr(),
                               dcarry.data_ptr() if dcarry is not None else 0,
-                              Nb, H, W, C, K, P, Q, R, S, stride, pad,
+                              Nb, H, W, C, Kp, P, Q, R, S, stride, pad,
                               current_stream_ptr())
             if ctx.cp:
                 # computed at padded C: slice the real channels back
@@ -202,22 +229,25 @@ class _ConvFn(torch.autograd.Function):
                 dx = dxs
         if ctx.needs_input_grad[1]:
             M = Nb * P * Q
-            split = _wgrad_split(M, K, C, R, S, stride, pad, P, Q)
-            partial = torch.empty(split * K * R * S * C,
+            split = _wgrad_split(M, Kp, C, R, S, stride, pad, P, Q)
+            partial = torch.empty(split * Kp * R * S * C,
                                   dtype=torch.float32, device=x.device)
             wt_tgt = ctx.gtgt[0]() if ctx.gtgt[0] is not None else None
-            if ctx.cp:
-                # padded-channel stem: wgrad at the padded C, then slice
-                # the real C_orig channels back into the (steal-target) dw
-                dwp = torch.empty_like(w).contiguous(memory_format=_CL)
-                lib.ps_conv_wgrad(dout.data_ptr(), x.data_ptr(),
+            if ctx.cp or padk:
+                # computed at padded C (stem) and/or padded K (ragged K):
+                # slice the real rows/channels back into the (steal) dw.
+                # NB: C here is already the padded value (x saved padded).
+                dwp = torch.empty((Kp, C, R, S), dtype=w.dtype,
+                                  device=w.device, memory_format=_CL)
+                lib.ps_conv_wgrad(doutp.data_ptr(), x.data_ptr(),
                                   partial.data_ptr(), dwp.data_ptr(),
-                                  Nb, H, W, ctx.cp, K, P, Q, R, S, stride,
+                                  Nb, H, W, C, Kp, P, Q, R, S, stride,
                                   pad, split, current_stream_ptr())
+                c_out = ctx.c_orig if ctx.cp else C
                 dw = (wt_tgt if wt_tgt is not None and wt_tgt.is_cuda
-                      else torch.empty((K, ctx.c_orig, R, S), dtype=w.dtype,
+                      else torch.empty((K, c_out, R, S), dtype=w.dtype,
                                        device=w.device, memory_format=_CL))
-                dw.copy_(dwp[:, :ctx.c_orig])
+                dw.copy_(dwp[:K, :c_out])
             else:
                 dw = (wt_tgt if wt_tgt is not None and wt_tgt.dtype == w.dtype
                       and wt_tgt.is_cuda

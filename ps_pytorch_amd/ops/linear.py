@@ -27,7 +27,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from . import require_lib, current_stream_ptr
-from .conv import _wgrad_split
+from .conv import _PADK, _pad_rows, _wgrad_split
 
 # A/B kill-switch: PS_LINEAR=0 routes through torch (Tensile) GEMMs.
 _ENABLED = os.environ.get('PS_LINEAR', '1') != '0'
@@ -68,25 +68,48 @@ class _LinearFn(torch.autograd.Function):
         N = w.shape[0]
         dy = dout.reshape(M, N).contiguous()
         dx = dw = db = None
+        # ragged-N (fc1's 500, classifier's 10): the dgrad/wgrad contraction
+        # runs over N — pad dy and wT rows to a 64-multiple so the b128 load
+        # path applies (see ops/conv.py _pad_rows)
+        padn = (_PADK and N % 8 != 0
+                and (ctx.needs_input_grad[0] or ctx.needs_input_grad[1]))
+        if padn:
+            Np = (N + 63) & ~63
+            dyp = _pad_rows(lib, dy, M, N, Np)
+        else:
+            Np, dyp = N, dy
         if ctx.needs_input_grad[0]:
             wt = torch.empty(K * N, dtype=w.dtype, device=w.device)
             lib.ps_wt_transpose(wt.data_ptr(), w.data_ptr(), N, K,
                                 current_stream_ptr())
+            if padn:
+                wt = _pad_rows(lib, wt, K, N, Np)
             dx = torch.empty_like(x)
-            lib.ps_conv_dgrad(dy.data_ptr(), wt.data_ptr(), dx.data_ptr(), 0,
-                              M, 1, 1, K, N, 1, 1, 1, 1, 1, 0,
+            lib.ps_conv_dgrad(dyp.data_ptr(), wt.data_ptr(), dx.data_ptr(), 0,
+                              M, 1, 1, K, Np, 1, 1, 1, 1, 1, 0,
                               current_stream_ptr())
             dx = dx.reshape(shape)
         if ctx.needs_input_grad[1]:
-            split = _wgrad_split(M, N, K, 1, 1, stride=1, pad=0, P=1, Q=1)
-            partial = torch.empty(split * N * K, dtype=torch.float32,
+            split = _wgrad_split(M, Np, K, 1, 1, stride=1, pad=0, P=1, Q=1)
+            partial = torch.empty(split * Np * K, dtype=torch.float32,
                                   device=x.device)
             wt_tgt = ctx.gtgt[0]() if ctx.gtgt[0] is not None else None
-            dw = (wt_tgt if wt_tgt is not None and wt_tgt.dtype == w.dtype
-                  and wt_tgt.is_cuda else torch.empty_like(w))
-            lib.ps_conv_wgrad(dy.data_ptr(), x.data_ptr(), partial.data_ptr(),
-                              dw.data_ptr(), M, 1, 1, K, N, 1, 1, 1, 1, 1, 0,
-                              split, current_stream_ptr())
+            if padn:
+                dwp = torch.empty((Np, K), dtype=w.dtype, device=w.device)
+                lib.ps_conv_wgrad(dyp.data_ptr(), x.data_ptr(),
+                                  partial.data_ptr(), dwp.data_ptr(),
+                                  M, 1, 1, K, Np, 1, 1, 1, 1, 1, 0,
+                                  split, current_stream_ptr())
+                dw = (wt_tgt if wt_tgt is not None and wt_tgt.is_cuda
+                      else torch.empty_like(w))
+                dw.copy_(dwp[:N])
+            else:
+                dw = (wt_tgt if wt_tgt is not None and wt_tgt.dtype == w.dtype
+                      and wt_tgt.is_cuda else torch.empty_like(w))
+                lib.ps_conv_wgrad(dy.data_ptr(), x.data_ptr(),
+                                  partial.data_ptr(), dw.data_ptr(),
+                                  M, 1, 1, K, N, 1, 1, 1, 1, 1, 0,
+                                  split, current_stream_ptr())
         if has_bias and ctx.needs_input_grad[2]:
             b_tgt = ctx.gtgt[1]() if ctx.gtgt[1] is not None else None
             db = (b_tgt if b_tgt is not None and b_tgt.dtype == dy.dtype
