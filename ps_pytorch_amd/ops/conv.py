@@ -261,11 +261,20 @@ class _ConvFn(torch.autograd.Function):
             db = (b_tgt if b_tgt is not None and b_tgt.dtype == dout.dtype
                   and b_tgt.is_cuda
                   else torch.empty(K, dtype=dout.dtype, device=dout.device))
-            bpart = torch.empty(512 * K, dtype=torch.float32,
+            bpart = torch.empty(512 * Kp, dtype=torch.float32,
                                 device=dout.device)
-            lib.ps_conv_bias_grad(db.data_ptr(), dout.data_ptr(),
-                                  bpart.data_ptr(), Nb * P * Q, K,
-                                  current_stream_ptr())
+            if padk:
+                # padded rows let colsum take its ushort8 path (ragged K
+                # rows fall back to per-element loads — conv.hip colsum)
+                dbp = torch.empty(Kp, dtype=dout.dtype, device=dout.device)
+                lib.ps_conv_bias_grad(dbp.data_ptr(), doutp.data_ptr(),
+                                      bpart.data_ptr(), Nb * P * Q, Kp,
+                                      current_stream_ptr())
+                db.copy_(dbp[:K])
+            else:
+                lib.ps_conv_bias_grad(db.data_ptr(), dout.data_ptr(),
+                                      bpart.data_ptr(), Nb * P * Q, K,
+                                      current_stream_ptr())
         return dx, dw, db, None, None
 
 

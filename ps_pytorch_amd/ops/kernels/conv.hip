@@ -2047,14 +2047,23 @@ extern "C" void ps_conv_fwd(
 #define FWD_BODY()                                                            \
     do {                                                                      \
         if (R * S > 1 && C == 4 && R * S * 4 <= 64) {   /* padded stem */     \
-            if (stride == 1) LAUNCH_GEMM_CV(128, 64, 1, true, 1);             \
+            if (K <= 32) {          /* skinny K (LeNet-class): TN=32 tile */  \
+                if (stride == 1) LAUNCH_GEMM_CV(128, 32, 1, true, 1);         \
+                else             LAUNCH_GEMM_CV(128, 32, 2, true, 1);         \
+            } else if (stride == 1) LAUNCH_GEMM_CV(128, 64, 1, true, 1);      \
             else             LAUNCH_GEMM_CV(128, 64, 2, true, 1);             \
         } else if (R * S > 1 && C == 4 && R * S * 4 <= 256) {                 \
-            if (stride == 1) LAUNCH_GEMM_CV(128, 64, 1, true, 2);             \
+            if (K <= 32) {                                                    \
+                if (stride == 1) LAUNCH_GEMM_CV(128, 32, 1, true, 2);         \
+                else             LAUNCH_GEMM_CV(128, 32, 2, true, 2);         \
+            } else if (stride == 1) LAUNCH_GEMM_CV(128, 64, 1, true, 2);      \
             else             LAUNCH_GEMM_CV(128, 64, 2, true, 2);             \
         } else if (R * S > 1 && (C & 7) == 0 && (C & 63)                      \
                    && R * S * C <= 768) {        /* padded/ragged C%8 */      \
-            if (stride == 1) LAUNCH_GEMM_CP(128, 64, 1, true, 2, 8);          \
+            if (K <= 32) {                                                    \
+                if (stride == 1) LAUNCH_GEMM_CP(128, 32, 1, true, 2, 8);      \
+                else             LAUNCH_GEMM_CP(128, 32, 2, true, 2, 8);      \
+            } else if (stride == 1) LAUNCH_GEMM_CP(128, 64, 1, true, 2, 8);   \
             else             LAUNCH_GEMM_CP(128, 64, 2, true, 2, 8);          \
         } else if (R * S > 1 && R * S * C <= 64) {  /* one-step: 1 buffer */  \
             if (stride == 1) LAUNCH_GEMM_NB(128, 64, 1, false, true, 1);      \

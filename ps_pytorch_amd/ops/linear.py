@@ -115,10 +115,18 @@ class _LinearFn(torch.autograd.Function):
             db = (b_tgt if b_tgt is not None and b_tgt.dtype == dy.dtype
                   and b_tgt.is_cuda
                   else torch.empty(N, dtype=dy.dtype, device=dy.device))
-            bpart = torch.empty(512 * N, dtype=torch.float32, device=dy.device)
-            lib.ps_conv_bias_grad(db.data_ptr(), dy.data_ptr(),
-                                  bpart.data_ptr(), M, N,
-                                  current_stream_ptr())
+            bpart = torch.empty(512 * Np, dtype=torch.float32,
+                                device=dy.device)
+            if padn:   # aligned rows: colsum's ushort8 path (see conv.py)
+                dbp = torch.empty(Np, dtype=dy.dtype, device=dy.device)
+                lib.ps_conv_bias_grad(dbp.data_ptr(), dyp.data_ptr(),
+                                      bpart.data_ptr(), M, Np,
+                                      current_stream_ptr())
+                db.copy_(dbp[:N])
+            else:
+                lib.ps_conv_bias_grad(db.data_ptr(), dy.data_ptr(),
+                                      bpart.data_ptr(), M, N,
+                                      current_stream_ptr())
         return dx, dw, db
 
 
