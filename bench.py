@@ -191,7 +191,10 @@ def main():
         value = imgs / elapsed
         dtype = 'bf16' if use_cuda else 'fp32'
         print(json.dumps({
-            "metric": "images/sec (whole node), ResNet-18/CIFAR-10 synchronous PS training",
+            "metric": (f"images/sec (whole node), "
+                       f"{'ResNet-18' if args.network == 'ResNet18' else args.network}"
+                       f"/{'CIFAR-10' if args.dataset == 'Cifar10' else args.dataset}"
+                       " synchronous PS training"),
             "value": value,
             "unit": "images/sec",
             "n_gpus": n_gpus,

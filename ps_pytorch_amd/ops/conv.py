@@ -86,7 +86,15 @@ def _wgrad_split(M: int, K: int, C: int, R: int, S: int,
 # conv.hip:70-101). Padding dout rows and wT rows to a 64-multiple with
 # zeros is bitwise-neutral (fp32 accum of +0.0 terms) and makes the whole
 # reduction take the AL=true path. PS_PADK=0 disables (A/B).
+# PS_PADK8=1: pad to an 8-multiple instead of 64 — rows stay 16-B aligned
+# (load16<false>'s uint4 branch) at much less padding traffic, but the
+# kernels run their AL=false instantiations.
 _PADK = os.environ.get('PS_PADK', '1') != '0'
+_PADK8 = os.environ.get('PS_PADK8', '0') != '0'
+
+
+def _pad_target(K: int) -> int:
+    return ((K + 7) & ~7) if _PADK8 else ((K + 63) & ~63)
 
 
 def _pad_rows(lib, src: torch.Tensor, nrows: int, K: int, Kp: int):
@@ -193,7 +201,7 @@ class _ConvFn(torch.autograd.Function):
         padk = (_PADK and K % 8 != 0
                 and (ctx.needs_input_grad[0] or ctx.needs_input_grad[1]))
         if padk:
-            Kp = (K + 63) & ~63
+            Kp = _pad_target(K)
             doutp = _pad_rows(lib, dout, Nb * P * Q, K, Kp)
         else:
             Kp, doutp = K, dout

@@ -27,7 +27,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from . import require_lib, current_stream_ptr
-from .conv import _PADK, _pad_rows, _wgrad_split
+from .conv import _PADK, _pad_rows, _pad_target, _wgrad_split
 
 # A/B kill-switch: PS_LINEAR=0 routes through torch (Tensile) GEMMs.
 _ENABLED = os.environ.get('PS_LINEAR', '1') != '0'
@@ -74,7 +74,7 @@ class _LinearFn(torch.autograd.Function):
         padn = (_PADK and N % 8 != 0
                 and (ctx.needs_input_grad[0] or ctx.needs_input_grad[1]))
         if padn:
-            Np = (N + 63) & ~63
+            Np = _pad_target(N)
             dyp = _pad_rows(lib, dy, M, N, Np)
         else:
             Np, dyp = N, dy
