@@ -26,6 +26,12 @@ class LeNet(nn.Module):
     def __init__(self, num_classes: int = 10, in_channels: int = 1):
         super().__init__()
         self.conv1 = PsConv2d(in_channels, 20, kernel_size=5)
+        # keep the conv1 -> relu -> pool -> conv2 pipeline 8-aligned on the
+        # kernel path: conv1 emits 24 channels (zero pad in weight space),
+        # conv2 absorbs the pre-padded input (see PsConv2d.out_pad). conv2
+        # itself stays at 50 — its output flattens into fc1's 800 features,
+        # where pad channels would shift feature positions.
+        self.conv1.out_pad = 24
         self.conv2 = PsConv2d(20, 50, kernel_size=5)
         # 28x28 -> conv5 -> 24 -> pool -> 12 -> conv5 -> 8 -> pool -> 4
         self.fc1 = PsLinear(50 * 4 * 4, 500)

@@ -329,15 +329,39 @@ def conv_with_passthrough(mod: nn.Conv2d, x: torch.Tensor):
 
 
 class PsConv2d(nn.Conv2d):
-    """nn.Conv2d whose GPU bf16 path runs the in-tree CDNA4 kernels."""
+    """nn.Conv2d whose GPU bf16 path runs the in-tree CDNA4 kernels.
+
+    Ragged-channel pipelines (LeNet's 20 -> 50) can stay 8-aligned end to
+    end via `out_pad` (plain attr, default 0, not in state_dict): on the
+    kernel path the conv emits out_pad channels by zero-padding its WEIGHT
+    and bias (tiny tensors, differentiable F.pad — grads slice back), so
+    the following relu/pool run on aligned rows and the next PsConv2d sees
+    a pre-padded input, which it absorbs by zero-padding its own weight's C
+    dim. The pad channels are exactly zero forward and get exactly-zero
+    grads; the activation-space pads (padc over multi-MB tensors each
+    step) disappear. Torch-fallback paths ignore out_pad and slice off any
+    pre-pad, so CPU/eval flows stay at the nominal channel counts.
+    """
+    out_pad = 0
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if _supported(x, self.weight, self.stride, self.padding,
+        w, b = self.weight, self.bias
+        if _supported(x, w, self.stride, self.padding,
                       self.dilation, self.groups):
-            out = _ConvFn.apply(x, self.weight, self.bias,
-                                self.stride[0], self.padding[0])
+            cin = x.shape[1]
+            if cin > self.in_channels:
+                # pre-padded input from an out_pad producer upstream
+                w = F.pad(w, (0, 0, 0, 0, 0, cin - self.in_channels))
+            if self.out_pad > self.out_channels:
+                kp = self.out_pad - self.out_channels
+                w = F.pad(w, (0, 0, 0, 0, 0, 0, 0, kp))
+                if b is not None:
+                    b = F.pad(b, (0, kp))
+            out = _ConvFn.apply(x, w, b, self.stride[0], self.padding[0])
             st = _pop_stats()
             if st is not None:
                 out._ps_bn_stats = st
             return out
+        if x.shape[1] != self.in_channels:   # fallback: drop pad channels
+            x = x[:, :self.in_channels]
         return super().forward(x)

@@ -139,3 +139,31 @@ def test_resnet18_gpu_loss_decreases():
     losses = [tr.train_step(x, y) for _ in range(40)]
     torch.cuda.synchronize()
     assert min(losses[-5:]) < 0.5 * losses[0], losses
+
+
+def test_lenet_padded_pipeline_grads_match_torch():
+    """conv1.out_pad=24 runs conv1 -> relu -> pool -> conv2 at padded
+    channels on the kernel path (ops/conv.py PsConv2d.out_pad). The pad
+    channels must be exactly zero in fwd and contribute exactly-zero grads:
+    outputs and weight grads must match a plain fp32 torch run (fallback
+    path, out_pad ignored) of the same parameters."""
+    import torch.nn.functional as F
+    from ps_pytorch_amd.models import build_model
+    torch.manual_seed(5)
+    m = build_model('LeNet', num_classes=10, in_channels=1)
+    ref = build_model('LeNet', num_classes=10, in_channels=1)
+    ref.load_state_dict(m.state_dict())
+    m = m.to('cuda').to(torch.bfloat16)
+    x = torch.randn(64, 1, 28, 28)
+    y = torch.randint(0, 10, (64,))
+    out = m(x.to('cuda', torch.bfloat16))
+    assert out.shape == (64, 10)
+    F.cross_entropy(out.float(), y.to('cuda')).backward()
+    rout = ref(x)                      # CPU fp32 reference
+    F.cross_entropy(rout, y).backward()
+    assert (out.float().cpu() - rout).abs().max() < 0.1
+    for (n, p), (_, rp) in zip(m.named_parameters(), ref.named_parameters()):
+        assert p.grad.shape == rp.grad.shape, n
+        g, rg = p.grad.float().cpu(), rp.grad
+        rel = (g - rg).norm() / (rg.norm() + 1e-12)
+        assert rel < 0.05, (n, float(rel))
