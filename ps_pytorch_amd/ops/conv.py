@@ -197,7 +197,10 @@ class _ConvFn(torch.autograd.Function):
         K, _, R, S = w.shape
         P, Q = dout.shape[2], dout.shape[3]
         dx = dw = db = None
-        # ragged-K: pad dout rows once, shared by dgrad and wgrad below
+        # ragged-K: pad dout rows once, shared by dgrad and wgrad below.
+        # K % 8 only — K%64-but-8-aligned douts (the out_pad pipeline's
+        # K=24) measured FASTER unpadded (3.59M vs 3.38M img/s): branchy
+        # 16-B-aligned loads beat eating 64/K x padc traffic.
         padk = (_PADK and K % 8 != 0
                 and (ctx.needs_input_grad[0] or ctx.needs_input_grad[1]))
         if padk:
