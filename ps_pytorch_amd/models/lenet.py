@@ -38,8 +38,12 @@ class LeNet(nn.Module):
         self.fc2 = PsLinear(500, num_classes)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = ps_max_pool2d(F.relu(self.conv1(x)), 2)
-        x = ps_max_pool2d(F.relu(self.conv2(x)), 2)
+        # conv -> pool -> relu -> conv -> pool -> relu -> fc -> fc, exactly
+        # the reference's op order (src/model_ops/lenet.py:24-35 — relu AFTER
+        # pool, and no activation between fc1 and fc2). pool-then-relu is
+        # also 4x cheaper on the relu: max commutes with monotonic relu.
+        x = F.relu(ps_max_pool2d(self.conv1(x), 2))
+        x = F.relu(ps_max_pool2d(self.conv2(x), 2))
         x = x.flatten(1)
-        x = F.relu(self.fc1(x))
+        x = self.fc1(x)
         return self.fc2(x)
