@@ -91,8 +91,23 @@ __global__ __launch_bounds__(256) void bn_stats_kernel(
     f32x8 sum = {0, 0, 0, 0, 0, 0, 0, 0};
     f32x8 sq = {0, 0, 0, 0, 0, 0, 0, 0};
     const long chunk = 8L * R;     // 8 rows in flight (was 4: 82% wait)
-    for (long base = (long)blockIdx.x * chunk + row_in_blk; base < M;
-         base += (long)gridDim.x * chunk) {
+    // guard-free main loop over FULL chunks: the per-row `if (r < M)` kept
+    // the 8 unrolled loads from issuing as one batch (each load sat behind
+    // a branch). Exactly one grid position can be partial (grid stride >=
+    // chunk), handled in the tail below.
+    const long gstride = (long)gridDim.x * chunk;
+    long blk = (long)blockIdx.x * chunk;
+    for (; blk + chunk <= M; blk += gstride) {
+        const long base = blk + row_in_blk;
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+            f32x8 v = VecIO<T>::load(x + (base + (long)u * R) * C + cg * 8);
+            sum += v;
+            sq += v * v;
+        }
+    }
+    if (blk < M) {
+        const long base = blk + row_in_blk;
 #pragma unroll
         for (int u = 0; u < 8; ++u) {
             long r = base + (long)u * R;
@@ -286,10 +301,31 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_kernel(
     f32x8 sum_dyx = {0, 0, 0, 0, 0, 0, 0, 0};
     // contiguous 8-row chunks per block iteration (16-deep REGRESSED
     // 467 -> 716 us/step: the wider span thrashes DRAM pages — same
-    // failure mode as round 1's gridDim-strided unroll)
+    // failure mode as round 1's gridDim-strided unroll).
+    // Guard-free main loop + single partial tail (cf. bn_stats_kernel).
     const long chunk = 8L * R;
-    for (long base = (long)blockIdx.x * chunk + row_in_blk; base < M;
-         base += (long)gridDim.x * chunk) {
+    const long gstride = (long)gridDim.x * chunk;
+    long blk = (long)blockIdx.x * chunk;
+    for (; blk + chunk <= M; blk += gstride) {
+        const long base = blk + row_in_blk;
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+            long r = base + (long)u * R;
+            long o = r * C + cg * 8;
+            f32x8 d = VecIO<T>::load(dy + o);
+            f32x8 xv = VecIO<T>::load(x + o);
+            if constexpr (RELU) {
+                unsigned char mb = mask[r * G + cg];
+#pragma unroll
+                for (int k = 0; k < 8; ++k)
+                    d[k] = (mb >> k) & 1 ? d[k] : 0.f;
+            }
+            sum_dy += d;
+            sum_dyx += d * (xv - mean) * invstd;
+        }
+    }
+    if (blk < M) {
+        const long base = blk + row_in_blk;
 #pragma unroll
         for (int u = 0; u < 8; ++u) {
             long r = base + (long)u * R;
