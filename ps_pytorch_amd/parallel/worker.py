@@ -12,6 +12,7 @@ HIP stream; program-order bucket launch on every rank replaces MPI tags.
 """
 from __future__ import annotations
 
+import os
 import time
 from typing import List, Optional
 
@@ -28,6 +29,10 @@ from ..utils.checkpoint import save_model_step
 from ..utils.logging import get_logger, worker_log_line
 
 logger = get_logger('ps_pytorch_amd.worker')
+
+# PS_HIP_TIMING=1: per-phase spans measured with HIP events (GPU execution
+# time) instead of host timers (enqueue time) — see _forward_backward.
+_HIP_TIMING = os.environ.get('PS_HIP_TIMING', '0') == '1'
 
 
 class DistributedWorker:
@@ -155,25 +160,51 @@ class DistributedWorker:
 
     def _forward_backward(self, data, target):
         """forward + backward with the straggler-abort protocol around it.
-        Returns (loss_or_None, killed)."""
+        Returns (loss_or_None, killed).
+
+        PS_HIP_TIMING=1: bracket the phases with HIP events instead of host
+        timers — host time.time() around async GPU work measures ENQUEUE,
+        not execution (SURVEY.md §5 tracing obligation). Events resolve in
+        _resolve_phase_timing() at log time (one sync in timing mode only).
+        """
         if self.ctrl is not None:
             self.ctrl.post()
         self._step_start = time.time()
         self.f_dur = self.b_dur = 0.0
+        hip_ev = None
+        if _HIP_TIMING and data.is_cuda:
+            hip_ev = [torch.cuda.Event(enable_timing=True) for _ in range(3)]
+            hip_ev[0].record()
         try:
             t0 = time.time()
             out = self.network(data)
             loss = ps_cross_entropy(out, target)
+            if hip_ev is not None:
+                hip_ev[1].record()
             self.f_dur = time.time() - t0
             t0 = time.time()
             loss.backward()
+            if hip_ev is not None:
+                hip_ev[2].record()
             self.b_dur = time.time() - t0
+            self._hip_ev = hip_ev
             if not self.cfg.overlap:
                 self._check_abort()
             return loss.detach(), False
         except StepKilled as e:
             logger.info('%s — aborting rest of backward', e)
+            self._hip_ev = None
             return None, True
+
+    def _resolve_phase_timing(self):
+        """In PS_HIP_TIMING mode, replace the host-measured forward/backward
+        spans with HIP-event-measured GPU execution times (seconds)."""
+        ev = getattr(self, '_hip_ev', None)
+        if ev is not None:
+            ev[2].synchronize()
+            self.f_dur = ev[0].elapsed_time(ev[1]) / 1e3
+            self.b_dur = ev[1].elapsed_time(ev[2]) / 1e3
+            self._hip_ev = None
 
     def train_step(self, data: torch.Tensor, target: torch.Tensor):
         """One synchronous PS step; returns detached loss (None if killed)."""
@@ -216,6 +247,7 @@ class DistributedWorker:
                 comm_dur = time.time() - t0
                 self.cur_step += 1
                 if self.cur_step % cfg.log_interval == 0:
+                    self._resolve_phase_timing()
                     logger.info(worker_log_line(
                         self.rank, self.cur_step, epoch,
                         batch_idx * cfg.batch_size, n_total,
