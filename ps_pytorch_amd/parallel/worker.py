@@ -14,10 +14,9 @@ from __future__ import annotations
 
 import os
 import time
-from typing import List, Optional
+from typing import Optional
 
 import torch
-import torch.nn.functional as F
 
 from ..ops.loss import cross_entropy as ps_cross_entropy
 
