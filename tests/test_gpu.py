@@ -178,3 +178,28 @@ def test_lenet_padded_pipeline_grads_match_torch():
         g, g2 = p.grad.float(), p2.grad.float()
         rel = (g - g2).norm() / (g2.norm() + 1e-12)
         assert rel < 2e-2, (n, float(rel))
+
+
+def test_hip_timing_event_resolution():
+    """PS_HIP_TIMING event bookkeeping: _resolve_phase_timing converts the
+    3 recorded events into positive forward/backward spans (duck-typed self
+    — the full worker needs a process group)."""
+    from ps_pytorch_amd.parallel.worker import DistributedWorker
+
+    class Host:
+        pass
+
+    h = Host()
+    h.f_dur = h.b_dur = 0.0
+    ev = [torch.cuda.Event(enable_timing=True) for _ in range(3)]
+    a = torch.randn(1024, 1024, device='cuda')
+    ev[0].record()
+    b = a @ a
+    ev[1].record()
+    c = b @ b
+    ev[2].record()
+    h._hip_ev = ev
+    DistributedWorker._resolve_phase_timing(h)
+    assert h.f_dur > 0.0 and h.b_dur > 0.0
+    assert h._hip_ev is None
+    del c
