@@ -81,3 +81,32 @@ def test_fused_wire_out():
     wire = torch.empty(n)
     fused_sgd_step(w, g, m, lr=0.1, momentum=0.9, wire_out=wire)
     assert torch.equal(wire, w)
+
+
+@pytest.mark.parametrize('opt_name', ['sgd', 'adam'])
+def test_region_sliced_steps_equal_full_step(opt_name):
+    """The pipelined PS updates bucket slices with region=(start, end) and
+    advance only on the first slice (parallel/ps.py _step_pipelined); the
+    result must be identical to one whole-buffer step."""
+    import torch
+    from ps_pytorch_amd.optim import FlatAdam, FlatSGD
+    torch.manual_seed(17)
+    n = 1000
+    g = torch.randn(n)
+    cuts = [0, 130, 512, 768, n]
+
+    def make(name, w):
+        return (FlatAdam(w, lr=1e-3) if name == 'adam'
+                else FlatSGD(w, lr=0.1, momentum=0.9, weight_decay=1e-4))
+
+    w_full = torch.randn(n)
+    w_sliced = w_full.clone()
+    full = make(opt_name, w_full)
+    sliced = make(opt_name, w_sliced)
+    for it in range(3):
+        full.step(g, grad_scale=0.5)
+        for i in range(len(cuts) - 1):
+            sliced.step(g, grad_scale=0.5, region=(cuts[i], cuts[i + 1]),
+                        advance=(i == 0))
+        g = g.roll(7) * 0.9   # vary grads across iterations
+    assert torch.equal(w_full, w_sliced)
