@@ -50,3 +50,24 @@ def test_resnet50_imagenet_stem():
 def test_unknown_network():
     with pytest.raises(ValueError):
         build_model('AlexNet')
+
+
+def test_psconv_fallback_slices_prepadded_input():
+    """On the torch fallback path (CPU here), a pre-padded input from an
+    out_pad producer upstream is sliced back to in_channels (ops/conv.py
+    PsConv2d.forward) so CPU flows stay at nominal channel counts."""
+    import torch
+    from ps_pytorch_amd.ops.conv import PsConv2d
+    torch.manual_seed(2)
+    conv = PsConv2d(20, 50, kernel_size=5)
+    x = torch.randn(2, 24, 12, 12)        # 4 pad channels of garbage
+    out_pad = conv(x)
+    out_ref = conv(x[:, :20])
+    assert out_pad.shape == out_ref.shape == (2, 50, 8, 8)
+    assert torch.equal(out_pad, out_ref)
+
+
+def test_pad_target_modes():
+    from ps_pytorch_amd.ops import conv as C
+    assert C._pad_target(20) == 64 and C._pad_target(50) == 64
+    assert C._pad_target(500) == 512 and C._pad_target(64) == 64
