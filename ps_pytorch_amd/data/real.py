@@ -216,9 +216,8 @@ class RealDataset:
         B, C, H, W = xb_u8.shape
         p = self.pad
         if p:
-            xf = xb_u8.float()
-            mode = self.pad_mode if self.pad_mode != 'constant' else 'constant'
-            xp = torch.nn.functional.pad(xf, (p, p, p, p), mode=mode)
+            xf = xb_u8.float()   # reflect pad is unsupported on uint8
+            xp = torch.nn.functional.pad(xf, (p, p, p, p), mode=self.pad_mode)
             oy = torch.randint(0, 2 * p + 1, (B,), generator=gen)
             ox = torch.randint(0, 2 * p + 1, (B,), generator=gen)
             oy = oy.to(xp.device)
