@@ -23,7 +23,6 @@ from typing import Optional
 
 import torch
 import torch.distributed as dist
-import torch.nn.functional as F
 
 from ..ops.loss import cross_entropy as ps_cross_entropy
 
