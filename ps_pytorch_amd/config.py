@@ -77,6 +77,11 @@ def add_fit_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
     parser.add_argument('--aggregation', type=str, default='collective',
                         help='collective (reduce-to-root) | gather (per-worker P2P, '
                              'enables arrival-order --num-aggregate selection)')
+    parser.add_argument('--optimizer', type=str, default='sgd',
+                        help='PS-side optimizer: sgd | adam (the reference '
+                             'ships optim/adam.py but hardwires SGD at '
+                             'sync_replicas_master_nn.py:126 — here both '
+                             'fused flat optimizers are selectable)')
     parser.add_argument('--engine', type=str, default='ps',
                         help='ps (1 PS + N-1 workers, the reference design) | '
                              'allreduce (collective DP on all N ranks — the '
@@ -119,12 +124,15 @@ class JobConfig:
     overlap: bool = True
     aggregation: str = 'collective'
     engine: str = 'ps'
+    optimizer: str = 'sgd'
 
     def __post_init__(self) -> None:
         if self.mode not in ('normal', 'kill', 'timeout'):
             raise ValueError(f"unknown --mode {self.mode!r}")
         if self.aggregation not in ('collective', 'gather'):
             raise ValueError(f"unknown --aggregation {self.aggregation!r}")
+        if self.optimizer not in ('sgd', 'adam'):
+            raise ValueError(f"unknown --optimizer {self.optimizer!r}")
         # --mode kill needs arrival-order fan-in: the PS's kill verdict fires
         # from the gather drain's first-k quota (parallel/ps.py); a collective
         # reduce has no arrival order and would never send a verdict, hanging

@@ -37,13 +37,15 @@ logger = get_logger('ps_pytorch_amd.ps')
 
 class ParameterServer:
     def __init__(self, cfg: JobConfig, rank: int, world: int,
-                 device: torch.device, optimizer: str = 'sgd'):
+                 device: torch.device, optimizer: Optional[str] = None):
         assert rank == 0, "the PS is rank 0"
         self.cfg = cfg
         self.rank = rank
         self.world = world
         self.device = device
-        self.optimizer_name = optimizer
+        # --optimizer (default sgd; the reference ships Adam but hardwires
+        # SGD, sync_replicas_master_nn.py:126) — explicit arg still wins
+        self.optimizer_name = optimizer or getattr(cfg, 'optimizer', 'sgd')
         self.compute_dtype = (torch.bfloat16
                               if (device.type == 'cuda' and cfg.compute_dtype == 'bf16')
                               else torch.float32)

@@ -132,3 +132,52 @@ def test_golden_step_bcast_pipeline_off(monkeypatch):
     got = torch.from_numpy(res[0])
     ref = _serial_reference()
     assert torch.allclose(got, ref, atol=1e-5, rtol=1e-5), (got - ref).abs().max()
+
+
+def _role_adam(rank: int, world: int, port: int):
+    """PS with --optimizer adam (the reference ships optim/adam.py but
+    hardwires SGD; here the flag wires the fused flat Adam in)."""
+    from ps_pytorch_amd.parallel.transport import init_distributed
+    from ps_pytorch_amd.parallel.ps import ParameterServer
+    from ps_pytorch_amd.parallel.worker import DistributedWorker
+    cfg = _cfg(True)
+    cfg.optimizer = 'adam'
+    env = init_distributed(backend='gloo')
+    if rank == 0:
+        ps = ParameterServer(cfg, rank, world, env['device'])
+        ps.build_model(10)
+        for _ in range(STEPS):
+            ps.step()
+        return ps.master_w[:ps.flat.total].clone()
+    w = DistributedWorker(cfg, rank, world, env['device'])
+    w.build_model(10)
+    xs, ys = _worker_batches(rank)
+    for i in range(STEPS):
+        w.train_step(xs[i], ys[i])
+    return None
+
+
+def test_golden_step_adam_optimizer():
+    from ps_pytorch_amd.optim import FlatAdam
+    results = run_dist(_role_adam, world=3)
+    got = torch.from_numpy(results[0])
+    # serial reference with FlatAdam, same per-worker batches
+    torch.manual_seed(SEED)
+    net = build_model('LeNet', num_classes=10, in_channels=1)
+    fs = FlatSpace(net)
+    fs.attach_grads()
+    master = fs.flat_w.detach().to(torch.float32).clone()
+    opt = FlatAdam(master, lr=LR)
+    data = {r: _worker_batches(r) for r in (1, 2)}
+    for step in range(STEPS):
+        grad_sum = torch.zeros_like(master)
+        for r in (1, 2):
+            fs.load_flat(master)
+            fs.zero_grads()
+            xs, ys = data[r]
+            F.cross_entropy(net(xs[step]).float(), ys[step]).backward()
+            grad_sum += fs.flat_g
+        opt.step(grad_sum, grad_scale=0.5)
+    ref = master[:fs.total]
+    assert torch.allclose(got, ref, atol=1e-5, rtol=1e-5), \
+        (got - ref).abs().max()
