@@ -156,6 +156,26 @@ class PSTransport:
             self._pending: dict = {}     # (bucket_idx, worker) -> Work
             if dist.is_initialized() and dist.get_backend(group) == 'nccl':
                 self._warmup_p2p()
+        if dist.is_initialized() and world > 1:
+            self._verify_layout()
+
+    def _verify_layout(self) -> None:
+        """All ranks must agree on the flat layout — bucket framing IS the
+        wire protocol (no per-message tags to catch a mismatch). Rank 0
+        broadcasts a fingerprint of FlatSpace.layout_signature(); a
+        divergent rank raises instead of silently mis-framing payloads."""
+        import zlib
+        sig = zlib.crc32(repr(self.flat.layout_signature()).encode())
+        dev = (self.device if dist.get_backend(self.group) == 'nccl'
+               else torch.device('cpu'))
+        t = torch.tensor([sig], dtype=torch.int64, device=dev)
+        mine = int(t)
+        dist.broadcast(t, src=PS_RANK, group=self.group)
+        if int(t) != mine:
+            raise RuntimeError(
+                f"rank {self.rank}: flat-layout fingerprint {mine:#x} does "
+                f"not match rank 0's {int(t):#x} — model/config divergence; "
+                f"bucket framing would corrupt payloads")
 
     def _warmup_p2p(self) -> None:
         """Open every PS<->worker NCCL P2P channel in one deterministic
