@@ -82,3 +82,29 @@ def test_layout_signature_deterministic():
     a = FlatSpace(build_model('ResNet18'))
     b = FlatSpace(build_model('ResNet18'))
     assert a.layout_signature() == b.layout_signature()
+
+
+def _role_layout_mismatch(rank: int, world: int, port: int):
+    """Rank 1 builds a different model: transport init must raise (layout
+    fingerprint broadcast, transport.py _verify_layout)."""
+    import torch
+    from ps_pytorch_amd.models import build_model
+    from ps_pytorch_amd.parallel.flat import FlatSpace
+    from ps_pytorch_amd.parallel.transport import PSTransport, init_distributed
+    env = init_distributed(backend='gloo')
+    torch.manual_seed(0)
+    net = build_model('LeNet' if rank == 0 else 'ResNet18', num_classes=10,
+                      in_channels=1 if rank == 0 else 3)
+    fs = FlatSpace(net)
+    try:
+        PSTransport(fs, torch.float32, env['device'], rank, world)
+        return 'no error'
+    except RuntimeError as e:
+        return 'raised' if 'fingerprint' in str(e) else f'other: {e}'
+
+
+def test_layout_mismatch_raises_at_transport_init():
+    from dist_utils import run_dist
+    res = run_dist(_role_layout_mismatch, world=2)
+    assert res[0] == 'no error'      # rank 0 is the reference fingerprint
+    assert res[1] == 'raised', res
