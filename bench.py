@@ -187,6 +187,15 @@ def main():
                      "compress_grad": "n/a (no comm at N=1)",
                      "overlap": "n/a (no comm at N=1)"}
 
+    if world > 1:
+        # orderly teardown: exiting with live comm threads makes gloo's C++
+        # layer abort sporadically at process exit ("terminate called
+        # without an active exception") — torchrun then reports the whole
+        # run failed even though the metric was printed
+        import torch.distributed as dist
+        dist.barrier(group=barrier_grp)
+        dist.destroy_process_group()
+
     if rank == 0:
         value = imgs / elapsed
         dtype = 'bf16' if use_cuda else 'fp32'

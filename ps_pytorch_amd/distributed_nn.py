@@ -60,6 +60,11 @@ def main(argv=None) -> None:
             cfg, rank=rank, num_shards=world - 1, device=device,
             dtype=worker.compute_dtype)
         worker.train(train_loader, test_loader)
+    # orderly teardown: exiting with live comm threads sporadically aborts
+    # in gloo's C++ layer ("terminate called without an active exception")
+    import torch.distributed as dist
+    if dist.is_initialized():
+        dist.destroy_process_group()
 
 
 if __name__ == '__main__':
