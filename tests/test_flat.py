@@ -1,5 +1,7 @@
 """FlatSpace: view aliasing, bucket framing, grad attachment."""
 import torch
+from hypothesis import given, settings
+from hypothesis import strategies as st
 import torch.nn.functional as F
 
 from ps_pytorch_amd.models import build_model
@@ -108,3 +110,40 @@ def test_layout_mismatch_raises_at_transport_init():
     res = run_dist(_role_layout_mismatch, world=2)
     assert res[0] == 'no error'      # rank 0 is the reference fingerprint
     assert res[1] == 'raised', res
+
+
+@given(
+    sizes=st.lists(st.integers(min_value=1, max_value=5000),
+                   min_size=1, max_size=20),
+    bucket_kb=st.integers(min_value=1, max_value=64),
+)
+@settings(max_examples=40, deadline=None)
+def test_partition_invariants_property(sizes, bucket_kb):
+    """For arbitrary parameter sizes and bucket budgets: buckets tile
+    [0, padded) contiguously, every boundary is 8-elem aligned, every
+    param belongs to exactly one bucket, and no param straddles one."""
+    import torch.nn as nn
+    from ps_pytorch_amd.parallel.flat import FlatSpace
+
+    class Blob(nn.Module):
+        def __init__(self):
+            super().__init__()
+            for i, n in enumerate(sizes):
+                setattr(self, f'p{i}', nn.Parameter(torch.zeros(n)))
+
+    fs = FlatSpace(Blob(), bucket_bytes=bucket_kb * 1024)
+    assert fs.padded % 8 == 0
+    pos = 0
+    seen = []
+    for b in fs.buckets:
+        assert b.start == pos and b.end > b.start
+        assert b.start % 8 == 0
+        seen += b.param_ids
+        pos = b.end
+    assert pos == fs.padded
+    assert seen == list(range(len(fs.params)))
+    for b in fs.buckets:
+        for pid in b.param_ids:
+            off = fs.offsets[pid]
+            assert b.start <= off
+            assert off + fs.params[pid].numel() <= b.end
