@@ -13,9 +13,7 @@ config 5) has the standard compute shape.
 """
 from __future__ import annotations
 
-import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from ..ops.conv import PsConv2d, conv_with_passthrough
 from ..ops.linear import PsLinear
