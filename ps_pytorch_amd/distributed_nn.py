@@ -9,7 +9,6 @@ Rank 0 becomes the PS; ranks 1..N-1 become workers (ref :109-125).
 """
 from __future__ import annotations
 
-import torch
 
 from .config import JobConfig, parse_args, num_classes_of
 from .data import prepare_data

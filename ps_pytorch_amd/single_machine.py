@@ -2,7 +2,6 @@
 BASELINE config 1: LeNet on MNIST, runs without a GPU."""
 from __future__ import annotations
 
-import torch
 
 from .config import JobConfig, parse_args
 from .data import prepare_data

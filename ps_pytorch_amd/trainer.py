@@ -54,7 +54,11 @@ class NNTrainer:
         self.flat = FlatSpace(net, bucket_bytes=int(cfg.bucket_mb * 2 ** 20))
         # f32 master copy + fused update; flat_g is the "wire" (local = trivially summed)
         self.master_w = self.flat.flat_w.detach().to(torch.float32).clone()
-        self.optimizer = FlatSGD(self.master_w, lr=cfg.lr, momentum=cfg.momentum)
+        if getattr(cfg, 'optimizer', 'sgd') == 'adam':
+            self.optimizer = FlatAdam(self.master_w, lr=cfg.lr)
+        else:
+            self.optimizer = FlatSGD(self.master_w, lr=cfg.lr,
+                                     momentum=cfg.momentum)
         # steal mode on GPU: Ps ops write grads into flat_g directly (no
         # per-param AccumulateGrad add kernels); view mode on CPU
         self.flat.attach_grads(steal=(self.device.type == 'cuda'))

@@ -31,3 +31,21 @@ def test_validate_runs():
                                   test_size=256)
     acc = tr.validate(test_loader)
     assert 0.0 <= acc <= 100.0
+
+
+def test_single_machine_adam_optimizer():
+    """--optimizer adam wires FlatAdam into the single-machine engine."""
+    import torch
+    from ps_pytorch_amd.config import JobConfig
+    from ps_pytorch_amd.optim import FlatAdam
+    from ps_pytorch_amd.trainer import NNTrainer
+    cfg = JobConfig(network='LeNet', dataset='MNIST', batch_size=8,
+                    lr=1e-3, optimizer='adam', compute_dtype='fp32')
+    tr = NNTrainer(cfg, device=torch.device('cpu'))
+    tr.build_model()
+    assert isinstance(tr.optimizer, FlatAdam)
+    x = torch.randn(8, 1, 28, 28)
+    y = torch.randint(0, 10, (8,))
+    before = tr.master_w.clone()
+    tr.train_step(x, y)
+    assert not torch.equal(before, tr.master_w)
