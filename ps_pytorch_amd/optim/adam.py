@@ -8,7 +8,6 @@ equivalent torch ops (the numerics reference).
 """
 from __future__ import annotations
 
-import math
 from typing import Optional
 
 import torch

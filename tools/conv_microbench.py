@@ -77,12 +77,12 @@ def main():
             if impl == 'ps':
                 def fwd():
                     lib.ps_conv_fwd(x.data_ptr(), w.data_ptr(), 0,
-                                    out.data_ptr(), Nb, H, W, C, K, P, P,
+                                    out.data_ptr(), 0, Nb, H, W, C, K, P, P,
                                     R, R, stride, pad, strm)
 
                 def bwd_x():
                     lib.ps_conv_dgrad(dout.data_ptr(), wt.data_ptr(),
-                                      dx.data_ptr(), Nb, H, W, C, K, P, P,
+                                      dx.data_ptr(), 0, Nb, H, W, C, K, P, P,
                                       R, R, stride, pad, strm)
 
                 def bwd_w():

@@ -49,10 +49,11 @@ def main():
     for _ in range(args.iters):
         if args.op == 'fwd':
             lib.ps_conv_fwd(x.data_ptr(), w.data_ptr(), 0, out.data_ptr(),
-                            Nb, H, W, C, K, P, P, R, R, stride, pad, strm)
+                            0, Nb, H, W, C, K, P, P, R, R, stride, pad, strm)
         elif args.op == 'dgrad':
             lib.ps_conv_dgrad(dout.data_ptr(), wt.data_ptr(), dx.data_ptr(),
-                              Nb, H, W, C, K, P, P, R, R, stride, pad, strm)
+                              0, Nb, H, W, C, K, P, P, R, R, stride, pad,
+                              strm)
         else:
             lib.ps_conv_wgrad(dout.data_ptr(), x.data_ptr(),
                               partial.data_ptr(), dw.data_ptr(),
