@@ -77,6 +77,11 @@ def add_fit_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
     parser.add_argument('--aggregation', type=str, default='collective',
                         help='collective (reduce-to-root) | gather (per-worker P2P, '
                              'enables arrival-order --num-aggregate selection)')
+    parser.add_argument('--resume-step', type=int, default=0,
+                        help='warm-start: load train-dir/model_step_<K> '
+                             'into the global model before training '
+                             '(parameters only — the reference checkpoints '
+                             'no optimizer state; 0 = fresh start)')
     parser.add_argument('--optimizer', type=str, default='sgd',
                         help='PS-side optimizer: sgd | adam (the reference '
                              'ships optim/adam.py but hardwires SGD at '
@@ -125,6 +130,7 @@ class JobConfig:
     aggregation: str = 'collective'
     engine: str = 'ps'
     optimizer: str = 'sgd'
+    resume_step: int = 0
 
     def __post_init__(self) -> None:
         if self.mode not in ('normal', 'kill', 'timeout'):

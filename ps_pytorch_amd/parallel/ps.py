@@ -78,6 +78,13 @@ class ParameterServer:
                                                    and cfg.aggregation == 'gather'))
         self.ctrl = (ControlPlane(self.rank, self.world)
                      if cfg.mode == 'kill' else None)
+        if cfg.resume_step:
+            # warm-start (beyond-reference: the reference never resumes):
+            # load model_step_<K> params into the live flat views, then the
+            # master copy below snapshots them; optimizer state starts fresh
+            from ..utils.checkpoint import load_model_step
+            load_model_step(net, cfg.train_dir, cfg.resume_step, strict=False)
+            logger.info('PS resumed parameters from step %d', cfg.resume_step)
         # f32 master copy + optimizer state in HBM
         self.master_w = self.flat.flat_w.detach().to(torch.float32).clone()
         if self.optimizer_name == 'adam':

@@ -52,6 +52,9 @@ class NNTrainer:
         net = prep_model(net, self.device, self.compute_dtype)
         self.network = net
         self.flat = FlatSpace(net, bucket_bytes=int(cfg.bucket_mb * 2 ** 20))
+        if getattr(cfg, 'resume_step', 0):
+            from .utils.checkpoint import load_model_step
+            load_model_step(net, cfg.train_dir, cfg.resume_step, strict=False)
         # f32 master copy + fused update; flat_g is the "wire" (local = trivially summed)
         self.master_w = self.flat.flat_w.detach().to(torch.float32).clone()
         if getattr(cfg, 'optimizer', 'sgd') == 'adam':

@@ -43,3 +43,34 @@ def test_bf16_model_checkpoint_saved_as_f32(tmp_path):
     sd = torch.load(model_step_path(str(tmp_path), 1), weights_only=True)
     assert all(v.dtype == torch.float32 for v in sd.values()
                if torch.is_floating_point(v))
+
+
+def test_resume_step_warm_start(tmp_path):
+    """--resume-step K: a fresh engine loads model_step_<K> parameters
+    (beyond-reference feature; the reference never resumes training)."""
+    import torch
+    from ps_pytorch_amd.config import JobConfig
+    from ps_pytorch_amd.trainer import NNTrainer
+    from ps_pytorch_amd.utils.checkpoint import save_model_step
+    cfg = JobConfig(network='LeNet', dataset='MNIST', batch_size=8, lr=0.05,
+                    momentum=0.9, compute_dtype='fp32',
+                    train_dir=str(tmp_path))
+    tr = NNTrainer(cfg, device=torch.device('cpu'))
+    tr.build_model()
+    x = torch.randn(8, 1, 28, 28)
+    y = torch.randint(0, 10, (8,))
+    for _ in range(2):
+        tr.train_step(x, y)
+    save_model_step(tr.network, str(tmp_path), 2)
+
+    cfg2 = JobConfig(network='LeNet', dataset='MNIST', batch_size=8, lr=0.05,
+                     momentum=0.9, compute_dtype='fp32',
+                     train_dir=str(tmp_path), resume_step=2, seed=999)
+    tr2 = NNTrainer(cfg2, device=torch.device('cpu'))
+    tr2.build_model()
+    # params match the checkpoint despite the different init seed
+    for (n, a), (_, b) in zip(tr.network.named_parameters(),
+                              tr2.network.named_parameters()):
+        assert torch.equal(a, b), n
+    assert torch.equal(tr2.master_w[:tr2.flat.total],
+                       tr.master_w[:tr.flat.total])
