@@ -7,6 +7,8 @@ import re
 import subprocess
 import sys
 
+from dist_utils import free_port
+
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
@@ -19,7 +21,7 @@ def test_cli_train_checkpoint_evaluate(tmp_path):
     out = subprocess.run(
         [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
          '--nproc-per-node', '2', '--master-addr', '127.0.0.1',
-         '--master-port', '29917', '--no-python', sys.executable,
+         '--master-port', str(free_port()), '--no-python', sys.executable,
          '-m', 'ps_pytorch_amd.distributed_nn', '--network', 'LeNet',
          '--dataset', 'MNIST', '--batch-size', '16', '--max-steps', '4',
          '--eval-freq', '2', '--train-dir', ck, '--compress-grad', 'None'],
