@@ -26,14 +26,36 @@ REF_B1K = {1: 1.0, 2: 1.9737, 4: 3.0871, 8: 5.1896, 16: 4.2394, 32: 2.4774}
 WORKER_RE = re.compile(r'Worker: \d+, Step: (\d+), .*Time Cost: ([0-9.]+),')
 
 
+def _bench_records(text):
+    """Yield bench-JSON dicts from a file's text: JSON-lines (one object
+    per line, bench.py's output), a whole-file array, or an object
+    wrapping a list (driver SCALE files)."""
+    text = text.strip()
+    if text.startswith('['):
+        yield from json.loads(text)
+        return
+    if text.startswith('{') and '\n' not in text:
+        d = json.loads(text)
+        if 'value' in d:
+            yield d
+        else:   # wrapper object: find the first list of bench records
+            for v in d.values():
+                if isinstance(v, list) and v and isinstance(v[0], dict):
+                    yield from v
+                    return
+        return
+    for line in text.splitlines():
+        line = line.strip()
+        if line.startswith('{'):
+            yield json.loads(line)
+
+
 def from_bench(paths):
     rows = []
     for p in paths:
         with open(p) as f:
-            for line in f:
-                line = line.strip()
-                if line.startswith('{'):
-                    d = json.loads(line)
+            for d in _bench_records(f.read()):
+                if 'value' in d:
                     rows.append((d['n_gpus'], d['value'], d['ms_per_step']))
     return sorted(rows)
 
