@@ -57,3 +57,24 @@ def test_run_backward_matches_plain_backward():
     out3 = x @ w
     run_backward(out3, torch.ones_like(out3))
     assert w.grad.abs().sum() > 0
+
+
+def test_speedup_bench_record_shapes(tmp_path):
+    """from_bench accepts JSON-lines (bench.py), arrays, and wrapper
+    objects (driver SCALE file shapes)."""
+    import os
+    sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), 'analysis'))
+    from speedup import from_bench
+    rec = {"n_gpus": 2, "value": 100.0, "ms_per_step": 1.0}
+    files = []
+    for i, text in enumerate([
+            json.dumps(rec),
+            json.dumps([rec, {**rec, "n_gpus": 4}]),
+            json.dumps(rec) + "\n" + json.dumps({**rec, "n_gpus": 8}),
+            json.dumps({"runs": [{**rec, "n_gpus": 1}]})]):
+        p = tmp_path / f"b{i}.json"
+        p.write_text(text)
+        files.append(str(p))
+    rows = from_bench(files)
+    assert [n for n, _, _ in rows] == [1, 2, 2, 4, 8]
