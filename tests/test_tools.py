@@ -77,4 +77,4 @@ def test_speedup_bench_record_shapes(tmp_path):
         p.write_text(text)
         files.append(str(p))
     rows = from_bench(files)
-    assert [n for n, _, _ in rows] == [1, 2, 2, 4, 8]
+    assert [n for n, _, _ in rows] == [1, 2, 2, 2, 4, 8]
