@@ -3,6 +3,11 @@
 Measures the BASELINE.json headline metric — images/sec (whole node) for the
 synchronous PS engine — at N GPUs (N=1: the single-machine engine; N>1:
 1 PS on rank 0 + N-1 workers, batch 1024 per worker, weak scaling).
+Synthetic data / random-init weights per BASELINE.json (no network for
+datasets); dtype bf16 (compute) with an f32 master; the timed region
+includes forward + backward + comm + the fused optimizer step every
+iteration. Expected scaling shape: rank 0 computes no samples, so
+value(N) ~ (N-1) x per-worker throughput (docs/RCCL_READINESS.md).
 
 Single GPU:      python bench.py --gpus 1 --steps 30 --warmup 5
 Multi GPU (driver launches):
