@@ -74,3 +74,49 @@ def test_resume_step_warm_start(tmp_path):
         assert torch.equal(a, b), n
     assert torch.equal(tr2.master_w[:tr2.flat.total],
                        tr.master_w[:tr.flat.total])
+
+
+def _role_bn_ckpt(rank: int, world: int, port: int, ckdir: str):
+    """BN-net checkpoint division of labor: worker rank 1 (not the PS)
+    saves ResNet/VGG so running stats come from a worker (ref
+    distributed_worker.py:175-177 / sync_replicas_master_nn.py:194-196)."""
+    from ps_pytorch_amd.config import JobConfig
+    from ps_pytorch_amd.data import prepare_data
+    from ps_pytorch_amd.parallel.transport import init_distributed
+    from ps_pytorch_amd.parallel.ps import ParameterServer
+    from ps_pytorch_amd.parallel.worker import DistributedWorker
+    cfg = JobConfig(network='ResNet18', dataset='Cifar10', batch_size=8,
+                    lr=0.05, momentum=0.9, seed=3, max_steps=2, eval_freq=2,
+                    train_dir=ckdir, compress_grad='None', wire_dtype='fp32',
+                    compute_dtype='fp32', log_interval=10 ** 9)
+    env = init_distributed(backend='gloo')
+    if rank == 0:
+        ps = ParameterServer(cfg, rank, world, env['device'])
+        ps.build_model(10)
+        ps.start()
+        return None
+    w = DistributedWorker(cfg, rank, world, env['device'])
+    w.build_model(10)
+    train_loader, _ = prepare_data(cfg, rank=rank, num_shards=world - 1,
+                                   device=env['device'],
+                                   dtype=w.compute_dtype,
+                                   train_size=64, test_size=16)
+    w.train(train_loader)
+    return None
+
+
+def test_bn_net_checkpoint_saved_by_worker(tmp_path):
+    import os
+    import torch
+    from dist_utils import run_dist
+    ck = str(tmp_path / 'ck')
+    os.makedirs(ck)
+    run_dist(_role_bn_ckpt, world=3, args=(ck,))
+    path = os.path.join(ck, 'model_step_2')
+    assert os.path.isfile(path), os.listdir(ck)
+    sd = torch.load(path, map_location='cpu', weights_only=True)
+    # full state_dict incl. BN buffers, with stats actually updated
+    assert 'bn1.running_mean' in sd
+    assert 'bn1.num_batches_tracked' in sd
+    assert int(sd['bn1.num_batches_tracked']) == 2
+    assert float(sd['bn1.running_var'].mean()) != 1.0
