@@ -120,3 +120,10 @@ def test_bn_net_checkpoint_saved_by_worker(tmp_path):
     assert 'bn1.num_batches_tracked' in sd
     assert int(sd['bn1.num_batches_tracked']) == 2
     assert float(sd['bn1.running_var'].mean()) != 1.0
+    # and the evaluator's buffer-aware strict load accepts it
+    from ps_pytorch_amd.config import JobConfig
+    from ps_pytorch_amd.evaluator import DistributedEvaluator
+    ev = DistributedEvaluator(JobConfig(network='ResNet18',
+                                        dataset='Cifar10', train_dir=ck))
+    ev._load_model(path)
+    assert int(ev.network.bn1.num_batches_tracked) == 2
