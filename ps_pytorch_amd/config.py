@@ -64,6 +64,9 @@ def add_fit_args(parser: argparse.ArgumentParser) -> argparse.ArgumentParser:
                              './<name>_data, then synthetic data')
     parser.add_argument('--enable-gpu', type=str2bool, nargs='?', const=True,
                         default=False, help='run compute on GPUs (one rank per GPU)')
+    parser.add_argument('--no-cuda', action='store_true', default=False,
+                        help='disables GPU training (overrides --enable-gpu; '
+                             'reference flag parity, distributed_nn.py:42)')
     # --- MI355X-native knobs (new; not in the reference) ---
     parser.add_argument('--wire-dtype', type=str, default='fp32',
                         help='fp32 | bf16 : uncompressed wire dtype (one declared '
@@ -160,6 +163,8 @@ class JobConfig:
     def from_args(cls, args: argparse.Namespace) -> "JobConfig":
         known = {f.name for f in dataclasses.fields(cls)}
         kw = {k: v for k, v in vars(args).items() if k in known}
+        if getattr(args, 'no_cuda', False):
+            kw['enable_gpu'] = False
         return cls(**kw)
 
 
