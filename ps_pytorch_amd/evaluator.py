@@ -100,7 +100,18 @@ class DistributedEvaluator:
 def main(argv=None) -> None:
     parser = argparse.ArgumentParser(description='ps_pytorch_amd evaluator')
     add_fit_args(parser)
+    # reference evaluator flag aliases (distributed_evaluator.py:45-50)
+    parser.add_argument('--eval-batch-size', type=int, default=0,
+                        help='validation batch size (alias of '
+                             '--test-batch-size; reference flag parity)')
+    parser.add_argument('--model-dir', type=str, default='',
+                        help='checkpoint directory (alias of --train-dir; '
+                             'reference flag parity)')
     args = parser.parse_args(argv)
+    if args.eval_batch_size:
+        args.test_batch_size = args.eval_batch_size
+    if args.model_dir:
+        args.train_dir = args.model_dir
     cfg = JobConfig.from_args(args)
     ev = DistributedEvaluator(cfg)
     _, test_loader = prepare_data(cfg, device=ev.device)
