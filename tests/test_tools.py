@@ -78,3 +78,14 @@ def test_speedup_bench_record_shapes(tmp_path):
         files.append(str(p))
     rows = from_bench(files)
     assert [n for n, _, _ in rows] == [1, 2, 2, 2, 4, 8]
+
+
+def test_worker_line_matches_reference_format():
+    """The worker log line is the de-facto metrics protocol (parsed by the
+    tuning parser and analysis); it must stay byte-identical to the
+    reference's template (src/distributed_worker.py:169)."""
+    from ps_pytorch_amd.utils.logging import WORKER_LINE
+    assert WORKER_LINE == (
+        'Worker: {}, Step: {}, Epoch: {} [{}/{} ({:.0f}%)], '
+        'Loss: {:.4f}, Time Cost: {:.4f}, FetchWeight: {:.4f}, '
+        'Forward: {:.4f}, Backward: {:.4f}, Comm Cost: {:.4f}')
