@@ -554,8 +554,15 @@ class ControlPlane:
                 and self._evt.is_set() and int(self._buf) == 1)
 
     def finish(self) -> None:
-        """End of step: consume this step's verdict (always arrives)."""
+        """End of step: consume this step's verdict (always arrives while
+        the PS lives). Bounded join — a dead PS otherwise hangs every
+        worker here forever (same liveness policy as PS_DRAIN_TIMEOUT)."""
         if getattr(self, '_thread', None) is not None:
-            self._thread.join()
+            limit = float(os.environ.get('PS_CTRL_TIMEOUT', '600'))
+            self._thread.join(limit)
+            if self._thread.is_alive():
+                raise RuntimeError(
+                    f"rank {self.rank}: no straggler verdict from the PS in "
+                    f"{limit:.0f}s (PS_CTRL_TIMEOUT) — PS likely crashed")
             self._thread = None
             self._evt = None
